@@ -1,0 +1,213 @@
+"""CLI: serve-native / serve-hf / serve-ollama / serve-hf-remote / register /
+bench — click group, same command names as the reference (bee2bee/__main__.py
+:30-123) plus native-engine extras.
+"""
+from __future__ import annotations
+
+import asyncio
+import os
+
+import click
+
+from .config import get_bootstrap_url, set_bootstrap_url
+from .utils import setup_logging
+
+
+@click.group()
+def cli() -> None:
+    """Bee2Bee-AMD: MI355X-native decentralized inference mesh."""
+    setup_logging()
+
+
+def _serve(backend: str, **kwargs) -> None:
+    from .mesh.node import run_mesh_node
+
+    bootstrap = get_bootstrap_url()
+    asyncio.run(run_mesh_node(bootstrap_link=bootstrap, backend=backend, **kwargs))
+
+
+@cli.command("serve-hf")
+@click.option("--model", default="distilgpt2", help="Model name (preset or HF id)")
+@click.option("--model-path", default=None, help="Local HF checkpoint dir (safetensors)")
+@click.option("--port", default=0, type=int, help="Mesh bind port")
+@click.option("--region", default="Auto", help="Region name")
+@click.option("--api-port", default=8000, type=int, help="FastAPI port")
+@click.option("--device", default=None, help="cuda:N / cpu (auto when omitted)")
+def serve_hf(model, model_path, port, region, api_port, device):
+    """Serve a model on the native MI355X engine (HF-checkpoint compatible).
+
+    Named serve-hf for reference-CLI compatibility; the execution engine is
+    the hand-written CDNA4 HIP runtime, not transformers."""
+    _serve(
+        "native",
+        model_name=model,
+        model_path=model_path,
+        port=port,
+        region=region,
+        api_port=api_port,
+        device=device,
+    )
+
+
+@cli.command("serve-native")
+@click.option("--model", default="llama3-8b", help="Model preset or HF id")
+@click.option("--model-path", default=None, help="Local HF checkpoint dir")
+@click.option("--port", default=0, type=int)
+@click.option("--region", default="Auto")
+@click.option("--api-port", default=8000, type=int)
+@click.option("--device", default=None)
+def serve_native(model, model_path, port, region, api_port, device):
+    """Serve a model on the native MI355X HIP engine (alias of serve-hf)."""
+    _serve(
+        "native",
+        model_name=model,
+        model_path=model_path,
+        port=port,
+        region=region,
+        api_port=api_port,
+        device=device,
+    )
+
+
+@cli.command("serve-ollama")
+@click.option("--model", default="llama3", help="Ollama model name")
+@click.option("--host", default="0.0.0.0", help="Bind host")
+@click.option("--port", default=0, type=int, help="Bind port")
+@click.option("--public-host", default=None, help="Public IP/hostname")
+@click.option("--region", default="Auto")
+@click.option("--api-port", default=8000, type=int)
+def serve_ollama(model, host, port, public_host, region, api_port):
+    """Serve a local Ollama model with P2P connectivity."""
+    _serve(
+        "ollama",
+        model_name=model,
+        host=host,
+        port=port,
+        announce_host=public_host,
+        region=region,
+        api_port=api_port,
+    )
+
+
+@cli.command("serve-hf-remote")
+@click.option("--model", default="meta-llama/Llama-2-7b-hf", help="HF model name")
+@click.option("--token", required=True, help="HF API token")
+@click.option("--region", default="Cloud")
+@click.option("--api-port", default=8000, type=int)
+def serve_hf_remote(model, token, region, api_port):
+    """Serve via the HF Inference API with a local FastAPI proxy."""
+    os.environ["HUGGING_FACE_HUB_TOKEN"] = token
+    _serve("hf_remote", model_name=model, region=region, api_port=api_port)
+
+
+@cli.command("config")
+@click.argument("key")
+@click.argument("value")
+def config_cmd(key, value):
+    """Set a config value (e.g. `config bootstrap_url ws://host:port`)."""
+    if key == "bootstrap_url":
+        set_bootstrap_url(value)
+    else:
+        from .config import load_config, save_config
+
+        cfg = load_config()
+        cfg[key] = value
+        save_config(cfg)
+    click.echo(f"set {key} = {value}")
+
+
+@cli.command("register")
+@click.option("--node-url", default=None, help="Specific node URL to register")
+@click.option("--network", default="connectit", help="Network name")
+@click.option("--region", prompt="Node Region", default="US-West")
+@click.option("--test/--no-test", default=True, help="Run handshake test")
+def register(node_url, network, region, test):
+    """Register a node in the global directory (Supabase / entrypoint)."""
+
+    async def _reg():
+        from .mesh.node import MeshNode
+        from .mesh.registry import RegistryClient
+
+        target_addr = node_url
+        peer_id = f"ext-{os.urandom(4).hex()}"
+        local_node = None
+        if not target_addr:
+            local_node = MeshNode(port=0, enable_nat=False)
+            await local_node.start()
+            target_addr = local_node.addr
+            peer_id = local_node.peer_id
+        click.echo(f"target region: {region}")
+        click.echo(f"node address:  {target_addr}")
+        if test and node_url:
+            # real handshake: open a WS, send hello, expect a hello back
+            import json
+
+            import aiohttp
+
+            from .mesh import wire
+
+            ok = False
+            try:
+                async with aiohttp.ClientSession() as session:
+                    async with session.ws_connect(
+                        node_url, max_msg_size=wire.MAX_FRAME, timeout=10
+                    ) as ws:
+                        await ws.send_str(
+                            json.dumps(
+                                wire.hello(peer_id, "", region, {}, {})
+                            )
+                        )
+                        msg = await asyncio.wait_for(ws.receive(), timeout=10)
+                        data = json.loads(msg.data)
+                        ok = data.get("type") == wire.HELLO
+            except Exception as e:
+                click.echo(f"handshake failed: {e}")
+            click.echo("handshake OK" if ok else "handshake FAILED")
+        reg = RegistryClient()
+        if reg.enabled:
+            await reg.sync_node(
+                peer_id=peer_id,
+                address=target_addr,
+                models=["manual-entry" if node_url else "system-test"],
+                tag=f"cli-{network}",
+                region=region,
+            )
+            click.echo("node registered")
+        else:
+            click.echo("registry unavailable (no credentials); skipped")
+        if local_node is not None:
+            await local_node.stop()
+
+    asyncio.run(_reg())
+
+
+@cli.command("bench")
+@click.option("--model", default="llama3-8b")
+@click.option("--batch", default=64, type=int)
+@click.option("--steps", default=32, type=int)
+@click.option("--warmup", default=8, type=int)
+def bench_cmd(model, batch, steps, warmup):
+    """Run the single-process decode benchmark (see bench.py for the full
+    multi-GPU contract)."""
+    import subprocess
+    import sys
+
+    subprocess.run(
+        [
+            sys.executable,
+            os.path.join(os.path.dirname(os.path.dirname(__file__)), "bench.py"),
+            "--model",
+            model,
+            "--batch",
+            str(batch),
+            "--steps",
+            str(steps),
+            "--warmup",
+            str(warmup),
+        ],
+        check=False,
+    )
+
+
+if __name__ == "__main__":
+    cli()

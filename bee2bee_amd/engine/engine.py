@@ -1,0 +1,445 @@
+"""InferenceEngine: continuous-batching serving front end over the Runner.
+
+Replaces the reference's per-request `transformers.generate()` thread
+(bee2bee/hf.py:84-108) with a single engine thread that owns the GPU:
+requests are admitted into a running batch (batched varlen prefill), decoded
+together one token per step (hipGraph-replayed on GPU), streamed out
+per-request, and retired individually. Paged KV blocks are recycled on
+retirement, so the 288 GB HBM pool is shared across the whole request mix.
+"""
+from __future__ import annotations
+
+import logging
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from ..models.spec import ModelSpec, resolve_spec
+from ..models.tokenizer import load_tokenizer
+from ..models.weights import ModelWeights
+from ..utils import new_id, report_engine_throughput
+from .graphs import DecodeGraphs, decode_slot_mapping
+from .kv import PagedKV
+from .runner import Runner
+from .sampler import SamplingParams, sample
+
+logger = logging.getLogger("bee2bee_amd.engine")
+
+_STREAM_END = object()
+
+
+@dataclass
+class GenerationRequest:
+    prompt_ids: List[int]
+    max_new_tokens: int = 128
+    sampling: SamplingParams = field(default_factory=SamplingParams)
+    stop_token_ids: tuple = ()
+    rid: str = field(default_factory=lambda: new_id("req"))
+    # filled by the engine
+    out_queue: "queue.Queue" = field(default_factory=queue.Queue)
+    submit_ts: float = field(default_factory=time.time)
+    first_token_ts: Optional[float] = None
+    done_ts: Optional[float] = None
+    error: Optional[str] = None
+    output_ids: List[int] = field(default_factory=list)
+
+
+class _Active:
+    __slots__ = ("req", "seq_id", "length")
+
+    def __init__(self, req: GenerationRequest, seq_id: int, length: int) -> None:
+        self.req = req
+        self.seq_id = seq_id
+        self.length = length  # tokens currently in KV cache
+
+
+class InferenceEngine:
+    def __init__(
+        self,
+        model: str | ModelSpec,
+        device: Optional[str] = None,
+        dtype: torch.dtype = torch.bfloat16,
+        model_path: Optional[str] = None,
+        max_batch: int = 64,
+        max_seq_len: Optional[int] = None,
+        use_graphs: Optional[bool] = None,
+        seed: int = 0,
+        kv_margin_blocks: int = 8,
+    ) -> None:
+        self.spec = model if isinstance(model, ModelSpec) else resolve_spec(model, model_path)
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        on_gpu = self.device.type == "cuda"
+        if not on_gpu and dtype is torch.bfloat16:
+            dtype = torch.float32  # CPU reference path runs fp32
+        self.dtype = dtype
+        self.max_batch = max_batch
+        self.max_seq_len = min(max_seq_len or 4096, self.spec.max_seq_len)
+        self.tokenizer = load_tokenizer(
+            model_path, self.spec.vocab_size, self.spec.bos_token_id, self.spec.eos_token_id
+        )
+
+        if on_gpu:
+            # fail loudly if the HIP extension is absent (no silent eager path)
+            from .. import ops
+
+            ops.require_hip()
+
+        logger.info(
+            "building %s (%.2fB params) on %s dtype=%s",
+            self.spec.name,
+            self.spec.n_params() / 1e9,
+            self.device,
+            dtype,
+        )
+        self.weights = ModelWeights(self.spec, self.device, dtype)
+        if model_path:
+            self.weights.load_hf(model_path)
+        else:
+            self.weights.random_init(seed=seed)
+
+        blocks_per_seq = -(-self.max_seq_len // 32)
+        n_blocks = max_batch * blocks_per_seq + kv_margin_blocks
+        self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks)
+        self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
+
+        self.use_graphs = on_gpu if use_graphs is None else (use_graphs and on_gpu)
+        self.graphs: Optional[DecodeGraphs] = None
+        if self.use_graphs:
+            self.graphs = DecodeGraphs(self.runner, max_batch, blocks_per_seq)
+        # scratch sequence backing pad rows of graph-bucket batches
+        self._scratch_seq = -1
+        self.kv.new_seq(self._scratch_seq)
+        self.kv.extend_seq(self._scratch_seq, 1)
+
+        self._gen = torch.Generator(device=self.device)
+        self._gen.manual_seed(seed)
+
+        self._pending: "queue.Queue[GenerationRequest]" = queue.Queue()
+        self._active: List[_Active] = []
+        self._next_seq = 0
+        self._stop = False
+        self._wake = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self._tok_window: List[tuple] = []  # (ts, n_tokens)
+        self.total_tokens = 0
+
+    # ------------------------------------------------------------ lifecycle
+
+    def start(self) -> None:
+        if self._thread is None:
+            self._thread = threading.Thread(target=self._loop, daemon=True, name="engine")
+            self._thread.start()
+
+    def shutdown(self) -> None:
+        self._stop = True
+        self._wake.set()
+        if self._thread is not None:
+            self._thread.join(timeout=30)
+            self._thread = None
+
+    # -------------------------------------------------------------- serving
+
+    def submit(self, req: GenerationRequest) -> GenerationRequest:
+        self.start()
+        self._pending.put(req)
+        self._wake.set()
+        return req
+
+    def generate(
+        self,
+        prompt_ids: List[int],
+        max_new_tokens: int = 128,
+        temperature: Optional[float] = 0.7,
+        stop_token_ids: tuple = (),
+        on_token=None,
+    ) -> GenerationRequest:
+        """Blocking convenience wrapper: submit + drain the stream."""
+        req = GenerationRequest(
+            prompt_ids=list(prompt_ids),
+            max_new_tokens=max_new_tokens,
+            sampling=SamplingParams.from_request(temperature),
+            stop_token_ids=tuple(stop_token_ids),
+        )
+        self.submit(req)
+        while True:
+            item = req.out_queue.get()
+            if item is _STREAM_END:
+                break
+            if on_token is not None:
+                on_token(item)
+        if req.error:
+            raise RuntimeError(req.error)
+        return req
+
+    def generate_text(
+        self,
+        prompt: str,
+        max_new_tokens: int = 128,
+        temperature: Optional[float] = 0.7,
+        on_text=None,
+    ) -> Dict[str, Any]:
+        """Text-level wrapper used by the mesh service."""
+        t0 = time.time()
+        ids = self.tokenizer.encode(prompt)
+        ids = ids[-(self.max_seq_len - max_new_tokens - 1) :]
+        emitted: List[str] = []
+        decoded_upto = 0
+
+        def _on_token(tok: int) -> None:
+            nonlocal decoded_upto
+            if on_text is None:
+                return
+            text = self.tokenizer.decode(req.output_ids)
+            delta = text[decoded_upto:]
+            # hold back partial unicode replacement char at the boundary
+            if delta and not delta.endswith("�"):
+                decoded_upto = len(text)
+                emitted.append(delta)
+                on_text(delta)
+
+        stop_ids = ()
+        eos = getattr(self.tokenizer, "eos_token_id", None)
+        if eos is not None:
+            stop_ids = (eos,)
+        req = GenerationRequest(
+            prompt_ids=ids,
+            max_new_tokens=max_new_tokens,
+            sampling=SamplingParams.from_request(temperature),
+            stop_token_ids=stop_ids,
+        )
+        self.submit(req)
+        while True:
+            item = req.out_queue.get()
+            if item is _STREAM_END:
+                break
+            _on_token(item)
+        if req.error:
+            raise RuntimeError(req.error)
+        text = self.tokenizer.decode(req.output_ids)
+        if on_text is not None and decoded_upto < len(text):
+            on_text(text[decoded_upto:])
+        return {
+            "text": text,
+            "tokens": len(req.output_ids),
+            "latency_ms": int((time.time() - t0) * 1000),
+            "ttft_ms": int(((req.first_token_ts or time.time()) - t0) * 1000),
+        }
+
+    # ---------------------------------------------------------- engine loop
+
+    def _loop(self) -> None:
+        while not self._stop:
+            try:
+                did_work = self._step()
+            except Exception as e:  # engine errors fail all active requests
+                logger.exception("engine step failed")
+                for a in self._active:
+                    a.req.error = str(e)
+                    a.req.out_queue.put(_STREAM_END)
+                    self.kv.free_seq(a.seq_id)
+                self._active.clear()
+                did_work = True
+            if not did_work:
+                self._wake.wait(timeout=0.05)
+                self._wake.clear()
+
+    def _step(self) -> bool:
+        admitted = self._admit()
+        if admitted:
+            self._prefill(admitted)
+        if self._active:
+            self._decode_once()
+            return True
+        return bool(admitted)
+
+    def _admit(self) -> List[_Active]:
+        admitted: List[_Active] = []
+        while len(self._active) + len(admitted) < self.max_batch:
+            try:
+                req = self._pending.get_nowait()
+            except queue.Empty:
+                break
+            need_len = min(
+                len(req.prompt_ids) + req.max_new_tokens, self.max_seq_len
+            )
+            blocks_needed = -(-need_len // self.kv.block_size)
+            if self.kv.free_blocks < blocks_needed:
+                # out of KV memory: push back and wait for retirements
+                self._pending.put(req)
+                break
+            seq_id = self._next_seq
+            self._next_seq += 1
+            self.kv.new_seq(seq_id)
+            admitted.append(_Active(req, seq_id, 0))
+        return admitted
+
+    @torch.no_grad()
+    def _prefill(self, admitted: List[_Active]) -> None:
+        ids_list, pos_list, slot_list, cu = [], [], [], [0]
+        for a in admitted:
+            p = a.req.prompt_ids or [self.spec.bos_token_id]
+            p = p[: self.max_seq_len - a.req.max_new_tokens - 1] or p[:1]
+            a.req.prompt_ids = p
+            self.kv.extend_seq(a.seq_id, len(p))
+            a.length = len(p)
+            ids_list.extend(p)
+            pos_list.extend(range(len(p)))
+            slot_list.extend(self.kv.slot_mapping(a.seq_id, range(len(p))))
+            cu.append(cu[-1] + len(p))
+        dev = self.device
+        input_ids = torch.tensor(ids_list, dtype=torch.int64, device=dev)
+        positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
+        slots = torch.tensor(slot_list, dtype=torch.int32, device=dev)
+        cu_seqlens = torch.tensor(cu, dtype=torch.int32, device=dev)
+        max_len = max(len(a.req.prompt_ids) for a in admitted)
+        hidden = self.runner.forward_prefill(
+            input_ids, positions, slots, cu_seqlens, max_len
+        )
+        last_rows = torch.tensor(
+            [c - 1 for c in cu[1:]], dtype=torch.int64, device=dev
+        )
+        logits = self.runner.lm_head(hidden[last_rows])
+        self._sample_and_emit(admitted, logits)
+        self._active.extend([a for a in admitted if a.req.done_ts is None])
+
+    @torch.no_grad()
+    def _decode_once(self) -> None:
+        B = len(self._active)
+        dev = self.device
+        for a in self._active:
+            self.kv.extend_seq(a.seq_id, a.length + 1)
+
+        last_ids = [
+            (a.req.output_ids[-1] if a.req.output_ids else a.req.prompt_ids[-1])
+            for a in self._active
+        ]
+        positions = [a.length for a in self._active]
+        seq_lens = [a.length + 1 for a in self._active]
+
+        if self.graphs is not None:
+            g = self.graphs
+            bucket = g.bucket_for(B)
+            g.input_ids[:B].copy_(
+                torch.tensor(last_ids, dtype=torch.int64), non_blocking=True
+            )
+            g.positions[:B].copy_(
+                torch.tensor(positions, dtype=torch.int32), non_blocking=True
+            )
+            g.seq_lens[:B].copy_(
+                torch.tensor(seq_lens, dtype=torch.int32), non_blocking=True
+            )
+            bt = self.kv.block_table([a.seq_id for a in self._active])
+            g.block_table[:B, : bt.shape[1]].copy_(bt, non_blocking=True)
+            if bucket > B:  # pad rows -> scratch sequence, length 1
+                sb = self.kv.block_table([self._scratch_seq])[0]
+                g.block_table[B:bucket, : sb.shape[0]].copy_(sb)
+                g.positions[B:bucket].zero_()
+                g.seq_lens[B:bucket].fill_(1)
+                g.input_ids[B:bucket].zero_()
+            logits = g.run(B)[:B]
+        else:
+            input_ids = torch.tensor(last_ids, dtype=torch.int64, device=dev)
+            pos_t = torch.tensor(positions, dtype=torch.int32, device=dev)
+            lens_t = torch.tensor(seq_lens, dtype=torch.int32, device=dev)
+            bt = self.kv.block_table([a.seq_id for a in self._active])
+            slots = decode_slot_mapping(bt, pos_t, self.kv.block_size)
+            hidden = self.runner.forward_decode(input_ids, pos_t, slots, bt, lens_t)
+            logits = self.runner.lm_head(hidden)
+
+        for a in self._active:
+            a.length += 1
+        self._sample_and_emit(self._active, logits)
+        done = [a for a in self._active if a.req.done_ts is not None]
+        for a in done:
+            self.kv.free_seq(a.seq_id)
+        self._active = [a for a in self._active if a.req.done_ts is None]
+
+    def _sample_and_emit(self, acts: List[_Active], logits: torch.Tensor) -> None:
+        # group rows by sampling params so each group is one sample() call
+        groups: Dict[tuple, List[int]] = {}
+        for i, a in enumerate(acts):
+            sp = a.req.sampling
+            groups.setdefault(
+                (sp.greedy, sp.temperature, sp.top_p, sp.top_k), []
+            ).append(i)
+        next_ids = torch.empty(len(acts), dtype=torch.int64)
+        for key, rows in groups.items():
+            sp = acts[rows[0]].req.sampling
+            idx = torch.tensor(rows, dtype=torch.int64, device=logits.device)
+            gen = self._gen if self.device.type == "cuda" else None
+            toks = sample(logits[idx], sp, generator=gen)
+            next_ids[torch.tensor(rows)] = toks.to("cpu")
+        now = time.time()
+        n_emitted = 0
+        for i, a in enumerate(acts):
+            tok = int(next_ids[i])
+            r = a.req
+            if r.first_token_ts is None:
+                r.first_token_ts = now
+            r.output_ids.append(tok)
+            r.out_queue.put(tok)
+            n_emitted += 1
+            if tok in r.stop_token_ids or len(r.output_ids) >= r.max_new_tokens:
+                r.done_ts = now
+                r.out_queue.put(_STREAM_END)
+        self._note_throughput(n_emitted, now)
+
+    def _note_throughput(self, n: int, now: float) -> None:
+        self.total_tokens += n
+        self._tok_window.append((now, n))
+        cutoff = now - 10.0
+        while self._tok_window and self._tok_window[0][0] < cutoff:
+            self._tok_window.pop(0)
+        span = now - self._tok_window[0][0] if len(self._tok_window) > 1 else 1.0
+        tps = sum(x[1] for x in self._tok_window) / max(span, 1e-3)
+        report_engine_throughput(tps)
+
+    # ------------------------------------------------ direct batch API (bench)
+
+    @torch.no_grad()
+    def bench_prefill(self, batch: int, prompt_len: int, seed: int = 1234) -> List[int]:
+        """Fill the KV cache with synthetic prompts; returns seq_ids."""
+        gen = torch.Generator().manual_seed(seed)
+        seq_ids = []
+        for b in range(batch):
+            seq_id = self._next_seq
+            self._next_seq += 1
+            self.kv.new_seq(seq_id)
+            seq_ids.append(seq_id)
+        ids = torch.randint(
+            4, self.spec.vocab_size, (batch, prompt_len), generator=gen
+        )
+        acts = []
+        for b, seq_id in enumerate(seq_ids):
+            req = GenerationRequest(
+                prompt_ids=ids[b].tolist(), max_new_tokens=10**9,
+                sampling=SamplingParams(greedy=True),
+            )
+            acts.append(_Active(req, seq_id, 0))
+        self._prefill_no_admit(acts)
+        self._bench_acts = acts
+        return seq_ids
+
+    @torch.no_grad()
+    def _prefill_no_admit(self, acts: List[_Active]) -> None:
+        # identical to _prefill but without touching the serving active set
+        saved = self._active
+        self._active = []
+        self._prefill(acts)
+        self._active = saved
+
+    @torch.no_grad()
+    def bench_decode_step(self) -> torch.Tensor:
+        """One decode step over the bench batch; returns sampled ids [B]."""
+        acts = self._bench_acts
+        saved = self._active
+        self._active = list(acts)
+        self._decode_once()
+        self._active = saved
+        return torch.tensor([a.req.output_ids[-1] for a in acts])

@@ -1,0 +1,195 @@
+"""Runner: executes the transformer layer stack with bee2bee_amd.ops.
+
+This is the replacement for the reference's `transformers.generate()` hot
+loop (bee2bee/hf.py:84-108): explicit prefill/decode forwards over fused
+weights, paged KV, and the HIP kernel set. Projection GEMMs go through
+F.linear (hipBLASLt); everything fused is ops.* (HIP on GPU).
+
+Pipeline parallelism: a Runner can own a contiguous layer range; the first
+stage embeds tokens, the last stage applies the final norm + lm_head
+(parallel/pp.py moves the hidden states between stages over RCCL).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from ..models.spec import ModelSpec
+from ..models.weights import ModelWeights
+from .kv import PagedKV
+
+
+class Runner:
+    def __init__(
+        self,
+        spec: ModelSpec,
+        weights: ModelWeights,
+        kv: PagedKV,
+        device: torch.device,
+        dtype: torch.dtype = torch.bfloat16,
+        layer_range: Optional[Tuple[int, int]] = None,
+    ) -> None:
+        self.spec = spec
+        self.weights = weights
+        self.kv = kv
+        self.device = device
+        self.dtype = dtype
+        self.layer_lo, self.layer_hi = layer_range or (0, spec.n_layers)
+        self.scale = spec.head_dim**-0.5
+        cos, sin = ops.rope_tables(
+            spec.max_seq_len, spec.head_dim, spec.rope_theta, device
+        )
+        self.rope_cos, self.rope_sin = cos, sin
+
+    @property
+    def is_first_stage(self) -> bool:
+        return self.layer_lo == 0
+
+    @property
+    def is_last_stage(self) -> bool:
+        return self.layer_hi == self.spec.n_layers
+
+    # ------------------------------------------------------------- forwards
+
+    def embed(self, input_ids: torch.Tensor) -> torch.Tensor:
+        return F.embedding(input_ids, self.weights.embed)
+
+    def lm_head(self, hidden: torch.Tensor) -> torch.Tensor:
+        normed = ops.rmsnorm(hidden, self.weights.final_norm, self.spec.rms_eps)
+        return F.linear(normed, self.weights.lm_head)
+
+    def _layer(
+        self,
+        layer_idx: int,
+        hidden: torch.Tensor,
+        residual: Optional[torch.Tensor],
+        positions: torch.Tensor,
+        slot_mapping: torch.Tensor,
+        attn_fn,
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        s = self.spec
+        lw = self.weights.layers[layer_idx]
+        if residual is None:
+            residual = hidden
+            normed = ops.rmsnorm(hidden, lw.attn_norm, s.rms_eps)
+        else:
+            normed, residual = ops.fused_add_rmsnorm(
+                hidden, residual, lw.attn_norm, s.rms_eps
+            )
+
+        qkv = F.linear(normed, lw.wqkv)
+        T = qkv.shape[0]
+        q, k, v = qkv.split([s.q_size, s.kv_size, s.kv_size], dim=-1)
+        q = q.view(T, s.n_heads, s.head_dim)
+        k = k.view(T, s.n_kv_heads, s.head_dim)
+        v = v.view(T, s.n_kv_heads, s.head_dim)
+        ops.rope_inplace(q, k, positions, self.rope_cos, self.rope_sin)
+        k_cache, v_cache = self.kv.layer(layer_idx)
+        ops.kv_cache_store(k, v, k_cache, v_cache, slot_mapping)
+        attn_out = attn_fn(layer_idx, q, k, v, k_cache, v_cache)
+        attn_out = F.linear(attn_out.reshape(T, s.q_size), lw.wo)
+
+        normed, residual = ops.fused_add_rmsnorm(
+            attn_out, residual, lw.mlp_norm, s.rms_eps
+        )
+        if s.is_moe:
+            mlp_out = self._moe_mlp(lw, normed)
+        else:
+            gate_up = F.linear(normed, lw.w_gate_up)
+            mlp_out = F.linear(ops.swiglu(gate_up), lw.w_down)
+        return mlp_out, residual
+
+    def _moe_mlp(self, lw, x: torch.Tensor) -> torch.Tensor:
+        """Top-k expert MLP. v1: per-expert gather/GEMM/scatter (dense GEMMs
+        through hipBLASLt); grouped-GEMM kernel lands with the MoE milestone."""
+        s = self.spec
+        logits = F.linear(x, lw.moe_gate)
+        weights, idx = ops.moe_topk_gate(logits, s.top_k_experts)  # [T,k]
+        out = torch.zeros_like(x, dtype=torch.float32)
+        for e in range(s.n_experts):
+            mask = idx == e  # [T, k]
+            if not bool(mask.any()):
+                continue
+            tok, kslot = mask.nonzero(as_tuple=True)
+            xe = x[tok]
+            ge = F.linear(xe, lw.moe_w_gate_up[e])
+            ye = F.linear(ops.swiglu(ge), lw.moe_w_down[e])
+            out.index_add_(0, tok, ye.float() * weights[tok, kslot, None])
+        return out.to(x.dtype)
+
+    def _run_layers(
+        self,
+        hidden: torch.Tensor,
+        positions: torch.Tensor,
+        slot_mapping: torch.Tensor,
+        attn_fn,
+    ) -> torch.Tensor:
+        residual = None
+        for layer_idx in range(self.layer_lo, self.layer_hi):
+            hidden, residual = self._layer(
+                layer_idx, hidden, residual, positions, slot_mapping, attn_fn
+            )
+        return hidden + residual if residual is not None else hidden
+
+    def forward_prefill(
+        self,
+        input_ids_or_hidden: torch.Tensor,
+        positions: torch.Tensor,
+        slot_mapping: torch.Tensor,
+        cu_seqlens: torch.Tensor,
+        max_seqlen: int,
+        block_table: Optional[torch.Tensor] = None,
+        seq_lens: Optional[torch.Tensor] = None,
+        query_lens: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """Full- or chunked-prompt forward. Returns final hidden [T, H]
+        (call lm_head on the rows you need). When seq_lens/query_lens are
+        given the chunk attends to the paged history (chunked prefill)."""
+        chunked = seq_lens is not None
+
+        def attn(layer_idx, q, k, v, k_cache, v_cache):
+            if chunked:
+                from ..ops import reference as _ref  # history path: paged
+
+                if q.is_cuda and ops.hip_available():
+                    return ops.require_hip().attn_prefill_paged(
+                        q, k_cache, v_cache, block_table, seq_lens, query_lens,
+                        cu_seqlens, self.scale,
+                    )
+                return _ref.attn_decode_with_history(
+                    q, k_cache, v_cache, block_table, seq_lens, query_lens, self.scale
+                )
+            return ops.attn_prefill(q, k, v, cu_seqlens, max_seqlen, self.scale)
+
+        hidden = (
+            self.embed(input_ids_or_hidden)
+            if self.is_first_stage
+            else input_ids_or_hidden
+        )
+        return self._run_layers(hidden, positions, slot_mapping, attn)
+
+    def forward_decode(
+        self,
+        input_ids_or_hidden: torch.Tensor,
+        positions: torch.Tensor,
+        slot_mapping: torch.Tensor,
+        block_table: torch.Tensor,
+        seq_lens: torch.Tensor,
+    ) -> torch.Tensor:
+        """One-token-per-sequence step. input [B] ids (or [B, H] hidden for
+        later pipeline stages); returns final hidden [B, H]."""
+
+        def attn(layer_idx, q, k, v, k_cache, v_cache):
+            return ops.attn_decode(
+                q, k_cache, v_cache, block_table, seq_lens, self.scale
+            )
+
+        hidden = (
+            self.embed(input_ids_or_hidden)
+            if self.is_first_stage
+            else input_ids_or_hidden
+        )
+        return self._run_layers(hidden, positions, slot_mapping, attn)

@@ -1,0 +1,199 @@
+"""Pure-PyTorch fp32 reference implementations of every engine op.
+
+These are the numerics ground truth that the HIP kernels are tested against
+(tests/test_ops_gpu.py compares each CDNA4 kernel to these at fp32), and the
+CPU execution path for GPU-less environments (plumbing tests, BASELINE
+config 1). On a GPU box the engine REQUIRES the HIP extension — see
+ops/__init__.py dispatch.
+
+Conventions (shared with the HIP kernels):
+  x        [T, H]            activations (T = tokens in batch/chunk)
+  q        [T, n_heads, hd]
+  k, v     [T, n_kv, hd]
+  k_cache  [n_blocks, n_kv, block_size, hd]   paged KV pool
+  block_table [B, max_blocks] int32
+  slot_mapping [T] int32     flat slot = block_id * block_size + offset
+  cos/sin  [max_len, hd/2]   fp32 RoPE tables (host-precomputed)
+RoPE is llama-style rotate_half: pair (i, i + hd/2).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    xf = x.float()
+    norm = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (norm * weight.float()).to(x.dtype)
+
+
+def fused_add_rmsnorm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    new_residual = (x.float() + residual.float()).to(x.dtype)
+    return rmsnorm(new_residual, weight, eps), new_residual
+
+
+def rope_tables(
+    max_len: int, head_dim: int, theta: float, device, dtype=torch.float32
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    inv_freq = 1.0 / (
+        theta ** (torch.arange(0, head_dim, 2, dtype=torch.float32) / head_dim)
+    )
+    t = torch.arange(max_len, dtype=torch.float32)
+    freqs = torch.outer(t, inv_freq)  # [max_len, hd/2]
+    return freqs.cos().to(device, dtype), freqs.sin().to(device, dtype)
+
+
+def rope_inplace(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    positions: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+) -> None:
+    hd = q.shape[-1]
+    c = cos[positions].unsqueeze(1)  # [T, 1, hd/2]
+    s = sin[positions].unsqueeze(1)
+    for t in (q, k):
+        # .float() is identity for fp32 tensors — clone so the first-half
+        # write cannot alias the second-half read
+        tf = t.detach().clone().float()
+        x1, x2 = tf[..., : hd // 2], tf[..., hd // 2 :]
+        t[..., : hd // 2] = (x1 * c - x2 * s).to(t.dtype)
+        t[..., hd // 2 :] = (x2 * c + x1 * s).to(t.dtype)
+
+
+def kv_cache_store(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    n_blocks, n_kv, block_size, hd = k_cache.shape
+    blk = torch.div(slot_mapping, block_size, rounding_mode="floor").long()
+    off = (slot_mapping % block_size).long()
+    k_cache[blk, :, off, :] = k.to(k_cache.dtype)
+    v_cache[blk, :, off, :] = v.to(v_cache.dtype)
+
+
+def attn_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_table: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    """Paged single-token attention, GQA. q [B, nq, hd] -> out [B, nq, hd]."""
+    B, nq, hd = q.shape
+    _nb, n_kv, block_size, _ = k_cache.shape
+    group = nq // n_kv
+    out = torch.empty_like(q)
+    for b in range(B):
+        L = int(seq_lens[b])
+        nblk = (L + block_size - 1) // block_size
+        blocks = block_table[b, :nblk].long()
+        keys = k_cache[blocks]  # [nblk, n_kv, bs, hd]
+        vals = v_cache[blocks]
+        keys = keys.permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
+        vals = vals.permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
+        qb = q[b].float().view(n_kv, group, hd)
+        scores = torch.einsum("kgd,kld->kgl", qb, keys.float()) * scale
+        probs = torch.softmax(scores, dim=-1)
+        ob = torch.einsum("kgl,kld->kgd", probs, vals.float())
+        out[b] = ob.reshape(nq, hd).to(q.dtype)
+    return out
+
+
+def attn_prefill(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cu_seqlens: torch.Tensor,
+    max_seqlen: int,
+    scale: float,
+    causal: bool = True,
+) -> torch.Tensor:
+    """Varlen full-prompt attention, GQA. q [T, nq, hd] -> [T, nq, hd]."""
+    T, nq, hd = q.shape
+    n_kv = k.shape[1]
+    group = nq // n_kv
+    out = torch.empty_like(q)
+    for i in range(cu_seqlens.numel() - 1):
+        s, e = int(cu_seqlens[i]), int(cu_seqlens[i + 1])
+        L = e - s
+        qi = q[s:e].float().view(L, n_kv, group, hd)
+        ki = k[s:e].float()
+        vi = v[s:e].float()
+        scores = torch.einsum("qkgd,lkd->kgql", qi, ki) * scale
+        if causal:
+            mask = torch.triu(
+                torch.ones(L, L, dtype=torch.bool, device=q.device), diagonal=1
+            )
+            scores.masked_fill_(mask, float("-inf"))
+        probs = torch.softmax(scores, dim=-1)
+        oi = torch.einsum("kgql,lkd->qkgd", probs, vi)
+        out[s:e] = oi.reshape(L, nq, hd).to(q.dtype)
+    return out
+
+
+def attn_decode_with_history(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_table: torch.Tensor,
+    seq_lens: torch.Tensor,
+    query_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    """Chunked-prefill attention: multiple new query tokens per sequence
+    attending to the full paged history (new tokens already stored in cache).
+
+    q [T, nq, hd] packed by sequence; seq_lens = total length incl. the new
+    chunk; query_lens = chunk length per sequence."""
+    T, nq, hd = q.shape
+    _nb, n_kv, block_size, _ = k_cache.shape
+    group = nq // n_kv
+    out = torch.empty_like(q)
+    t0 = 0
+    for b in range(query_lens.numel()):
+        QL = int(query_lens[b])
+        L = int(seq_lens[b])
+        nblk = (L + block_size - 1) // block_size
+        blocks = block_table[b, :nblk].long()
+        keys = (
+            k_cache[blocks].permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
+        )
+        vals = (
+            v_cache[blocks].permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
+        )
+        qb = q[t0 : t0 + QL].float().view(QL, n_kv, group, hd)
+        scores = torch.einsum("qkgd,kld->kgql", qb, keys.float()) * scale
+        # causal within the chunk: query j (absolute pos L-QL+j) sees keys <= pos
+        qpos = torch.arange(L - QL, L, device=q.device).view(1, 1, QL, 1)
+        kpos = torch.arange(L, device=q.device).view(1, 1, 1, L)
+        scores.masked_fill_(kpos > qpos, float("-inf"))
+        probs = torch.softmax(scores, dim=-1)
+        ob = torch.einsum("kgql,kld->qkgd", probs, vals.float())
+        out[t0 : t0 + QL] = ob.reshape(QL, nq, hd).to(q.dtype)
+        t0 += QL
+    return out
+
+
+def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
+    g, u = gate_up.float().chunk(2, dim=-1)
+    return (torch.nn.functional.silu(g) * u).to(gate_up.dtype)
+
+
+def moe_topk_gate(
+    logits: torch.Tensor, top_k: int
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Softmax-after-topk gating (Mixtral). logits [T, E] ->
+    (weights [T, k] fp32, indices [T, k] int32)."""
+    vals, idx = torch.topk(logits.float(), top_k, dim=-1)
+    weights = torch.softmax(vals, dim=-1)
+    return weights, idx.to(torch.int32)

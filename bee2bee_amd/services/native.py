@@ -1,0 +1,131 @@
+"""NativeEngineService — the MI355X HIP engine behind the mesh service
+interface.
+
+This is the row-for-row replacement of the reference's HFService
+(bee2bee/services.py:27-116): same metadata/execute/execute_stream/load_sync
+contract and JSON-lines streaming shape, but the compute is the in-process
+InferenceEngine (CDNA4 kernels + hipGraph decode) instead of
+transformers.generate(). Fixes reference quirk Q6: `tokens` counts NEW
+tokens only (the reference re-encoded the full decoded text, prompt
+included).
+"""
+from __future__ import annotations
+
+import json
+import queue
+import threading
+import time
+from typing import Any, Dict, Iterator, Optional
+
+from .base import BaseService, ServiceError
+
+
+class NativeEngineService(BaseService):
+    def __init__(
+        self,
+        model_name: str,
+        price_per_token: float = 0.0,
+        max_new_tokens: int = 2048,
+        model_path: Optional[str] = None,
+        device: Optional[str] = None,
+        max_batch: int = 16,
+        max_seq_len: Optional[int] = None,
+        wire_name: str = "hf",
+    ) -> None:
+        # wire_name defaults to "hf" so reference peers route to us unchanged
+        super().__init__(wire_name)
+        self.model_name = model_name
+        self.price_per_token = price_per_token
+        self.max_new_tokens = max_new_tokens
+        self.model_path = model_path
+        self.device = device
+        self.max_batch = max_batch
+        self.max_seq_len = max_seq_len
+        self.engine = None
+
+    def load_sync(self) -> None:
+        try:
+            from ..engine.engine import InferenceEngine
+
+            self.engine = InferenceEngine(
+                self.model_name,
+                device=self.device,
+                model_path=self.model_path,
+                max_batch=self.max_batch,
+                max_seq_len=self.max_seq_len,
+            )
+            self.engine.start()
+        except Exception as e:
+            raise ServiceError(f"failed to load model: {e}") from e
+
+    def get_metadata(self) -> Dict[str, Any]:
+        meta = {
+            "models": [self.model_name],
+            "price_per_token": self.price_per_token,
+            "max_new_tokens": self.max_new_tokens,
+            "backend": "bee2bee-amd-native",
+        }
+        if self.engine is not None:
+            meta["arch"] = self.engine.spec.name
+            meta["device"] = str(self.engine.device)
+        return meta
+
+    def _check(self, params: Dict[str, Any]):
+        if self.engine is None:
+            raise ServiceError("Model not loaded")
+        prompt = params.get("prompt")
+        if not prompt:
+            raise ServiceError("Missing prompt")
+        max_new = int(params.get("max_new_tokens", self.max_new_tokens))
+        max_new = max(1, min(max_new, self.max_new_tokens))
+        temperature = float(params.get("temperature", 0.7))
+        return prompt, max_new, temperature
+
+    def execute(self, params: Dict[str, Any]) -> Dict[str, Any]:
+        prompt, max_new, temperature = self._check(params)
+        try:
+            t0 = time.time()
+            res = self.engine.generate_text(prompt, max_new, temperature)
+            latency_ms = int((time.time() - t0) * 1000.0)
+            tokens = res["tokens"]
+            return {
+                "text": res["text"],
+                "tokens": tokens,
+                "latency_ms": latency_ms,
+                "ttft_ms": res.get("ttft_ms"),
+                "price_per_token": self.price_per_token,
+                "cost": self.price_per_token * tokens,
+                "backend": "bee2bee-amd-native",
+            }
+        except ServiceError:
+            raise
+        except Exception as e:
+            raise ServiceError(str(e)) from e
+
+    def execute_stream(self, params: Dict[str, Any]) -> Iterator[str]:
+        prompt, max_new, temperature = self._check(params)
+        chunks: "queue.Queue" = queue.Queue()
+        DONE = object()
+
+        def _run() -> None:
+            try:
+                self.engine.generate_text(
+                    prompt, max_new, temperature, on_text=chunks.put
+                )
+                chunks.put(DONE)
+            except Exception as e:  # noqa: BLE001
+                chunks.put(e)
+
+        t = threading.Thread(target=_run, daemon=True)
+        t.start()
+        while True:
+            item = chunks.get()
+            if item is DONE:
+                break
+            if isinstance(item, Exception):
+                yield json.dumps(
+                    {"status": "error", "message": f"Stream error: {item}"}
+                ) + "\n"
+                return
+            yield json.dumps({"text": item}) + "\n"
+        yield json.dumps({"done": True}) + "\n"

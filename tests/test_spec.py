@@ -1,0 +1,54 @@
+import json
+
+from bee2bee_amd.models.spec import PRESETS, resolve_spec, spec_from_hf_config
+
+
+def test_llama3_8b_param_count():
+    spec = PRESETS["llama3-8b"]
+    # Llama-3-8B is 8.03B parameters
+    assert abs(spec.n_params() / 1e9 - 8.03) < 0.1
+
+
+def test_llama3_70b_param_count():
+    spec = PRESETS["llama3-70b"]
+    assert abs(spec.n_params() / 1e9 - 70.6) < 1.0
+
+
+def test_mixtral_param_counts():
+    spec = PRESETS["mixtral-8x7b"]
+    assert abs(spec.n_params() / 1e9 - 46.7) < 1.0
+    # ~12.9B active per token (2 of 8 experts)
+    assert abs(spec.active_params_per_token() / 1e9 - 12.9) < 0.5
+
+
+def test_alias_resolution():
+    assert resolve_spec("meta-llama/Meta-Llama-3-8B").name == "llama3-8b"
+    assert resolve_spec("HuggingFaceH4/zephyr-7b-beta").name == "zephyr-7b"
+    assert resolve_spec("mistralai/Mixtral-8x7B-v0.1").is_moe
+
+
+def test_unknown_falls_back_to_demo():
+    spec = resolve_spec("distilgpt2")
+    assert spec.hidden_size == 768  # demo spec under the requested name
+    assert spec.name == "distilgpt2"
+
+
+def test_hf_config_parsing(tmp_path):
+    cfg = {
+        "vocab_size": 32000,
+        "hidden_size": 4096,
+        "intermediate_size": 14336,
+        "num_hidden_layers": 32,
+        "num_attention_heads": 32,
+        "num_key_value_heads": 8,
+        "rope_theta": 10000.0,
+        "rms_norm_eps": 1e-5,
+        "max_position_embeddings": 32768,
+        "num_local_experts": 8,
+        "num_experts_per_tok": 2,
+    }
+    (tmp_path / "config.json").write_text(json.dumps(cfg))
+    spec = spec_from_hf_config(str(tmp_path))
+    assert spec.n_experts == 8 and spec.head_dim == 128
+    spec2 = resolve_spec("whatever", model_path=str(tmp_path))
+    assert spec2.n_experts == 8

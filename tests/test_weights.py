@@ -1,0 +1,54 @@
+import torch
+
+from bee2bee_amd.models.spec import PRESETS
+from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+
+def test_random_init_shapes():
+    spec = PRESETS["tiny"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init()
+    assert w.embed.shape == (spec.vocab_size, spec.hidden_size)
+    assert w.lm_head is w.embed  # tied
+    lw = w.layers[0]
+    assert lw.wqkv.shape == (spec.q_size + 2 * spec.kv_size, spec.hidden_size)
+    assert lw.w_gate_up.shape == (2 * spec.intermediate_size, spec.hidden_size)
+
+
+def test_hf_roundtrip(tmp_path):
+    """save_hf (unfused HF names) -> load_hf (refused) must be identical —
+    the checkpoint-compatibility guarantee."""
+    spec = PRESETS["tiny"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(seed=3)
+    save_hf(w, str(tmp_path))
+    assert (tmp_path / "model.safetensors").exists()
+    assert (tmp_path / "config.json").exists()
+
+    w2 = ModelWeights(spec, torch.device("cpu"), torch.float32).load_hf(str(tmp_path))
+    assert torch.equal(w.embed, w2.embed)
+    for a, b in zip(w.layers, w2.layers):
+        assert torch.equal(a.wqkv, b.wqkv)
+        assert torch.equal(a.wo, b.wo)
+        assert torch.equal(a.w_gate_up, b.w_gate_up)
+        assert torch.equal(a.w_down, b.w_down)
+
+
+def test_moe_roundtrip(tmp_path):
+    spec = PRESETS["tiny-moe"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(seed=5)
+    save_hf(w, str(tmp_path))
+    w2 = ModelWeights(spec, torch.device("cpu"), torch.float32).load_hf(str(tmp_path))
+    for a, b in zip(w.layers, w2.layers):
+        assert torch.equal(a.moe_gate, b.moe_gate)
+        assert torch.equal(a.moe_w_gate_up, b.moe_w_gate_up)
+        assert torch.equal(a.moe_w_down, b.moe_w_down)
+
+
+def test_layer_range_partial_load():
+    spec = PRESETS["tiny"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(
+        layer_range=(1, 2)
+    )
+    assert w.layers[0].wqkv is None
+    assert w.layers[1].wqkv is not None
+    assert w.embed is None  # not first stage
+    assert w.final_norm is not None  # last stage
