@@ -435,6 +435,75 @@ class InferenceEngine:
         self._active = saved
 
     @torch.no_grad()
+    def bench_setup(self, batch: int, prompt_len: int, steps_budget: int,
+                    seed: int = 1234) -> None:
+        """Prefill a synthetic batch and preallocate KV for the whole run so
+        the timed loop is pure device work (graph replay + sampling + buffer
+        bumps; zero host round-trips per step)."""
+        assert batch <= self.max_batch
+        seq_ids = self.bench_prefill(batch, prompt_len, seed=seed)
+        for sid in seq_ids:
+            self.kv.extend_seq(sid, prompt_len + steps_budget + 1)
+        dev = self.device
+        acts = self._bench_acts
+        first = torch.tensor(
+            [a.req.output_ids[-1] for a in acts], dtype=torch.int64, device=dev
+        )
+        bt = self.kv.block_table([a.seq_id for a in acts])
+        if self.graphs is not None:
+            g = self.graphs
+            assert g.bucket_for(batch) == batch or batch in g.buckets(), (
+                "bench batch should be a graph bucket"
+            )
+            g.input_ids[:batch].copy_(first)
+            g.positions[:batch].fill_(prompt_len)
+            g.seq_lens[:batch].fill_(prompt_len + 1)
+            g.block_table[:batch, : bt.shape[1]].copy_(bt)
+            bucket = g.bucket_for(batch)
+            if bucket > batch:
+                sb = self.kv.block_table([self._scratch_seq])[0]
+                g.block_table[batch:bucket, : sb.shape[0]].copy_(sb)
+                g.positions[batch:bucket].zero_()
+                g.seq_lens[batch:bucket].fill_(1)
+                g.input_ids[batch:bucket].zero_()
+            self._bench_state = ("graph", batch)
+            self.graphs.run(batch)  # capture outside the timed region
+            g.positions[:batch].fill_(prompt_len)
+            g.seq_lens[:batch].fill_(prompt_len + 1)
+            g.input_ids[:batch].copy_(first)
+        else:
+            self._bench_eager = {
+                "ids": first.clone(),
+                "pos": torch.full((batch,), prompt_len, dtype=torch.int32, device=dev),
+                "lens": torch.full((batch,), prompt_len + 1, dtype=torch.int32, device=dev),
+                "bt": bt,
+            }
+            self._bench_state = ("eager", batch)
+
+    @torch.no_grad()
+    def bench_step(self) -> None:
+        """One fully device-side greedy decode step over the bench batch."""
+        mode, B = self._bench_state
+        if mode == "graph":
+            g = self.graphs
+            logits = g.run(B)[:B]
+            next_ids = torch.argmax(logits, dim=-1)
+            g.input_ids[:B].copy_(next_ids)
+            g.positions[:B].add_(1)
+            g.seq_lens[:B].add_(1)
+        else:
+            st = self._bench_eager
+            slots = decode_slot_mapping(st["bt"], st["pos"], self.kv.block_size)
+            hidden = self.runner.forward_decode(
+                st["ids"], st["pos"], slots, st["bt"], st["lens"]
+            )
+            logits = self.runner.lm_head(hidden)
+            st["ids"].copy_(torch.argmax(logits, dim=-1))
+            st["pos"].add_(1)
+            st["lens"].add_(1)
+        self.total_tokens += B
+
+    @torch.no_grad()
     def bench_decode_step(self) -> torch.Tensor:
         """One decode step over the bench batch; returns sampled ids [B]."""
         acts = self._bench_acts

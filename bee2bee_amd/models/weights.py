@@ -69,11 +69,14 @@ class ModelWeights:
         embed only on the first stage, head/final norm only on the last."""
         s = self.spec
         lo, hi = layer_range or (0, s.n_layers)
-        gen = torch.Generator(device="cpu").manual_seed(seed)
+        # generate directly on the target device (an 8B randn on host would
+        # serialize startup; device-side randn fills 16 GB in ~a second)
+        gen = torch.Generator(device=self.device).manual_seed(seed)
 
         def rnd(*shape: int, std: float) -> torch.Tensor:
-            t = torch.randn(*shape, generator=gen, dtype=torch.float32)
-            return (t * std).to(self.device, self.dtype)
+            t = torch.randn(*shape, generator=gen, dtype=torch.float32,
+                            device=self.device)
+            return (t * std).to(self.dtype)
 
         std = 0.02
         proj_std = std / max(1.0, (2 * s.n_layers) ** 0.5)

@@ -1,0 +1,200 @@
+// Memory-bound fused kernels: RMSNorm, fused add+RMSNorm, RoPE, SwiGLU,
+// paged-KV store. All bf16 I/O vectorized as short8 (16 B/lane — guide G13:
+// hipcc does not auto-vectorize bf16 loads; scalar bf16 is ~2x slower).
+//
+// These replace what the reference ran inside transformers.generate()
+// (bee2bee/hf.py:84-108) — each is one HBM round-trip, fused.
+#include "common.h"
+
+// ---------------------------------------------------------------- rmsnorm
+// One workgroup per row. Row cached in LDS as fp32 between the two passes
+// so the input is read from HBM exactly once.
+template <bool FUSED_ADD>
+__global__ void rmsnorm_kernel(
+    const unsigned short* __restrict__ x,      // [T, H]
+    const unsigned short* __restrict__ resid,  // [T, H] or null
+    const unsigned short* __restrict__ w,      // [H]
+    unsigned short* __restrict__ y,            // [T, H]
+    unsigned short* __restrict__ resid_out,    // [T, H] or null
+    int H, float eps) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    float* row = reinterpret_cast<float*>(smem_raw);       // [H]
+    __shared__ float red[8];
+
+    const long t = blockIdx.x;
+    const unsigned short* xr = x + t * (long)H;
+    const unsigned short* rr = FUSED_ADD ? resid + t * (long)H : nullptr;
+    unsigned short* ro = FUSED_ADD ? resid_out + t * (long)H : nullptr;
+
+    float sumsq = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        float v[8];
+        load_bf16x8(xr + i, v);
+        if (FUSED_ADD) {
+            float r[8];
+            load_bf16x8(rr + i, r);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) v[j] += r[j];
+            store_bf16x8(ro + i, v);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            row[i + j] = v[j];
+            sumsq += v[j] * v[j];
+        }
+    }
+    // block reduction: wave-level then cross-wave through LDS
+    sumsq = wave_sum(sumsq);
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = sumsq;
+    __syncthreads();
+    const int nw = blockDim.x / WAVE;
+    float total = 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) total += (i < nw) ? red[i] : 0.f;
+    const float inv = rsqrtf(total / (float)H + eps);
+
+    unsigned short* yr = y + t * (long)H;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        float wv[8], o[8];
+        load_bf16x8(w + i, wv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o[j] = row[i + j] * inv * wv[j];
+        store_bf16x8(yr + i, o);
+    }
+}
+
+// ------------------------------------------------------------------- rope
+// Llama rotate-half RoPE, in place on strided q/k views of the fused qkv
+// projection. One wave per (token, head); lane owns pair (d, d+hd/2).
+// cos/sin tables are host-precomputed fp32 (guide App. B: no device trig).
+__global__ void rope_kernel(
+    unsigned short* __restrict__ q,  // [T, nq, hd], token stride sq
+    unsigned short* __restrict__ k,  // [T, nkv, hd], token stride sk
+    const int* __restrict__ pos,     // [T]
+    const float* __restrict__ cos_t, // [max_len, hd/2]
+    const float* __restrict__ sin_t,
+    int T, int nq, int nkv, int hd, long sq, long sk) {
+    const int glob_wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int heads = nq + nkv;
+    if (glob_wave >= T * heads) return;
+    const int t = glob_wave / heads;
+    const int h = glob_wave % heads;
+    unsigned short* base = (h < nq) ? q + (long)t * sq + (long)h * hd
+                                    : k + (long)t * sk + (long)(h - nq) * hd;
+    const int half = hd / 2;
+    const long tab = (long)pos[t] * half;
+    for (int d = lane; d < half; d += WAVE) {
+        float c = cos_t[tab + d];
+        float s = sin_t[tab + d];
+        float x1 = bf2f(base[d]);
+        float x2 = bf2f(base[d + half]);
+        base[d] = f2bf(x1 * c - x2 * s);
+        base[d + half] = f2bf(x2 * c + x1 * s);
+    }
+}
+
+// ---------------------------------------------------------- kv_cache_store
+// Scatter new K/V rows into the paged pool. One wave per (token, kv head).
+__global__ void kv_store_kernel(
+    const unsigned short* __restrict__ k,  // [T, nkv, hd], token stride sk
+    const unsigned short* __restrict__ v,
+    unsigned short* __restrict__ k_cache,  // [nb, nkv, bs, hd]
+    unsigned short* __restrict__ v_cache,
+    const int* __restrict__ slots,         // [T]
+    int T, int nkv, int hd, int bs, long sk, long sv) {
+    const int glob_wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (glob_wave >= T * nkv) return;
+    const int t = glob_wave / nkv;
+    const int h = glob_wave % nkv;
+    const int slot = slots[t];
+    const int blk = slot / bs, off = slot % bs;
+    const long dst = (((long)blk * nkv + h) * bs + off) * hd;
+    const unsigned short* ks = k + (long)t * sk + (long)h * hd;
+    const unsigned short* vs = v + (long)t * sv + (long)h * hd;
+    for (int d = lane * 2; d + 1 < hd; d += WAVE * 2) {
+        *reinterpret_cast<short2v*>(k_cache + dst + d) =
+            *reinterpret_cast<const short2v*>(ks + d);
+        *reinterpret_cast<short2v*>(v_cache + dst + d) =
+            *reinterpret_cast<const short2v*>(vs + d);
+    }
+    if (hd % 2) { /* head dims are even for all supported models */ }
+}
+
+// ----------------------------------------------------------------- swiglu
+// out[t, i] = silu(gu[t, i]) * gu[t, I + i]; grid-stride, short8 loads.
+__global__ void swiglu_kernel(
+    const unsigned short* __restrict__ gu,  // [T, 2I]
+    unsigned short* __restrict__ out,       // [T, I]
+    long T, long I) {
+    const long total = T * I / 8;
+    for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+         idx += (long)gridDim.x * blockDim.x) {
+        const long t = idx / (I / 8);
+        const long i = (idx % (I / 8)) * 8;
+        float g[8], u[8], o[8];
+        load_bf16x8(gu + t * 2 * I + i, g);
+        load_bf16x8(gu + t * 2 * I + I + i, u);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float sig = 1.f / (1.f + __expf(-g[j]));
+            o[j] = g[j] * sig * u[j];
+        }
+        store_bf16x8(out + t * I + i, o);
+    }
+}
+
+// ------------------------------------------------------------- launchers
+extern "C" {
+
+void launch_rmsnorm(const unsigned short* x, const unsigned short* w,
+                    unsigned short* y, long T, int H, float eps,
+                    hipStream_t stream) {
+    const int smem = H * 4;
+    hipLaunchKernelGGL((rmsnorm_kernel<false>), dim3(T), dim3(256), smem,
+                       stream, x, nullptr, w, y, nullptr, H, eps);
+}
+
+void launch_fused_add_rmsnorm(const unsigned short* x,
+                              const unsigned short* resid,
+                              const unsigned short* w, unsigned short* y,
+                              unsigned short* resid_out, long T, int H,
+                              float eps, hipStream_t stream) {
+    const int smem = H * 4;
+    hipLaunchKernelGGL((rmsnorm_kernel<true>), dim3(T), dim3(256), smem,
+                       stream, x, resid, w, y, resid_out, H, eps);
+}
+
+void launch_rope(unsigned short* q, unsigned short* k, const int* pos,
+                 const float* cos_t, const float* sin_t, int T, int nq,
+                 int nkv, int hd, long sq, long sk, hipStream_t stream) {
+    const long waves = (long)T * (nq + nkv);
+    const long blocks = (waves + 3) / 4;
+    hipLaunchKernelGGL(rope_kernel, dim3(blocks), dim3(256), 0, stream, q, k,
+                       pos, cos_t, sin_t, T, nq, nkv, hd, sq, sk);
+}
+
+void launch_kv_store(const unsigned short* k, const unsigned short* v,
+                     unsigned short* k_cache, unsigned short* v_cache,
+                     const int* slots, int T, int nkv, int hd, int bs,
+                     long sk, long sv, hipStream_t stream) {
+    const long waves = (long)T * nkv;
+    const long blocks = (waves + 3) / 4;
+    hipLaunchKernelGGL(kv_store_kernel, dim3(blocks), dim3(256), 0, stream, k,
+                       v, k_cache, v_cache, slots, T, nkv, hd, bs, sk, sv);
+}
+
+void launch_swiglu(const unsigned short* gu, unsigned short* out, long T,
+                   long I, hipStream_t stream) {
+    const long total = T * I / 8;
+    long blocks = (total + 255) / 256;
+    if (blocks > 2048) blocks = 2048;  // grid-stride (guide G11)
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(swiglu_kernel, dim3(blocks), dim3(256), 0, stream, gu,
+                       out, T, I);
+}
+
+}  // extern "C"

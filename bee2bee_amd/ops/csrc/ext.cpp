@@ -1,0 +1,204 @@
+// Torch bindings for the bee2bee_amd CDNA4 kernel set.
+//
+// Validation lives here (shape/dtype/device checks) so the kernels stay
+// branch-free. All entry points are hipGraph-capture safe: no allocation
+// beyond torch's caching allocator, no synchronization, no host reads.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+using torch::Tensor;
+
+extern "C" {
+void launch_rmsnorm(const unsigned short*, const unsigned short*,
+                    unsigned short*, long, int, float, hipStream_t);
+void launch_fused_add_rmsnorm(const unsigned short*, const unsigned short*,
+                              const unsigned short*, unsigned short*,
+                              unsigned short*, long, int, float, hipStream_t);
+void launch_rope(unsigned short*, unsigned short*, const int*, const float*,
+                 const float*, int, int, int, int, long, long, hipStream_t);
+void launch_kv_store(const unsigned short*, const unsigned short*,
+                     unsigned short*, unsigned short*, const int*, int, int,
+                     int, int, long, long, hipStream_t);
+void launch_swiglu(const unsigned short*, unsigned short*, long, long,
+                   hipStream_t);
+void launch_attn_decode(const unsigned short*, const unsigned short*,
+                        const unsigned short*, const int*, const int*,
+                        unsigned short*, int, int, int, int, int, int, long,
+                        float, hipStream_t);
+void launch_attn_prefill(const unsigned short*, const unsigned short*,
+                         const unsigned short*, const int*, unsigned short*,
+                         int, int, int, int, long, long, long, int, float,
+                         int, hipStream_t);
+void launch_mfma_probe(const unsigned short*, const unsigned short*, float*,
+                       hipStream_t);
+}
+
+namespace {
+
+inline hipStream_t stream() {
+    return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+inline const unsigned short* bf16p(const Tensor& t) {
+    return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+
+inline unsigned short* bf16p_mut(Tensor& t) {
+    return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+void check_bf16(const Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+
+Tensor rmsnorm(const Tensor& x, const Tensor& w, double eps) {
+    check_bf16(x, "x");
+    check_bf16(w, "w");
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+    const long T = x.numel() / x.size(-1);
+    const int H = x.size(-1);
+    TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+    Tensor y = torch::empty_like(x);
+    launch_rmsnorm(bf16p(x), bf16p(w), bf16p_mut(y), T, H, (float)eps,
+                   stream());
+    return y;
+}
+
+std::vector<Tensor> fused_add_rmsnorm(const Tensor& x, const Tensor& resid,
+                                      const Tensor& w, double eps) {
+    check_bf16(x, "x");
+    check_bf16(resid, "resid");
+    TORCH_CHECK(x.is_contiguous() && resid.is_contiguous() && w.is_contiguous());
+    TORCH_CHECK(x.sizes() == resid.sizes());
+    const long T = x.numel() / x.size(-1);
+    const int H = x.size(-1);
+    TORCH_CHECK(H % 8 == 0);
+    Tensor y = torch::empty_like(x);
+    Tensor resid_out = torch::empty_like(x);
+    launch_fused_add_rmsnorm(bf16p(x), bf16p(resid), bf16p(w), bf16p_mut(y),
+                             bf16p_mut(resid_out), T, H, (float)eps, stream());
+    return {y, resid_out};
+}
+
+void rope_inplace(Tensor& q, Tensor& k, const Tensor& pos, const Tensor& cos_t,
+                  const Tensor& sin_t) {
+    check_bf16(q, "q");
+    check_bf16(k, "k");
+    TORCH_CHECK(pos.scalar_type() == torch::kInt32);
+    TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32);
+    const int T = q.size(0);
+    const int nq = q.size(1), nkv = k.size(1), hd = q.size(2);
+    TORCH_CHECK(k.size(2) == hd && k.size(0) == T);
+    // heads/dims contiguous within a token; token stride may be larger
+    TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == hd);
+    TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == hd);
+    TORCH_CHECK(cos_t.size(1) == hd / 2 && cos_t.is_contiguous() &&
+                sin_t.is_contiguous());
+    launch_rope(bf16p_mut(q), bf16p_mut(k), pos.data_ptr<int>(),
+                cos_t.data_ptr<float>(), sin_t.data_ptr<float>(), T, nq, nkv,
+                hd, q.stride(0), k.stride(0), stream());
+}
+
+void kv_cache_store(const Tensor& k, const Tensor& v, Tensor& k_cache,
+                    Tensor& v_cache, const Tensor& slots) {
+    check_bf16(k, "k");
+    check_bf16(k_cache, "k_cache");
+    TORCH_CHECK(slots.scalar_type() == torch::kInt32);
+    const int T = k.size(0), nkv = k.size(1), hd = k.size(2);
+    const int bs = k_cache.size(2);
+    TORCH_CHECK(k_cache.size(1) == nkv && k_cache.size(3) == hd);
+    TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+    TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == hd);
+    TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == hd);
+    TORCH_CHECK(hd % 2 == 0);
+    launch_kv_store(bf16p(k), bf16p(v), bf16p_mut(k_cache), bf16p_mut(v_cache),
+                    slots.data_ptr<int>(), T, nkv, hd, bs, k.stride(0),
+                    v.stride(0), stream());
+}
+
+Tensor swiglu(const Tensor& gu) {
+    check_bf16(gu, "gate_up");
+    TORCH_CHECK(gu.is_contiguous());
+    const long I = gu.size(-1) / 2;
+    const long T = gu.numel() / gu.size(-1);
+    TORCH_CHECK(I % 8 == 0);
+    Tensor out = torch::empty({gu.size(0), I}, gu.options());
+    launch_swiglu(bf16p(gu), bf16p_mut(out), T, I, stream());
+    return out;
+}
+
+Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
+                   const Tensor& v_cache, const Tensor& block_table,
+                   const Tensor& seq_lens, double scale) {
+    check_bf16(q, "q");
+    check_bf16(k_cache, "k_cache");
+    TORCH_CHECK(block_table.scalar_type() == torch::kInt32 &&
+                seq_lens.scalar_type() == torch::kInt32);
+    TORCH_CHECK(block_table.is_contiguous());
+    const int B = q.size(0), nq = q.size(1), hd = q.size(2);
+    const int nkv = k_cache.size(1), bs = k_cache.size(2);
+    const int W = block_table.size(1);
+    TORCH_CHECK(hd <= 128 && hd % 2 == 0, "decode kernel supports hd<=128");
+    TORCH_CHECK(nq % nkv == 0);
+    const int G = nq / nkv;
+    TORCH_CHECK(G == 1 || G == 2 || G == 4 || G == 8 || G == 16,
+                "unsupported GQA group size ", G);
+    TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == hd);
+    Tensor out = torch::empty({B, nq, hd}, q.options());
+    launch_attn_decode(bf16p(q), bf16p(k_cache), bf16p(v_cache),
+                       block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                       bf16p_mut(out), B, nkv, G, W, bs, hd, q.stride(0),
+                       (float)scale, stream());
+    return out;
+}
+
+Tensor attn_prefill(const Tensor& q, const Tensor& k, const Tensor& v,
+                    const Tensor& cu_seqlens, long max_seqlen, double scale,
+                    bool causal) {
+    check_bf16(q, "q");
+    check_bf16(k, "k");
+    TORCH_CHECK(cu_seqlens.scalar_type() == torch::kInt32 &&
+                cu_seqlens.is_contiguous());
+    const int T = q.size(0), nq = q.size(1), hd = q.size(2);
+    const int nkv = k.size(1);
+    const int nseq = cu_seqlens.size(0) - 1;
+    TORCH_CHECK(nq % nkv == 0);
+    TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == hd);
+    TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == hd);
+    TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == hd);
+    Tensor out = torch::empty({T, nq, hd}, q.options());
+    launch_attn_prefill(bf16p(q), bf16p(k), bf16p(v),
+                        cu_seqlens.data_ptr<int>(), bf16p_mut(out), nseq, nq,
+                        nkv, hd, q.stride(0), k.stride(0), v.stride(0),
+                        (int)max_seqlen, (float)scale, causal ? 1 : 0,
+                        stream());
+    return out;
+}
+
+Tensor mfma_probe(const Tensor& A, const Tensor& B) {
+    check_bf16(A, "A");
+    check_bf16(B, "B");
+    TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}));
+    TORCH_CHECK(B.sizes() == torch::IntArrayRef({32, 16}));
+    TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+    Tensor C = torch::empty({16, 16},
+                            A.options().dtype(torch::kFloat32));
+    launch_mfma_probe(bf16p(A), bf16p(B), C.data_ptr<float>(), stream());
+    return C;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16)");
+    m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+          "residual += x; y = rmsnorm(residual)");
+    m.def("rope_inplace", &rope_inplace, "llama RoPE in place");
+    m.def("kv_cache_store", &kv_cache_store, "paged KV scatter");
+    m.def("swiglu", &swiglu, "silu(g) * u");
+    m.def("attn_decode", &attn_decode, "paged GQA decode attention");
+    m.def("attn_prefill", &attn_prefill, "varlen causal flash prefill");
+    m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+}

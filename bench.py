@@ -1,0 +1,137 @@
+"""Flagship benchmark: Llama-3-8B bf16 greedy decode throughput (output
+tokens/sec) on N MI355X GPUs, one engine/peer per GPU (replica mesh — weak
+scaling, matching BASELINE.json's "Llama-3-8B across 1/2/4/8 peers").
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched via torch.distributed.run with one rank per GPU over RCCL. Does W
+untimed warmup steps, times exactly K steps bracketed by barrier +
+torch.cuda.synchronize() on both sides, takes the MAX elapsed over ranks,
+and rank 0 prints ONE JSON line.
+
+A step = one decode iteration of a fixed per-GPU batch (default 64 seqs at
+prompt length 1024): full layer stack on the HIP kernels via hipGraph
+replay + greedy sampling + paged-KV append. Weights are random-init of the
+real architecture; prompts synthetic (no network for checkpoints).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=32)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--batch", type=int, default=64)
+    ap.add_argument("--prompt-len", type=int, default=1024)
+    ap.add_argument("--no-graphs", action="store_true")
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = world_size if world_size > 1 else args.gpus
+
+    on_gpu = torch.cuda.is_available()
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(
+            backend="nccl" if on_gpu else "gloo",
+            rank=rank,
+            world_size=world_size,
+        )
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+    else:
+        device = "cpu"
+
+    from bee2bee_amd.engine.engine import InferenceEngine
+
+    budget = args.warmup + args.steps + 4
+    engine = InferenceEngine(
+        args.model,
+        device=device,
+        max_batch=args.batch,
+        max_seq_len=args.prompt_len + budget + 1,
+        use_graphs=(not args.no_graphs) and on_gpu,
+        seed=1234 + rank,
+    )
+    spec = engine.spec
+
+    engine.bench_setup(args.batch, args.prompt_len, budget)
+
+    def sync() -> None:
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        engine.bench_step()
+    sync()
+
+    if dist:
+        dist.barrier()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        engine.bench_step()
+    sync()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_tps = args.batch * args.steps * n_gpus / elapsed
+
+    if rank == 0:
+        result = {
+            "metric": "output tokens/sec (Llama-3-8B bf16 greedy decode, "
+                      "replica peers)",
+            "value": round(total_tps, 1),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16" if on_gpu else "fp32-cpu-fallback",
+            "data": "synthetic",
+            "config": {
+                "model": spec.name,
+                "global_batch": args.batch * n_gpus,
+                "seq_len": args.prompt_len,
+                "parallelism": f"replica-dp{n_gpus}",
+                "decode_graphs": bool(engine.graphs is not None),
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    engine.shutdown()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,186 @@
+"""Numerics tests: each CDNA4 HIP kernel vs the plain-PyTorch fp32 reference
+(ops/reference.py) on identical bf16 inputs. Run on an MI355X via gpurun."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():  # allow collection on CPU boxes
+    pytest.skip("requires MI355X", allow_module_level=True)
+
+from bee2bee_amd import ops
+from bee2bee_amd.ops import reference as R
+
+DEV = "cuda:0"
+HIP = ops.require_hip()  # loud failure if the extension didn't load
+
+
+def assert_close(a: torch.Tensor, b: torch.Tensor, atol=2e-2, rtol=2e-2, msg=""):
+    a = a.float().cpu()
+    b = b.float().cpu()
+    diff = (a - b).abs()
+    tol = atol + rtol * b.abs()
+    bad = diff > tol
+    assert not bool(bad.any()), (
+        f"{msg}: {int(bad.sum())}/{bad.numel()} elements out of tolerance; "
+        f"max diff {float(diff.max()):.4f}"
+    )
+
+
+def test_extension_is_native():
+    """The loaded module must be the in-tree .so (not a silent fallback)."""
+    assert "_bee2bee_hip" in HIP.__file__
+    assert "bee2bee_amd/ops" in HIP.__file__
+
+
+def test_mfma_fragment_map():
+    """Hardware check of the A/B/C fragment layouts assumed by the prefill
+    kernel — asymmetric operands so a transposed C-write cannot pass."""
+    torch.manual_seed(0)
+    A = (torch.randn(16, 32) * 0.5).bfloat16().to(DEV)
+    B = (torch.arange(32 * 16).reshape(32, 16).float() * 0.01 - 2.0).bfloat16().to(DEV)
+    C = HIP.mfma_probe(A, B)
+    expect = A.float() @ B.float()
+    assert_close(C, expect, atol=5e-2, rtol=5e-2, msg="mfma 16x16x32 layout")
+
+
+@pytest.mark.parametrize("T,H", [(1, 64), (7, 4096), (64, 4096), (33, 8192), (5, 14336)])
+def test_rmsnorm(T, H):
+    torch.manual_seed(1)
+    x = torch.randn(T, H, device=DEV).bfloat16()
+    w = torch.randn(H, device=DEV).bfloat16()
+    got = ops.rmsnorm(x, w, 1e-5)
+    ref = R.rmsnorm(x, w, 1e-5)
+    assert_close(got, ref, msg=f"rmsnorm {T}x{H}")
+
+
+@pytest.mark.parametrize("T,H", [(16, 4096), (3, 2048)])
+def test_fused_add_rmsnorm(T, H):
+    torch.manual_seed(2)
+    x = torch.randn(T, H, device=DEV).bfloat16()
+    r = torch.randn(T, H, device=DEV).bfloat16()
+    w = torch.randn(H, device=DEV).bfloat16()
+    got_y, got_r = ops.fused_add_rmsnorm(x, r, w, 1e-5)
+    ref_y, ref_r = R.fused_add_rmsnorm(x, r, w, 1e-5)
+    assert_close(got_r, ref_r, msg="residual")
+    assert_close(got_y, ref_y, msg="normed")
+
+
+@pytest.mark.parametrize("nq,nkv,hd", [(32, 8, 128), (32, 8, 64), (4, 2, 16)])
+def test_rope_strided(nq, nkv, hd):
+    """RoPE applied in place on strided views of a fused qkv tensor."""
+    torch.manual_seed(3)
+    T = 9
+    qkv = torch.randn(T, (nq + 2 * nkv) * hd, device=DEV).bfloat16()
+    qkv_ref = qkv.clone()
+    cos, sin = R.rope_tables(256, hd, 500000.0, DEV)
+    pos = torch.randint(0, 250, (T,), device=DEV, dtype=torch.int32)
+
+    def views(t):
+        q, k, _v = t.split([nq * hd, nkv * hd, nkv * hd], dim=-1)
+        return q.view(T, nq, hd), k.view(T, nkv, hd)
+
+    q, k = views(qkv)
+    ops.rope_inplace(q, k, pos, cos, sin)
+    qr, kr = views(qkv_ref)
+    R.rope_inplace(qr, kr, pos, cos, sin)
+    assert_close(q, qr, msg="rope q")
+    assert_close(k, kr, msg="rope k")
+
+
+def test_kv_cache_store():
+    torch.manual_seed(4)
+    T, nkv, hd, bs, nb = 50, 8, 128, 32, 12
+    qkv = torch.randn(T, 2 * nkv * hd, device=DEV).bfloat16()
+    k = qkv[:, : nkv * hd].view(T, nkv, hd)
+    v = qkv[:, nkv * hd :].view(T, nkv, hd)
+    slots = torch.randperm(nb * bs, device=DEV)[:T].to(torch.int32)
+    kc = torch.zeros(nb, nkv, bs, hd, device=DEV).bfloat16()
+    vc = torch.zeros_like(kc)
+    kc_ref, vc_ref = kc.clone(), vc.clone()
+    ops.kv_cache_store(k, v, kc, vc, slots)
+    R.kv_cache_store(k, v, kc_ref, vc_ref, slots)
+    assert torch.equal(kc, kc_ref)
+    assert torch.equal(vc, vc_ref)
+
+
+@pytest.mark.parametrize("T,I", [(4, 14336), (17, 128)])
+def test_swiglu(T, I):
+    torch.manual_seed(5)
+    gu = torch.randn(T, 2 * I, device=DEV).bfloat16()
+    got = ops.swiglu(gu)
+    ref = R.swiglu(gu)
+    assert_close(got, ref, msg="swiglu")
+
+
+@pytest.mark.parametrize(
+    "B,nkv,G,hd,bs,maxlen",
+    [
+        (3, 8, 4, 128, 32, 500),   # llama3-8b shape
+        (2, 8, 8, 128, 32, 300),   # llama3-70b shape
+        (2, 8, 4, 64, 32, 129),    # llama3.2-1b shape
+        (2, 2, 2, 16, 32, 70),     # tiny
+        (1, 1, 1, 128, 32, 33),
+    ],
+)
+def test_attn_decode(B, nkv, G, hd, bs, maxlen):
+    torch.manual_seed(6)
+    nq = nkv * G
+    lens = torch.randint(1, maxlen + 1, (B,), dtype=torch.int32)
+    lens[0] = maxlen
+    W = (maxlen + bs - 1) // bs
+    nb = B * W + 1
+    perm = torch.randperm(nb - 1) + 1  # block 0 unused: catches offset bugs
+    block_table = perm[: B * W].reshape(B, W).to(torch.int32).to(DEV)
+    kc = torch.randn(nb, nkv, bs, hd, device=DEV).bfloat16()
+    vc = torch.randn(nb, nkv, bs, hd, device=DEV).bfloat16()
+    q = torch.randn(B, nq, hd, device=DEV).bfloat16()
+    scale = hd**-0.5
+    lens_dev = lens.to(DEV)
+    got = ops.attn_decode(q, kc, vc, block_table, lens_dev, scale)
+    ref = R.attn_decode(q, kc, vc, block_table, lens_dev, scale)
+    assert_close(got, ref, msg=f"attn_decode B{B} G{G} hd{hd}")
+
+
+@pytest.mark.parametrize(
+    "lens,nq,nkv,hd",
+    [
+        ([70, 200], 32, 8, 128),   # mfma path, llama3-8b shape
+        ([64], 32, 8, 128),        # exactly one tile
+        ([65], 4, 4, 128),         # tile boundary +1
+        ([10, 300, 1], 32, 8, 64), # mfma hd=64
+        ([33, 50], 4, 2, 16),      # basic fallback path
+    ],
+)
+def test_attn_prefill(lens, nq, nkv, hd):
+    torch.manual_seed(7)
+    T = sum(lens)
+    cu_list = [0]
+    for ln in lens:
+        cu_list.append(cu_list[-1] + ln)
+    cu = torch.tensor(cu_list, dtype=torch.int32).to(DEV)
+    qkv = torch.randn(T, (nq + 2 * nkv) * hd, device=DEV).bfloat16() * 0.5
+    q = qkv[:, : nq * hd].view(T, nq, hd)
+    k = qkv[:, nq * hd : (nq + nkv) * hd].view(T, nkv, hd)
+    v = qkv[:, (nq + nkv) * hd :].view(T, nkv, hd)
+    scale = hd**-0.5
+    got = ops.attn_prefill(q, k, v, cu, max(lens), scale, True)
+    ref = R.attn_prefill(q, k, v, cu, max(lens), scale, True)
+    assert_close(got, ref, atol=3e-2, rtol=3e-2, msg=f"prefill {lens} hd{hd}")
+
+
+def test_attn_decode_long_context():
+    """Multi-chunk online softmax across many 256-key passes."""
+    torch.manual_seed(8)
+    B, nkv, G, hd, bs = 2, 4, 4, 128, 32
+    L = 2048
+    W = L // bs
+    nb = B * W + 1
+    bt = (torch.arange(1, B * W + 1).reshape(B, W)).to(torch.int32).to(DEV)
+    kc = torch.randn(nb, nkv, bs, hd, device=DEV).bfloat16()
+    vc = torch.randn(nb, nkv, bs, hd, device=DEV).bfloat16()
+    q = torch.randn(B, nkv * G, hd, device=DEV).bfloat16()
+    lens = torch.tensor([L, L - 17], dtype=torch.int32, device=DEV)
+    got = ops.attn_decode(q, kc, vc, bt, lens, hd**-0.5)
+    ref = R.attn_decode(q, kc, vc, bt, lens, hd**-0.5)
+    assert_close(got, ref, msg="long decode")
