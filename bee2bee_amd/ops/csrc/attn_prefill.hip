@@ -114,10 +114,16 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
     const int li = lane & 15;
 
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-    // K tile [PF_TN][HD+PAD], V^T tile [HD][PF_TN+PAD], both bf16
-    unsigned short* k_s = reinterpret_cast<unsigned short*>(smem_raw);
-    unsigned short* vt_s = k_s + PF_TN * (HD + PF_PAD);
+    // K tile [PF_TN][HD+PAD], V^T tile [HD][PF_TN+PAD], both bf16.
+    // The K buffer is later reused as the P tile [PF_TM][PF_TN+PAD], so the
+    // V^T carve starts past max(K tile, P tile) — at HD=64 the P tile is the
+    // larger one and a K-sized carve would let P overwrite V^T.
     constexpr int VT_STRIDE = PF_TN + PF_PAD;
+    constexpr int K_ELEMS = PF_TN * (HD + PF_PAD);
+    constexpr int P_ELEMS = PF_TM * VT_STRIDE;
+    constexpr int VT_OFF = (K_ELEMS > P_ELEMS ? K_ELEMS : P_ELEMS);
+    unsigned short* k_s = reinterpret_cast<unsigned short*>(smem_raw);
+    unsigned short* vt_s = k_s + VT_OFF;
 
     // ---- load this wave's Q rows into A-fragments (registers), scaled
     bf16x8 a_q[KSTEPS];
@@ -293,6 +299,7 @@ extern "C" void launch_attn_prefill(
         // LDS: max(K tile, P tile) + V^T tile, bf16
         const int k_bytes = PF_TN * (hd + PF_PAD) * 2;
         const int p_bytes = PF_TM * (PF_TN + PF_PAD) * 2;
+        // must match VT_OFF in the kernel: V^T starts past max(K, P)
         const int vt_bytes = hd * (PF_TN + PF_PAD) * 2;
         const int smem = (k_bytes > p_bytes ? k_bytes : p_bytes) + vt_bytes;
         if (hd == 128)

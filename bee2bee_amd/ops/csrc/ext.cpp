@@ -24,8 +24,8 @@ void launch_swiglu(const unsigned short*, unsigned short*, long, long,
                    hipStream_t);
 void launch_attn_decode(const unsigned short*, const unsigned short*,
                         const unsigned short*, const int*, const int*,
-                        unsigned short*, int, int, int, int, int, int, long,
-                        float, hipStream_t);
+                        float*, float*, unsigned short*, int, int, int, int,
+                        int, int, int, long, float, hipStream_t);
 void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          const unsigned short*, const int*, unsigned short*,
                          int, int, int, int, long, long, long, int, float,
@@ -147,9 +147,16 @@ Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
                 "unsupported GQA group size ", G);
     TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == hd);
     Tensor out = torch::empty({B, nq, hd}, q.options());
+    // flash-decoding split: one partial per 256-key chunk, then combine
+    const int kDecChunk = 256;  // keep in sync with DEC_CHUNK
+    const int C = (W * bs + kDecChunk - 1) / kDecChunk;
+    auto fopt = q.options().dtype(torch::kFloat32);
+    Tensor part_o = torch::empty({B, nkv, C, G, hd}, fopt);
+    Tensor part_ml = torch::empty({B, nkv, C, G, 2}, fopt);
     launch_attn_decode(bf16p(q), bf16p(k_cache), bf16p(v_cache),
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                       bf16p_mut(out), B, nkv, G, W, bs, hd, q.stride(0),
+                       part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
+                       bf16p_mut(out), B, nkv, G, W, bs, hd, C, q.stride(0),
                        (float)scale, stream());
     return out;
 }

@@ -35,7 +35,18 @@ def main() -> None:
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--prompt-len", type=int, default=1024)
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--tunableop", action="store_true",
+                    help="tune hipBLASLt GEMM algorithm selection first")
     args = ap.parse_args()
+
+    if args.tunableop:
+        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+        os.environ.setdefault(
+            "PYTORCH_TUNABLEOP_FILENAME",
+            os.path.join("gpurun_out", "tunableop_%d.csv"),
+        )
+        os.makedirs("gpurun_out", exist_ok=True)
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
