@@ -100,6 +100,8 @@ class DecodeGraphs:
         B = self.bucket_for(batch)
         if B not in self._graphs:
             self._capture(B)
-        else:
-            self._graphs[B].replay()
+        # ALWAYS consume a replay, never the capture run itself: library
+        # GEMMs may pick a different algorithm while capturing, so replayed
+        # steps would not be bit-identical with the capture step otherwise.
+        self._graphs[B].replay()
         return self.logits[B]

@@ -1,19 +1,17 @@
-// Paged GQA decode attention — flash-decoding with key-chunk splitting.
+// Paged GQA decode attention — three-stage flash-decoding.
 //
-// The decode step is memory-bound on KV reads (per key: 2*hd*2 B moved,
-// ~O(G*hd) VALU — far under the VALU roof). The first version used one
-// workgroup per (seq, kv-head): at B=64, nkv=8 that is 512 workgroups = 2
-// per CU = 2 waves/SIMD — not enough latency hiding (measured 178 us/layer
-// vs the ~43 us HBM roofline). This version splits the key axis into
-// 256-key chunks, one workgroup per (seq, kv-head, chunk), each writing an
-// unnormalized partial (m, l, o) that a small combine kernel merges — the
-// MI355X needs >>256 workgroups to fill its 8 XCDs.
-//
-// Phase 1 (scores): lanes-over-keys — lane l owns one key; it streams the
-//   key row in 16 B pieces and accumulates G dot products against q held
-//   in LDS (no per-key cross-lane reduction).
-// Phase 2 (PV): lanes-over-dims — lane owns 2 output dims; V rows read as
-//   4 B/lane x 64 lanes = one coalesced 256 B row; P broadcast from LDS.
+// The decode step is memory-bound on KV reads (per key: 2*hd*2 B moved).
+// A single fused kernel could not reach the HBM roofline: its score phase
+// (lanes-over-keys) and PV phase (lanes-over-dims) have opposing register
+// needs, and together they pushed VGPRs past 100 -> 2 waves/SIMD -> too few
+// bytes in flight (measured 1.7-2.5 TB/s vs 6.0 TB/s for each access
+// pattern probed in isolation — see probe.hip bw_* kernels). Splitting into
+// per-chunk kernels keeps each at <=64 VGPRs with deep independent-load
+// batches:
+//   1. attn_scores:  S = q.K^T + chunk-level softmax -> p, (m,l) scratch
+//   2. attn_pv:      partial_o = p.V per chunk (p from scratch, LDS-staged)
+//   3. attn_combine: merge chunk partials, normalize, write bf16 out
+// Scratch p traffic is ~3% of KV traffic.
 //
 // Replaces: the reference's decode path inside transformers.generate()
 // (bee2bee/hf.py:84-108). Numerics reference: ops/reference.py attn_decode.
@@ -23,14 +21,42 @@
 #define DEC_WAVES 4
 #define DEC_CHUNK 256  // keys per workgroup (DEC_WAVES x 64)
 
+// ------------------------------------------------------------ stage 1: S
+// Whole K row staged into registers (HD/8 independent 16 B loads in one
+// burst), then G dot products against the LDS-held q. Compile-time HD keeps
+// the row array in registers (a runtime bound would spill it to scratch).
+template <int G, int HD>
+__device__ __forceinline__ void score_row(const unsigned short* kr,
+                                          const unsigned short* q_s,
+                                          float* s) {
+    short8 krow[HD / 8];
+#pragma unroll
+    for (int ii = 0; ii < HD / 8; ++ii)
+        krow[ii] = *reinterpret_cast<const short8*>(kr + ii * 8);
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+#pragma unroll
+        for (int ii = 0; ii < HD / 8; ++ii) {
+            const short8 qraw =
+                *reinterpret_cast<const short8*>(q_s + g * HD + ii * 8);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                s[g] = fmaf(bf2f((unsigned short)krow[ii][j]),
+                            bf2f((unsigned short)qraw[j]), s[g]);
+        }
+        // stop the scheduler hoisting every g's LDS q-reads to the top
+        // (otherwise G=4/HD=128 allocates 256 VGPRs -> 1 wave/SIMD)
+        __builtin_amdgcn_sched_barrier(0);
+    }
+}
+
 template <int G>
-__global__ __launch_bounds__(DEC_BLOCK) void attn_decode_chunk_kernel(
+__global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     const unsigned short* __restrict__ q,        // [B, nq, hd] (strided)
     const unsigned short* __restrict__ k_cache,  // [nb, nkv, bs, hd]
-    const unsigned short* __restrict__ v_cache,
     const int* __restrict__ block_table,         // [B, W]
     const int* __restrict__ seq_lens,            // [B]
-    float* __restrict__ part_o,                  // [B, nkv, C, G, hd]
+    float* __restrict__ p_out,                   // [B, nkv, C, CHUNK, G]
     float* __restrict__ part_ml,                 // [B, nkv, C, G, 2]
     int nkv, int W, int bs, int hd, int C, long q_stride, float scale) {
     const int b = blockIdx.x;
@@ -38,24 +64,24 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_chunk_kernel(
     const int chunk = blockIdx.z;
     const int L = seq_lens[b];
     const int start = chunk * DEC_CHUNK;
-    if (start >= L) return;  // inactive chunk: combine never reads it
-    const int nq = nkv * G;
+    if (start >= L) return;
     const int lane = threadIdx.x % WAVE;
     const int wid = threadIdx.x / WAVE;
 
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-    float* q_s = reinterpret_cast<float*>(smem_raw);  // [G * hd]
-    float* p_s = q_s + G * hd;                        // [DEC_WAVES][G][WAVE]
-    float* merge = p_s + DEC_WAVES * G * WAVE;        // [DEC_WAVES][G][hd+2]
+    unsigned short* q_s = reinterpret_cast<unsigned short*>(smem_raw);  // [G*hd]
+    float* red = reinterpret_cast<float*>(smem_raw + ((G * hd * 2 + 15) & ~15));
+    // red: [DEC_WAVES][G][2] wave-level m / sum
 
     for (int i = threadIdx.x; i < G * hd; i += DEC_BLOCK) {
         const int g = i / hd, d = i % hd;
-        q_s[i] = bf2f(q[(long)b * q_stride + (kvh * G + g) * (long)hd + d]) * scale;
+        q_s[i] = f2bf(
+            bf2f(q[(long)b * q_stride + (kvh * G + g) * (long)hd + d]) * scale);
     }
     __syncthreads();
 
     const int* bt = block_table + (long)b * W;
-    const int key = start + wid * WAVE + lane;  // one key per lane
+    const int key = start + threadIdx.x;  // one key per thread
     const bool valid = key < L;
 
     float s[G];
@@ -67,68 +93,171 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_chunk_kernel(
             k_cache + (((long)page * nkv + kvh) * bs + key % bs) * hd;
 #pragma unroll
         for (int g = 0; g < G; ++g) s[g] = 0.f;
-        for (int d = 0; d < hd; d += 8) {
-            float kv[8];
-            load_bf16x8(kr + d, kv);
+        if (hd == 128) {
+            score_row<G, 128>(kr, q_s, s);
+        } else if (hd == 64) {
+            score_row<G, 64>(kr, q_s, s);
+        } else {
+            for (int d = 0; d < hd; d += 8) {  // tail models (hd 16/24/...)
+                const short8 kraw = *reinterpret_cast<const short8*>(kr + d);
 #pragma unroll
-            for (int g = 0; g < G; ++g) {
-                const float* qg = q_s + g * hd + d;
+                for (int g = 0; g < G; ++g) {
+                    const short8 qraw =
+                        *reinterpret_cast<const short8*>(q_s + g * hd + d);
 #pragma unroll
-                for (int j = 0; j < 8; ++j) s[g] = fmaf(kv[j], qg[j], s[g]);
+                    for (int j = 0; j < 8; ++j)
+                        s[g] = fmaf(bf2f((unsigned short)kraw[j]),
+                                    bf2f((unsigned short)qraw[j]), s[g]);
+                }
             }
         }
     }
 
-    // per-wave softmax over this wave's 64 keys
-    float m[G], lsum[G];
-    float* my_p = p_s + (wid * G) * WAVE;
+    // chunk-level softmax: wave max -> LDS -> chunk max; then exp + sums
+    float wmax[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) wmax[g] = wave_max(s[g]);
+    if (lane == 0) {
+#pragma unroll
+        for (int g = 0; g < G; ++g) red[(wid * G + g) * 2] = wmax[g];
+    }
+    __syncthreads();
+    float M[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-        m[g] = wave_max(s[g]);
-        const float p = valid ? __expf(s[g] - m[g]) : 0.f;
-        lsum[g] = wave_sum(p);
-        my_p[g * WAVE + lane] = p;
+        float m = red[g * 2];
+        for (int w2 = 1; w2 < DEC_WAVES; ++w2)
+            m = fmaxf(m, red[(w2 * G + g) * 2]);
+        M[g] = m;
     }
+    float p[G], wsum[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+        p[g] = valid ? __expf(s[g] - M[g]) : 0.f;
+        wsum[g] = wave_sum(p[g]);
+    }
+    __syncthreads();  // red reuse (second slot written below)
+    if (lane == 0) {
+#pragma unroll
+        for (int g = 0; g < G; ++g) red[(wid * G + g) * 2 + 1] = wsum[g];
+    }
+    // coalesced p write: [key][G], 4*G B per thread, consecutive
+    {
+        float* prow =
+            p_out + ((((long)b * nkv + kvh) * C + chunk) * DEC_CHUNK +
+                     threadIdx.x) * G;
+#pragma unroll
+        for (int g = 0; g < G; ++g) prow[g] = p[g];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float* pml = part_ml + ((((long)b * nkv + kvh) * C + chunk) * G) * 2;
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+            float lsum = 0.f;
+            for (int w2 = 0; w2 < DEC_WAVES; ++w2)
+                lsum += red[(w2 * G + g) * 2 + 1];
+            pml[g * 2] = M[g];
+            pml[g * 2 + 1] = lsum;
+        }
+    }
+}
 
-    // phase 2: lane owns dims (d0, d0+1); iterate this wave's keys
+// ----------------------------------------------------------- stage 2: PV
+template <int G>
+__global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
+    const unsigned short* __restrict__ v_cache,  // [nb, nkv, bs, hd]
+    const int* __restrict__ block_table,
+    const int* __restrict__ seq_lens,
+    const float* __restrict__ p_in,              // [B, nkv, C, CHUNK, G]
+    float* __restrict__ part_o,                  // [B, nkv, C, G, hd]
+    int nkv, int W, int bs, int hd, int C) {
+    const int b = blockIdx.x;
+    const int kvh = blockIdx.y;
+    const int chunk = blockIdx.z;
+    const int L = seq_lens[b];
+    const int start = chunk * DEC_CHUNK;
+    if (start >= L) return;
+    const int lane = threadIdx.x % WAVE;
+    const int wid = threadIdx.x / WAVE;
+
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    float* p_s = reinterpret_cast<float*>(smem_raw);  // [CHUNK][G]
+    long* voff_s = reinterpret_cast<long*>(p_s + DEC_CHUNK * G);  // [CHUNK]
+    float* merge = reinterpret_cast<float*>(voff_s + DEC_CHUNK);  // [Wv][G][hd]
+
+    // stage p for the whole chunk into LDS (coalesced global read)
+    {
+        const float* psrc =
+            p_in + (((long)b * nkv + kvh) * C + chunk) * DEC_CHUNK * G;
+        for (int i = threadIdx.x; i < DEC_CHUNK * G / 4; i += DEC_BLOCK)
+            reinterpret_cast<f32x4*>(p_s)[i] =
+                reinterpret_cast<const f32x4*>(psrc)[i];
+    }
+    // per-thread V row offset (one chained block-table load per key)
+    {
+        const int* bt = block_table + (long)b * W;
+        const int key = start + threadIdx.x;
+        long voff = 0;
+        if (key < L) {
+            const int page = bt[key / bs];
+            voff = (((long)page * nkv + kvh) * bs + key % bs) * hd;
+        }
+        voff_s[threadIdx.x] = voff;
+    }
+    __syncthreads();
+
+    // wave wid accumulates keys [wid*64, wid*64+64) of the chunk;
+    // lane owns output dims (d0, d0+1)
     const int d0 = lane * 2;
-    const int wave_start = start + wid * WAVE;
-    const int nkeys = min(WAVE, L - wave_start);
+    const int wave_start = wid * WAVE;
+    const int nkeys = min(WAVE, L - (start + wave_start));
     float o0[G], o1[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) o0[g] = o1[g] = 0.f;
     if (d0 < hd && nkeys > 0) {
-        for (int t = 0; t < nkeys; ++t) {
-            const int tkey = wave_start + t;
-            const int page = bt[tkey / bs];
-            const unsigned short* vr =
-                v_cache + (((long)page * nkv + kvh) * bs + tkey % bs) * hd;
-            const short2v vv = *reinterpret_cast<const short2v*>(vr + d0);
+        const long* voffs = voff_s + wave_start;
+        const float* pw = p_s + wave_start * G;
+        int t = 0;
+        for (; t + 16 <= nkeys; t += 16) {  // 16 V loads in flight (4 KB/wave)
+            short2v vv[16];
+#pragma unroll
+            for (int j = 0; j < 16; ++j)
+                vv[j] = *reinterpret_cast<const short2v*>(
+                    v_cache + voffs[t + j] + d0);
+#pragma unroll
+            for (int j = 0; j < 16; ++j) {
+                const float v0 = bf2f((unsigned short)vv[j][0]);
+                const float v1 = bf2f((unsigned short)vv[j][1]);
+                const float* pt = pw + (t + j) * G;
+#pragma unroll
+                for (int g = 0; g < G; ++g) {
+                    o0[g] = fmaf(pt[g], v0, o0[g]);
+                    o1[g] = fmaf(pt[g], v1, o1[g]);
+                }
+            }
+        }
+        for (; t < nkeys; ++t) {
+            const short2v vv =
+                *reinterpret_cast<const short2v*>(v_cache + voffs[t] + d0);
             const float v0 = bf2f((unsigned short)vv[0]);
             const float v1 = bf2f((unsigned short)vv[1]);
+            const float* pt = pw + t * G;
 #pragma unroll
             for (int g = 0; g < G; ++g) {
-                const float p = my_p[g * WAVE + t];
-                o0[g] = fmaf(p, v0, o0[g]);
-                o1[g] = fmaf(p, v1, o1[g]);
+                o0[g] = fmaf(pt[g], v0, o0[g]);
+                o1[g] = fmaf(pt[g], v1, o1[g]);
             }
         }
     }
 
-    // cross-wave merge through LDS -> one partial per chunk
-    float* mw = merge + wid * G * (hd + 2);
+    // all waves used the same chunk max, so cross-wave merge is a plain sum
+    float* mw = merge + wid * G * hd;
     if (d0 < hd) {
 #pragma unroll
         for (int g = 0; g < G; ++g) {
-            mw[g * (hd + 2) + d0] = o0[g];
-            mw[g * (hd + 2) + d0 + 1] = o1[g];
-        }
-    }
-    if (lane == 0) {
-#pragma unroll
-        for (int g = 0; g < G; ++g) {
-            mw[g * (hd + 2) + hd] = m[g];
-            mw[g * (hd + 2) + hd + 1] = lsum[g];
+            mw[g * hd + d0] = o0[g];
+            mw[g * hd + d0 + 1] = o1[g];
         }
     }
     __syncthreads();
@@ -136,32 +265,19 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_chunk_kernel(
         const long base = (((long)b * nkv + kvh) * C + chunk) * G;
 #pragma unroll
         for (int g = 0; g < G; ++g) {
-            float M = -1e30f;
-#pragma unroll
-            for (int w2 = 0; w2 < DEC_WAVES; ++w2)
-                M = fmaxf(M, merge[(w2 * G + g) * (hd + 2) + hd]);
-            float Ltot = 0.f, acc0 = 0.f, acc1 = 0.f;
-#pragma unroll
+            float a0 = 0.f, a1 = 0.f;
             for (int w2 = 0; w2 < DEC_WAVES; ++w2) {
-                const float* row = merge + (w2 * G + g) * (hd + 2);
-                const float f = __expf(row[hd] - M);
-                Ltot += f * row[hd + 1];
-                acc0 += f * row[d0];
-                acc1 += f * row[d0 + 1];
+                a0 += merge[(w2 * G + g) * hd + d0];
+                a1 += merge[(w2 * G + g) * hd + d0 + 1];
             }
             float* po = part_o + (base + g) * hd;
-            po[d0] = acc0;
-            po[d0 + 1] = acc1;
-            if (lane == 0) {
-                float* pml = part_ml + (base + g) * 2;
-                pml[0] = M;
-                pml[1] = Ltot;
-            }
+            po[d0] = a0;
+            po[d0 + 1] = a1;
         }
     }
 }
 
-// merge the per-chunk partials into the final normalized output
+// ------------------------------------------------------ stage 3: combine
 template <int G>
 __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
     const float* __restrict__ part_o,   // [B, nkv, C, G, hd]
@@ -176,7 +292,6 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
     const int nc = (L + DEC_CHUNK - 1) / DEC_CHUNK;
     const long base = ((long)b * nkv + kvh) * C;
 
-    // threads cover (g, d) pairs
     for (int i = threadIdx.x; i < G * hd; i += DEC_BLOCK) {
         const int g = i / hd, d = i % hd;
         float M = -1e30f;
@@ -197,19 +312,23 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
 extern "C" void launch_attn_decode(
     const unsigned short* q, const unsigned short* k_cache,
     const unsigned short* v_cache, const int* block_table,
-    const int* seq_lens, float* part_o, float* part_ml,
+    const int* seq_lens, float* p_buf, float* part_o, float* part_ml,
     unsigned short* out, int B, int nkv, int G, int W, int bs, int hd, int C,
     long q_stride, float scale, hipStream_t stream) {
     dim3 grid(B, nkv, C);
     dim3 cgrid(B, nkv);
-    const int smem =
-        (G * hd + DEC_WAVES * G * WAVE + DEC_WAVES * G * (hd + 2)) * 4;
+    const int smem_s = ((G * hd * 2 + 15) & ~15) + DEC_WAVES * G * 2 * 4;
+    const int smem_pv =
+        DEC_CHUNK * G * 4 + DEC_CHUNK * 8 + DEC_WAVES * G * hd * 4;
 #define LAUNCH(GG)                                                             \
     do {                                                                       \
-        hipLaunchKernelGGL(attn_decode_chunk_kernel<GG>, grid,                 \
-                           dim3(DEC_BLOCK), smem, stream, q, k_cache,          \
-                           v_cache, block_table, seq_lens, part_o, part_ml,    \
-                           nkv, W, bs, hd, C, q_stride, scale);                \
+        hipLaunchKernelGGL(attn_scores_kernel<GG>, grid, dim3(DEC_BLOCK),      \
+                           smem_s, stream, q, k_cache, block_table, seq_lens,  \
+                           p_buf, part_ml, nkv, W, bs, hd, C, q_stride,        \
+                           scale);                                             \
+        hipLaunchKernelGGL(attn_pv_kernel<GG>, grid, dim3(DEC_BLOCK),          \
+                           smem_pv, stream, v_cache, block_table, seq_lens,    \
+                           p_buf, part_o, nkv, W, bs, hd, C);                  \
         hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,              \
                            dim3(DEC_BLOCK), 0, stream, part_o, part_ml,        \
                            seq_lens, out, nkv, hd, C);                         \

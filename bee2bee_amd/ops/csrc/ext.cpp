@@ -24,14 +24,19 @@ void launch_swiglu(const unsigned short*, unsigned short*, long, long,
                    hipStream_t);
 void launch_attn_decode(const unsigned short*, const unsigned short*,
                         const unsigned short*, const int*, const int*,
-                        float*, float*, unsigned short*, int, int, int, int,
-                        int, int, int, long, float, hipStream_t);
+                        float*, float*, float*, unsigned short*, int, int,
+                        int, int, int, int, int, long, float, hipStream_t);
 void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          const unsigned short*, const int*, unsigned short*,
                          int, int, int, int, long, long, long, int, float,
                          int, hipStream_t);
 void launch_mfma_probe(const unsigned short*, const unsigned short*, float*,
                        hipStream_t);
+void launch_bw_linear(const unsigned short*, long, float*, hipStream_t);
+void launch_bw_rowperlane(const unsigned short*, long, int, int, float*,
+                          hipStream_t);
+void launch_bw_rowperinstr(const unsigned short*, long, int, int, float*,
+                           hipStream_t);
 }
 
 namespace {
@@ -151,11 +156,13 @@ Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
     const int kDecChunk = 256;  // keep in sync with DEC_CHUNK
     const int C = (W * bs + kDecChunk - 1) / kDecChunk;
     auto fopt = q.options().dtype(torch::kFloat32);
+    Tensor p_buf = torch::empty({B, nkv, C, kDecChunk, G}, fopt);
     Tensor part_o = torch::empty({B, nkv, C, G, hd}, fopt);
     Tensor part_ml = torch::empty({B, nkv, C, G, 2}, fopt);
     launch_attn_decode(bf16p(q), bf16p(k_cache), bf16p(v_cache),
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                       part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
+                       p_buf.data_ptr<float>(), part_o.data_ptr<float>(),
+                       part_ml.data_ptr<float>(),
                        bf16p_mut(out), B, nkv, G, W, bs, hd, C, q.stride(0),
                        (float)scale, stream());
     return out;
@@ -196,6 +203,20 @@ Tensor mfma_probe(const Tensor& A, const Tensor& B) {
     return C;
 }
 
+void bw_probe(const Tensor& pool, int64_t mode, int64_t hd, int64_t arg) {
+    check_bf16(pool, "pool");
+    Tensor sink = torch::zeros({1}, pool.options().dtype(torch::kFloat32));
+    const long n = pool.numel();
+    if (mode == 0)
+        launch_bw_linear(bf16p(pool), n, sink.data_ptr<float>(), stream());
+    else if (mode == 1)
+        launch_bw_rowperlane(bf16p(pool), n / hd, hd, (int)arg,
+                             sink.data_ptr<float>(), stream());
+    else
+        launch_bw_rowperinstr(bf16p(pool), n / hd, hd, (int)arg,
+                              sink.data_ptr<float>(), stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -208,4 +229,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_decode", &attn_decode, "paged GQA decode attention");
     m.def("attn_prefill", &attn_prefill, "varlen causal flash prefill");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+    m.def("bw_probe", &bw_probe, "bandwidth pattern probe");
 }

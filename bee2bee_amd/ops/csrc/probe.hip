@@ -26,3 +26,102 @@ extern "C" void launch_mfma_probe(const unsigned short* A,
                                   hipStream_t stream) {
     hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream, A, B, C);
 }
+
+// ---------------------------------------------------------------------------
+// Bandwidth probes over a bf16 pool — used to locate the decode-attention
+// bandwidth ceiling (scripts/bench_attn.py --probe). Each accumulates a
+// checksum so loads cannot be DCE'd.
+
+// (a) linear roofline: grid-stride dwordx4
+__global__ void bw_linear_kernel(const unsigned short* __restrict__ src,
+                                 long n_elems, float* __restrict__ sink) {
+    float acc = 0.f;
+    const long stride = (long)gridDim.x * blockDim.x * 8;
+    for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+         i < n_elems; i += stride) {
+        short8 v = *reinterpret_cast<const short8*>(src + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc += (float)(short)v[j];
+    }
+    if (acc == 1234.5678f) sink[0] = acc;  // never true; keeps loads live
+}
+
+// (b) phase-1 pattern: lane owns one 256 B row, reads it in 16 B pieces.
+// rows_per_block keys per workgroup (4 waves x 64), row stride = hd elems.
+__global__ void bw_rowperlane_kernel(const unsigned short* __restrict__ src,
+                                     long n_rows, int hd, int batch8,
+                                     float* __restrict__ sink) {
+    const long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (row >= n_rows) return;
+    const unsigned short* r = src + row * hd;
+    float acc = 0.f;
+    if (batch8) {
+        for (int d = 0; d + 64 <= hd; d += 64) {
+            short8 raw[8];
+#pragma unroll
+            for (int ii = 0; ii < 8; ++ii)
+                raw[ii] = *reinterpret_cast<const short8*>(r + d + ii * 8);
+#pragma unroll
+            for (int ii = 0; ii < 8; ++ii)
+#pragma unroll
+                for (int j = 0; j < 8; ++j) acc += (float)(short)raw[ii][j];
+        }
+    } else {
+        for (int d = 0; d < hd; d += 8) {
+            short8 v = *reinterpret_cast<const short8*>(r + d);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) acc += (float)(short)v[j];
+        }
+    }
+    if (acc == 1234.5678f) sink[0] = acc;
+}
+
+// (c) phase-2 pattern: whole wave reads one row per instruction (4 B/lane),
+// iterating rows in batches of 8 independent loads.
+__global__ void bw_rowperinstr_kernel(const unsigned short* __restrict__ src,
+                                      long n_rows, int hd, int rows_per_wg,
+                                      float* __restrict__ sink) {
+    const int lane = threadIdx.x % WAVE;
+    const int wid = threadIdx.x / WAVE;
+    const int waves = blockDim.x / WAVE;
+    const long base = (long)blockIdx.x * rows_per_wg;
+    const int d0 = (lane * 2) % hd;
+    float a0 = 0.f, a1 = 0.f;
+    for (long t0 = base + wid * 8; t0 < base + rows_per_wg;
+         t0 += (long)waves * 8) {
+        short2v vv[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const long row = (t0 + j < n_rows) ? t0 + j : 0;
+            vv[j] = *reinterpret_cast<const short2v*>(src + row * hd + d0);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            a0 += (float)(short)vv[j][0];
+            a1 += (float)(short)vv[j][1];
+        }
+    }
+    if (a0 + a1 == 1234.5678f) sink[0] = a0;
+}
+
+extern "C" {
+void launch_bw_linear(const unsigned short* src, long n, float* sink,
+                      hipStream_t st) {
+    long blocks = (n / 8 + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    hipLaunchKernelGGL(bw_linear_kernel, dim3(blocks), dim3(256), 0, st, src,
+                       n, sink);
+}
+void launch_bw_rowperlane(const unsigned short* src, long n_rows, int hd,
+                          int batch8, float* sink, hipStream_t st) {
+    long blocks = (n_rows + 255) / 256;
+    hipLaunchKernelGGL(bw_rowperlane_kernel, dim3(blocks), dim3(256), 0, st,
+                       src, n_rows, hd, batch8, sink);
+}
+void launch_bw_rowperinstr(const unsigned short* src, long n_rows, int hd,
+                           int rows_per_wg, float* sink, hipStream_t st) {
+    long blocks = (n_rows + rows_per_wg - 1) / rows_per_wg;
+    hipLaunchKernelGGL(bw_rowperinstr_kernel, dim3(blocks), dim3(256), 0, st,
+                       src, n_rows, hd, rows_per_wg, sink);
+}
+}
