@@ -32,7 +32,7 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--batch", type=int, default=64)
+    ap.add_argument("--batch", type=int, default=256)
     ap.add_argument("--prompt-len", type=int, default=1024)
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--tunableop", action="store_true",
