@@ -108,7 +108,11 @@ class InferenceEngine:
         self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks)
         self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
 
-        self.use_graphs = on_gpu if use_graphs is None else (use_graphs and on_gpu)
+        # MoE routing (topk + gather/scatter) has data-dependent shapes and
+        # is not yet capture-safe; grouped-GEMM fixed-shape dispatch will
+        # lift this restriction
+        graphs_ok = on_gpu and not self.spec.is_moe
+        self.use_graphs = graphs_ok if use_graphs is None else (use_graphs and graphs_ok)
         self.graphs: Optional[DecodeGraphs] = None
         if self.use_graphs:
             self.graphs = DecodeGraphs(self.runner, max_batch, blocks_per_seq)

@@ -69,11 +69,16 @@ class ModelWeights:
         embed only on the first stage, head/final norm only on the last."""
         s = self.spec
         lo, hi = layer_range or (0, s.n_layers)
-        # generate directly on the target device (an 8B randn on host would
-        # serialize startup; device-side randn fills 16 GB in ~a second)
-        gen = torch.Generator(device=self.device).manual_seed(seed)
+        import zlib
 
-        def rnd(*shape: int, std: float) -> torch.Tensor:
+        # Generate directly on the target device (an 8B randn on host would
+        # serialize startup). Each tensor gets its own generator seeded by
+        # (seed, tensor name) so a pipeline stage materializing only layers
+        # [lo, hi) draws IDENTICAL values to a full single-process init —
+        # the PP-vs-single-process equivalence tests rely on this.
+        def rnd(name: str, *shape: int, std: float) -> torch.Tensor:
+            tseed = (seed * 1000003 + zlib.crc32(name.encode())) % (2**31)
+            gen = torch.Generator(device=self.device).manual_seed(tseed)
             t = torch.randn(*shape, generator=gen, dtype=torch.float32,
                             device=self.device)
             return (t * std).to(self.dtype)
@@ -81,32 +86,38 @@ class ModelWeights:
         std = 0.02
         proj_std = std / max(1.0, (2 * s.n_layers) ** 0.5)
         if lo == 0:
-            self.embed = rnd(s.vocab_size, s.hidden_size, std=std)
+            self.embed = rnd("embed", s.vocab_size, s.hidden_size, std=std)
         if hi == s.n_layers:
             self.final_norm = torch.ones(
                 s.hidden_size, device=self.device, dtype=self.dtype
             )
-            if s.tie_embeddings and lo == 0:
-                self.lm_head = self.embed
+            if s.tie_embeddings:
+                self.lm_head = (
+                    self.embed
+                    if self.embed is not None
+                    else rnd("embed", s.vocab_size, s.hidden_size, std=std)
+                )
             else:
-                self.lm_head = rnd(s.vocab_size, s.hidden_size, std=std)
+                self.lm_head = rnd("lm_head", s.vocab_size, s.hidden_size, std=std)
         for i in range(lo, hi):
             lw = self.layers[i]
             lw.attn_norm = torch.ones(s.hidden_size, device=self.device, dtype=self.dtype)
             lw.mlp_norm = torch.ones(s.hidden_size, device=self.device, dtype=self.dtype)
-            lw.wqkv = rnd(s.q_size + 2 * s.kv_size, s.hidden_size, std=std)
-            lw.wo = rnd(s.hidden_size, s.q_size, std=proj_std)
+            lw.wqkv = rnd(f"l{i}.wqkv", s.q_size + 2 * s.kv_size, s.hidden_size, std=std)
+            lw.wo = rnd(f"l{i}.wo", s.hidden_size, s.q_size, std=proj_std)
             if s.is_moe:
-                lw.moe_gate = rnd(s.n_experts, s.hidden_size, std=std)
+                lw.moe_gate = rnd(f"l{i}.gate", s.n_experts, s.hidden_size, std=std)
                 lw.moe_w_gate_up = rnd(
-                    s.n_experts, 2 * s.intermediate_size, s.hidden_size, std=std
+                    f"l{i}.w_gu", s.n_experts, 2 * s.intermediate_size,
+                    s.hidden_size, std=std
                 )
                 lw.moe_w_down = rnd(
-                    s.n_experts, s.hidden_size, s.intermediate_size, std=proj_std
+                    f"l{i}.w_dn", s.n_experts, s.hidden_size,
+                    s.intermediate_size, std=proj_std
                 )
             else:
-                lw.w_gate_up = rnd(2 * s.intermediate_size, s.hidden_size, std=std)
-                lw.w_down = rnd(s.hidden_size, s.intermediate_size, std=proj_std)
+                lw.w_gate_up = rnd(f"l{i}.w_gu", 2 * s.intermediate_size, s.hidden_size, std=std)
+                lw.w_down = rnd(f"l{i}.w_dn", s.hidden_size, s.intermediate_size, std=proj_std)
         return self
 
     # ---------------------------------------------------------- safetensors

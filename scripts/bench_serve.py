@@ -1,0 +1,151 @@
+"""End-to-end serving benchmark: p50/p95 request latency + tokens/sec
+through the FULL stack (HTTP gateway -> mesh node -> service -> engine).
+
+This measures the second half of BASELINE.json's metric ("output tokens/sec
++ p50 e2e request latency"): concurrent clients POST /generate against a
+live node serving the native engine, exactly like an external user.
+
+Usage:
+  python scripts/bench_serve.py --model llama3-8b --clients 16 \
+      --requests 64 --max-new 64 [--device cuda:0] [--prompt-len 512]
+Prints one JSON line with latency percentiles and aggregate throughput.
+"""
+import argparse
+import asyncio
+import json
+import os
+import random
+import statistics
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+async def run() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--clients", type=int, default=16)
+    ap.add_argument("--requests", type=int, default=64)
+    ap.add_argument("--max-new", type=int, default=64)
+    ap.add_argument("--prompt-len", type=int, default=512, help="chars of prompt")
+    ap.add_argument("--max-batch", type=int, default=32)
+    ap.add_argument("--api-port", type=int, default=18321)
+    args = ap.parse_args()
+
+    import aiohttp
+    import uvicorn
+
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.mesh.node import MeshNode
+    from bee2bee_amd.services.native import NativeEngineService
+
+    node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+    await node.start()
+    node.api_port = args.api_port
+    svc = NativeEngineService(
+        args.model, device=args.device, max_batch=args.max_batch,
+        max_seq_len=2048,
+    )
+    loop = asyncio.get_running_loop()
+    t0 = time.time()
+    await loop.run_in_executor(None, svc.load_sync)
+    load_s = time.time() - t0
+    await node.add_service(svc)
+
+    gateway_api.node = node
+    config = uvicorn.Config(
+        gateway_api.app, host="127.0.0.1", port=args.api_port, log_level="error"
+    )
+    server = uvicorn.Server(config)
+    asyncio.get_running_loop().create_task(server.serve())
+    await asyncio.sleep(1.0)
+
+    rng = random.Random(0)
+    words = ["alpha", "beta", "gamma", "delta", "mesh", "gpu", "tensor", "ring"]
+
+    def mk_prompt() -> str:
+        out = []
+        while sum(len(w) + 1 for w in out) < args.prompt_len:
+            out.append(rng.choice(words))
+        return " ".join(out)
+
+    lat: list = []
+    ttft: list = []
+    tokens_total = 0
+    sem = asyncio.Semaphore(args.clients)
+    url = f"http://127.0.0.1:{args.api_port}/generate"
+
+    async def one_request(session) -> None:
+        nonlocal tokens_total
+        async with sem:
+            t_start = time.perf_counter()
+            first = None
+            async with session.post(
+                url,
+                json={
+                    "prompt": mk_prompt(),
+                    "max_new_tokens": args.max_new,
+                    "temperature": 0.0,
+                    "stream": True,
+                },
+            ) as resp:
+                assert resp.status == 200, await resp.text()
+                n_tok = 0
+                async for line in resp.content:
+                    if not line.strip():
+                        continue
+                    if first is None:
+                        first = time.perf_counter()
+                    try:
+                        d = json.loads(line)
+                    except Exception:
+                        continue
+                    if d.get("done"):
+                        break
+                    if "text" in d:
+                        n_tok += 1
+            t_end = time.perf_counter()
+            lat.append(t_end - t_start)
+            if first is not None:
+                ttft.append(first - t_start)
+            tokens_total += args.max_new
+
+    t_bench = time.perf_counter()
+    async with aiohttp.ClientSession() as session:
+        # small warmup
+        await one_request(session)
+        lat.clear(); ttft.clear()
+        tokens = tokens_total = 0
+        await asyncio.gather(*(one_request(session) for _ in range(args.requests)))
+    wall = time.perf_counter() - t_bench
+
+    lat.sort()
+    def pct(xs, p):
+        return xs[min(len(xs) - 1, int(p * len(xs)))] if xs else None
+
+    result = {
+        "metric": "e2e request latency + tokens/sec via /generate (streaming)",
+        "model": args.model,
+        "device": str(svc.engine.device),
+        "requests": args.requests,
+        "concurrency": args.clients,
+        "max_new_tokens": args.max_new,
+        "p50_s": round(pct(lat, 0.50), 3),
+        "p95_s": round(pct(lat, 0.95), 3),
+        "ttft_p50_s": round(pct(sorted(ttft), 0.50), 3) if ttft else None,
+        "tokens_per_sec": round(args.requests * args.max_new / wall, 1),
+        "wall_s": round(wall, 2),
+        "model_load_s": round(load_s, 1),
+        "data": "synthetic prompts, random-init weights",
+    }
+    print(json.dumps(result), flush=True)
+    svc.engine.shutdown()
+    await node.stop()
+    server.should_exit = True
+    await asyncio.sleep(0.2)
+
+
+if __name__ == "__main__":
+    asyncio.run(run())
