@@ -194,18 +194,26 @@ class InferenceEngine:
         ids = ids[-(self.max_seq_len - max_new_tokens - 1) :]
         emitted: List[str] = []
         decoded_upto = 0
+        held = 0
 
         def _on_token(tok: int) -> None:
-            nonlocal decoded_upto
+            nonlocal decoded_upto, held
             if on_text is None:
                 return
             text = self.tokenizer.decode(req.output_ids)
             delta = text[decoded_upto:]
-            # hold back partial unicode replacement char at the boundary
-            if delta and not delta.endswith("�"):
-                decoded_upto = len(text)
-                emitted.append(delta)
-                on_text(delta)
+            if not delta:
+                return
+            # hold back a possibly-incomplete multibyte char at the boundary,
+            # but never more than 3 tokens (random-weight output is mostly
+            # U+FFFD and would otherwise buffer the whole stream)
+            if delta.endswith("�") and held < 3:
+                held += 1
+                return
+            held = 0
+            decoded_upto = len(text)
+            emitted.append(delta)
+            on_text(delta)
 
         stop_ids = ()
         eos = getattr(self.tokenizer, "eos_token_id", None)
