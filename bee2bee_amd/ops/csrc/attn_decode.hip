@@ -163,6 +163,158 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     }
 }
 
+
+// MFMA variant of stage 1 (hd 64/128): the G dot products per key run on
+// the matrix pipe as S^T tiles = K_tile(16 keys) x q^T(16 cols, G used).
+// A-fragment rows map exactly to "lane reads its key row's 16 B segment",
+// so K streams from HBM straight into MFMA operands — no LDS staging, no
+// VALU fma chains, ~70 VGPRs -> high occupancy.
+// Fragment maps (verified on HW by tests/test_ops_gpu.py::test_mfma_fragment_map):
+//   A (16x32): lane = row (l&15), k = (l>>4)*8+e
+//   B (32x16): lane = col (l&15), k = (l>>4)*8+e
+//   C (16x16): lane = col (l&15), row = (l>>4)*4+r
+template <int G, int HD>
+__global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
+    const unsigned short* __restrict__ q,
+    const unsigned short* __restrict__ k_cache,
+    const int* __restrict__ block_table,
+    const int* __restrict__ seq_lens,
+    float* __restrict__ p_out,
+    float* __restrict__ part_ml,
+    int nkv, int W, int bs, int C, long q_stride, float scale) {
+    constexpr int KSTEPS = HD / 32;
+    constexpr int TILES = WAVE / 16;  // 4 key-tiles of 16 per wave
+    const int b = blockIdx.x;
+    const int kvh = blockIdx.y;
+    const int chunk = blockIdx.z;
+    const int L = seq_lens[b];
+    const int start = chunk * DEC_CHUNK;
+    if (start >= L) return;
+    const int lane = threadIdx.x % WAVE;
+    const int wid = threadIdx.x / WAVE;
+    const int lg = lane >> 4;  // lane group 0..3
+    const int li = lane & 15;
+
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    unsigned short* q_s = reinterpret_cast<unsigned short*>(smem_raw);  // [G*HD]
+    float* red = reinterpret_cast<float*>(smem_raw + ((G * HD * 2 + 15) & ~15));
+    // red: [DEC_WAVES][G][2]
+
+    for (int i = threadIdx.x; i < G * HD; i += DEC_BLOCK) {
+        const int g = i / HD, d = i % HD;
+        q_s[i] = f2bf(
+            bf2f(q[(long)b * q_stride + (kvh * G + g) * (long)HD + d]) * scale);
+    }
+    __syncthreads();
+
+    // B-fragments: col = q head (li), k = dim lg*8+e within each kstep
+    bf16x8 bq[KSTEPS];
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+        if (li < G) {
+            bq[ks] = *reinterpret_cast<const bf16x8*>(
+                q_s + li * HD + ks * 32 + lg * 8);
+        } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) bq[ks][e] = (__bf16)0.f;
+        }
+    }
+
+    const int* bt = block_table + (long)b * W;
+    const int wave_key0 = start + wid * WAVE;  // this wave covers 64 keys
+
+    // scores: per tile of 16 keys, A streams from the cache rows
+    f32x4 acc[TILES];
+#pragma unroll
+    for (int t = 0; t < TILES; ++t) {
+        const int key = wave_key0 + t * 16 + li;  // A row = li
+        const unsigned short* kr = nullptr;
+        if (key < L) {
+            const int page = bt[key / bs];
+            kr = k_cache + (((long)page * nkv + kvh) * bs + key % bs) * HD;
+        }
+        f32x4 c{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < KSTEPS; ++ks) {
+            bf16x8 a;
+            if (kr != nullptr) {
+                a = *reinterpret_cast<const bf16x8*>(kr + ks * 32 + lg * 8);
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e) a[e] = (__bf16)0.f;
+            }
+            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bq[ks], c, 0, 0, 0);
+        }
+        acc[t] = c;
+    }
+
+    // this lane holds scores for g = li, keys (t*16 + lg*4 + r)
+    float sv[TILES * 4];
+    float lmax = -1e30f;
+#pragma unroll
+    for (int t = 0; t < TILES; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int key = wave_key0 + t * 16 + lg * 4 + r;
+            float v = (key < L) ? acc[t][r] : -1e30f;
+            sv[t * 4 + r] = v;
+            lmax = fmaxf(lmax, v);
+        }
+    }
+    // reduce over the 4 lane groups holding the same g (stride-16 lanes)
+    lmax = fmaxf(lmax, __shfl_xor(lmax, 16));
+    lmax = fmaxf(lmax, __shfl_xor(lmax, 32));
+    if (lg == 0 && li < G) red[(wid * G + li) * 2] = lmax;
+    __syncthreads();
+    float M = -1e30f;  // chunk max for this lane's g
+    if (li < G) {
+#pragma unroll
+        for (int w2 = 0; w2 < DEC_WAVES; ++w2)
+            M = fmaxf(M, red[(w2 * G + li) * 2]);
+    }
+    float lsum = 0.f;
+#pragma unroll
+    for (int i = 0; i < TILES * 4; ++i) {
+        const float pv = (sv[i] <= -1e29f) ? 0.f : __expf(sv[i] - M);
+        sv[i] = pv;
+        lsum += pv;
+    }
+    lsum += __shfl_xor(lsum, 16);
+    lsum += __shfl_xor(lsum, 32);
+    __syncthreads();  // red slot-0 reads done
+    if (lg == 0 && li < G) red[(wid * G + li) * 2 + 1] = lsum;
+
+    // p write-out: [key][G]; lanes with the same (t, r) write G consecutive
+    // floats for 4 key-groups -> 4 x 64 B segments per instruction
+    if (li < G) {
+        float* pbase =
+            p_out + (((long)b * nkv + kvh) * C + chunk) * DEC_CHUNK * G;
+#pragma unroll
+        for (int t = 0; t < TILES; ++t) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int key = wid * WAVE + t * 16 + lg * 4 + r;
+                pbase[key * G + li] = sv[t * 4 + r];
+            }
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float* pml = part_ml + ((((long)b * nkv + kvh) * C + chunk) * G) * 2;
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+            float ls = 0.f;
+            for (int w2 = 0; w2 < DEC_WAVES; ++w2)
+                ls += red[(w2 * G + g) * 2 + 1];
+            float mg = -1e30f;
+            for (int w2 = 0; w2 < DEC_WAVES; ++w2)
+                mg = fmaxf(mg, red[(w2 * G + g) * 2]);
+            pml[g * 2] = mg;
+            pml[g * 2 + 1] = ls;
+        }
+    }
+}
+
 // ----------------------------------------------------------- stage 2: PV
 template <int G>
 __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
@@ -322,10 +474,21 @@ extern "C" void launch_attn_decode(
         DEC_CHUNK * G * 4 + DEC_CHUNK * 8 + DEC_WAVES * G * hd * 4;
 #define LAUNCH(GG)                                                             \
     do {                                                                       \
-        hipLaunchKernelGGL(attn_scores_kernel<GG>, grid, dim3(DEC_BLOCK),      \
-                           smem_s, stream, q, k_cache, block_table, seq_lens,  \
-                           p_buf, part_ml, nkv, W, bs, hd, C, q_stride,        \
-                           scale);                                             \
+        if (hd == 128)                                                         \
+            hipLaunchKernelGGL((attn_scores_mfma_kernel<GG, 128>), grid,       \
+                               dim3(DEC_BLOCK), smem_s, stream, q, k_cache,    \
+                               block_table, seq_lens, p_buf, part_ml, nkv, W,  \
+                               bs, C, q_stride, scale);                        \
+        else if (hd == 64)                                                     \
+            hipLaunchKernelGGL((attn_scores_mfma_kernel<GG, 64>), grid,        \
+                               dim3(DEC_BLOCK), smem_s, stream, q, k_cache,    \
+                               block_table, seq_lens, p_buf, part_ml, nkv, W,  \
+                               bs, C, q_stride, scale);                        \
+        else                                                                   \
+            hipLaunchKernelGGL(attn_scores_kernel<GG>, grid, dim3(DEC_BLOCK),  \
+                               smem_s, stream, q, k_cache, block_table,        \
+                               seq_lens, p_buf, part_ml, nkv, W, bs, hd, C,    \
+                               q_stride, scale);                               \
         hipLaunchKernelGGL(attn_pv_kernel<GG>, grid, dim3(DEC_BLOCK),          \
                            smem_pv, stream, v_cache, block_table, seq_lens,    \
                            p_buf, part_o, nkv, W, bs, hd, C);                  \
