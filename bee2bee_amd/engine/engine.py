@@ -126,6 +126,9 @@ class InferenceEngine:
 
         self._pending: "queue.Queue[GenerationRequest]" = queue.Queue()
         self._active: List[_Active] = []
+        self._dec_seqs = None  # device-state cache key (active seq ids)
+        self._last_sampled: Optional[torch.Tensor] = None
+        self._eager_state = None
         self._next_seq = 0
         self._stop = False
         self._wake = threading.Event()
@@ -322,46 +325,91 @@ class InferenceEngine:
 
     @torch.no_grad()
     def _decode_once(self) -> None:
+        """One decode step over the active set.
+
+        Host work is incremental: the device-side ids/positions/lens/block
+        tables persist across steps and are bumped in place; full rebuilds
+        happen only when the active set changes membership or a sequence
+        crosses a KV-block boundary (every block_size tokens)."""
         B = len(self._active)
         dev = self.device
+        blocks_changed = False
         for a in self._active:
+            nblk0 = (a.length + self.kv.block_size - 1) // self.kv.block_size
             self.kv.extend_seq(a.seq_id, a.length + 1)
+            nblk1 = (a.length + self.kv.block_size) // self.kv.block_size
+            blocks_changed |= nblk1 != nblk0
 
-        last_ids = [
-            (a.req.output_ids[-1] if a.req.output_ids else a.req.prompt_ids[-1])
-            for a in self._active
-        ]
-        positions = [a.length for a in self._active]
-        seq_lens = [a.length + 1 for a in self._active]
+        seqs = [a.seq_id for a in self._active]
+        stable = (
+            seqs == self._dec_seqs
+            and self._last_sampled is not None
+            and self._last_sampled.shape[0] >= B
+        )
 
         if self.graphs is not None:
             g = self.graphs
             bucket = g.bucket_for(B)
-            g.input_ids[:B].copy_(
-                torch.tensor(last_ids, dtype=torch.int64), non_blocking=True
-            )
-            g.positions[:B].copy_(
-                torch.tensor(positions, dtype=torch.int32), non_blocking=True
-            )
-            g.seq_lens[:B].copy_(
-                torch.tensor(seq_lens, dtype=torch.int32), non_blocking=True
-            )
-            bt = self.kv.block_table([a.seq_id for a in self._active])
-            g.block_table[:B, : bt.shape[1]].copy_(bt, non_blocking=True)
-            if bucket > B:  # pad rows -> scratch sequence, length 1
-                sb = self.kv.block_table([self._scratch_seq])[0]
-                g.block_table[B:bucket, : sb.shape[0]].copy_(sb)
-                g.positions[B:bucket].zero_()
-                g.seq_lens[B:bucket].fill_(1)
-                g.input_ids[B:bucket].zero_()
+            if stable:
+                g.input_ids[:B].copy_(self._last_sampled[:B])
+                g.positions[:B].add_(1)
+                g.seq_lens[:B].add_(1)
+                if blocks_changed:
+                    bt = self.kv.block_table(seqs)
+                    g.block_table[:B, : bt.shape[1]].copy_(bt, non_blocking=True)
+            else:
+                last_ids = [
+                    (a.req.output_ids[-1] if a.req.output_ids else a.req.prompt_ids[-1])
+                    for a in self._active
+                ]
+                g.input_ids[:B].copy_(
+                    torch.tensor(last_ids, dtype=torch.int64), non_blocking=True
+                )
+                g.positions[:B].copy_(
+                    torch.tensor([a.length for a in self._active],
+                                 dtype=torch.int32), non_blocking=True
+                )
+                g.seq_lens[:B].copy_(
+                    torch.tensor([a.length + 1 for a in self._active],
+                                 dtype=torch.int32), non_blocking=True
+                )
+                bt = self.kv.block_table(seqs)
+                g.block_table[:B, : bt.shape[1]].copy_(bt, non_blocking=True)
+                if bucket > B:  # pad rows -> scratch sequence, length 1
+                    sb = self.kv.block_table([self._scratch_seq])[0]
+                    g.block_table[B:bucket, : sb.shape[0]].copy_(sb)
+                    g.positions[B:bucket].zero_()
+                    g.seq_lens[B:bucket].fill_(1)
+                    g.input_ids[B:bucket].zero_()
+                self._dec_seqs = list(seqs)
             logits = g.run(B)[:B]
         else:
-            input_ids = torch.tensor(last_ids, dtype=torch.int64, device=dev)
-            pos_t = torch.tensor(positions, dtype=torch.int32, device=dev)
-            lens_t = torch.tensor(seq_lens, dtype=torch.int32, device=dev)
-            bt = self.kv.block_table([a.seq_id for a in self._active])
-            slots = decode_slot_mapping(bt, pos_t, self.kv.block_size)
-            hidden = self.runner.forward_decode(input_ids, pos_t, slots, bt, lens_t)
+            st = self._eager_state
+            if stable and st is not None:
+                st["ids"].copy_(self._last_sampled[:B])
+                st["pos"].add_(1)
+                st["lens"].add_(1)
+                if blocks_changed:
+                    st["bt"] = self.kv.block_table(seqs)
+            else:
+                last_ids = [
+                    (a.req.output_ids[-1] if a.req.output_ids else a.req.prompt_ids[-1])
+                    for a in self._active
+                ]
+                st = {
+                    "ids": torch.tensor(last_ids, dtype=torch.int64, device=dev),
+                    "pos": torch.tensor([a.length for a in self._active],
+                                        dtype=torch.int32, device=dev),
+                    "lens": torch.tensor([a.length + 1 for a in self._active],
+                                         dtype=torch.int32, device=dev),
+                    "bt": self.kv.block_table(seqs),
+                }
+                self._eager_state = st
+                self._dec_seqs = list(seqs)
+            slots = decode_slot_mapping(st["bt"], st["pos"], self.kv.block_size)
+            hidden = self.runner.forward_decode(
+                st["ids"], st["pos"], slots, st["bt"], st["lens"]
+            )
             logits = self.runner.lm_head(hidden)
 
         for a in self._active:
@@ -370,7 +418,9 @@ class InferenceEngine:
         done = [a for a in self._active if a.req.done_ts is not None]
         for a in done:
             self.kv.free_seq(a.seq_id)
-        self._active = [a for a in self._active if a.req.done_ts is None]
+        if done:
+            self._active = [a for a in self._active if a.req.done_ts is None]
+            self._dec_seqs = None  # membership changed
 
     def _sample_and_emit(self, acts: List[_Active], logits: torch.Tensor) -> None:
         # group rows by sampling params so each group is one sample() call
@@ -380,13 +430,17 @@ class InferenceEngine:
             groups.setdefault(
                 (sp.greedy, sp.temperature, sp.top_p, sp.top_k), []
             ).append(i)
-        next_ids = torch.empty(len(acts), dtype=torch.int64)
+        next_dev = torch.empty(
+            len(acts), dtype=torch.int64, device=logits.device
+        )
         for key, rows in groups.items():
             sp = acts[rows[0]].req.sampling
             idx = torch.tensor(rows, dtype=torch.int64, device=logits.device)
             gen = self._gen if self.device.type == "cuda" else None
             toks = sample(logits[idx], sp, generator=gen)
-            next_ids[torch.tensor(rows)] = toks.to("cpu")
+            next_dev[idx] = toks
+        self._last_sampled = next_dev  # feeds the next step without H2D
+        next_ids = next_dev.cpu()  # the one host sync per step (emission)
         now = time.time()
         n_emitted = 0
         for i, a in enumerate(acts):

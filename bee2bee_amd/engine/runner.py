@@ -39,6 +39,9 @@ class Runner:
         self.dtype = dtype
         self.layer_lo, self.layer_hi = layer_range or (0, spec.n_layers)
         self.scale = spec.head_dim**-0.5
+        # optional expert parallelism (parallel/ep.py); when set, MoE layers
+        # dispatch tokens over RCCL all-to-all to the expert owners
+        self.ep = None
         cos, sin = ops.rope_tables(
             spec.max_seq_len, spec.head_dim, spec.rope_theta, device
         )
@@ -106,6 +109,12 @@ class Runner:
         """Top-k expert MLP. v1: per-expert gather/GEMM/scatter (dense GEMMs
         through hipBLASLt); grouped-GEMM kernel lands with the MoE milestone."""
         s = self.spec
+        if self.ep is not None:
+            lo = self.ep.e_lo
+            hi = lo + self.ep.local_e
+            return self.ep.forward(
+                x, lw.moe_gate, lw.moe_w_gate_up[lo:hi], lw.moe_w_down[lo:hi]
+            )
         logits = F.linear(x, lw.moe_gate)
         weights, idx = ops.moe_topk_gate(logits, s.top_k_experts)  # [T,k]
         out = torch.zeros_like(x, dtype=torch.float32)

@@ -1,0 +1,71 @@
+"""Expert parallelism on CPU: gloo world_size 2, all-to-all dispatch must
+reproduce the single-process MoE MLP."""
+import multiprocessing as mp
+import pickle
+
+import pytest
+import torch
+
+from bee2bee_amd.models.spec import PRESETS
+
+SEED = 33
+
+
+def _single_reference():
+    from bee2bee_amd.engine.kv import PagedKV
+    from bee2bee_amd.engine.runner import Runner
+    from bee2bee_amd.models.weights import ModelWeights
+
+    spec = PRESETS["tiny-moe"]
+    torch.manual_seed(0)
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(SEED)
+    kv = PagedKV(spec, torch.device("cpu"), torch.float32, n_blocks=8)
+    runner = Runner(spec, w, kv, torch.device("cpu"), torch.float32)
+    x = torch.randn(7, spec.hidden_size, generator=torch.Generator().manual_seed(9))
+    return runner._moe_mlp(w.layers[0], x), x
+
+
+def _ep_worker(rank: int, world: int, port: int, out_path: str) -> None:
+    import torch.distributed as dist
+
+    from bee2bee_amd.models.weights import ModelWeights
+    from bee2bee_amd.parallel.ep import ExpertParallelMoE
+
+    dist.init_process_group(
+        backend="gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world,
+    )
+    try:
+        spec = PRESETS["tiny-moe"]
+        w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(SEED)
+        lw = w.layers[0]
+        ep = ExpertParallelMoE(spec.n_experts, spec.top_k_experts)
+        x = torch.randn(
+            7, spec.hidden_size, generator=torch.Generator().manual_seed(9)
+        )
+        lo, hi = ep.e_lo, ep.e_lo + ep.local_e
+        out = ep.forward(x, lw.moe_gate, lw.moe_w_gate_up[lo:hi], lw.moe_w_down[lo:hi])
+        if rank == 0:
+            with open(out_path, "wb") as f:
+                pickle.dump(out, f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_ep2_matches_single(tmp_path):
+    ref, _x = _single_reference()
+    out_path = str(tmp_path / "ep_out.pkl")
+    ctx = mp.get_context("spawn")
+    procs = [
+        ctx.Process(target=_ep_worker, args=(r, 2, 29617, out_path))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=200)
+        assert p.exitcode == 0
+    with open(out_path, "rb") as f:
+        ep_out = pickle.load(f)
+    assert torch.allclose(ep_out, ref, atol=1e-5), (ep_out - ref).abs().max()
