@@ -69,7 +69,12 @@ class InferenceEngine:
         use_graphs: Optional[bool] = None,
         seed: int = 0,
         kv_margin_blocks: int = 8,
+        max_prefill_tokens: int = 4096,
     ) -> None:
+        # cap on prompt tokens prefill-batched per step: bounds time-to-first
+        # -token for requests behind a burst (they decode while later
+        # arrivals prefill)
+        self.max_prefill_tokens = max_prefill_tokens
         self.spec = model if isinstance(model, ModelSpec) else resolve_spec(model, model_path)
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
@@ -275,7 +280,10 @@ class InferenceEngine:
 
     def _admit(self) -> List[_Active]:
         admitted: List[_Active] = []
+        admit_tokens = 0
         while len(self._active) + len(admitted) < self.max_batch:
+            if admitted and admit_tokens >= self.max_prefill_tokens:
+                break
             try:
                 req = self._pending.get_nowait()
             except queue.Empty:
@@ -292,6 +300,7 @@ class InferenceEngine:
             self._next_seq += 1
             self.kv.new_seq(seq_id)
             admitted.append(_Active(req, seq_id, 0))
+            admit_tokens += len(req.prompt_ids)
         return admitted
 
     @torch.no_grad()
