@@ -87,7 +87,7 @@ __global__ __launch_bounds__(64) void attn_prefill_basic(
 #define PF_QROWS 16   // q rows per wave
 #define PF_WAVES 4
 #define PF_TM (PF_QROWS * PF_WAVES)  // 64-row q tile per workgroup
-#define PF_TN 32                     // kv tile
+#define PF_TN 64                     // kv tile (4 x 16-key fragments)
 #define PF_PAD 8                     // LDS row padding (bf16 elems)
 
 template <int HD>
@@ -183,10 +183,10 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
         }
         __syncthreads();
 
-        // ---- S = Q K^T : two 16-key column fragments
-        f32x4 sfrag[2];
+        // ---- S = Q K^T : PF_TN/16 16-key column fragments
+        f32x4 sfrag[PF_TN / 16];
 #pragma unroll
-        for (int f = 0; f < 2; ++f) {
+        for (int f = 0; f < PF_TN / 16; ++f) {
             f32x4 acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int ks = 0; ks < KSTEPS; ++ks) {
@@ -204,7 +204,7 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
 #pragma unroll
         for (int r = 0; r < 4; ++r) pmax[r] = -1e30f;
 #pragma unroll
-        for (int f = 0; f < 2; ++f) {
+        for (int f = 0; f < PF_TN / 16; ++f) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int qrow = q0 + wid * PF_QROWS + lg * 4 + r;
@@ -234,7 +234,7 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
         unsigned short* p_s = k_s;  // reuse; careful with sizes (TM*(TN+8))
         __syncthreads();  // everyone done reading K before overwrite
 #pragma unroll
-        for (int f = 0; f < 2; ++f) {
+        for (int f = 0; f < PF_TN / 16; ++f) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 float pv = __expf(sfrag[f][r] - m_run[r]);
@@ -263,14 +263,20 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
         // (p_s writes above are within-wave for our rows; but other waves
         // share the LDS buffer -> sync so K-tile reuse is safe)
         __syncthreads();
-        bf16x8 a_p = *reinterpret_cast<const bf16x8*>(
-            p_s + (wid * PF_QROWS + li) * VT_STRIDE + lg * 8);
+        bf16x8 a_p[PF_TN / 32];  // A k-dim = keys: 32 keys per fragment
+#pragma unroll
+        for (int ks2 = 0; ks2 < PF_TN / 32; ++ks2)
+            a_p[ks2] = *reinterpret_cast<const bf16x8*>(
+                p_s + (wid * PF_QROWS + li) * VT_STRIDE + ks2 * 32 + lg * 8);
 #pragma unroll
         for (int nb = 0; nb < HD / 16; ++nb) {
-            bf16x8 bv = *reinterpret_cast<const bf16x8*>(
-                vt_s + (nb * 16 + li) * VT_STRIDE + lg * 8);
-            o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, bv,
-                                                                o_acc[nb], 0, 0, 0);
+#pragma unroll
+            for (int ks2 = 0; ks2 < PF_TN / 32; ++ks2) {
+                bf16x8 bv = *reinterpret_cast<const bf16x8*>(
+                    vt_s + (nb * 16 + li) * VT_STRIDE + ks2 * 32 + lg * 8);
+                o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_p[ks2], bv, o_acc[nb], 0, 0, 0);
+            }
         }
     }
 

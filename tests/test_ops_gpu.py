@@ -184,3 +184,27 @@ def test_attn_decode_long_context():
     got = ops.attn_decode(q, kc, vc, bt, lens, hd**-0.5)
     ref = R.attn_decode(q, kc, vc, bt, lens, hd**-0.5)
     assert_close(got, ref, msg="long decode")
+
+
+def test_attn_prefill_stress():
+    """Large-shape prefill stress (8k blocks in flight): watches for the
+    intermittent fault seen once on one box; checks finiteness + one
+    sequence against the reference."""
+    torch.manual_seed(9)
+    nq, nkv, hd = 32, 8, 128
+    B, L = 8, 2048
+    T = B * L
+    q = (torch.randn(T, nq, hd, device=DEV) * 0.5).bfloat16()
+    k = (torch.randn(T, nkv, hd, device=DEV) * 0.5).bfloat16()
+    v = (torch.randn(T, nkv, hd, device=DEV) * 0.5).bfloat16()
+    cu = torch.arange(0, T + 1, L, dtype=torch.int32, device=DEV)
+    scale = hd**-0.5
+    for _ in range(10):
+        out = ops.attn_prefill(q, k, v, cu, L, scale, True)
+    torch.cuda.synchronize()
+    assert torch.isfinite(out.float()).all()
+    ref1 = R.attn_prefill(
+        q[:L], k[:L], v[:L],
+        torch.tensor([0, L], dtype=torch.int32, device=DEV), L, scale, True,
+    )
+    assert_close(out[:L], ref1, atol=3e-2, rtol=3e-2, msg="prefill stress seq0")
