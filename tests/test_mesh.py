@@ -243,3 +243,28 @@ def test_disconnect_reaps_provider():
         await a.stop()
 
     asyncio.run(run())
+
+
+def test_example_mesh_client():
+    """The standalone example client (examples/mesh_client.py) speaks the
+    wire protocol against a live node."""
+    import sys
+    sys.path.insert(0, ".")
+    from examples.mesh_client import MeshClient
+
+    async def run():
+        node = await _start_node()
+        await node.add_service(EchoService())
+        client = MeshClient()
+        await client.connect(f"ws://127.0.0.1:{node.port}")
+        assert "hf" in client.providers
+        res = await client.generate("ping", model="echo-model", max_new_tokens=8)
+        assert res["text"] == "echo:ping"
+        chunks = []
+        await client.generate("a b c", model="echo-model", stream=True,
+                              on_chunk=chunks.append)
+        assert "".join(chunks).strip() == "echo:a b c"
+        await client.close()
+        await node.stop()
+
+    asyncio.run(run())
