@@ -439,24 +439,30 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
     int nkv, int hd, int C) {
     const int b = blockIdx.x;
     const int kvh = blockIdx.y;
+    const int g = blockIdx.z;  // one block per (seq, kv head, q head)
     const int nq = nkv * G;
     const int L = seq_lens[b];
     const int nc = (L + DEC_CHUNK - 1) / DEC_CHUNK;
     const long base = ((long)b * nkv + kvh) * C;
 
-    for (int i = threadIdx.x; i < G * hd; i += DEC_BLOCK) {
-        const int g = i / hd, d = i % hd;
-        float M = -1e30f;
+    // stage this head's (m, l) pairs once
+    __shared__ float ml_s[64][2];  // C is far below 64 in practice
+    for (int c = threadIdx.x; c < nc; c += DEC_BLOCK) {
+        ml_s[c][0] = part_ml[((base + c) * G + g) * 2];
+        ml_s[c][1] = part_ml[((base + c) * G + g) * 2 + 1];
+    }
+    __syncthreads();
+    float M = -1e30f;
+    for (int c = 0; c < nc; ++c) M = fmaxf(M, ml_s[c][0]);
+    float Ltot = 0.f;
+    for (int c = 0; c < nc; ++c) Ltot += __expf(ml_s[c][0] - M) * ml_s[c][1];
+    const float inv = (Ltot > 0.f) ? 1.f / Ltot : 0.f;
+
+    for (int d = threadIdx.x; d < hd; d += DEC_BLOCK) {
+        float acc = 0.f;
         for (int c = 0; c < nc; ++c)
-            M = fmaxf(M, part_ml[((base + c) * G + g) * 2]);
-        float Ltot = 0.f, acc = 0.f;
-        for (int c = 0; c < nc; ++c) {
-            const float* ml = part_ml + ((base + c) * G + g) * 2;
-            const float f = __expf(ml[0] - M);
-            Ltot += f * ml[1];
-            acc += f * part_o[((base + c) * G + g) * hd + d];
-        }
-        const float inv = (Ltot > 0.f) ? 1.f / Ltot : 0.f;
+            acc += __expf(ml_s[c][0] - M) *
+                   part_o[((base + c) * G + g) * hd + d];
         out[((long)b * nq + kvh * G + g) * hd + d] = f2bf(acc * inv);
     }
 }
@@ -468,7 +474,7 @@ extern "C" void launch_attn_decode(
     unsigned short* out, int B, int nkv, int G, int W, int bs, int hd, int C,
     long q_stride, float scale, hipStream_t stream) {
     dim3 grid(B, nkv, C);
-    dim3 cgrid(B, nkv);
+    dim3 cgrid(B, nkv, G);
     const int smem_s = ((G * hd * 2 + 15) & ~15) + DEC_WAVES * G * 2 * 4;
     const int smem_pv =
         DEC_CHUNK * G * 4 + DEC_CHUNK * 8 + DEC_WAVES * G * hd * 4;
