@@ -113,10 +113,12 @@ class InferenceEngine:
         self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks)
         self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
 
-        # MoE routing (topk + gather/scatter) has data-dependent shapes and
-        # is not yet capture-safe; grouped-GEMM fixed-shape dispatch will
-        # lift this restriction
-        graphs_ok = on_gpu and not self.spec.is_moe
+        # MoE decode uses the dense all-experts path (static shapes), so it
+        # is capture-safe as long as the decode batch stays under the dense
+        # threshold — enforced by max_batch here
+        graphs_ok = on_gpu and (
+            not self.spec.is_moe or max_batch <= Runner.MOE_DENSE_MAX_TOKENS
+        )
         self.use_graphs = graphs_ok if use_graphs is None else (use_graphs and graphs_ok)
         self.graphs: Optional[DecodeGraphs] = None
         if self.use_graphs:

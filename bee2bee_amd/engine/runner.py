@@ -105,9 +105,16 @@ class Runner:
             mlp_out = F.linear(ops.swiglu(gate_up), lw.w_down)
         return mlp_out, residual
 
+    # below this many tokens the MoE layer runs all experts densely: at
+    # decode batch sizes every expert's weights stream from HBM anyway
+    # (bandwidth-bound), a batched GEMM reads them exactly once with full
+    # MFMA efficiency, and every shape is static -> hipGraph-capturable
+    # (the per-expert gather loop was 768 tiny launches per Mixtral step)
+    MOE_DENSE_MAX_TOKENS = 256
+
     def _moe_mlp(self, lw, x: torch.Tensor) -> torch.Tensor:
-        """Top-k expert MLP. v1: per-expert gather/GEMM/scatter (dense GEMMs
-        through hipBLASLt); grouped-GEMM kernel lands with the MoE milestone."""
+        """Top-k expert MLP: dense all-experts bmm for decode-sized batches,
+        per-expert gather/GEMM/scatter for prefill-sized ones."""
         s = self.spec
         if self.ep is not None:
             lo = self.ep.e_lo
@@ -117,8 +124,21 @@ class Runner:
             )
         logits = F.linear(x, lw.moe_gate)
         weights, idx = ops.moe_topk_gate(logits, s.top_k_experts)  # [T,k]
+        T, H = x.shape
+        E, I = s.n_experts, s.intermediate_size
+
+        if T <= self.MOE_DENSE_MAX_TOKENS:
+            x_e = x.unsqueeze(0).expand(E, T, H)
+            gu = torch.bmm(x_e, lw.moe_w_gate_up.transpose(1, 2))  # [E,T,2I]
+            act = ops.swiglu(gu.reshape(E * T, 2 * I)).view(E, T, I)
+            y = torch.bmm(act, lw.moe_w_down.transpose(1, 2))  # [E,T,H]
+            wfull = torch.zeros(T, E, dtype=torch.float32, device=x.device)
+            wfull.scatter_(1, idx.long(), weights)
+            out = torch.einsum("eth,te->th", y.float(), wfull)
+            return out.to(x.dtype)
+
         out = torch.zeros_like(x, dtype=torch.float32)
-        for e in range(s.n_experts):
+        for e in range(E):
             mask = idx == e  # [T, k]
             if not bool(mask.any()):
                 continue
