@@ -26,8 +26,11 @@ class ByteTokenizer:
         return ([self.bos_token_id] if add_bos else []) + ids
 
     def decode(self, ids: List[int]) -> str:
+        # total decode: ids past the byte range fold back onto bytes, so
+        # every sampled id yields text (a real tokenizer maps every id to a
+        # string; synthetic serving benchmarks rely on that for streaming)
         data = bytes(
-            i - self.n_special for i in ids if self.n_special <= i < 256 + self.n_special
+            (i - self.n_special) % 256 for i in ids if i >= self.n_special
         )
         return data.decode("utf-8", errors="replace")
 

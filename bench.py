@@ -39,9 +39,13 @@ def main() -> None:
                     help="skip hipBLASLt algorithm tuning")
     args = ap.parse_args()
 
+    from bee2bee_amd.models.spec import resolve_spec
+
+    _spec_probe = resolve_spec(args.model)
     # Tune hipBLASLt GEMM algorithm selection during (untimed) setup/warmup:
-    # worth ~4% on the skinny decode projections. Off on CPU.
-    if not args.no_tunableop and torch.cuda.is_available():
+    # worth ~4% on the skinny decode projections. Off on CPU and for MoE
+    # (tuning the batched expert GEMM shapes takes minutes).
+    if not args.no_tunableop and torch.cuda.is_available() and not _spec_probe.is_moe:
         os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
         os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
         os.environ.setdefault(
