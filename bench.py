@@ -122,7 +122,7 @@ def main() -> None:
 
     if rank == 0:
         result = {
-            "metric": "output tokens/sec (Llama-3-8B bf16 greedy decode, "
+            "metric": f"output tokens/sec ({spec.name} bf16 greedy decode, "
                       "replica peers)",
             "value": round(total_tps, 1),
             "unit": "tokens/s",
