@@ -96,17 +96,21 @@ class PipelineEngine:
         )
 
     # ------------------------------------------------------------ transport
+    # RCCL moves device tensors directly (xGMI); gloo (CPU tests, or a
+    # single-GPU multi-process test) stages through host memory.
+
+    @property
+    def _wire_gpu(self) -> bool:
+        return self.device.type == "cuda" and dist.get_backend(self.group) == "nccl"
 
     def _send(self, t: torch.Tensor, dst: int) -> None:
-        if t.device.type == "cuda":
-            dist.send(t.contiguous(), dst, group=self.group)
-        else:
-            dist.send(t.contiguous().cpu(), dst, group=self.group)
+        t = t.contiguous()
+        dist.send(t if self._wire_gpu else t.cpu(), dst, group=self.group)
 
     def _recv(self, shape, dtype, src: int) -> torch.Tensor:
         buf = torch.empty(
             shape, dtype=dtype,
-            device=self.device if self.device.type == "cuda" else "cpu",
+            device=self.device if self._wire_gpu else "cpu",
         )
         dist.recv(buf, src, group=self.group)
         return buf.to(self.device)
@@ -115,10 +119,7 @@ class PipelineEngine:
         src = self.world - 1
         if ids is None:
             ids = torch.zeros(B, dtype=torch.int64)
-        buf = ids.cpu() if self.device.type != "cuda" else ids.to(self.device)
-        if self.device.type == "cuda":
-            dist.broadcast(buf, src, group=self.group)
-            return buf
+        buf = ids.to(self.device) if self._wire_gpu else ids.cpu()
         dist.broadcast(buf, src, group=self.group)
         return buf
 
