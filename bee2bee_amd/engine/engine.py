@@ -205,18 +205,25 @@ class InferenceEngine:
         emitted: List[str] = []
         decoded_upto = 0
         held = 0
+        n_seen = 0
+        # emit the first token at once (TTFT), then flush every few tokens:
+        # per-token queue/HTTP hops dominate the serving path at high
+        # concurrency (dozens of thread wakeups per decode step otherwise)
+        flush_every = 4
 
         def _on_token(tok: int) -> None:
-            nonlocal decoded_upto, held
+            nonlocal decoded_upto, held, n_seen
             if on_text is None:
+                return
+            n_seen += 1
+            if n_seen > 1 and n_seen % flush_every != 1:
                 return
             text = self.tokenizer.decode(req.output_ids)
             delta = text[decoded_upto:]
             if not delta:
                 return
             # hold back a possibly-incomplete multibyte char at the boundary,
-            # but never more than 3 tokens (random-weight output is mostly
-            # U+FFFD and would otherwise buffer the whole stream)
+            # but never more than 3 flushes
             if delta.endswith("�") and held < 3:
                 held += 1
                 return
