@@ -45,6 +45,7 @@ class GenerationRequest:
     first_token_ts: Optional[float] = None
     done_ts: Optional[float] = None
     error: Optional[str] = None
+    cancelled: bool = False
     output_ids: List[int] = field(default_factory=list)
 
 
@@ -197,8 +198,12 @@ class InferenceEngine:
         max_new_tokens: int = 128,
         temperature: Optional[float] = 0.7,
         on_text=None,
+        stop: Optional[List[str]] = None,
+        repetition_penalty: float = 1.0,
     ) -> Dict[str, Any]:
-        """Text-level wrapper used by the mesh service."""
+        """Text-level wrapper used by the mesh service. `stop` strings
+        truncate the output at the first occurrence (reference stop-word
+        scan, bee2bee/hf.py:111-136)."""
         t0 = time.time()
         ids = self.tokenizer.encode(prompt)
         ids = ids[-(self.max_seq_len - max_new_tokens - 1) :]
@@ -236,10 +241,12 @@ class InferenceEngine:
         eos = getattr(self.tokenizer, "eos_token_id", None)
         if eos is not None:
             stop_ids = (eos,)
+        sp = SamplingParams.from_request(temperature)
+        sp.repetition_penalty = repetition_penalty
         req = GenerationRequest(
             prompt_ids=ids,
             max_new_tokens=max_new_tokens,
-            sampling=SamplingParams.from_request(temperature),
+            sampling=sp,
             stop_token_ids=stop_ids,
         )
         self.submit(req)
@@ -251,6 +258,13 @@ class InferenceEngine:
         if req.error:
             raise RuntimeError(req.error)
         text = self.tokenizer.decode(req.output_ids)
+        if stop:
+            cut = min(
+                (text.find(w) for w in stop if w and text.find(w) >= 0),
+                default=-1,
+            )
+            if cut >= 0:
+                text = text[:cut]
         if on_text is not None and decoded_upto < len(text):
             on_text(text[decoded_upto:])
         return {
@@ -455,7 +469,23 @@ class InferenceEngine:
             sp = acts[rows[0]].req.sampling
             idx = torch.tensor(rows, dtype=torch.int64, device=logits.device)
             gen = self._gen if self.device.type == "cuda" else None
-            toks = sample(logits[idx], sp, generator=gen)
+            grp_logits = logits[idx]
+            if sp.repetition_penalty != 1.0:
+                from .sampler import apply_repetition_penalty
+
+                width = max(
+                    (len(acts[r].req.output_ids) + len(acts[r].req.prompt_ids))
+                    for r in rows
+                )
+                prev = torch.full((len(rows), width), -1, dtype=torch.int64)
+                for j, r in enumerate(rows):
+                    ids = acts[r].req.prompt_ids + acts[r].req.output_ids
+                    prev[j, : len(ids)] = torch.tensor(ids, dtype=torch.int64)
+                grp_logits = apply_repetition_penalty(
+                    grp_logits.float(), prev.to(grp_logits.device),
+                    sp.repetition_penalty,
+                )
+            toks = sample(grp_logits, sp, generator=gen)
             next_dev[idx] = toks
         self._last_sampled = next_dev  # feeds the next step without H2D
         next_ids = next_dev.cpu()  # the one host sync per step (emission)
@@ -469,7 +499,8 @@ class InferenceEngine:
             r.output_ids.append(tok)
             r.out_queue.put(tok)
             n_emitted += 1
-            if tok in r.stop_token_ids or len(r.output_ids) >= r.max_new_tokens:
+            if (r.cancelled or tok in r.stop_token_ids
+                    or len(r.output_ids) >= r.max_new_tokens):
                 r.done_ts = now
                 r.out_queue.put(_STREAM_END)
         self._note_throughput(n_emitted, now)

@@ -79,13 +79,17 @@ class NativeEngineService(BaseService):
         max_new = int(params.get("max_new_tokens", self.max_new_tokens))
         max_new = max(1, min(max_new, self.max_new_tokens))
         temperature = float(params.get("temperature", 0.7))
-        return prompt, max_new, temperature
+        extra = {
+            "stop": params.get("stop"),
+            "repetition_penalty": float(params.get("repetition_penalty", 1.0)),
+        }
+        return prompt, max_new, temperature, extra
 
     def execute(self, params: Dict[str, Any]) -> Dict[str, Any]:
-        prompt, max_new, temperature = self._check(params)
+        prompt, max_new, temperature, extra = self._check(params)
         try:
             t0 = time.time()
-            res = self.engine.generate_text(prompt, max_new, temperature)
+            res = self.engine.generate_text(prompt, max_new, temperature, **extra)
             latency_ms = int((time.time() - t0) * 1000.0)
             tokens = res["tokens"]
             return {
@@ -103,14 +107,14 @@ class NativeEngineService(BaseService):
             raise ServiceError(str(e)) from e
 
     def execute_stream(self, params: Dict[str, Any]) -> Iterator[str]:
-        prompt, max_new, temperature = self._check(params)
+        prompt, max_new, temperature, extra = self._check(params)
         chunks: "queue.Queue" = queue.Queue()
         DONE = object()
 
         def _run() -> None:
             try:
                 self.engine.generate_text(
-                    prompt, max_new, temperature, on_text=chunks.put
+                    prompt, max_new, temperature, on_text=chunks.put, **extra
                 )
                 chunks.put(DONE)
             except Exception as e:  # noqa: BLE001
