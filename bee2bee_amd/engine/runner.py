@@ -181,13 +181,16 @@ class Runner:
 
         def attn(layer_idx, q, k, v, k_cache, v_cache):
             if chunked:
-                from ..ops import reference as _ref  # history path: paged
-
-                if q.is_cuda and ops.hip_available():
-                    return ops.require_hip().attn_prefill_paged(
-                        q, k_cache, v_cache, block_table, seq_lens, query_lens,
-                        cu_seqlens, self.scale,
+                # chunked prefill against paged history: CPU reference only
+                # for now (the engine prefills whole prompts; a paged-history
+                # HIP prefill kernel is queued for the next round)
+                if q.is_cuda:
+                    raise NotImplementedError(
+                        "chunked prefill is not yet implemented on the HIP "
+                        "path; prefill whole prompts (engine default)"
                     )
+                from ..ops import reference as _ref
+
                 return _ref.attn_decode_with_history(
                     q, k_cache, v_cache, block_table, seq_lens, query_lens, self.scale
                 )
