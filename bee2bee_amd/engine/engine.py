@@ -32,6 +32,35 @@ logger = logging.getLogger("bee2bee_amd.engine")
 _STREAM_END = object()
 
 
+class TextStreamDecoder:
+    """Incrementally turns a growing token list into text deltas: emits the
+    first token at once (TTFT), then every `flush_every` tokens, holding
+    back a possibly-incomplete multibyte char (at most 3 flushes)."""
+
+    def __init__(self, tokenizer, flush_every: int = 4) -> None:
+        self.tokenizer = tokenizer
+        self.flush_every = flush_every
+        self.decoded_upto = 0
+        self.held = 0
+        self.n_seen = 0
+
+    def delta(self, output_ids: List[int], final: bool = False) -> str:
+        if not final:
+            self.n_seen += 1
+            if self.n_seen > 1 and self.n_seen % self.flush_every != 1:
+                return ""
+        text = self.tokenizer.decode(output_ids)
+        d = text[self.decoded_upto :]
+        if not d:
+            return ""
+        if not final and d.endswith("\ufffd") and self.held < 3:
+            self.held += 1
+            return ""
+        self.held = 0
+        self.decoded_upto = len(text)
+        return d
+
+
 @dataclass
 class GenerationRequest:
     prompt_ids: List[int]
@@ -47,6 +76,10 @@ class GenerationRequest:
     error: Optional[str] = None
     cancelled: bool = False
     output_ids: List[int] = field(default_factory=list)
+    # optional: called from the engine thread as on_emit(token_id, done);
+    # when set, tokens are NOT pushed to out_queue (async consumers use this
+    # to avoid one blocking thread per request)
+    on_emit: Optional[Any] = None
 
 
 class _Active:
@@ -284,7 +317,13 @@ class InferenceEngine:
                 logger.exception("engine step failed")
                 for a in self._active:
                     a.req.error = str(e)
-                    a.req.out_queue.put(_STREAM_END)
+                    if a.req.on_emit is not None:
+                        try:
+                            a.req.on_emit(None, True)
+                        except Exception:
+                            pass
+                    else:
+                        a.req.out_queue.put(_STREAM_END)
                     self.kv.free_seq(a.seq_id)
                 self._active.clear()
                 did_work = True
@@ -497,12 +536,20 @@ class InferenceEngine:
             if r.first_token_ts is None:
                 r.first_token_ts = now
             r.output_ids.append(tok)
-            r.out_queue.put(tok)
             n_emitted += 1
-            if (r.cancelled or tok in r.stop_token_ids
-                    or len(r.output_ids) >= r.max_new_tokens):
+            done = (r.cancelled or tok in r.stop_token_ids
+                    or len(r.output_ids) >= r.max_new_tokens)
+            if done:
                 r.done_ts = now
-                r.out_queue.put(_STREAM_END)
+            if r.on_emit is not None:
+                try:
+                    r.on_emit(tok, done)
+                except Exception:
+                    logger.exception("on_emit callback failed")
+            else:
+                r.out_queue.put(tok)
+                if done:
+                    r.out_queue.put(_STREAM_END)
         self._note_throughput(n_emitted, now)
 
     def _note_throughput(self, n: int, now: float) -> None:

@@ -192,6 +192,10 @@ async def chat(req: ChatRequest):
                 "temperature": req.temperature or 0.7,
             }
             if req.stream:
+                if hasattr(svc, "execute_stream_async"):
+                    return StreamingResponse(
+                        svc.execute_stream_async(params), media_type="text/plain"
+                    )
                 # pump the sync generator in a worker thread
                 queue: asyncio.Queue = asyncio.Queue()
 

@@ -403,7 +403,18 @@ class MeshNode:
 
         if svc is not None:
             try:
-                if data.get("stream"):
+                if data.get("stream") and hasattr(svc, "execute_stream_async"):
+                    async for chunk_raw in svc.execute_stream_async(params):
+                        try:
+                            text = json.loads(chunk_raw).get("text", "")
+                        except Exception:
+                            text = ""
+                        if text:
+                            await self._send(ws, wire.gen_chunk(rid, text))
+                    closing = {"text": "", "backend": "bee2bee-amd"}
+                    await self._send(ws, wire.gen_success(rid, closing))
+                    await self._send(ws, wire.gen_result(rid, closing))  # Q1
+                elif data.get("stream"):
                     # pump the (sync, blocking) stream generator from a thread
                     queue: asyncio.Queue = asyncio.Queue()
 

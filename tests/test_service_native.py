@@ -46,3 +46,33 @@ def test_execute_stream_jsonlines(svc):
     streamed = "".join(p.get("text", "") for p in parsed[:-1])
     buffered = svc.execute({"prompt": "abc", "max_new_tokens": 4, "temperature": 0.0})
     assert streamed == buffered["text"]
+
+
+def test_execute_stream_async(svc):
+    import asyncio
+
+    async def run():
+        lines = []
+        async for line in svc.execute_stream_async(
+            {"prompt": "abc", "max_new_tokens": 6, "temperature": 0.0}
+        ):
+            lines.append(line)
+        return lines
+
+    lines = asyncio.run(run())
+    parsed = [json.loads(l) for l in lines]
+    assert parsed[-1] == {"done": True}
+    streamed = "".join(p.get("text", "") for p in parsed[:-1])
+    buffered = svc.execute({"prompt": "abc", "max_new_tokens": 6, "temperature": 0.0})
+    assert streamed == buffered["text"]
+
+
+def test_stop_strings(svc):
+    full = svc.execute({"prompt": "xy", "max_new_tokens": 8, "temperature": 0.0})
+    if len(full["text"]) >= 3:
+        stopw = full["text"][1:3]
+        res = svc.execute(
+            {"prompt": "xy", "max_new_tokens": 8, "temperature": 0.0,
+             "stop": [stopw]}
+        )
+        assert stopw not in res["text"]
