@@ -100,6 +100,7 @@ class MeshNode:
         self.pieces: Dict[str, Dict[str, Any]] = {}
 
         self._lock = asyncio.Lock()
+        self._bootstrap_addrs: List[str] = []  # reconnect targets
         self._pending: Dict[str, asyncio.Future] = {}
         self._chunk_cbs: Dict[str, Callable[[str], None]] = {}
         self._running = False
@@ -118,7 +119,10 @@ class MeshNode:
         app.router.add_get("/ws", self._ws_endpoint)
         self._runner = web.AppRunner(app)
         await self._runner.setup()
-        self._site = web.TCPSite(self._runner, self.host, self.port)
+        # short shutdown grace: don't hang stop() on live WS handlers
+        self._site = web.TCPSite(
+            self._runner, self.host, self.port, shutdown_timeout=2.0
+        )
         await self._site.start()
         self._running = True
 
@@ -201,26 +205,36 @@ class MeshNode:
         else:
             addrs = [link_or_addr]
         for addr in addrs:
+            if addr not in self._bootstrap_addrs:
+                self._bootstrap_addrs.append(addr)
+        for addr in addrs:
             try:
                 await self._connect_peer(addr)
                 return
             except Exception as e:
                 logger.warning("bootstrap connect failed %s: %s", addr, e)
-        logger.error("all bootstrap connections failed")
+        logger.error("all bootstrap connections failed (will keep retrying)")
 
     async def _connect_peer(self, addr: str) -> None:
         if addr == self.addr:
             return
         assert self._session is not None, "node not started"
         try:
-            ws = await self._session.ws_connect(
-                addr, max_msg_size=wire.MAX_FRAME, heartbeat=None
+            ws = await asyncio.wait_for(
+                self._session.ws_connect(
+                    addr, max_msg_size=wire.MAX_FRAME, heartbeat=None
+                ),
+                timeout=10.0,
             )
         except Exception as e:
             # wss→ws fallback for local/dev SSL mismatch (reference :354-361)
             if addr.startswith("wss://"):
-                ws = await self._session.ws_connect(
-                    addr.replace("wss://", "ws://"), max_msg_size=wire.MAX_FRAME
+                ws = await asyncio.wait_for(
+                    self._session.ws_connect(
+                        addr.replace("wss://", "ws://"),
+                        max_msg_size=wire.MAX_FRAME,
+                    ),
+                    timeout=10.0,
                 )
             else:
                 raise IOError(f"could not connect to {addr}: {e}") from e
@@ -714,11 +728,27 @@ class MeshNode:
         while self._monitor_active and self._running:
             try:
                 await self._run_health_checks()
+                await self._reconnect_bootstraps()
                 if self.registry.enabled:
                     await self.sync_with_registry()
             except Exception:
                 logger.exception("monitoring error")
             await asyncio.sleep(interval)
+
+    async def _reconnect_bootstraps(self) -> None:
+        """Elastic recovery: re-dial bootstrap peers whose connection
+        dropped (the reference's JS bridge reconnects after 5 s,
+        app/api/bridge.js:83-95; its Python node never did)."""
+        for addr in list(self._bootstrap_addrs):
+            alive = any(
+                p.addr == addr and not p.ws.closed for p in self.peers.values()
+            )
+            if not alive:
+                try:
+                    await self._connect_peer(addr)
+                    logger.info("reconnected bootstrap %s", addr)
+                except Exception as e:
+                    logger.debug("bootstrap %s still down: %s", addr, e)
 
     async def _run_health_checks(self) -> None:
         from ..utils import now_ms

@@ -268,3 +268,27 @@ def test_example_mesh_client():
         await node.stop()
 
     asyncio.run(run())
+
+
+def test_bootstrap_reconnect():
+    """If the bootstrap peer restarts, the node re-dials it from the
+    monitoring loop (elastic recovery)."""
+
+    async def run():
+        b = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await b.start()
+        fixed_port = b.port
+        a = await _start_node()
+        await a.connect_bootstrap(b.addr)
+        await _wait_for(lambda: b.peer_id in a.peers)
+        await b.stop()
+        await _wait_for(lambda: b.peer_id not in a.peers, timeout=8)
+        # restart the bootstrap on the same port
+        b2 = MeshNode(host="127.0.0.1", port=fixed_port, enable_nat=False)
+        await b2.start()
+        await a._reconnect_bootstraps()
+        await _wait_for(lambda: b2.peer_id in a.peers, timeout=8)
+        await a.stop()
+        await b2.stop()
+
+    asyncio.run(run())
