@@ -69,3 +69,65 @@ def test_ep2_matches_single(tmp_path):
     with open(out_path, "rb") as f:
         ep_out = pickle.load(f)
     assert torch.allclose(ep_out, ref, atol=1e-5), (ep_out - ref).abs().max()
+
+
+def _moe_engine_worker(rank, world, port, out_path):
+    import torch.distributed as dist
+
+    from bee2bee_amd.parallel.moe_engine import MoEEngine
+
+    dist.init_process_group(
+        backend="gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world,
+    )
+    try:
+        eng = MoEEngine("tiny-moe", device="cpu", max_batch=4, max_seq_len=64,
+                        seed=SEED)
+        prompts = [[rank * 10 + 1, 2, 3], [rank * 10 + 4, 5]]
+        outs = eng.generate(prompts, 5)
+        with open(f"{out_path}.{rank}", "wb") as f:
+            pickle.dump((prompts, outs), f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_moe_engine_matches_single(tmp_path):
+    """Each EP rank's generations equal the single-process engine run on
+    the same prompts (fp32 exact)."""
+    import pickle as pkl
+
+    out_path = str(tmp_path / "moe_eng")
+    ctx = mp.get_context("spawn")
+    procs = [
+        ctx.Process(target=_moe_engine_worker, args=(r, 2, 29641, out_path))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    single = InferenceEngine("tiny-moe", device="cpu", max_batch=4,
+                             max_seq_len=64, seed=SEED)
+    try:
+        for r in range(2):
+            with open(f"{out_path}.{r}", "rb") as f:
+                prompts, ep_outs = pkl.load(f)
+            for prompt, ep_out in zip(prompts, ep_outs):
+                req = GenerationRequest(
+                    prompt_ids=list(prompt), max_new_tokens=5,
+                    sampling=SamplingParams(greedy=True),
+                )
+                single.submit(req)
+                while True:
+                    item = req.out_queue.get(timeout=60)
+                    if not isinstance(item, int):
+                        break
+                assert req.output_ids == ep_out, (r, prompt)
+    finally:
+        single.shutdown()
