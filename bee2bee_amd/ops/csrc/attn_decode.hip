@@ -371,14 +371,14 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
         const long* voffs = voff_s + wave_start;
         const float* pw = p_s + wave_start * G;
         int t = 0;
-        for (; t + 16 <= nkeys; t += 16) {  // 16 V loads in flight (4 KB/wave)
-            short2v vv[16];
+        for (; t + 32 <= nkeys; t += 32) {  // 32 V loads in flight (8 KB/wave)
+            short2v vv[32];
 #pragma unroll
-            for (int j = 0; j < 16; ++j)
+            for (int j = 0; j < 32; ++j)
                 vv[j] = *reinterpret_cast<const short2v*>(
                     v_cache + voffs[t + j] + d0);
 #pragma unroll
-            for (int j = 0; j < 16; ++j) {
+            for (int j = 0; j < 32; ++j) {
                 const float v0 = bf2f((unsigned short)vv[j][0]);
                 const float v1 = bf2f((unsigned short)vv[j][1]);
                 const float* pt = pw + (t + j) * G;

@@ -292,3 +292,41 @@ def test_bootstrap_reconnect():
         await b2.stop()
 
     asyncio.run(run())
+
+
+def test_reference_quirk_messages_handled():
+    """A peer sending reference-shaped frames (task_id ids, max_tokens,
+    no-services hello) gets a correct response."""
+
+    async def run():
+        import aiohttp
+
+        node = await _start_node()
+        await node.add_service(EchoService())
+        async with aiohttp.ClientSession() as session:
+            async with session.ws_connect(f"ws://127.0.0.1:{node.port}") as ws:
+                await ws.send_str(json.dumps(
+                    {"type": "hello", "peer_id": "ref-peer", "addr": ""}
+                ))
+                # drain hello/peer_list/ping
+                await ws.send_str(json.dumps({
+                    "type": "gen_request", "task_id": "t-1", "svc": "hf",
+                    "prompt": "ref style", "max_tokens": 16,
+                }))
+                deadline = asyncio.get_event_loop().time() + 10
+                got = None
+                while asyncio.get_event_loop().time() < deadline:
+                    msg = await asyncio.wait_for(ws.receive(), timeout=10)
+                    data = json.loads(msg.data)
+                    if data.get("type") == "ping":
+                        await ws.send_str(json.dumps(
+                            {"type": "pong", "ts": data.get("ts")}
+                        ))
+                    if data.get("type") == "gen_success" and data.get("rid") == "t-1":
+                        got = data
+                        break
+                assert got is not None, "no gen_success for task_id request"
+                assert got["text"] == "echo:ref style"
+        await node.stop()
+
+    asyncio.run(run())

@@ -34,3 +34,32 @@ def test_request_id_legacy():
 
 def test_terminal_types():
     assert set(wire.TERMINAL_TYPES) == {"gen_result", "gen_success", "gen_error"}
+
+
+def test_reference_shaped_messages_parse():
+    """Messages exactly as the reference emits them must normalize cleanly
+    (quirk coverage: legacy task_id ids, max_tokens-only, no-services
+    hello)."""
+    # reference _handle_gen_request reads rid or task_id, max_tokens default
+    # 2048, temperature default 0.7 (p2p_runtime.py:574-586)
+    ref_req = {
+        "type": "gen_request",
+        "task_id": "task-42",
+        "svc": "hf",
+        "model": "llama3",
+        "prompt": "hi",
+        "max_tokens": 128,
+        "temperature": 0.2,
+        "stream": True,
+    }
+    assert wire.request_id(ref_req) == "task-42"
+    p = wire.request_params(ref_req)
+    assert p["max_new_tokens"] == 128 and p["temperature"] == 0.2
+
+    # reference buffered provider answers with gen_success (Q1): it must be
+    # terminal for us
+    assert "gen_success" in wire.TERMINAL_TYPES
+
+    # hello without services/metrics (early-handshake JS bridge shape)
+    bare = {"type": "hello", "peer_id": "p1", "addr": "ws://x:1"}
+    assert bare.get("services") is None  # nodes must tolerate absence
