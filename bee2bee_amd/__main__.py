@@ -181,6 +181,33 @@ def register(node_url, network, region, test):
     asyncio.run(_reg())
 
 
+@cli.command("seed-model")
+@click.option("--name", required=True, help="Checkpoint name to publish")
+@click.option("--path", required=True, help="HF checkpoint dir to seed")
+@click.option("--port", default=0, type=int)
+def seed_model(name, path, port):
+    """Seed a checkpoint's pieces to the mesh (torrent-style distribution)."""
+
+    async def _run():
+        from .mesh.dht import DHTNode
+        from .mesh.node import MeshNode
+        from .mesh.weightshare import seed_checkpoint
+
+        dht = DHTNode()
+        await dht.start()
+        node = MeshNode(port=port, enable_nat=False)
+        await node.start()
+        manifest = await seed_checkpoint(node, dht, name, path)
+        total = sum(f["bytes"] for f in manifest["files"])
+        click.echo(f"seeding '{name}': {len(manifest['files'])} files, "
+                   f"{total / 1e6:.1f} MB at {node.addr}")
+        click.echo("Ctrl+C to stop seeding")
+        while True:
+            await asyncio.sleep(15)
+
+    asyncio.run(_run())
+
+
 @cli.command("bench")
 @click.option("--model", default="llama3-8b")
 @click.option("--batch", default=64, type=int)

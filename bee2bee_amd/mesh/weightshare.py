@@ -1,0 +1,110 @@
+"""Torrent-style model-weight distribution over the mesh.
+
+The reference's tagline capability ("torrent-like LLM deployment",
+README.md.backup:4) existed as disconnected primitives — pieces.py content
+shards, dht.py piece announcements, and STUBBED piece transfer handlers
+(p2p_runtime.py:675-683). This module completes the loop:
+
+  seed_checkpoint(node, dht, dir)  -> split every file into hash-verified
+                                      pieces, serve them, announce to DHT
+  fetch_checkpoint(node, dht, ...) -> discover providers, pull pieces over
+                                      the wire protocol, verify hashes,
+                                      reassemble the checkpoint dir
+
+A fetched checkpoint is a normal HF-format dir that NativeEngineService /
+ModelWeights.load_hf consumes. Piece size defaults to 8 MiB (frames are
+base64 inside the 32 MiB WS limit).
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import Dict, List, Optional
+
+from ..utils import sha256_hex_bytes
+from . import dht as dht_mod
+from .pieces import piece_hashes, split_pieces, verify_and_reassemble
+
+DEFAULT_PIECE = 8 * 1024 * 1024
+
+
+def manifest_key(name: str) -> str:
+    return f"manifest:{name}"
+
+
+async def seed_checkpoint(
+    node,
+    dht: dht_mod.DHTNode,
+    name: str,
+    directory: str,
+    piece_size: int = DEFAULT_PIECE,
+) -> Dict:
+    """Split every file of a checkpoint dir into pieces, register them with
+    the node for serving, and publish a manifest + provider records."""
+    files = []
+    for fname in sorted(os.listdir(directory)):
+        path = os.path.join(directory, fname)
+        if not os.path.isfile(path):
+            continue
+        with open(path, "rb") as f:
+            data = f.read()
+        pieces = split_pieces(data, piece_size) if data else [b""]
+        hashes = piece_hashes(pieces)
+        content_hash = sha256_hex_bytes(data)
+        node.share_pieces(content_hash, pieces)
+        await dht_mod.announce_piece(dht, content_hash, node.addr)
+        files.append(
+            {
+                "name": fname,
+                "bytes": len(data),
+                "content_hash": content_hash,
+                "piece_hashes": hashes,
+                "piece_size": piece_size,
+            }
+        )
+    manifest = {"name": name, "files": files}
+    await dht.set(manifest_key(name), json.dumps(manifest))
+    return manifest
+
+
+async def fetch_checkpoint(
+    node,
+    dht: dht_mod.DHTNode,
+    name: str,
+    out_dir: str,
+    provider_peer_id: Optional[str] = None,
+) -> str:
+    """Pull a seeded checkpoint from the mesh into out_dir (hash-verified).
+
+    Providers are discovered per content hash from the DHT; the caller can
+    pin a peer id instead (it must already be connected)."""
+    raw = await dht.get(manifest_key(name))
+    if not raw:
+        raise FileNotFoundError(f"no manifest for '{name}' in DHT")
+    manifest = json.loads(raw)
+    os.makedirs(out_dir, exist_ok=True)
+    for entry in manifest["files"]:
+        pid = provider_peer_id
+        if pid is None:
+            providers = await dht_mod.find_providers(dht, entry["content_hash"])
+            # map announced addrs back to connected peers
+            for p, peer in node.peers.items():
+                if peer.addr in providers:
+                    pid = p
+                    break
+        if pid is None:
+            raise RuntimeError(
+                f"no connected provider for {entry['name']} "
+                f"({entry['content_hash'][:12]})"
+            )
+        pieces: List[bytes] = []
+        for i in range(len(entry["piece_hashes"])):
+            pieces.append(
+                await node.request_piece(pid, entry["content_hash"], i)
+            )
+        data = verify_and_reassemble(pieces, entry["piece_hashes"])
+        if sha256_hex_bytes(data) != entry["content_hash"]:
+            raise ValueError(f"content hash mismatch for {entry['name']}")
+        with open(os.path.join(out_dir, entry["name"]), "wb") as f:
+            f.write(data)
+    return out_dir
