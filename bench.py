@@ -46,13 +46,19 @@ def main() -> None:
     # worth ~4% on the skinny decode projections. Off on CPU and for MoE
     # (tuning the batched expert GEMM shapes takes minutes).
     if not args.no_tunableop and torch.cuda.is_available() and not _spec_probe.is_moe:
-        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
-        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
-        os.environ.setdefault(
-            "PYTORCH_TUNABLEOP_FILENAME",
-            os.path.join("gpurun_out", "tunableop_%d.csv"),
+        tune_dir = os.path.join(
+            os.path.dirname(os.path.abspath(__file__)), "gpurun_out"
         )
-        os.makedirs("gpurun_out", exist_ok=True)
+        try:
+            os.makedirs(tune_dir, exist_ok=True)
+            os.environ.setdefault(
+                "PYTORCH_TUNABLEOP_FILENAME",
+                os.path.join(tune_dir, "tunableop_%d.csv"),
+            )
+            os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+            os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+        except OSError:
+            pass  # read-only checkout: run untuned
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
