@@ -147,9 +147,9 @@ class InferenceEngine:
         self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks)
         self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
 
-        # MoE decode uses the dense all-experts path (static shapes), so it
-        # is capture-safe as long as the decode batch stays under the dense
-        # threshold — enforced by max_batch here
+        # MoE graphs only when the decode batch stays within the dense
+        # all-experts path (static shapes); the sorted/padded path has a
+        # data-dependent buffer size and runs eagerly
         graphs_ok = on_gpu and (
             not self.spec.is_moe or max_batch <= Runner.MOE_DENSE_MAX_TOKENS
         )
