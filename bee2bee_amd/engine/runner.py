@@ -110,7 +110,7 @@ class Runner:
     # (bandwidth-bound), a batched GEMM reads them exactly once with full
     # MFMA efficiency, and every shape is static -> hipGraph-capturable
     # (the per-expert gather loop was 768 tiny launches per Mixtral step)
-    MOE_DENSE_MAX_TOKENS = 64
+    MOE_DENSE_MAX_TOKENS = 256
     # above the dense threshold: sort token-slots by expert into a padded
     # [E, maxcount, H] buffer and run ONE bmm per projection — weights read
     # once per expert, flops proportional to routed tokens (vs E x T dense);
