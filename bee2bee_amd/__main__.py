@@ -181,6 +181,25 @@ def register(node_url, network, region, test):
     asyncio.run(_reg())
 
 
+@cli.command("export-model")
+@click.option("--model", required=True, help="Model preset or HF id")
+@click.option("--out", required=True, help="Output checkpoint dir")
+@click.option("--seed", default=0, type=int)
+def export_model(model, out, seed):
+    """Write an HF-format safetensors checkpoint (random-init of the named
+    architecture when offline) — pairs with seed-model for mesh
+    distribution."""
+    import torch
+
+    from .models.spec import resolve_spec
+    from .models.weights import ModelWeights, save_hf
+
+    spec = resolve_spec(model)
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(seed)
+    save_hf(w, out)
+    click.echo(f"wrote {spec.name} ({spec.n_params() / 1e9:.2f}B params) to {out}")
+
+
 @cli.command("seed-model")
 @click.option("--name", required=True, help="Checkpoint name to publish")
 @click.option("--path", required=True, help="HF checkpoint dir to seed")

@@ -119,3 +119,39 @@ def test_chat_no_service_error(client):
     )
     body = r.json()
     assert body["status"] == "error"
+
+
+def test_chat_p2p_fallback_to_remote_provider(client):
+    """No local service for the model -> the gateway resolves a remote mesh
+    provider and relays the request."""
+    import asyncio
+
+    from tests.test_mesh import EchoService, _wait_for
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.mesh.node import MeshNode
+
+    node = gateway_api.node
+    loop = node._tasks[0].get_loop()  # the node's running loop (uvicorn's)
+
+    async def setup():
+        provider = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await provider.start()
+        await provider.add_service(EchoService(model="remote-model"))
+        await node.connect_bootstrap(provider.addr)
+        await _wait_for(lambda: provider.peer_id in node.providers)
+        return provider
+
+    fut = asyncio.run_coroutine_threadsafe(setup(), loop)
+    provider = fut.result(timeout=30)
+    try:
+        r = client.post(
+            "/chat",
+            headers={"X-API-KEY": "secret-key"},
+            json={"prompt": "over the mesh", "model": "remote-model"},
+        )
+        body = r.json()
+        assert body["status"] == "ok", body
+        assert body["text"] == "echo:over the mesh"
+        assert body["metadata"]["engine"] == "bee2bee-amd-p2p"
+    finally:
+        asyncio.run_coroutine_threadsafe(provider.stop(), loop).result(timeout=15)
