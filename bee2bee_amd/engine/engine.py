@@ -147,12 +147,9 @@ class InferenceEngine:
         self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks)
         self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
 
-        # MoE graphs only when the decode batch stays within the dense
-        # all-experts path (static shapes); the sorted/padded path has a
-        # data-dependent buffer size and runs eagerly
-        graphs_ok = on_gpu and (
-            not self.spec.is_moe or max_batch <= Runner.MOE_DENSE_MAX_TOKENS
-        )
+        # both MoE decode paths (dense all-experts and sorted/grouped-GEMM)
+        # are static-shaped, so MoE captures like the dense models
+        graphs_ok = on_gpu
         self.use_graphs = graphs_ok if use_graphs is None else (use_graphs and graphs_ok)
         self.graphs: Optional[DecodeGraphs] = None
         if self.use_graphs:

@@ -110,12 +110,12 @@ class Runner:
     # (bandwidth-bound), a batched GEMM reads them exactly once with full
     # MFMA efficiency, and every shape is static -> hipGraph-capturable
     # (the per-expert gather loop was 768 tiny launches per Mixtral step)
-    MOE_DENSE_MAX_TOKENS = 256
-    # above the dense threshold: sort token-slots by expert into a padded
-    # [E, maxcount, H] buffer and run ONE bmm per projection — weights read
-    # once per expert, flops proportional to routed tokens (vs E x T dense);
-    # maxcount is data-dependent, so this path is not graph-captured
-    MOE_PADDED_MAX_TOKENS = 100_000
+    MOE_DENSE_MAX_TOKENS = 64
+    # above the dense threshold: sort token-slots by expert and run the
+    # grouped-GEMM HIP kernel per projection — weights read once per expert,
+    # flops proportional to routed tokens, segment sizes stay on device
+    # (static shapes -> graph-capturable, no host sync)
+    MOE_GROUPED_MAX_TOKENS = 100_000
 
     def _moe_mlp(self, lw, x: torch.Tensor) -> torch.Tensor:
         """Top-k expert MLP: dense all-experts bmm for decode-sized batches,
@@ -142,28 +142,25 @@ class Runner:
             out = torch.einsum("eth,te->th", y.float(), wfull)
             return out.to(x.dtype)
 
-        if T <= self.MOE_PADDED_MAX_TOKENS:
+        if T <= self.MOE_GROUPED_MAX_TOKENS:
             k = s.top_k_experts
             S = T * k
             flat_e = idx.reshape(-1).to(torch.int64)  # slot s -> expert
             counts = torch.bincount(flat_e, minlength=E)
-            offs = torch.cumsum(counts, 0) - counts  # exclusive prefix
+            offs_ex = torch.cumsum(counts, 0)
+            offsets = torch.zeros(E + 1, dtype=torch.int32, device=x.device)
+            offsets[1:] = offs_ex.to(torch.int32)
             order = torch.argsort(flat_e, stable=True)
-            sorted_e = flat_e[order]
-            pos = torch.arange(S, device=x.device) - offs[sorted_e]
-            C = int(counts.max())  # host sync; bounds the padded buffer
             tok_of_slot = (
                 torch.arange(S, device=x.device, dtype=torch.int64) // k
             )[order]
-            padded = torch.zeros(E, C, H, dtype=x.dtype, device=x.device)
-            padded[sorted_e, pos] = x[tok_of_slot]
-            gu = torch.bmm(padded, lw.moe_w_gate_up.transpose(1, 2))
-            act = ops.swiglu(gu.reshape(E * C, 2 * I)).view(E, C, I)
-            y = torch.bmm(act, lw.moe_w_down.transpose(1, 2))  # [E, C, H]
-            y_slot = y[sorted_e, pos].float()  # [S, H] in sorted order
+            x_sorted = x[tok_of_slot].contiguous()  # [S, H]
+            gu = ops.grouped_gemm(x_sorted, lw.moe_w_gate_up, offsets)
+            act = ops.swiglu(gu)
+            y = ops.grouped_gemm(act, lw.moe_w_down, offsets)  # [S, H]
             w_slot = weights.reshape(-1)[order].unsqueeze(-1)
             out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
-            out.index_add_(0, tok_of_slot, y_slot * w_slot)
+            out.index_add_(0, tok_of_slot, y.float() * w_slot)
             return out.to(x.dtype)
 
         out = torch.zeros_like(x, dtype=torch.float32)

@@ -153,3 +153,13 @@ def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
 def moe_topk_gate(logits: torch.Tensor, top_k: int):
     # gating math is tiny ([T, E]); torch ops are fine on both devices
     return reference.moe_topk_gate(logits, top_k)
+
+
+def grouped_gemm(
+    x: torch.Tensor, w: torch.Tensor, offsets: torch.Tensor
+) -> torch.Tensor:
+    """Per-expert segment GEMM out[seg_e] = x[seg_e] @ w[e].T — the MoE
+    expert projection (segment sizes stay on device; graph-capturable)."""
+    if _use_hip(x):
+        return require_hip().grouped_gemm(x, w, offsets)
+    return reference.grouped_gemm(x, w, offsets)

@@ -32,6 +32,9 @@ void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          int, hipStream_t);
 void launch_mfma_probe(const unsigned short*, const unsigned short*, float*,
                        hipStream_t);
+void launch_grouped_gemm(const unsigned short*, const unsigned short*,
+                         const int*, unsigned short*, int, int, int, int,
+                         hipStream_t);
 void launch_bw_linear(const unsigned short*, long, float*, hipStream_t);
 void launch_bw_rowperlane(const unsigned short*, long, int, int, float*,
                           hipStream_t);
@@ -192,6 +195,24 @@ Tensor attn_prefill(const Tensor& q, const Tensor& k, const Tensor& v,
     return out;
 }
 
+Tensor grouped_gemm(const Tensor& x, const Tensor& w, const Tensor& offs) {
+    check_bf16(x, "x");
+    check_bf16(w, "w");
+    TORCH_CHECK(offs.scalar_type() == torch::kInt32 && offs.is_contiguous());
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+    const int S = x.size(0), K = x.size(1);
+    const int E = w.size(0), N = w.size(1);
+    TORCH_CHECK(w.size(2) == K, "K mismatch");
+    TORCH_CHECK(offs.size(0) == E + 1);
+    TORCH_CHECK(N % 64 == 0, "N must be a multiple of 64");
+    TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
+    Tensor out = torch::empty({S, N}, x.options());
+    if (S > 0)
+        launch_grouped_gemm(bf16p(x), bf16p(w), offs.data_ptr<int>(),
+                            bf16p_mut(out), E, S, N, K, stream());
+    return out;
+}
+
 Tensor mfma_probe(const Tensor& A, const Tensor& B) {
     check_bf16(A, "A");
     check_bf16(B, "B");
@@ -230,5 +251,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_decode", &attn_decode, "paged GQA decode attention");
     m.def("attn_prefill", &attn_prefill, "varlen causal flash prefill");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+    m.def("grouped_gemm", &grouped_gemm, "per-expert segment GEMM (MoE)");
     m.def("bw_probe", &bw_probe, "bandwidth pattern probe");
 }

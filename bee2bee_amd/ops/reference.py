@@ -197,3 +197,19 @@ def moe_topk_gate(
     vals, idx = torch.topk(logits.float(), top_k, dim=-1)
     weights = torch.softmax(vals, dim=-1)
     return weights, idx.to(torch.int32)
+
+
+def grouped_gemm(
+    x: torch.Tensor,        # [S, K] (rows sorted by expert)
+    w: torch.Tensor,        # [E, N, K]
+    offsets: torch.Tensor,  # [E+1] int32
+) -> torch.Tensor:
+    """Per-expert segment GEMM: out[offs[e]:offs[e+1]] = x_seg @ w[e].T"""
+    S, K = x.shape
+    E, N, _ = w.shape
+    out = torch.zeros(S, N, dtype=x.dtype, device=x.device)
+    for e in range(E):
+        lo, hi = int(offsets[e]), int(offsets[e + 1])
+        if hi > lo:
+            out[lo:hi] = (x[lo:hi].float() @ w[e].float().t()).to(x.dtype)
+    return out

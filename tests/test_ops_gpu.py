@@ -208,3 +208,34 @@ def test_attn_prefill_stress():
         torch.tensor([0, L], dtype=torch.int32, device=DEV), L, scale, True,
     )
     assert_close(out[:L], ref1, atol=3e-2, rtol=3e-2, msg="prefill stress seq0")
+
+
+@pytest.mark.parametrize(
+    "S,E,N,K,skew",
+    [
+        (512, 8, 28672, 4096, "even"),    # mixtral gate_up shape
+        (512, 8, 4096, 14336, "even"),    # mixtral down shape
+        (300, 8, 1024, 4096, "skewed"),   # ragged segments incl. empty
+        (7, 4, 128, 64, "tiny"),
+    ],
+)
+def test_grouped_gemm(S, E, N, K, skew):
+    torch.manual_seed(11)
+    x = (torch.randn(S, K, device=DEV) * 0.3).bfloat16()
+    w = (torch.randn(E, N, K, device=DEV) * 0.05).bfloat16()
+    if skew == "even":
+        counts = torch.full((E,), S // E, dtype=torch.int64)
+        counts[-1] += S - int(counts.sum())
+    elif skew == "skewed":
+        counts = torch.zeros(E, dtype=torch.int64)
+        counts[0] = S - 5
+        counts[3] = 5  # others empty
+    else:
+        counts = torch.tensor([3, 0, 4, 0][:E], dtype=torch.int64)
+    assert int(counts.sum()) == S
+    offs = torch.zeros(E + 1, dtype=torch.int32)
+    offs[1:] = counts.cumsum(0).to(torch.int32)
+    offs = offs.to(DEV)
+    got = ops.grouped_gemm(x, w, offs)
+    ref = R.grouped_gemm(x, w, offs)
+    assert_close(got, ref, atol=3e-2, rtol=3e-2, msg=f"grouped {S}x{N}x{K}")
