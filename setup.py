@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "attn_decode.hip"),
         os.path.join(CSRC, "attn_prefill.hip"),
         os.path.join(CSRC, "probe.hip"),
+        os.path.join(CSRC, "grouped_gemm.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
