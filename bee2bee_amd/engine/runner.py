@@ -143,24 +143,30 @@ class Runner:
             return out.to(x.dtype)
 
         if T <= self.MOE_GROUPED_MAX_TOKENS:
+            # counting sort from one-hot cumsums: every op here (one_hot,
+            # cumsum, gather, index_copy) is hipGraph-capture-safe —
+            # torch.bincount/argsort are not
             k = s.top_k_experts
             S = T * k
             flat_e = idx.reshape(-1).to(torch.int64)  # slot s -> expert
-            counts = torch.bincount(flat_e, minlength=E)
-            offs_ex = torch.cumsum(counts, 0)
+            oh = F.one_hot(flat_e, E)  # [S, E]
+            counts = oh.sum(0)
+            offs_excl = counts.cumsum(0) - counts
             offsets = torch.zeros(E + 1, dtype=torch.int32, device=x.device)
-            offsets[1:] = offs_ex.to(torch.int32)
-            order = torch.argsort(flat_e, stable=True)
-            tok_of_slot = (
-                torch.arange(S, device=x.device, dtype=torch.int64) // k
-            )[order]
-            x_sorted = x[tok_of_slot].contiguous()  # [S, H]
+            offsets[1:] = counts.cumsum(0).to(torch.int32)
+            rank = (oh.cumsum(0) - oh).gather(1, flat_e.unsqueeze(1)).squeeze(1)
+            pos = (offs_excl[flat_e] + rank).to(torch.int64)  # dest row
+            tok = torch.arange(S, device=x.device, dtype=torch.int64) // k
+            x_sorted = torch.empty(S, H, dtype=x.dtype, device=x.device)
+            x_sorted.index_copy_(0, pos, x[tok])
             gu = ops.grouped_gemm(x_sorted, lw.moe_w_gate_up, offsets)
             act = ops.swiglu(gu)
             y = ops.grouped_gemm(act, lw.moe_w_down, offsets)  # [S, H]
-            w_slot = weights.reshape(-1)[order].unsqueeze(-1)
+            contrib = y[pos].float()  # back to slot order
             out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
-            out.index_add_(0, tok_of_slot, y.float() * w_slot)
+            out.index_add_(
+                0, tok, contrib * weights.reshape(-1).unsqueeze(-1)
+            )
             return out.to(x.dtype)
 
         out = torch.zeros_like(x, dtype=torch.float32)
