@@ -147,9 +147,11 @@ class InferenceEngine:
         self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks)
         self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
 
-        # both MoE decode paths (dense all-experts and sorted/grouped-GEMM)
-        # are static-shaped, so MoE captures like the dense models
-        graphs_ok = on_gpu
+        # MoE captures only in the dense all-experts range (static shapes);
+        # the padded-bmm path above it has a data-dependent buffer size
+        graphs_ok = on_gpu and (
+            not self.spec.is_moe or max_batch <= Runner.MOE_DENSE_MAX_TOKENS
+        )
         self.use_graphs = graphs_ok if use_graphs is None else (use_graphs and graphs_ok)
         self.graphs: Optional[DecodeGraphs] = None
         if self.use_graphs:

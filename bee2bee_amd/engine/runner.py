@@ -110,7 +110,7 @@ class Runner:
     # (bandwidth-bound), a batched GEMM reads them exactly once with full
     # MFMA efficiency, and every shape is static -> hipGraph-capturable
     # (the per-expert gather loop was 768 tiny launches per Mixtral step)
-    MOE_DENSE_MAX_TOKENS = 64
+    MOE_DENSE_MAX_TOKENS = 256
     # above the dense threshold: sort token-slots by expert and run the
     # grouped-GEMM HIP kernel per projection — weights read once per expert,
     # flops proportional to routed tokens, segment sizes stay on device
@@ -146,23 +146,40 @@ class Runner:
             # counting sort from one-hot cumsums: every op here (one_hot,
             # cumsum, gather, index_copy) is hipGraph-capture-safe —
             # torch.bincount/argsort are not
+            import os as _os
+
             k = s.top_k_experts
             S = T * k
             flat_e = idx.reshape(-1).to(torch.int64)  # slot s -> expert
             oh = F.one_hot(flat_e, E)  # [S, E]
             counts = oh.sum(0)
             offs_excl = counts.cumsum(0) - counts
-            offsets = torch.zeros(E + 1, dtype=torch.int32, device=x.device)
-            offsets[1:] = counts.cumsum(0).to(torch.int32)
             rank = (oh.cumsum(0) - oh).gather(1, flat_e.unsqueeze(1)).squeeze(1)
             pos = (offs_excl[flat_e] + rank).to(torch.int64)  # dest row
             tok = torch.arange(S, device=x.device, dtype=torch.int64) // k
-            x_sorted = torch.empty(S, H, dtype=x.dtype, device=x.device)
-            x_sorted.index_copy_(0, pos, x[tok])
-            gu = ops.grouped_gemm(x_sorted, lw.moe_w_gate_up, offsets)
-            act = ops.swiglu(gu)
-            y = ops.grouped_gemm(act, lw.moe_w_down, offsets)  # [S, H]
-            contrib = y[pos].float()  # back to slot order
+            if _os.environ.get("BEE2BEE_MOE_GROUPED") == "1":
+                # in-development grouped-GEMM kernel path (capture-safe,
+                # device-side segment sizes); currently ~1.7 TB/s of W vs
+                # hipBLASLt bmm's ~3.1 — see docs/ROADMAP.md
+                offsets = torch.zeros(E + 1, dtype=torch.int32, device=x.device)
+                offsets[1:] = counts.cumsum(0).to(torch.int32)
+                x_sorted = torch.empty(S, H, dtype=x.dtype, device=x.device)
+                x_sorted.index_copy_(0, pos, x[tok])
+                gu = ops.grouped_gemm(x_sorted, lw.moe_w_gate_up, offsets)
+                act = ops.swiglu(gu)
+                y = ops.grouped_gemm(act, lw.moe_w_down, offsets)  # [S, H]
+                contrib = y[pos].float()  # back to slot order
+            else:
+                # padded per-expert bmm: weights read once per expert, flops
+                # proportional to max routed count; counts.max() is one host
+                # sync per MoE layer (so no hipGraph above the dense range)
+                C = int(counts.max())
+                padded = torch.zeros(E, C, H, dtype=x.dtype, device=x.device)
+                padded[flat_e, rank] = x[tok]
+                gu = torch.bmm(padded, lw.moe_w_gate_up.transpose(1, 2))
+                act = ops.swiglu(gu.reshape(E * C, 2 * I)).view(E, C, I)
+                y = torch.bmm(act, lw.moe_w_down.transpose(1, 2))  # [E,C,H]
+                contrib = y[flat_e, rank].float()  # [S, H] slot order
             out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
             out.index_add_(
                 0, tok, contrib * weights.reshape(-1).unsqueeze(-1)

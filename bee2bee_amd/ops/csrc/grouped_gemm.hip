@@ -12,11 +12,11 @@
 // Decode-shaped design (M per expert ~10^1-10^3, N/K in the thousands):
 //   grid (N/64, E), one workgroup = 4 waves = 64 output columns of one
 //   expert; the wave's 16-column W panel streams from HBM exactly ONCE
-//   while up to 16 row-tiles (256 rows) accumulate in registers; x K-tiles
-//   are staged cooperatively in LDS (padded rows -> conflict-free
-//   ds_read_b128) and reused by all 4 waves. Experts with more than 256
-//   rows loop in 256-row passes (W is re-read per pass; at decode batches
-//   a pass covers everything).
+//   while up to 8 row-tiles (128 rows) accumulate in registers. A-fragments
+//   read straight from global memory — x is a few MB and L2-resident, so
+//   staging it through LDS only added two barriers per 32-deep K-step
+//   (measured 2x slower than the barrier-free form at decode sizes).
+//   Experts with more rows loop in 128-row passes (W re-read per pass).
 //
 // Numerics reference: ops/reference.py grouped_gemm.
 #include "common.h"
@@ -24,9 +24,10 @@
 #define GG_BLOCK 256
 #define GG_WAVES 4
 #define GG_BN 64          // output cols per workgroup (16 per wave)
-#define GG_MROWS 256      // rows per pass (16 m-tiles)
+#define GG_MROWS 64       // rows per pass (VGPR-bound: 128 rows -> 252 VGPRs)
 #define GG_MT (GG_MROWS / 16)
 #define GG_BK 32          // K step (one mfma_16x16x32)
+#define GG_KG 128         // K-group staged per barrier pair (4 K-steps)
 #define GG_XPAD 8         // LDS row padding (bf16 elems)
 
 __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
@@ -46,13 +47,10 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
     const int lg = lane >> 4;
     const int li = lane & 15;
 
-    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-    // x K-tile: [GG_MROWS][GG_BK + GG_XPAD] bf16 (80 B rows -> the 16-lane
-    // ds_read_b128 groups land on distinct banks)
-    unsigned short* x_s = reinterpret_cast<unsigned short*>(smem_raw);
-    constexpr int XS = GG_BK + GG_XPAD;
-
     const unsigned short* wp = w + ((long)e * N + n0 + wid * 16 + li) * K;
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    unsigned short* x_s = reinterpret_cast<unsigned short*>(smem_raw);
+    constexpr int XS = GG_KG + GG_XPAD;  // 272 B rows: conflict-free b128
 
     for (int pass = m_lo; pass < m_hi; pass += GG_MROWS) {
         const int m_cnt = min(GG_MROWS, m_hi - pass);
@@ -62,38 +60,46 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
 #pragma unroll
         for (int t = 0; t < GG_MT; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-        for (int k0 = 0; k0 < K; k0 += GG_BK) {
+        // K-groups of 128 dims: the x tile stages cooperatively into LDS
+        // (8 independent 16 B loads in flight per thread), the wave's B
+        // panel loads 4 fragments per group — one barrier pair per 32
+        // MFMAs. (Register-resident A hoisted to 250 VGPRs; per-step LDS
+        // staging was barrier-bound: this is the middle ground.)
+        for (int k0 = 0; k0 < K; k0 += GG_KG) {
             __syncthreads();
-            // stage the x K-tile: thread i covers row i/2, dims (i%2)*16
-            for (int i = tid; i < GG_MROWS * 2; i += GG_BLOCK) {
-                const int r = i / 2;
-                const int d = (i % 2) * 16;
-                short8 v{};
-                if (r < m_cnt)
-                    v = *reinterpret_cast<const short8*>(
-                        x + (long)(pass + r) * K + k0 + d);
-                *reinterpret_cast<short8*>(x_s + r * XS + d) = v;
-                *reinterpret_cast<short8*>(x_s + r * XS + d + 8) =
-                    (r < m_cnt)
-                        ? *reinterpret_cast<const short8*>(
-                              x + (long)(pass + r) * K + k0 + d + 8)
-                        : short8{};
-            }
-            __syncthreads();
-
-            // B fragment: this wave's 16 W rows (= output cols), k-seg
-            const bf16x8 b =
-                *reinterpret_cast<const bf16x8*>(wp + k0 + lg * 8);
-            // static unroll with a guard: a runtime-indexed acc[t] would be
-            // demoted to scratch (register arrays need constant indices)
-#pragma unroll
-            for (int t = 0; t < GG_MT; ++t) {
-                if (t < n_mt) {
-                    const bf16x8 a = *reinterpret_cast<const bf16x8*>(
-                        x_s + (t * 16 + li) * XS + lg * 8);
-                    acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a, b, acc[t], 0, 0, 0);
+            for (int i = tid; i < GG_MROWS * (GG_KG / 16); i += GG_BLOCK) {
+                const int r = i / (GG_KG / 16);
+                const int d = (i % (GG_KG / 16)) * 16;
+                short8 v0{}, v1{};
+                if (r < m_cnt) {
+                    const unsigned short* src =
+                        x + (long)(pass + r) * K + k0 + d;
+                    v0 = *reinterpret_cast<const short8*>(src);
+                    v1 = *reinterpret_cast<const short8*>(src + 8);
                 }
+                *reinterpret_cast<short8*>(x_s + r * XS + d) = v0;
+                *reinterpret_cast<short8*>(x_s + r * XS + d + 8) = v1;
+            }
+            bf16x8 bg[GG_KG / GG_BK];
+#pragma unroll
+            for (int g = 0; g < GG_KG / GG_BK; ++g)
+                bg[g] = *reinterpret_cast<const bf16x8*>(
+                    wp + k0 + g * GG_BK + lg * 8);
+            __syncthreads();
+#pragma unroll
+            for (int g = 0; g < GG_KG / GG_BK; ++g) {
+#pragma unroll
+                for (int t = 0; t < GG_MT; ++t) {
+                    if (t < n_mt) {
+                        const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                            x_s + (t * 16 + li) * XS + g * GG_BK + lg * 8);
+                        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a, bg[g], acc[t], 0, 0, 0);
+                    }
+                }
+                // fence the scheduler per K-step: without it all 32 LDS
+                // A-fragments are hoisted live -> 252 VGPRs, 1 wave/SIMD
+                __builtin_amdgcn_sched_barrier(0);
             }
         }
 
@@ -116,7 +122,7 @@ extern "C" void launch_grouped_gemm(
     const unsigned short* x, const unsigned short* w, const int* offs,
     unsigned short* out, int E, int S, int N, int K, hipStream_t stream) {
     dim3 grid(N / GG_BN, E);
-    const int smem = GG_MROWS * (GG_BK + GG_XPAD) * 2;
+    const int smem = GG_MROWS * (GG_KG + GG_XPAD) * 2;
     hipLaunchKernelGGL(grouped_gemm_kernel, grid, dim3(GG_BLOCK), smem,
                        stream, x, w, offs, out, N, K);
 }

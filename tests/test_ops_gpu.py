@@ -216,7 +216,7 @@ def test_attn_prefill_stress():
         (512, 8, 28672, 4096, "even"),    # mixtral gate_up shape
         (512, 8, 4096, 14336, "even"),    # mixtral down shape
         (300, 8, 1024, 4096, "skewed"),   # ragged segments incl. empty
-        (7, 4, 128, 64, "tiny"),
+        (7, 4, 128, 128, "tiny"),
     ],
 )
 def test_grouped_gemm(S, E, N, K, skew):
