@@ -223,28 +223,33 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     const int* bt = block_table + (long)b * W;
     const int wave_key0 = start + wid * WAVE;  // this wave covers 64 keys
 
-    // scores: per tile of 16 keys, A streams from the cache rows
+    // scores: all TILES*KSTEPS A-fragments issue before any MFMA (probe-
+    // style load batching: 16 independent 16 B loads in flight per wave)
     f32x4 acc[TILES];
+    bf16x8 afr[TILES][KSTEPS];
 #pragma unroll
     for (int t = 0; t < TILES; ++t) {
         const int key = wave_key0 + t * 16 + li;  // A row = li
-        const unsigned short* kr = nullptr;
+        const unsigned short* kr;
         if (key < L) {
             const int page = bt[key / bs];
             kr = k_cache + (((long)page * nkv + kvh) * bs + key % bs) * HD;
+        } else {
+            // safe dummy row; masked later via the key<L score mask
+            kr = k_cache;
         }
+#pragma unroll
+        for (int ks = 0; ks < KSTEPS; ++ks)
+            afr[t][ks] =
+                *reinterpret_cast<const bf16x8*>(kr + ks * 32 + lg * 8);
+    }
+#pragma unroll
+    for (int t = 0; t < TILES; ++t) {
         f32x4 c{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int ks = 0; ks < KSTEPS; ++ks) {
-            bf16x8 a;
-            if (kr != nullptr) {
-                a = *reinterpret_cast<const bf16x8*>(kr + ks * 32 + lg * 8);
-            } else {
-#pragma unroll
-                for (int e = 0; e < 8; ++e) a[e] = (__bf16)0.f;
-            }
-            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bq[ks], c, 0, 0, 0);
-        }
+        for (int ks = 0; ks < KSTEPS; ++ks)
+            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[t][ks], bq[ks], c,
+                                                        0, 0, 0);
         acc[t] = c;
     }
 

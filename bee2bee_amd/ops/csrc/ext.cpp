@@ -40,6 +40,8 @@ void launch_bw_rowperlane(const unsigned short*, long, int, int, float*,
                           hipStream_t);
 void launch_bw_rowperinstr(const unsigned short*, long, int, int, float*,
                            hipStream_t);
+void launch_bw_paged(const unsigned short*, const int*, long, int, int,
+                     float*, hipStream_t);
 }
 
 namespace {
@@ -225,6 +227,15 @@ Tensor mfma_probe(const Tensor& A, const Tensor& B) {
     return C;
 }
 
+void bw_probe_paged(const Tensor& pool, const Tensor& table, int64_t hd,
+                    int64_t rows_per_page) {
+    check_bf16(pool, "pool");
+    Tensor sink = torch::zeros({1}, pool.options().dtype(torch::kFloat32));
+    launch_bw_paged(bf16p(pool), table.data_ptr<int>(), pool.numel() / hd,
+                    (int)hd, (int)rows_per_page, sink.data_ptr<float>(),
+                    stream());
+}
+
 void bw_probe(const Tensor& pool, int64_t mode, int64_t hd, int64_t arg) {
     check_bf16(pool, "pool");
     Tensor sink = torch::zeros({1}, pool.options().dtype(torch::kFloat32));
@@ -253,4 +264,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
     m.def("grouped_gemm", &grouped_gemm, "per-expert segment GEMM (MoE)");
     m.def("bw_probe", &bw_probe, "bandwidth pattern probe");
+    m.def("bw_probe_paged", &bw_probe_paged, "paged row-per-lane probe");
 }

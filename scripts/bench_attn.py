@@ -82,7 +82,7 @@ def main():
     _ = out
 
 
-if __name__ == "__main__" and "probe" not in sys.argv[1:2]:
+if __name__ == "__main__" and sys.argv[1:2] not in (["probe"], ["paged"]):
     main()
 
 
@@ -114,4 +114,34 @@ def probe():
 
 if __name__ == "__main__" and "probe" in sys.argv[1:2]:
     probe()
+    sys.exit(0)
+
+
+def probe_paged():
+    from bee2bee_amd import ops
+
+    hip = ops.require_hip()
+    dev = "cuda:0"
+    hd, rpp = 128, 256  # 64 KB pages of 256 rows (like 32-key KV blocks x8 heads)
+    pool = torch.randn(1 << 29, device=dev).bfloat16()
+    n_pages = pool.numel() // hd // rpp
+    nbytes = pool.numel() * 2
+    for name, perm in (("sequential pages", False), ("permuted pages", True)):
+        table = (
+            torch.randperm(n_pages, device=dev)
+            if perm else torch.arange(n_pages, device=dev)
+        ).to(torch.int32)
+        for _ in range(2):
+            hip.bw_probe_paged(pool, table, hd, rpp)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10):
+            hip.bw_probe_paged(pool, table, hd, rpp)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 10
+        print(f"paged {name:18s} {nbytes / dt / 1e12:6.2f} TB/s")
+
+
+if __name__ == "__main__" and "paged" in sys.argv[1:2]:
+    probe_paged()
     sys.exit(0)
