@@ -186,10 +186,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     constexpr int TILES = WAVE / 16;  // 4 key-tiles of 16 per wave
     const int b = blockIdx.x;
     const int kvh = blockIdx.y;
-    const int chunk = blockIdx.z;
     const int L = seq_lens[b];
-    const int start = chunk * DEC_CHUNK;
-    if (start >= L) return;
+    if ((int)blockIdx.z * DEC_CHUNK >= L) return;
     const int lane = threadIdx.x % WAVE;
     const int wid = threadIdx.x / WAVE;
     const int lg = lane >> 4;  // lane group 0..3
@@ -221,6 +219,11 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     }
 
     const int* bt = block_table + (long)b * W;
+    // loop chunks with stride gridDim.z: the q staging amortizes and the
+    // per-chunk softmax tail overlaps other chunks' K bursts (single-chunk
+    // blocks measured ~66% memory-idle from cold-start/tail phases)
+    for (int chunk = blockIdx.z; chunk * DEC_CHUNK < L; chunk += gridDim.z) {
+    const int start = chunk * DEC_CHUNK;
     const int wave_key0 = start + wid * WAVE;  // this wave covers 64 keys
 
     // scores: all TILES*KSTEPS A-fragments issue before any MFMA (probe-
@@ -318,6 +321,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
             pml[g * 2 + 1] = ls;
         }
     }
+    __syncthreads();  // red reused by the next chunk iteration
+    }  // chunk loop
 }
 
 // ----------------------------------------------------------- stage 2: PV
@@ -331,10 +336,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     int nkv, int W, int bs, int hd, int C) {
     const int b = blockIdx.x;
     const int kvh = blockIdx.y;
-    const int chunk = blockIdx.z;
     const int L = seq_lens[b];
-    const int start = chunk * DEC_CHUNK;
-    if (start >= L) return;
+    if ((int)blockIdx.z * DEC_CHUNK >= L) return;
     const int lane = threadIdx.x % WAVE;
     const int wid = threadIdx.x / WAVE;
 
@@ -343,6 +346,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     long* voff_s = reinterpret_cast<long*>(p_s + DEC_CHUNK * G);  // [CHUNK]
     float* merge = reinterpret_cast<float*>(voff_s + DEC_CHUNK);  // [Wv][G][hd]
 
+    for (int chunk = blockIdx.z; chunk * DEC_CHUNK < L; chunk += gridDim.z) {
+    const int start = chunk * DEC_CHUNK;
     // stage p for the whole chunk into LDS (coalesced global read)
     {
         const float* psrc =
@@ -432,6 +437,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
             po[d0 + 1] = a1;
         }
     }
+    __syncthreads();  // LDS buffers reused by the next chunk iteration
+    }  // chunk loop
 }
 
 // ------------------------------------------------------ stage 3: combine
@@ -478,7 +485,12 @@ extern "C" void launch_attn_decode(
     const int* seq_lens, float* p_buf, float* part_o, float* part_ml,
     unsigned short* out, int B, int nkv, int G, int W, int bs, int hd, int C,
     long q_stride, float scale, hipStream_t stream) {
-    dim3 grid(B, nkv, C);
+    // enough blocks to fill the chip, but chunks loop within a block so the
+    // cold-start/tail phases amortize over multiple 64 KB K/V bursts
+    int Z = (4096 + B * nkv - 1) / (B * nkv);
+    if (Z > C) Z = C;
+    if (Z < 1) Z = 1;
+    dim3 grid(B, nkv, Z);
     dim3 cgrid(B, nkv, G);
     const int smem_s = ((G * hd * 2 + 15) & ~15) + DEC_WAVES * G * 2 * 4;
     const int smem_pv =
