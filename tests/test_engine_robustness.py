@@ -1,0 +1,100 @@
+"""Engine serving robustness: KV exhaustion queueing, oversized prompts,
+cancellation, malformed mesh frames."""
+import json
+import queue
+import time
+
+import pytest
+
+from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+from bee2bee_amd.engine.sampler import SamplingParams
+
+
+@pytest.fixture(scope="module")
+def small_engine():
+    # tiny KV pool: 2 seqs of 64 max
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=64, seed=5)
+    yield eng
+    eng.shutdown()
+
+
+def _req(n=4, prompt=(1, 2, 3), max_new=None):
+    return GenerationRequest(
+        prompt_ids=list(prompt), max_new_tokens=max_new or n,
+        sampling=SamplingParams(greedy=True),
+    )
+
+
+def _drain(r, timeout=120):
+    while True:
+        item = r.out_queue.get(timeout=timeout)
+        if not isinstance(item, int):
+            break
+
+
+def test_burst_beyond_batch_queues(small_engine):
+    """6 requests into a max_batch=2 engine: all must complete."""
+    reqs = [small_engine.submit(_req(4, (i, i + 1))) for i in range(6)]
+    for r in reqs:
+        _drain(r)
+        assert r.error is None and len(r.output_ids) == 4
+
+
+def test_oversized_prompt_truncated(small_engine):
+    r = small_engine.submit(_req(4, tuple(range(1, 200))))
+    _drain(r)
+    assert r.error is None
+    assert len(r.output_ids) == 4
+
+
+def test_cancellation_stops_generation(small_engine):
+    r = _req(4, (5, 6), max_new=10_000)
+    small_engine.submit(r)
+    # wait for the first token, then cancel
+    tok = r.out_queue.get(timeout=60)
+    assert isinstance(tok, int)
+    r.cancelled = True
+    t0 = time.time()
+    _drain(r, timeout=60)
+    assert time.time() - t0 < 30
+    assert len(r.output_ids) < 10_000
+
+
+def test_empty_prompt_gets_bos(small_engine):
+    r = small_engine.submit(_req(3, ()))
+    _drain(r)
+    assert r.error is None and len(r.output_ids) == 3
+
+
+def test_malformed_wire_frames_do_not_crash_node():
+    import asyncio
+
+    import aiohttp
+
+    from bee2bee_amd.mesh.node import MeshNode
+
+    async def run():
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        async with aiohttp.ClientSession() as session:
+            async with session.ws_connect(f"ws://127.0.0.1:{node.port}") as ws:
+                await ws.send_str("this is not json")
+                await ws.send_str(json.dumps({"no": "type"}))
+                await ws.send_str(json.dumps({"type": "gen_request"}))  # no rid
+                await ws.send_str(json.dumps({"type": "pong"}))  # no ts
+                await ws.send_str(json.dumps({"type": "hello"}))  # no peer_id
+                await ws.send_str(json.dumps(
+                    {"type": "piece_request", "hash": None, "index": "x"}
+                ))
+                await asyncio.sleep(0.3)
+        # node still serves new connections afterwards
+        async with aiohttp.ClientSession() as session:
+            async with session.ws_connect(f"ws://127.0.0.1:{node.port}") as ws:
+                await ws.send_str(json.dumps({
+                    "type": "hello", "peer_id": "p2", "addr": "",
+                }))
+                msg = await asyncio.wait_for(ws.receive(), timeout=5)
+                assert json.loads(msg.data).get("type") == "hello"
+        await node.stop()
+
+    asyncio.run(run())
