@@ -122,24 +122,27 @@ def probe_paged():
 
     hip = ops.require_hip()
     dev = "cuda:0"
-    hd, rpp = 128, 256  # 64 KB pages of 256 rows (like 32-key KV blocks x8 heads)
+    hd = 128
     pool = torch.randn(1 << 29, device=dev).bfloat16()
-    n_pages = pool.numel() // hd // rpp
     nbytes = pool.numel() * 2
-    for name, perm in (("sequential pages", False), ("permuted pages", True)):
-        table = (
-            torch.randperm(n_pages, device=dev)
-            if perm else torch.arange(n_pages, device=dev)
-        ).to(torch.int32)
-        for _ in range(2):
-            hip.bw_probe_paged(pool, table, hd, rpp)
-        torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        for _ in range(10):
-            hip.bw_probe_paged(pool, table, hd, rpp)
-        torch.cuda.synchronize()
-        dt = (time.perf_counter() - t0) / 10
-        print(f"paged {name:18s} {nbytes / dt / 1e12:6.2f} TB/s")
+    for rpp in (256, 64, 32, 16):
+        n_pages = pool.numel() // hd // rpp
+        for name, perm in (("seq", False), ("perm", True)):
+            table = (
+                torch.randperm(n_pages, device=dev)
+                if perm else torch.arange(n_pages, device=dev)
+            ).to(torch.int32)
+            for _ in range(2):
+                hip.bw_probe_paged(pool, table, hd, rpp)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(10):
+                hip.bw_probe_paged(pool, table, hd, rpp)
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / 10
+            print(f"paged rows/page={rpp:4d} {name:5s}"
+                  f" ({rpp * hd * 2 // 1024:3d} KB contig) "
+                  f"{nbytes / dt / 1e12:6.2f} TB/s")
 
 
 if __name__ == "__main__" and "paged" in sys.argv[1:2]:
