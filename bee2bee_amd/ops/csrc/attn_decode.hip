@@ -56,7 +56,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     const unsigned short* __restrict__ k_cache,  // [nb, nkv, bs, hd]
     const int* __restrict__ block_table,         // [B, W]
     const int* __restrict__ seq_lens,            // [B]
-    float* __restrict__ p_out,                   // [B, nkv, C, CHUNK, G]
+    unsigned short* __restrict__ p_out,          // [B, nkv, C, CHUNK, G] bf16
     float* __restrict__ part_ml,                 // [B, nkv, C, G, 2]
     int nkv, int W, int bs, int hd, int C, long q_stride, float scale) {
     const int b = blockIdx.x;
@@ -133,7 +133,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     float p[G], wsum[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-        p[g] = valid ? __expf(s[g] - M[g]) : 0.f;
+        // round to bf16 BEFORE summing so l matches the p that PV consumes
+        p[g] = valid ? bf2f(f2bf(__expf(s[g] - M[g]))) : 0.f;
         wsum[g] = wave_sum(p[g]);
     }
     __syncthreads();  // red reuse (second slot written below)
@@ -141,13 +142,14 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
 #pragma unroll
         for (int g = 0; g < G; ++g) red[(wid * G + g) * 2 + 1] = wsum[g];
     }
-    // coalesced p write: [key][G], 4*G B per thread, consecutive
+    // coalesced p write: [key][G] bf16 (p in [0,1]; bf16 halves the
+    // G-proportional scratch traffic)
     {
-        float* prow =
+        unsigned short* prow =
             p_out + ((((long)b * nkv + kvh) * C + chunk) * DEC_CHUNK +
                      threadIdx.x) * G;
 #pragma unroll
-        for (int g = 0; g < G; ++g) prow[g] = p[g];
+        for (int g = 0; g < G; ++g) prow[g] = f2bf(p[g]);
     }
     __syncthreads();
     if (threadIdx.x == 0) {
@@ -179,7 +181,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     const unsigned short* __restrict__ k_cache,
     const int* __restrict__ block_table,
     const int* __restrict__ seq_lens,
-    float* __restrict__ p_out,
+    unsigned short* __restrict__ p_out,  // bf16
     float* __restrict__ part_ml,
     int nkv, int W, int bs, int C, long q_stride, float scale) {
     constexpr int KSTEPS = HD / 32;
@@ -283,7 +285,9 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     float lsum = 0.f;
 #pragma unroll
     for (int i = 0; i < TILES * 4; ++i) {
-        const float pv = (sv[i] <= -1e29f) ? 0.f : __expf(sv[i] - M);
+        // round to bf16 BEFORE summing so l matches the p that PV consumes
+        const float pv =
+            (sv[i] <= -1e29f) ? 0.f : bf2f(f2bf(__expf(sv[i] - M)));
         sv[i] = pv;
         lsum += pv;
     }
@@ -295,14 +299,14 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     // p write-out: [key][G]; lanes with the same (t, r) write G consecutive
     // floats for 4 key-groups -> 4 x 64 B segments per instruction
     if (li < G) {
-        float* pbase =
+        unsigned short* pbase =
             p_out + (((long)b * nkv + kvh) * C + chunk) * DEC_CHUNK * G;
 #pragma unroll
         for (int t = 0; t < TILES; ++t) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int key = wid * WAVE + t * 16 + lg * 4 + r;
-                pbase[key * G + li] = sv[t * 4 + r];
+                pbase[key * G + li] = f2bf(sv[t * 4 + r]);
             }
         }
     }
@@ -331,7 +335,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     const unsigned short* __restrict__ v_cache,  // [nb, nkv, bs, hd]
     const int* __restrict__ block_table,
     const int* __restrict__ seq_lens,
-    const float* __restrict__ p_in,              // [B, nkv, C, CHUNK, G]
+    const unsigned short* __restrict__ p_in,     // [B,nkv,C,CHUNK,G] bf16
     float* __restrict__ part_o,                  // [B, nkv, C, G, hd]
     int nkv, int W, int bs, int hd, int C) {
     const int b = blockIdx.x;
@@ -348,13 +352,17 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
 
     for (int chunk = blockIdx.z; chunk * DEC_CHUNK < L; chunk += gridDim.z) {
     const int start = chunk * DEC_CHUNK;
-    // stage p for the whole chunk into LDS (coalesced global read)
+    // stage p for the whole chunk into LDS, widening bf16 -> fp32 so the
+    // inner fma loop reads full-rate f32x4
     {
-        const float* psrc =
+        const unsigned short* psrc =
             p_in + (((long)b * nkv + kvh) * C + chunk) * DEC_CHUNK * G;
-        for (int i = threadIdx.x; i < DEC_CHUNK * G / 4; i += DEC_BLOCK)
-            reinterpret_cast<f32x4*>(p_s)[i] =
-                reinterpret_cast<const f32x4*>(psrc)[i];
+        for (int i = threadIdx.x; i < DEC_CHUNK * G / 8; i += DEC_BLOCK) {
+            float v[8];
+            load_bf16x8(psrc + i * 8, v);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) p_s[i * 8 + j] = v[j];
+        }
     }
     // per-thread V row offset (one chained block-table load per key)
     {
@@ -482,7 +490,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
 extern "C" void launch_attn_decode(
     const unsigned short* q, const unsigned short* k_cache,
     const unsigned short* v_cache, const int* block_table,
-    const int* seq_lens, float* p_buf, float* part_o, float* part_ml,
+    const int* seq_lens, unsigned short* p_buf, float* part_o, float* part_ml,
     unsigned short* out, int B, int nkv, int G, int W, int bs, int hd, int C,
     long q_stride, float scale, hipStream_t stream) {
     // enough blocks to fill the chip, but chunks loop within a block so the

@@ -24,8 +24,9 @@ void launch_swiglu(const unsigned short*, unsigned short*, long, long,
                    hipStream_t);
 void launch_attn_decode(const unsigned short*, const unsigned short*,
                         const unsigned short*, const int*, const int*,
-                        float*, float*, float*, unsigned short*, int, int,
-                        int, int, int, int, int, long, float, hipStream_t);
+                        unsigned short*, float*, float*, unsigned short*,
+                        int, int, int, int, int, int, int, long, float,
+                        hipStream_t);
 void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          const unsigned short*, const int*, unsigned short*,
                          int, int, int, int, long, long, long, int, float,
@@ -162,12 +163,12 @@ Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
     const int C = (W * bs + kDecChunk - 1) / kDecChunk;
     TORCH_CHECK(C <= 64, "decode supports up to 16384-token contexts for now");
     auto fopt = q.options().dtype(torch::kFloat32);
-    Tensor p_buf = torch::empty({B, nkv, C, kDecChunk, G}, fopt);
+    Tensor p_buf = torch::empty({B, nkv, C, kDecChunk, G}, q.options());
     Tensor part_o = torch::empty({B, nkv, C, G, hd}, fopt);
     Tensor part_ml = torch::empty({B, nkv, C, G, 2}, fopt);
     launch_attn_decode(bf16p(q), bf16p(k_cache), bf16p(v_cache),
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                       p_buf.data_ptr<float>(), part_o.data_ptr<float>(),
+                       bf16p_mut(p_buf), part_o.data_ptr<float>(),
                        part_ml.data_ptr<float>(),
                        bf16p_mut(out), B, nkv, G, W, bs, hd, C, q.stride(0),
                        (float)scale, stream());
