@@ -97,6 +97,20 @@ PRESETS: Dict[str, ModelSpec] = {
         max_seq_len=512,
         tie_embeddings=True,
     ),
+    # 3-layer variant so a 3-stage pipeline has a pure-middle rank to test
+    "tiny3": ModelSpec(
+        name="tiny3",
+        vocab_size=512,
+        hidden_size=64,
+        intermediate_size=128,
+        n_layers=3,
+        n_heads=4,
+        n_kv_heads=2,
+        head_dim=16,
+        rope_theta=10000.0,
+        max_seq_len=512,
+        tie_embeddings=True,
+    ),
     "tiny-moe": ModelSpec(
         name="tiny-moe",
         vocab_size=512,

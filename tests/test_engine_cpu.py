@@ -152,3 +152,30 @@ def test_kv_blocks_recycled(engine):
             break
         time.sleep(0.02)
     assert engine.kv.free_blocks == free0
+
+
+def test_engine_serves_hf_checkpoint(tmp_path):
+    """export-model style checkpoint -> engine(model_path) must reproduce
+    the random-init engine exactly (the HF-checkpoint serving path)."""
+    import torch
+
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+    spec = PRESETS["tiny"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(23)
+    save_hf(w, str(tmp_path))
+
+    eng_ckpt = InferenceEngine("tiny", device="cpu", model_path=str(tmp_path),
+                               max_batch=2, max_seq_len=64)
+    eng_rand = InferenceEngine("tiny", device="cpu", max_batch=2,
+                               max_seq_len=64, seed=23)
+    try:
+        r1 = eng_ckpt.submit(_greedy_req([4, 5, 6], 5))
+        _drain(r1)
+        r2 = eng_rand.submit(_greedy_req([4, 5, 6], 5))
+        _drain(r2)
+        assert r1.output_ids == r2.output_ids
+    finally:
+        eng_ckpt.shutdown()
+        eng_rand.shutdown()

@@ -60,7 +60,8 @@ def test_partial_random_init_matches_full():
     assert torch.equal(full.lm_head, part.lm_head)
 
 
-def _pp_worker(rank: int, world: int, port: int, out_path: str) -> None:
+def _pp_worker(rank: int, world: int, port: int, out_path: str,
+               model: str = "tiny") -> None:
     import torch.distributed as dist
 
     from bee2bee_amd.parallel.pp import PipelineEngine
@@ -73,7 +74,7 @@ def _pp_worker(rank: int, world: int, port: int, out_path: str) -> None:
     )
     try:
         eng = PipelineEngine(
-            "tiny", device="cpu", max_batch=4, max_seq_len=64, seed=SEED
+            model, device="cpu", max_batch=4, max_seq_len=64, seed=SEED
         )
         outs = eng.generate(PROMPTS, N_NEW)
         if rank == world - 1:
@@ -83,11 +84,11 @@ def _pp_worker(rank: int, world: int, port: int, out_path: str) -> None:
         dist.destroy_process_group()
 
 
-def _single_process_reference():
+def _single_process_reference(model: str = "tiny"):
     from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
     from bee2bee_amd.engine.sampler import SamplingParams
 
-    eng = InferenceEngine("tiny", device="cpu", max_batch=4, max_seq_len=64, seed=SEED)
+    eng = InferenceEngine(model, device="cpu", max_batch=4, max_seq_len=64, seed=SEED)
     try:
         outs = []
         for p in PROMPTS:
@@ -107,14 +108,19 @@ def _single_process_reference():
 
 
 @pytest.mark.timeout(300)
-def test_pp2_matches_single_process(tmp_path):
-    """2-stage pipeline greedy decode == single-process engine (fp32 exact)."""
+@pytest.mark.parametrize(
+    "world,port,model", [(2, 29611, "tiny"), (3, 29612, "tiny3")]
+)
+def test_pp_matches_single_process(tmp_path, world, port, model):
+    """2- and 3-stage pipeline greedy decode == single-process engine
+    (fp32 exact). world=3 on the 3-layer tiny3 exercises a pure-middle
+    rank (no embedding, no lm_head) — the shape every interior rank of
+    a llama3-70b/8 deployment has."""
     out_path = str(tmp_path / "pp_out.pkl")
-    port = 29611
     ctx = mp.get_context("spawn")
     procs = [
-        ctx.Process(target=_pp_worker, args=(r, 2, port, out_path))
-        for r in range(2)
+        ctx.Process(target=_pp_worker, args=(r, world, port, out_path, model))
+        for r in range(world)
     ]
     for p in procs:
         p.start()
@@ -123,5 +129,5 @@ def test_pp2_matches_single_process(tmp_path):
         assert p.exitcode == 0, f"pp worker failed (exit {p.exitcode})"
     with open(out_path, "rb") as f:
         pp_outs = pickle.load(f)
-    ref = _single_process_reference()
+    ref = _single_process_reference(model)
     assert pp_outs == ref
