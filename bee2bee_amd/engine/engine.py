@@ -83,12 +83,13 @@ class GenerationRequest:
 
 
 class _Active:
-    __slots__ = ("req", "seq_id", "length")
+    __slots__ = ("req", "seq_id", "length", "prefilled")
 
     def __init__(self, req: GenerationRequest, seq_id: int, length: int) -> None:
         self.req = req
         self.seq_id = seq_id
         self.length = length  # tokens currently in KV cache
+        self.prefilled = 0  # prompt tokens prefetched so far (chunked prefill)
 
 
 class InferenceEngine:
@@ -166,6 +167,10 @@ class InferenceEngine:
 
         self._pending: "queue.Queue[GenerationRequest]" = queue.Queue()
         self._active: List[_Active] = []
+        # admitted requests whose prompt is longer than one prefill budget:
+        # they prefill one chunk per step (vs the paged history) while the
+        # active set keeps decoding — chunked prefill
+        self._prefilling: List[_Active] = []
         self._dec_seqs = None  # device-state cache key (active seq ids)
         self._last_sampled: Optional[torch.Tensor] = None
         self._eager_state = None
@@ -314,7 +319,7 @@ class InferenceEngine:
                 did_work = self._step()
             except Exception as e:  # engine errors fail all active requests
                 logger.exception("engine step failed")
-                for a in self._active:
+                for a in self._active + self._prefilling:
                     a.req.error = str(e)
                     if a.req.on_emit is not None:
                         try:
@@ -325,6 +330,7 @@ class InferenceEngine:
                         a.req.out_queue.put(_STREAM_END)
                     self.kv.free_seq(a.seq_id)
                 self._active.clear()
+                self._prefilling.clear()
                 did_work = True
             if not did_work:
                 self._wake.wait(timeout=0.05)
@@ -332,12 +338,13 @@ class InferenceEngine:
 
     def _step(self) -> bool:
         admitted = self._admit()
-        if admitted:
-            self._prefill(admitted)
+        worked = bool(admitted) or bool(self._prefilling)
+        if worked:
+            self._prefill(self._prefilling + admitted)
         if self._active:
             self._decode_once()
             return True
-        return bool(admitted)
+        return worked
 
     def _admit(self) -> List[_Active]:
         admitted: List[_Active] = []
@@ -365,33 +372,84 @@ class InferenceEngine:
         return admitted
 
     @torch.no_grad()
-    def _prefill(self, admitted: List[_Active]) -> None:
-        ids_list, pos_list, slot_list, cu = [], [], [], [0]
-        for a in admitted:
-            p = a.req.prompt_ids or [self.spec.bos_token_id]
-            p = p[: self.max_seq_len - a.req.max_new_tokens - 1] or p[:1]
-            a.req.prompt_ids = p
-            self.kv.extend_seq(a.seq_id, len(p))
-            a.length = len(p)
-            ids_list.extend(p)
-            pos_list.extend(range(len(p)))
-            slot_list.extend(self.kv.slot_mapping(a.seq_id, range(len(p))))
-            cu.append(cu[-1] + len(p))
+    def _prefill(self, queue_: List[_Active]) -> None:
+        """Prefill up to max_prefill_tokens prompt tokens across the queued
+        requests. Whole prompts take the packed varlen fast path; prompts
+        longer than the budget prefill in chunks against their paged history
+        (ops.attn_prefill_paged) across successive steps."""
+        budget = self.max_prefill_tokens
+        batch: List[tuple] = []  # (active, start, take)
+        n_consumed = 0
+        for a in queue_:
+            if budget <= 0:
+                break
+            if a.prefilled == 0:
+                p = a.req.prompt_ids or [self.spec.bos_token_id]
+                p = p[: self.max_seq_len - a.req.max_new_tokens - 1] or p[:1]
+                a.req.prompt_ids = p
+            take = min(len(a.req.prompt_ids) - a.prefilled, budget)
+            batch.append((a, a.prefilled, take))
+            budget -= take
+            n_consumed += 1
+        self._prefilling = [
+            a
+            for a, start, take in batch
+            if start + take < len(a.req.prompt_ids)
+        ] + [t for t in queue_[n_consumed:]]
+
         dev = self.device
+        ids_list, pos_list, slot_list, cu = [], [], [], [0]
+        for a, start, take in batch:
+            p = a.req.prompt_ids
+            self.kv.extend_seq(a.seq_id, start + take)
+            a.prefilled = a.length = start + take
+            ids_list.extend(p[start : start + take])
+            pos_list.extend(range(start, start + take))
+            slot_list.extend(
+                self.kv.slot_mapping(a.seq_id, range(start, start + take))
+            )
+            cu.append(cu[-1] + take)
         input_ids = torch.tensor(ids_list, dtype=torch.int64, device=dev)
         positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
         slots = torch.tensor(slot_list, dtype=torch.int32, device=dev)
         cu_seqlens = torch.tensor(cu, dtype=torch.int32, device=dev)
-        max_len = max(len(a.req.prompt_ids) for a in admitted)
-        hidden = self.runner.forward_prefill(
-            input_ids, positions, slots, cu_seqlens, max_len
+        max_len = max(take for _, _, take in batch)
+        whole = all(
+            start == 0 and take == len(a.req.prompt_ids)
+            for a, start, take in batch
         )
+        if whole:
+            hidden = self.runner.forward_prefill(
+                input_ids, positions, slots, cu_seqlens, max_len
+            )
+        else:
+            bt = self.kv.block_table([a.seq_id for a, _, _ in batch])
+            seq_lens = torch.tensor(
+                [start + take for _, start, take in batch],
+                dtype=torch.int32, device=dev,
+            )
+            query_lens = torch.tensor(
+                [take for _, _, take in batch], dtype=torch.int32, device=dev
+            )
+            hidden = self.runner.forward_prefill(
+                input_ids, positions, slots, cu_seqlens, max_len,
+                block_table=bt, seq_lens=seq_lens, query_lens=query_lens,
+            )
+        # sample only for requests whose whole prompt is now in the cache
+        done_idx = [
+            i
+            for i, (a, start, take) in enumerate(batch)
+            if start + take == len(a.req.prompt_ids)
+        ]
+        if not done_idx:
+            return
+        completed = [batch[i][0] for i in done_idx]
         last_rows = torch.tensor(
-            [c - 1 for c in cu[1:]], dtype=torch.int64, device=dev
+            [cu[i + 1] - 1 for i in done_idx], dtype=torch.int64, device=dev
         )
         logits = self.runner.lm_head(hidden[last_rows])
-        self._sample_and_emit(admitted, logits)
-        self._active.extend([a for a in admitted if a.req.done_ts is None])
+        self._sample_and_emit(completed, logits)
+        self._active.extend([a for a in completed if a.req.done_ts is None])
 
     @torch.no_grad()
     def _decode_once(self) -> None:

@@ -230,18 +230,13 @@ class Runner:
 
         def attn(layer_idx, q, k, v, k_cache, v_cache):
             if chunked:
-                # chunked prefill against paged history: CPU reference only
-                # for now (the engine prefills whole prompts; a paged-history
-                # HIP prefill kernel is queued for the next round)
-                if q.is_cuda:
-                    raise NotImplementedError(
-                        "chunked prefill is not yet implemented on the HIP "
-                        "path; prefill whole prompts (engine default)"
-                    )
-                from ..ops import reference as _ref
-
-                return _ref.attn_decode_with_history(
-                    q, k_cache, v_cache, block_table, seq_lens, query_lens, self.scale
+                # chunked prefill: this chunk's K/V were just stored, so the
+                # kernel reads K/V exclusively from the paged cache and masks
+                # causally at absolute positions (HIP attn_prefill_mfma
+                # <HD, PAGED=true>; reference attn_decode_with_history on CPU)
+                return ops.attn_prefill_paged(
+                    q, k_cache, v_cache, block_table, seq_lens, cu_seqlens,
+                    max_seqlen, self.scale,
                 )
             return ops.attn_prefill(q, k, v, cu_seqlens, max_seqlen, self.scale)
 

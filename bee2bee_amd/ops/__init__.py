@@ -144,6 +144,28 @@ def attn_prefill(
     return reference.attn_prefill(q, k, v, cu_seqlens, max_seqlen, scale, causal)
 
 
+def attn_prefill_paged(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_table: torch.Tensor,
+    seq_lens: torch.Tensor,
+    cu_q: torch.Tensor,
+    max_qlen: int,
+    scale: float,
+) -> torch.Tensor:
+    """Chunked prefill: packed query chunks (cu_q) attend to the full paged
+    history (seq_lens includes the chunk; its K/V are already stored)."""
+    if _use_hip(q):
+        return require_hip().attn_prefill_paged(
+            q, k_cache, v_cache, block_table, seq_lens, cu_q, max_qlen, scale
+        )
+    query_lens = cu_q[1:] - cu_q[:-1]
+    return reference.attn_decode_with_history(
+        q, k_cache, v_cache, block_table, seq_lens, query_lens, scale
+    )
+
+
 def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
     if _use_hip(gate_up):
         return require_hip().swiglu(gate_up)

@@ -1,5 +1,12 @@
 // Varlen causal prefill attention (flash-style, no S x S materialization).
 //
+// Both kernels are templated over PAGED: when true, K/V come from the paged
+// KV cache ([nb, nkv, bs, hd] + block table) instead of the packed fresh
+// projections, and each query chunk attends to the full cached history
+// (chunked prefill: query row j of sequence b has absolute position
+// hist + j where hist = seq_lens[b] - query_len[b]). The chunk's own K/V
+// were stored to the cache by kv_store before this kernel runs.
+//
 // Two kernels:
 //  * attn_prefill_mfma — the production path for head_dim 64/128: per
 //    workgroup 4 waves x 16 q-rows (64-row Q tile), 32-key K/V tiles staged
@@ -15,30 +22,36 @@
 #include "common.h"
 
 // ------------------------------------------------------------------ basic
+template <bool PAGED>
 __global__ __launch_bounds__(64) void attn_prefill_basic(
     const unsigned short* __restrict__ q,  // [T, nq, hd]
-    const unsigned short* __restrict__ k,  // [T, nkv, hd]
+    const unsigned short* __restrict__ k,  // [T,nkv,hd] | PAGED: [nb,nkv,bs,hd]
     const unsigned short* __restrict__ v,
-    const int* __restrict__ cu,            // [nseq+1]
+    const int* __restrict__ cu,            // [nseq+1] (query tokens)
+    const int* __restrict__ block_table,   // PAGED only: [nseq, W]
+    const int* __restrict__ seq_lens,      // PAGED only: total len incl chunk
     unsigned short* __restrict__ out,      // [T, nq, hd]
-    int nq, int nkv, int hd, long s_q, long s_k, long s_v, float scale,
-    int causal) {
+    int nq, int nkv, int hd, long s_q, long s_k, long s_v, int W, int bs,
+    float scale, int causal) {
     const int seq = blockIdx.y;
     const int h = blockIdx.x;
     const int kvh = h / (nq / nkv);
     const int s0 = cu[seq], s1 = cu[seq + 1];
-    const int L = s1 - s0;
+    const int QL = s1 - s0;
+    const int L = PAGED ? seq_lens[seq] : QL;
+    const int hist = L - QL;
+    const int* bt = PAGED ? block_table + (long)seq * W : nullptr;
     const int lane = threadIdx.x;
 
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* q_s = reinterpret_cast<float*>(smem_raw);  // [hd]
     float* p_s = q_s + hd;                            // [WAVE]
 
-    for (int r = 0; r < L; ++r) {  // one q row at a time
+    for (int r = 0; r < QL; ++r) {  // one q row at a time
         for (int d = lane; d < hd; d += WAVE)
             q_s[d] = bf2f(q[(long)(s0 + r) * s_q + (long)h * hd + d]) * scale;
         __builtin_amdgcn_s_barrier();
-        const int kmax = causal ? r + 1 : L;
+        const int kmax = causal ? hist + r + 1 : L;
         float m = -1e30f, lsum = 0.f, o0 = 0.f, o1 = 0.f;
         const int d0 = lane * 2;
         for (int base = 0; base < kmax; base += WAVE) {
@@ -46,7 +59,12 @@ __global__ __launch_bounds__(64) void attn_prefill_basic(
             float s = -1e30f;
             if (key < kmax) {
                 s = 0.f;
-                const unsigned short* kr = k + (long)(s0 + key) * s_k + (long)kvh * hd;
+                const unsigned short* kr;
+                if (PAGED)
+                    kr = k + (((long)bt[key / bs] * nkv + kvh) * bs +
+                              key % bs) * hd;
+                else
+                    kr = k + (long)(s0 + key) * s_k + (long)kvh * hd;
                 for (int d = 0; d < hd; ++d) s = fmaf(bf2f(kr[d]), q_s[d], s);
             }
             float cmax = wave_max(s);
@@ -61,8 +79,13 @@ __global__ __launch_bounds__(64) void attn_prefill_basic(
             const int nk = min(WAVE, kmax - base);
             if (d0 < hd) {
                 for (int t = 0; t < nk; ++t) {
-                    const unsigned short* vr =
-                        v + (long)(s0 + base + t) * s_v + (long)kvh * hd;
+                    const int vk = base + t;
+                    const unsigned short* vr;
+                    if (PAGED)
+                        vr = v + (((long)bt[vk / bs] * nkv + kvh) * bs +
+                                  vk % bs) * hd;
+                    else
+                        vr = v + (long)(s0 + vk) * s_v + (long)kvh * hd;
                     o0 = fmaf(p_s[t], bf2f(vr[d0]), o0);
                     o1 = fmaf(p_s[t], bf2f(vr[d0 + 1]), o1);
                 }
@@ -90,22 +113,28 @@ __global__ __launch_bounds__(64) void attn_prefill_basic(
 #define PF_TN 64                     // kv tile (4 x 16-key fragments)
 #define PF_PAD 8                     // LDS row padding (bf16 elems)
 
-template <int HD>
+template <int HD, bool PAGED>
 __global__ __launch_bounds__(256) void attn_prefill_mfma(
     const unsigned short* __restrict__ q,
-    const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ k,  // PAGED: k_cache [nb,nkv,bs,hd]
     const unsigned short* __restrict__ v,
-    const int* __restrict__ cu,
+    const int* __restrict__ cu,            // [nseq+1] query tokens
+    const int* __restrict__ block_table,   // PAGED only
+    const int* __restrict__ seq_lens,      // PAGED only
     unsigned short* __restrict__ out,
-    int nq, int nkv, long s_q, long s_k, long s_v, float scale) {
+    int nq, int nkv, long s_q, long s_k, long s_v, int W, int bs,
+    float scale) {
     constexpr int KSTEPS = HD / 32;
     const int seq = blockIdx.z;
     const int h = blockIdx.y;
     const int kvh = h / (nq / nkv);
     const int s0 = cu[seq], s1 = cu[seq + 1];
-    const int L = s1 - s0;
-    const int q0 = blockIdx.x * PF_TM;  // q tile base (position in seq)
-    if (q0 >= L) return;
+    const int QL = s1 - s0;                // query rows in this chunk
+    const int L = PAGED ? seq_lens[seq] : QL;  // keys incl. cached history
+    const int hist = L - QL;
+    const int* bt = PAGED ? block_table + (long)seq * W : nullptr;
+    const int q0 = blockIdx.x * PF_TM;  // q tile base (chunk-local row)
+    if (q0 >= QL) return;
 
     const int tid = threadIdx.x;
     const int lane = tid % WAVE;
@@ -128,7 +157,7 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
     // ---- load this wave's Q rows into A-fragments (registers), scaled
     bf16x8 a_q[KSTEPS];
     const int my_qrow = q0 + wid * PF_QROWS + li;  // A row = li
-    const bool row_ok = my_qrow < L;
+    const bool row_ok = my_qrow < QL;
     {
         const unsigned short* qr = q + (long)(s0 + my_qrow) * s_q + (long)h * HD;
 #pragma unroll
@@ -156,7 +185,8 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
 #pragma unroll
     for (int nb = 0; nb < HD / 16; ++nb) o_acc[nb] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    const int kv_end = min(L, q0 + PF_TM);  // causal: keys < q-tile end
+    // causal: keys up to the last absolute q position of this tile
+    const int kv_end = min(L, hist + q0 + PF_TM);
 
     for (int kb = 0; kb < kv_end; kb += PF_TN) {
         // ---- stage K tile [32][HD] and V^T tile [HD][32]
@@ -170,10 +200,17 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
                 const int gk = kb + key;
                 short8 kraw{}, vraw{};
                 if (gk < kv_end) {
-                    kraw = *reinterpret_cast<const short8*>(
-                        k + (long)(s0 + gk) * s_k + (long)kvh * HD + d);
-                    vraw = *reinterpret_cast<const short8*>(
-                        v + (long)(s0 + gk) * s_v + (long)kvh * HD + d);
+                    if (PAGED) {
+                        const long koff = (((long)bt[gk / bs] * nkv + kvh) *
+                                           bs + gk % bs) * HD + d;
+                        kraw = *reinterpret_cast<const short8*>(k + koff);
+                        vraw = *reinterpret_cast<const short8*>(v + koff);
+                    } else {
+                        kraw = *reinterpret_cast<const short8*>(
+                            k + (long)(s0 + gk) * s_k + (long)kvh * HD + d);
+                        vraw = *reinterpret_cast<const short8*>(
+                            v + (long)(s0 + gk) * s_v + (long)kvh * HD + d);
+                    }
                 }
                 *reinterpret_cast<short8*>(k_s + key * (HD + PF_PAD) + d) = kraw;
 #pragma unroll
@@ -209,7 +246,7 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
             for (int r = 0; r < 4; ++r) {
                 const int qrow = q0 + wid * PF_QROWS + lg * 4 + r;
                 const int key = kb + f * 16 + li;
-                if (key > qrow || qrow >= L || key >= kv_end)
+                if (key > hist + qrow || qrow >= QL || key >= kv_end)
                     sfrag[f][r] = -1e30f;
                 pmax[r] = fmaxf(pmax[r], sfrag[f][r]);
             }
@@ -285,13 +322,21 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
         const int qrow = orow_base + lg * 4 + r;
-        if (qrow >= L) continue;
+        if (qrow >= QL) continue;
         const float inv = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
         unsigned short* orow = out + ((long)(s0 + qrow) * nq + h) * HD;
 #pragma unroll
         for (int nb = 0; nb < HD / 16; ++nb)
             orow[nb * 16 + li] = f2bf(o_acc[nb][r] * inv);
     }
+}
+
+static int prefill_mfma_smem(int hd) {
+    // LDS: max(K tile, P tile) + V^T tile, bf16; must match VT_OFF
+    const int k_bytes = PF_TN * (hd + PF_PAD) * 2;
+    const int p_bytes = PF_TM * (PF_TN + PF_PAD) * 2;
+    const int vt_bytes = hd * (PF_TN + PF_PAD) * 2;
+    return (k_bytes > p_bytes ? k_bytes : p_bytes) + vt_bytes;
 }
 
 extern "C" void launch_attn_prefill(
@@ -302,23 +347,53 @@ extern "C" void launch_attn_prefill(
     if ((hd == 64 || hd == 128) && causal) {
         const int tiles = (max_seqlen + PF_TM - 1) / PF_TM;
         dim3 grid(tiles, nq, nseq);
-        // LDS: max(K tile, P tile) + V^T tile, bf16
-        const int k_bytes = PF_TN * (hd + PF_PAD) * 2;
-        const int p_bytes = PF_TM * (PF_TN + PF_PAD) * 2;
-        // must match VT_OFF in the kernel: V^T starts past max(K, P)
-        const int vt_bytes = hd * (PF_TN + PF_PAD) * 2;
-        const int smem = (k_bytes > p_bytes ? k_bytes : p_bytes) + vt_bytes;
+        const int smem = prefill_mfma_smem(hd);
         if (hd == 128)
-            hipLaunchKernelGGL(attn_prefill_mfma<128>, grid, dim3(256), smem,
-                               stream, q, k, v, cu, out, nq, nkv, s_q, s_k, s_v, scale);
+            hipLaunchKernelGGL((attn_prefill_mfma<128, false>), grid,
+                               dim3(256), smem, stream, q, k, v, cu, nullptr,
+                               nullptr, out, nq, nkv, s_q, s_k, s_v, 0, 0,
+                               scale);
         else
-            hipLaunchKernelGGL(attn_prefill_mfma<64>, grid, dim3(256), smem,
-                               stream, q, k, v, cu, out, nq, nkv, s_q, s_k, s_v, scale);
+            hipLaunchKernelGGL((attn_prefill_mfma<64, false>), grid, dim3(256),
+                               smem, stream, q, k, v, cu, nullptr, nullptr,
+                               out, nq, nkv, s_q, s_k, s_v, 0, 0, scale);
         return;
     }
     dim3 grid(nq, nseq);
     const int smem = (hd + WAVE) * 4;
-    hipLaunchKernelGGL(attn_prefill_basic, grid, dim3(WAVE), smem, stream, q,
-                       k, v, cu, out, nq, nkv, hd, s_q, s_k, s_v, scale,
-                       causal);
+    hipLaunchKernelGGL(attn_prefill_basic<false>, grid, dim3(WAVE), smem,
+                       stream, q, k, v, cu, nullptr, nullptr, out, nq, nkv,
+                       hd, s_q, s_k, s_v, 0, 0, scale, causal);
+}
+
+// Chunked prefill: query chunks (packed varlen, cu) attend to the full
+// paged history; the chunk's K/V are already in the cache (kv_store ran
+// first), so K/V come exclusively from the paged pool.
+extern "C" void launch_attn_prefill_paged(
+    const unsigned short* q, const unsigned short* k_cache,
+    const unsigned short* v_cache, const int* cu, const int* block_table,
+    const int* seq_lens, unsigned short* out, int nseq, int nq, int nkv,
+    int hd, long s_q, int W, int bs, int max_qlen, float scale,
+    hipStream_t stream) {
+    if (hd == 64 || hd == 128) {
+        const int tiles = (max_qlen + PF_TM - 1) / PF_TM;
+        dim3 grid(tiles, nq, nseq);
+        const int smem = prefill_mfma_smem(hd);
+        if (hd == 128)
+            hipLaunchKernelGGL((attn_prefill_mfma<128, true>), grid,
+                               dim3(256), smem, stream, q, k_cache, v_cache,
+                               cu, block_table, seq_lens, out, nq, nkv, s_q,
+                               0, 0, W, bs, scale);
+        else
+            hipLaunchKernelGGL((attn_prefill_mfma<64, true>), grid, dim3(256),
+                               smem, stream, q, k_cache, v_cache, cu,
+                               block_table, seq_lens, out, nq, nkv, s_q, 0, 0,
+                               W, bs, scale);
+        return;
+    }
+    dim3 grid(nq, nseq);
+    const int smem = (hd + WAVE) * 4;
+    hipLaunchKernelGGL(attn_prefill_basic<true>, grid, dim3(WAVE), smem,
+                       stream, q, k_cache, v_cache, cu, block_table, seq_lens,
+                       out, nq, nkv, hd, s_q, 0, 0, W, bs, scale, 1);
 }

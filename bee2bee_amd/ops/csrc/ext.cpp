@@ -31,6 +31,10 @@ void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          const unsigned short*, const int*, unsigned short*,
                          int, int, int, int, long, long, long, int, float,
                          int, hipStream_t);
+void launch_attn_prefill_paged(const unsigned short*, const unsigned short*,
+                               const unsigned short*, const int*, const int*,
+                               const int*, unsigned short*, int, int, int,
+                               int, long, int, int, int, float, hipStream_t);
 void launch_mfma_probe(const unsigned short*, const unsigned short*, float*,
                        hipStream_t);
 void launch_grouped_gemm(const unsigned short*, const unsigned short*,
@@ -198,6 +202,35 @@ Tensor attn_prefill(const Tensor& q, const Tensor& k, const Tensor& v,
     return out;
 }
 
+Tensor attn_prefill_paged(const Tensor& q, const Tensor& k_cache,
+                          const Tensor& v_cache, const Tensor& block_table,
+                          const Tensor& seq_lens, const Tensor& cu_q,
+                          long max_qlen, double scale) {
+    check_bf16(q, "q");
+    check_bf16(k_cache, "k_cache");
+    TORCH_CHECK(block_table.scalar_type() == torch::kInt32 &&
+                seq_lens.scalar_type() == torch::kInt32 &&
+                cu_q.scalar_type() == torch::kInt32);
+    TORCH_CHECK(block_table.is_contiguous() && cu_q.is_contiguous());
+    const int T = q.size(0), nq = q.size(1), hd = q.size(2);
+    const int nkv = k_cache.size(1), bs = k_cache.size(2);
+    const int W = block_table.size(1);
+    const int nseq = cu_q.size(0) - 1;
+    TORCH_CHECK(nq % nkv == 0);
+    TORCH_CHECK(hd % 8 == 0, "paged prefill needs hd % 8 == 0");
+    TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == hd);
+    TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+    TORCH_CHECK(block_table.size(0) == nseq && seq_lens.size(0) == nseq);
+    Tensor out = torch::empty({T, nq, hd}, q.options());
+    launch_attn_prefill_paged(bf16p(q), bf16p(k_cache), bf16p(v_cache),
+                              cu_q.data_ptr<int>(),
+                              block_table.data_ptr<int>(),
+                              seq_lens.data_ptr<int>(), bf16p_mut(out), nseq,
+                              nq, nkv, hd, q.stride(0), W, bs, (int)max_qlen,
+                              (float)scale, stream());
+    return out;
+}
+
 Tensor grouped_gemm(const Tensor& x, const Tensor& w, const Tensor& offs) {
     check_bf16(x, "x");
     check_bf16(w, "w");
@@ -262,6 +295,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("swiglu", &swiglu, "silu(g) * u");
     m.def("attn_decode", &attn_decode, "paged GQA decode attention");
     m.def("attn_prefill", &attn_prefill, "varlen causal flash prefill");
+    m.def("attn_prefill_paged", &attn_prefill_paged,
+          "chunked prefill vs paged history");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
     m.def("grouped_gemm", &grouped_gemm, "per-expert segment GEMM (MoE)");
     m.def("bw_probe", &bw_probe, "bandwidth pattern probe");

@@ -179,3 +179,55 @@ def test_engine_serves_hf_checkpoint(tmp_path):
     finally:
         eng_ckpt.shutdown()
         eng_rand.shutdown()
+
+
+def test_chunked_prefill_matches_whole():
+    """A prompt longer than the prefill budget must prefill in chunks
+    against the paged history and produce exactly the whole-prefill output
+    (runner chunked path -> ops.attn_prefill_paged)."""
+    prompt = [(i * 7 + 3) % 500 for i in range(50)]
+    eng_small = InferenceEngine("tiny", device="cpu", max_batch=4,
+                                max_seq_len=128, seed=7, max_prefill_tokens=16)
+    try:
+        r1 = eng_small.submit(_greedy_req(list(prompt), 6))
+        _drain(r1)
+        r2 = engine_whole = InferenceEngine(
+            "tiny", device="cpu", max_batch=4, max_seq_len=128, seed=7
+        )
+        try:
+            rw = engine_whole.submit(_greedy_req(list(prompt), 6))
+            _drain(rw)
+            assert r1.output_ids == rw.output_ids
+            assert len(r1.output_ids) == 6
+        finally:
+            engine_whole.shutdown()
+    finally:
+        eng_small.shutdown()
+
+
+def test_chunked_prefill_interleaves_with_decode():
+    """While a long prompt prefills chunk-by-chunk, already-active requests
+    keep decoding and all outputs stay solo-equal."""
+    long_prompt = [(i * 11 + 1) % 500 for i in range(60)]
+    short = [9, 8, 7]
+    ref = InferenceEngine("tiny", device="cpu", max_batch=4,
+                          max_seq_len=128, seed=7)
+    try:
+        solo_long = ref.submit(_greedy_req(list(long_prompt), 5))
+        _drain(solo_long)
+        solo_short = ref.submit(_greedy_req(list(short), 8))
+        _drain(solo_short)
+    finally:
+        ref.shutdown()
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=4,
+                          max_seq_len=128, seed=7, max_prefill_tokens=8)
+    try:
+        rs = eng.submit(_greedy_req(list(short), 8))
+        rl = eng.submit(_greedy_req(list(long_prompt), 5))
+        _drain(rs)
+        _drain(rl)
+        assert rs.output_ids == solo_short.output_ids
+        assert rl.output_ids == solo_long.output_ids
+    finally:
+        eng.shutdown()

@@ -239,3 +239,43 @@ def test_grouped_gemm(S, E, N, K, skew):
     got = ops.grouped_gemm(x, w, offs)
     ref = R.grouped_gemm(x, w, offs)
     assert_close(got, ref, atol=3e-2, rtol=3e-2, msg=f"grouped {S}x{N}x{K}")
+
+
+@pytest.mark.parametrize(
+    "hist_q,nq,nkv,hd",
+    [
+        ([(200, 70), (0, 64), (31, 33)], 32, 8, 128),  # llama3-8b, mixed hist
+        ([(128, 64)], 32, 8, 64),                      # mfma hd=64, aligned
+        ([(100, 1), (5, 90)], 4, 4, 128),              # 1-token chunk + big
+        ([(40, 20), (0, 10)], 4, 2, 16),               # basic fallback path
+    ],
+)
+def test_attn_prefill_paged(hist_q, nq, nkv, hd):
+    """Chunked prefill vs paged history: HIP attn_prefill_paged must match
+    the fp32 reference (attn_decode_with_history). hist=0 rows double-check
+    the degenerate whole-prompt case through the paged path."""
+    torch.manual_seed(11)
+    bs = 32
+    B = len(hist_q)
+    seq_lens = [h + ql for h, ql in hist_q]
+    W = (max(seq_lens) + bs - 1) // bs
+    nb = B * W + 1
+    perm = torch.randperm(nb - 1) + 1
+    bt = perm[: B * W].reshape(B, W).to(torch.int32).to(DEV)
+    kc = (torch.randn(nb, nkv, bs, hd, device=DEV) * 0.5).bfloat16()
+    vc = (torch.randn(nb, nkv, bs, hd, device=DEV) * 0.5).bfloat16()
+    T = sum(ql for _, ql in hist_q)
+    q = (torch.randn(T, nq, hd, device=DEV) * 0.5).bfloat16()
+    cu_list = [0]
+    for _, ql in hist_q:
+        cu_list.append(cu_list[-1] + ql)
+    cu = torch.tensor(cu_list, dtype=torch.int32, device=DEV)
+    lens_dev = torch.tensor(seq_lens, dtype=torch.int32, device=DEV)
+    qlens = torch.tensor([ql for _, ql in hist_q], dtype=torch.int32, device=DEV)
+    scale = hd**-0.5
+    got = ops.attn_prefill_paged(
+        q, kc, vc, bt, lens_dev, cu, max(ql for _, ql in hist_q), scale
+    )
+    ref = R.attn_decode_with_history(q, kc, vc, bt, lens_dev, qlens, scale)
+    assert_close(got, ref, atol=3e-2, rtol=3e-2,
+                 msg=f"prefill_paged {hist_q} hd{hd}")
