@@ -122,11 +122,14 @@ class Runner:
         per-expert gather/GEMM/scatter for prefill-sized ones."""
         s = self.spec
         if self.ep is not None:
-            lo = self.ep.e_lo
-            hi = lo + self.ep.local_e
-            return self.ep.forward(
-                x, lw.moe_gate, lw.moe_w_gate_up[lo:hi], lw.moe_w_down[lo:hi]
-            )
+            if self.weights.expert_range is not None:
+                # weights hold only this rank's expert shard (EP sharding)
+                w_gu, w_dn = lw.moe_w_gate_up, lw.moe_w_down
+            else:
+                lo = self.ep.e_lo
+                hi = lo + self.ep.local_e
+                w_gu, w_dn = lw.moe_w_gate_up[lo:hi], lw.moe_w_down[lo:hi]
+            return self.ep.forward(x, lw.moe_gate, w_gu, w_dn)
         logits = F.linear(x, lw.moe_gate)
         weights, idx = ops.moe_topk_gate(logits, s.top_k_experts)  # [T,k]
         T, H = x.shape

@@ -57,18 +57,31 @@ class ModelWeights:
         self.final_norm: Optional[torch.Tensor] = None  # [hidden]
         self.lm_head: Optional[torch.Tensor] = None  # [vocab, hidden]
         self.layers = [LayerWeights() for _ in range(spec.n_layers)]
+        # expert-parallel shard: when set, MoE expert tensors hold only
+        # experts [lo, hi) (gate stays replicated); see parallel/ep.py
+        self.expert_range: Optional[Tuple[int, int]] = None
 
     # -------------------------------------------------------------- random
 
     @torch.no_grad()
-    def random_init(self, seed: int = 0, layer_range: Optional[Tuple[int, int]] = None) -> "ModelWeights":
+    def random_init(
+        self,
+        seed: int = 0,
+        layer_range: Optional[Tuple[int, int]] = None,
+        expert_range: Optional[Tuple[int, int]] = None,
+    ) -> "ModelWeights":
         """Random weights of the real architecture (synthetic serving — the
         benchmark contract requires random-init weights of the named model).
 
         layer_range (for pipeline stages): only materialize layers [lo, hi);
-        embed only on the first stage, head/final norm only on the last."""
+        embed only on the first stage, head/final norm only on the last.
+        expert_range (for expert parallelism): only materialize that expert
+        shard — drawn per expert, so a shard holds exactly the values the
+        full init would (EP-vs-single equivalence tests rely on this)."""
         s = self.spec
         lo, hi = layer_range or (0, s.n_layers)
+        self.expert_range = expert_range
+        e_lo, e_hi = expert_range or (0, getattr(s, "n_experts", 0) or 0)
         import zlib
 
         # Generate directly on the target device (an 8B randn on host would
@@ -107,14 +120,16 @@ class ModelWeights:
             lw.wo = rnd(f"l{i}.wo", s.hidden_size, s.q_size, std=proj_std)
             if s.is_moe:
                 lw.moe_gate = rnd(f"l{i}.gate", s.n_experts, s.hidden_size, std=std)
-                lw.moe_w_gate_up = rnd(
-                    f"l{i}.w_gu", s.n_experts, 2 * s.intermediate_size,
-                    s.hidden_size, std=std
-                )
-                lw.moe_w_down = rnd(
-                    f"l{i}.w_dn", s.n_experts, s.hidden_size,
-                    s.intermediate_size, std=proj_std
-                )
+                lw.moe_w_gate_up = torch.stack([
+                    rnd(f"l{i}.w_gu.e{e}", 2 * s.intermediate_size,
+                        s.hidden_size, std=std)
+                    for e in range(e_lo, e_hi)
+                ])
+                lw.moe_w_down = torch.stack([
+                    rnd(f"l{i}.w_dn.e{e}", s.hidden_size,
+                        s.intermediate_size, std=proj_std)
+                    for e in range(e_lo, e_hi)
+                ])
             else:
                 lw.w_gate_up = rnd(f"l{i}.w_gu", 2 * s.intermediate_size, s.hidden_size, std=std)
                 lw.w_down = rnd(f"l{i}.w_dn", s.hidden_size, s.intermediate_size, std=proj_std)
@@ -123,10 +138,19 @@ class ModelWeights:
     # ---------------------------------------------------------- safetensors
 
     @torch.no_grad()
-    def load_hf(self, model_path: str, layer_range: Optional[Tuple[int, int]] = None) -> "ModelWeights":
-        """Load HF safetensors shards, fusing qkv / gate|up on the fly."""
+    def load_hf(
+        self,
+        model_path: str,
+        layer_range: Optional[Tuple[int, int]] = None,
+        expert_range: Optional[Tuple[int, int]] = None,
+    ) -> "ModelWeights":
+        """Load HF safetensors shards, fusing qkv / gate|up on the fly.
+        expert_range: keep only that expert shard on device (EP ranks)."""
         s = self.spec
         lo, hi = layer_range or (0, s.n_layers)
+        self.expert_range = expert_range
+        e_lo, e_hi = expert_range or (0, getattr(s, "n_experts", 0) or 0)
+        n_local_e = e_hi - e_lo
         want_embed = lo == 0
         want_head = hi == s.n_layers
 
@@ -194,19 +218,21 @@ class ModelWeights:
                 lw.moe_gate = to_dev(tensor)
             elif parts[3] == "block_sparse_moe" and parts[4] == "experts":
                 e = int(parts[5])
+                if not (e_lo <= e < e_hi):
+                    continue  # expert owned by another EP rank
                 w = parts[6]  # w1 (gate), w2 (down), w3 (up)
                 key = f"moe.{i}"
                 pending.setdefault(key, {})[f"{w}.{e}"] = to_dev(tensor)
                 p = pending[key]
-                if len(p) == 3 * s.n_experts:
+                if len(p) == 3 * n_local_e:
                     lw.moe_w_gate_up = torch.stack(
                         [
                             torch.cat([p[f"w1.{e2}"], p[f"w3.{e2}"]], dim=0)
-                            for e2 in range(s.n_experts)
+                            for e2 in range(e_lo, e_hi)
                         ]
                     )
                     lw.moe_w_down = torch.stack(
-                        [p[f"w2.{e2}"] for e2 in range(s.n_experts)]
+                        [p[f"w2.{e2}"] for e2 in range(e_lo, e_hi)]
                     )
                     pending.pop(key)
 

@@ -64,11 +64,17 @@ class MoEEngine:
             ops.require_hip()
 
         self.max_seq_len = min(max_seq_len, self.spec.max_seq_len)
+        # expert sharding: this rank only materializes its E/world experts
+        # (an 8-way Mixtral rank holds 1/8 of the expert bytes; gate and the
+        # attention stack stay replicated)
+        local_e = self.spec.n_experts // self.world
+        e_lo = self.rank * local_e
+        expert_range = (e_lo, e_lo + local_e)
         self.weights = ModelWeights(self.spec, self.device, dtype)
         if model_path:
-            self.weights.load_hf(model_path)
+            self.weights.load_hf(model_path, expert_range=expert_range)
         else:
-            self.weights.random_init(seed=seed)
+            self.weights.random_init(seed=seed, expert_range=expert_range)
         blocks_per_seq = -(-self.max_seq_len // 32)
         self.kv = PagedKV(
             self.spec, self.device, dtype,

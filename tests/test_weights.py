@@ -52,3 +52,52 @@ def test_layer_range_partial_load():
     assert w.layers[1].wqkv is not None
     assert w.embed is None  # not first stage
     assert w.final_norm is not None  # last stage
+
+
+def test_expert_range_shard_matches_full():
+    """An EP rank's expert shard must hold exactly the full init's values
+    for its expert slice (per-expert seeded draws), for both random_init
+    and the HF checkpoint loader."""
+    import torch
+
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+    spec = PRESETS["tiny-moe"]
+    dev = torch.device("cpu")
+    full = ModelWeights(spec, dev, torch.float32).random_init(3)
+    shard = ModelWeights(spec, dev, torch.float32).random_init(
+        3, expert_range=(2, 4)
+    )
+    assert shard.expert_range == (2, 4)
+    assert shard.layers[0].moe_w_gate_up.shape[0] == 2
+    assert torch.equal(
+        shard.layers[0].moe_w_gate_up, full.layers[0].moe_w_gate_up[2:4]
+    )
+    assert torch.equal(
+        shard.layers[1].moe_w_down, full.layers[1].moe_w_down[2:4]
+    )
+    # gate replicated
+    assert torch.equal(shard.layers[0].moe_gate, full.layers[0].moe_gate)
+
+
+def test_expert_range_load_hf(tmp_path):
+    import torch
+
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+    spec = PRESETS["tiny-moe"]
+    dev = torch.device("cpu")
+    full = ModelWeights(spec, dev, torch.float32).random_init(3)
+    save_hf(full, str(tmp_path))
+    shard = ModelWeights(spec, dev, torch.float32).load_hf(
+        str(tmp_path), expert_range=(1, 3)
+    )
+    assert shard.layers[0].moe_w_gate_up.shape[0] == 2
+    assert torch.allclose(
+        shard.layers[0].moe_w_gate_up, full.layers[0].moe_w_gate_up[1:3]
+    )
+    assert torch.allclose(
+        shard.layers[1].moe_w_down, full.layers[1].moe_w_down[1:3]
+    )
