@@ -330,13 +330,20 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
 }
 
 // ----------------------------------------------------------- stage 2: PV
-template <int G>
+// When FOLD (launcher sets it iff gridDim.z == 1, i.e. every chunk of this
+// (b, kvh) runs serially in THIS workgroup — always true at flagship batch
+// sizes where B*nkv >= 4096), the kernel carries the running flash-decoding
+// merge (M, l, acc) across chunks in registers and writes the normalized
+// bf16 output directly: no part_o round-trip, no combine kernel.
+template <int G, bool FOLD>
 __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     const unsigned short* __restrict__ v_cache,  // [nb, nkv, bs, hd]
     const int* __restrict__ block_table,
     const int* __restrict__ seq_lens,
     const unsigned short* __restrict__ p_in,     // [B,nkv,C,CHUNK,G] bf16
     float* __restrict__ part_o,                  // [B, nkv, C, G, hd]
+    const float* __restrict__ part_ml,           // [B, nkv, C, G, 2]
+    unsigned short* __restrict__ out,            // [B, nq, hd] (FOLD only)
     int nkv, int W, int bs, int hd, int C) {
     const int b = blockIdx.x;
     const int kvh = blockIdx.y;
@@ -349,6 +356,18 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     float* p_s = reinterpret_cast<float*>(smem_raw);  // [CHUNK][G]
     long* voff_s = reinterpret_cast<long*>(p_s + DEC_CHUNK * G);  // [CHUNK]
     float* merge = reinterpret_cast<float*>(voff_s + DEC_CHUNK);  // [Wv][G][hd]
+    // FOLD running state lives in LDS so the hot loop stays at ~64 VGPRs
+    // (a register version measured 101 VGPRs -> 5 waves/SIMD). Only wave 0
+    // touches it, in wave-lockstep, so no barriers are needed around it.
+    float* run = merge + DEC_WAVES * G * hd;   // [G][hd] running acc
+    float* runml = run + G * hd;               // [G][2] running (m, l)
+    if (FOLD && wid == 0) {
+        for (int i = lane; i < G * hd; i += WAVE) run[i] = 0.f;
+        if (lane < G) {
+            runml[lane * 2] = -1e30f;
+            runml[lane * 2 + 1] = 0.f;
+        }
+    }
 
     for (int chunk = blockIdx.z; chunk * DEC_CHUNK < L; chunk += gridDim.z) {
     const int start = chunk * DEC_CHUNK;
@@ -440,13 +459,189 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
                 a0 += merge[(w2 * G + g) * hd + d0];
                 a1 += merge[(w2 * G + g) * hd + d0 + 1];
             }
-            float* po = part_o + (base + g) * hd;
-            po[d0] = a0;
-            po[d0 + 1] = a1;
+            if (FOLD) {
+                // online merge with this chunk's (m, l) from the scores pass
+                const float mc = part_ml[(base + g) * 2];
+                const float lc = part_ml[(base + g) * 2 + 1];
+                const float mold = runml[g * 2];
+                const float mn = fmaxf(mold, mc);
+                const float alpha = __expf(mold - mn);
+                const float beta = __expf(mc - mn);
+                run[g * hd + d0] = run[g * hd + d0] * alpha + a0 * beta;
+                run[g * hd + d0 + 1] =
+                    run[g * hd + d0 + 1] * alpha + a1 * beta;
+                if (lane == 0) {
+                    runml[g * 2] = mn;
+                    runml[g * 2 + 1] = runml[g * 2 + 1] * alpha + lc * beta;
+                }
+            } else {
+                float* po = part_o + (base + g) * hd;
+                po[d0] = a0;
+                po[d0 + 1] = a1;
+            }
         }
     }
     __syncthreads();  // LDS buffers reused by the next chunk iteration
     }  // chunk loop
+
+    if (FOLD && wid == 0) {
+        const int d0f = lane * 2;
+        if (d0f < hd) {
+            const int nq = nkv * G;
+#pragma unroll
+            for (int g = 0; g < G; ++g) {
+                const float lr = runml[g * 2 + 1];
+                const float inv = (lr > 0.f) ? 1.f / lr : 0.f;
+                unsigned short* orow =
+                    out + ((long)b * nq + kvh * G + g) * hd;
+                orow[d0f] = f2bf(run[g * hd + d0f] * inv);
+                orow[d0f + 1] = f2bf(run[g * hd + d0f + 1] * inv);
+            }
+        }
+    }
+}
+
+// ------------------------------------------- stage 2 (MFMA variant, hd 128)
+// P.V on the matrix pipe: out[g][d] = sum_k P[g][k] V[k][d] as 16x16x32
+// MFMA tiles (A = P rows staged transposed [g][key] in LDS, B = V^T staged
+// [d][key] in LDS). The waves SPLIT THE 128 OUTPUT DIMS (2 dim-groups of 16
+// each) and share one cooperative 32-key V subtile per iteration, so there
+// is no cross-wave merge and the per-wave state is 2 C-fragments (8 VGPRs).
+// Per lane per 32 keys this is ~25 ops vs ~420 for the VALU PV (2G-fma
+// chains + G LDS reads per key) — the G-proportional tail the roadmap
+// attributed the 4.2-4.7 TB/s plateau to.
+// vt row stride 40 elems: 80 B rows keep ds_read_b128 16 B-aligned and hit
+// 16 distinct banks at the 64-dword b128 modulus (bank math in the CDNA4
+// guide §2); the b16 scatter writes are 2-way conflicted (acceptable).
+#define PVM_PAD 8
+#define PVM_STRIDE (32 + PVM_PAD)
+
+template <int G>
+__global__ __launch_bounds__(DEC_BLOCK) void attn_pv_mfma_kernel(
+    const unsigned short* __restrict__ v_cache,  // [nb, nkv, bs, 128]
+    const int* __restrict__ block_table,
+    const int* __restrict__ seq_lens,
+    const unsigned short* __restrict__ p_in,     // [B,nkv,C,CHUNK,G] bf16
+    float* __restrict__ part_o,                  // [B, nkv, C, G, 128]
+    int nkv, int W, int bs, int C) {
+    constexpr int HD = 128;
+    constexpr int PT_STRIDE = DEC_CHUNK + 8;
+    const int b = blockIdx.x;
+    const int kvh = blockIdx.y;
+    const int L = seq_lens[b];
+    if ((int)blockIdx.z * DEC_CHUNK >= L) return;
+    const int tid = threadIdx.x;
+    const int lane = tid % WAVE;
+    const int wid = tid / WAVE;
+    const int lg = lane >> 4;
+    const int li = lane & 15;
+
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    unsigned short* p_t = reinterpret_cast<unsigned short*>(smem_raw);
+    const int off_v = (G * PT_STRIDE * 2 + 15) & ~15;
+    long* voff_s = reinterpret_cast<long*>(smem_raw + off_v);  // [CHUNK]
+    unsigned short* vt =
+        reinterpret_cast<unsigned short*>(smem_raw + off_v + DEC_CHUNK * 8);
+    // vt: double-buffered [2][HD][PVM_STRIDE]
+
+    for (int chunk = blockIdx.z; chunk * DEC_CHUNK < L; chunk += gridDim.z) {
+        const int start = chunk * DEC_CHUNK;
+        // stage p transposed ([key][G] -> [g][key]) and the V row offsets
+        {
+            const unsigned short* psrc =
+                p_in + (((long)b * nkv + kvh) * C + chunk) * DEC_CHUNK * G;
+            for (int i = tid; i < DEC_CHUNK * G; i += DEC_BLOCK)
+                p_t[(i % G) * PT_STRIDE + i / G] = psrc[i];
+            const int* btb = block_table + (long)b * W;
+            const int key = start + tid;
+            long vo = 0;
+            if (key < L)
+                vo = (((long)btb[key / bs] * nkv + kvh) * bs + key % bs) * HD;
+            voff_s[tid] = vo;
+        }
+        __syncthreads();
+
+        const int nkeys = min(DEC_CHUNK, L - start);
+        const int nsub = (nkeys + 31) / 32;
+        f32x4 acc[2];
+        acc[0] = acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+        // Cooperative 32-key x 128-dim V subtiles, transposed through LDS.
+        // T14 register staging (guide): loads for subtile t+2 issue BEFORE
+        // the LDS write of t+1 waits on t+1's loads, so each load has a full
+        // iteration (barrier + 2 MFMAs + epilogue traffic) to land. Two
+        // register sets = 2 subtiles of load latency in flight per thread.
+        // Thread t owns key t/8, dims [(t%8)*16, +16): two 16 B loads,
+        // sixteen b16 column writes.
+        const int st_key = tid >> 3;
+        const int st_d = (tid & 7) * 16;
+        short8 ra[2], rb[2];
+#define PVM_LOAD(SUB, SET)                                                     \
+    do {                                                                       \
+        const int k_ = (SUB) * 32 + st_key;                                    \
+        ra[SET] = short8{};                                                    \
+        rb[SET] = short8{};                                                    \
+        if (k_ < nkeys) {                                                      \
+            const unsigned short* vr_ = v_cache + voff_s[k_] + st_d;           \
+            ra[SET] = *reinterpret_cast<const short8*>(vr_);                   \
+            rb[SET] = *reinterpret_cast<const short8*>(vr_ + 8);               \
+        }                                                                      \
+    } while (0)
+#define PVM_WRITE(SET, BUF)                                                    \
+    do {                                                                       \
+        unsigned short* w_ = vt + (BUF) * HD * PVM_STRIDE + st_key;            \
+        _Pragma("unroll") for (int e = 0; e < 8; ++e)                          \
+            w_[(st_d + e) * PVM_STRIDE] = (unsigned short)ra[SET][e];          \
+        _Pragma("unroll") for (int e = 0; e < 8; ++e)                          \
+            w_[(st_d + 8 + e) * PVM_STRIDE] = (unsigned short)rb[SET][e];      \
+    } while (0)
+
+// one 32-key step: LDS-write reg set, prefetch 2 subtiles ahead, barrier,
+// 2 MFMAs. SET/BUF are LITERALS so the compiler tracks which outstanding
+// loads each write actually needs (a runtime-indexed set forced
+// vmcnt(0) at every write: measured 2.1 TB/s vs 4.1)
+#define PVM_STEP(SUB, SET)                                                     \
+    do {                                                                       \
+        PVM_WRITE(SET, SET);                                                   \
+        if ((SUB) + 2 < nsub) PVM_LOAD((SUB) + 2, SET);                        \
+        __syncthreads();                                                       \
+        const unsigned short* vb_ = vt + (SET) * HD * PVM_STRIDE;              \
+        const bf16x8 ap_ = *reinterpret_cast<const bf16x8*>(                   \
+            p_t + prow + (SUB) * 32 + lg * 8);                                 \
+        _Pragma("unroll") for (int dgi = 0; dgi < 2; ++dgi) {                  \
+            const bf16x8 bv_ = *reinterpret_cast<const bf16x8*>(               \
+                vb_ + ((wid * 2 + dgi) * 16 + li) * PVM_STRIDE + lg * 8);      \
+            acc[dgi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                \
+                ap_, bv_, acc[dgi], 0, 0, 0);                                  \
+        }                                                                      \
+    } while (0)
+
+        PVM_LOAD(0, 0);
+        if (nsub > 1) PVM_LOAD(1, 1);
+        const int prow = (li < G ? li : G - 1) * PT_STRIDE;
+        int sub = 0;
+        for (; sub + 1 < nsub; sub += 2) {
+            PVM_STEP(sub, 0);
+            PVM_STEP(sub + 1, 1);
+        }
+        if (sub < nsub) PVM_STEP(sub, 0);
+#undef PVM_STEP
+#undef PVM_LOAD
+#undef PVM_WRITE
+
+        // C layout: lane holds rows g = lg*4 + r, col d = dgroup*16 + li
+        const long base = (((long)b * nkv + kvh) * C + chunk) * G;
+#pragma unroll
+        for (int dgi = 0; dgi < 2; ++dgi) {
+            const int d = (wid * 2 + dgi) * 16 + li;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int g = lg * 4 + r;
+                if (g < G) part_o[(base + g) * HD + d] = acc[dgi][r];
+            }
+        }
+        __syncthreads();  // p_t / voff_s reused by the next chunk
+    }
 }
 
 // ------------------------------------------------------ stage 3: combine
@@ -501,8 +696,20 @@ extern "C" void launch_attn_decode(
     dim3 grid(B, nkv, Z);
     dim3 cgrid(B, nkv, G);
     const int smem_s = ((G * hd * 2 + 15) & ~15) + DEC_WAVES * G * 2 * 4;
-    const int smem_pv =
-        DEC_CHUNK * G * 4 + DEC_CHUNK * 8 + DEC_WAVES * G * hd * 4;
+    const int smem_pv = DEC_CHUNK * G * 4 + DEC_CHUNK * 8 +
+                        DEC_WAVES * G * hd * 4 +
+                        G * hd * 4 + G * 8;  // + FOLD running state
+    const int smem_pvm = ((G * (DEC_CHUNK + 8) * 2 + 15) & ~15) +
+                         DEC_CHUNK * 8 + 2 * hd * PVM_STRIDE * 2;
+    // MFMA PV (attn_pv_mfma_kernel) measured 4.2 TB/s vs the VALU PV's
+    // 4.5 at B768/G4: the barrier-paced LDS pipeline loses more latency
+    // hiding than the matrix pipe saves (PV is latency-, not VALU-bound).
+    // Kept as opt-in infrastructure for future tr_b16 staging work.
+    static int pv_mfma = -1;
+    if (pv_mfma < 0) {
+        const char* e = getenv("BEE2BEE_PV_MFMA");
+        pv_mfma = (e != nullptr && e[0] == '1') ? 1 : 0;
+    }
 #define LAUNCH(GG)                                                             \
     do {                                                                       \
         if (hd == 128)                                                         \
@@ -520,12 +727,28 @@ extern "C" void launch_attn_decode(
                                smem_s, stream, q, k_cache, block_table,        \
                                seq_lens, p_buf, part_ml, nkv, W, bs, hd, C,    \
                                q_stride, scale);                               \
-        hipLaunchKernelGGL(attn_pv_kernel<GG>, grid, dim3(DEC_BLOCK),          \
-                           smem_pv, stream, v_cache, block_table, seq_lens,    \
-                           p_buf, part_o, nkv, W, bs, hd, C);                  \
-        hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,              \
-                           dim3(DEC_BLOCK), 0, stream, part_o, part_ml,        \
-                           seq_lens, out, nkv, hd, C);                         \
+        if (hd == 128 && pv_mfma) {                                            \
+            hipLaunchKernelGGL(attn_pv_mfma_kernel<GG>, grid,                  \
+                               dim3(DEC_BLOCK), smem_pvm, stream, v_cache,     \
+                               block_table, seq_lens, p_buf, part_o, nkv, W,   \
+                               bs, C);                                         \
+            hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,          \
+                               dim3(DEC_BLOCK), 0, stream, part_o, part_ml,    \
+                               seq_lens, out, nkv, hd, C);                     \
+        } else if (Z == 1) {                                                   \
+            hipLaunchKernelGGL((attn_pv_kernel<GG, true>), grid,               \
+                               dim3(DEC_BLOCK), smem_pv, stream, v_cache,      \
+                               block_table, seq_lens, p_buf, part_o, part_ml,  \
+                               out, nkv, W, bs, hd, C);                        \
+        } else {                                                               \
+            hipLaunchKernelGGL((attn_pv_kernel<GG, false>), grid,              \
+                               dim3(DEC_BLOCK), smem_pv, stream, v_cache,      \
+                               block_table, seq_lens, p_buf, part_o,           \
+                               part_ml, out, nkv, W, bs, hd, C);               \
+            hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,          \
+                               dim3(DEC_BLOCK), 0, stream, part_o, part_ml,    \
+                               seq_lens, out, nkv, hd, C);                     \
+        }                                                                      \
     } while (0)
     switch (G) {
         case 1: LAUNCH(1); break;
