@@ -105,11 +105,21 @@ class InferenceEngine:
         seed: int = 0,
         kv_margin_blocks: int = 8,
         max_prefill_tokens: int = 4096,
+        spec_decode: bool = False,
+        spec_k: int = 4,
+        spec_ngram: int = 2,
     ) -> None:
         # cap on prompt tokens prefill-batched per step: bounds time-to-first
         # -token for requests behind a burst (they decode while later
         # arrivals prefill)
         self.max_prefill_tokens = max_prefill_tokens
+        # speculative decoding (prompt-lookup / n-gram proposer + exact
+        # greedy verification over the paged-history prefill path); output-
+        # invariant: verified tokens equal plain greedy decode
+        self.spec_decode = spec_decode
+        self.spec_k = spec_k
+        self.spec_ngram = spec_ngram
+        self.spec_stats = {"steps": 0, "proposed": 0, "accepted": 0}
         self.spec = model if isinstance(model, ModelSpec) else resolve_spec(model, model_path)
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
@@ -150,7 +160,7 @@ class InferenceEngine:
 
         # MoE captures only in the dense all-experts range (static shapes);
         # the padded-bmm path above it has a data-dependent buffer size
-        graphs_ok = on_gpu and (
+        graphs_ok = on_gpu and not spec_decode and (
             not self.spec.is_moe or max_batch <= Runner.MOE_DENSE_MAX_TOKENS
         )
         self.use_graphs = graphs_ok if use_graphs is None else (use_graphs and graphs_ok)
@@ -342,7 +352,10 @@ class InferenceEngine:
         if worked:
             self._prefill(self._prefilling + admitted)
         if self._active:
-            self._decode_once()
+            if self.spec_decode:
+                self._decode_spec_once()
+            else:
+                self._decode_once()
             return True
         return worked
 
@@ -549,6 +562,135 @@ class InferenceEngine:
         if done:
             self._active = [a for a in self._active if a.req.done_ts is None]
             self._dec_seqs = None  # membership changed
+
+    # ------------------------------------------------- speculative decoding
+
+    def _propose(self, ids: List[int], k: int) -> List[int]:
+        """Prompt-lookup proposer: continuation of the most recent earlier
+        occurrence of the trailing n-gram (no draft model needed)."""
+        n = self.spec_ngram
+        if k <= 0 or len(ids) <= n:
+            return []
+        tail = ids[-n:]
+        best: List[int] = []
+        for i in range(len(ids) - n - 1, -1, -1):
+            if ids[i : i + n] == tail:
+                cont = ids[i + n : i + n + k]
+                if len(cont) > len(best):
+                    best = cont
+                if len(best) == k:
+                    break
+        return best
+
+    @torch.no_grad()
+    def _decode_spec_once(self) -> None:
+        """One speculative step: each greedy sequence verifies its pending
+        token plus up to spec_k proposed tokens in ONE forward through the
+        chunked-prefill (paged-history) path; the accepted prefix is exactly
+        what plain greedy decode would emit (verification is exact)."""
+        dev = self.device
+        acts = self._active
+        ids_list, pos_list, slot_list, cu = [], [], [], [0]
+        props: List[List[int]] = []
+        for a in acts:
+            r = a.req
+            ctx = r.prompt_ids + r.output_ids
+            prop: List[int] = []
+            if r.sampling.greedy and r.sampling.repetition_penalty == 1.0:
+                room = self.max_seq_len - (a.length + 1) - 1
+                rem = r.max_new_tokens - len(r.output_ids) - 1
+                kcap = min(self.spec_k, room, rem)
+                prop = self._propose(ctx, kcap)
+            toks = [ctx[-1]] + prop
+            props.append(prop)
+            self.kv.extend_seq(a.seq_id, a.length + len(toks))
+            ids_list.extend(toks)
+            pos_list.extend(range(a.length, a.length + len(toks)))
+            slot_list.extend(
+                self.kv.slot_mapping(a.seq_id, range(a.length, a.length + len(toks)))
+            )
+            cu.append(cu[-1] + len(toks))
+        input_ids = torch.tensor(ids_list, dtype=torch.int64, device=dev)
+        positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
+        slots = torch.tensor(slot_list, dtype=torch.int32, device=dev)
+        cu_t = torch.tensor(cu, dtype=torch.int32, device=dev)
+        qlens = [cu[i + 1] - cu[i] for i in range(len(acts))]
+        bt = self.kv.block_table([a.seq_id for a in acts])
+        seq_lens = torch.tensor(
+            [a.length + q for a, q in zip(acts, qlens)],
+            dtype=torch.int32, device=dev,
+        )
+        qlens_t = torch.tensor(qlens, dtype=torch.int32, device=dev)
+        hidden = self.runner.forward_prefill(
+            input_ids, positions, slots, cu_t, max(qlens),
+            block_table=bt, seq_lens=seq_lens, query_lens=qlens_t,
+        )
+        logits = self.runner.lm_head(hidden)
+
+        greedy_idx = [i for i, a in enumerate(acts)
+                      if a.req.sampling.greedy
+                      and a.req.sampling.repetition_penalty == 1.0]
+        nong = [a for i, a in enumerate(acts) if i not in set(greedy_idx)]
+        if greedy_idx:
+            argmax = logits.argmax(dim=-1).cpu()
+        now = time.time()
+        n_emitted = 0
+        for i in greedy_idx:
+            a, r, prop = acts[i], acts[i].req, props[i]
+            self.spec_stats["proposed"] += len(prop)
+            emitted: List[int] = []
+            j = 0
+            while True:
+                tok = int(argmax[cu[i] + j])
+                emitted.append(tok)
+                if j < len(prop) and tok == prop[j] and tok not in r.stop_token_ids:
+                    j += 1
+                    continue
+                break
+            self.spec_stats["accepted"] += j
+            # rewind the KV length past rejected proposals (their stored
+            # keys get overwritten; attention never reads past seq_len)
+            a.length += len(emitted)
+            self.kv.extend_seq(a.seq_id, a.length)
+            for tok in emitted:
+                if r.done_ts is not None:
+                    break
+                if r.first_token_ts is None:
+                    r.first_token_ts = now
+                r.output_ids.append(tok)
+                n_emitted += 1
+                done = (r.cancelled or tok in r.stop_token_ids
+                        or len(r.output_ids) >= r.max_new_tokens)
+                if done:
+                    r.done_ts = now
+                if r.on_emit is not None:
+                    try:
+                        r.on_emit(tok, done)
+                    except Exception:
+                        logger.exception("on_emit callback failed")
+                else:
+                    r.out_queue.put(tok)
+                    if done:
+                        r.out_queue.put(_STREAM_END)
+        if n_emitted:
+            self._note_throughput(n_emitted, now)
+        if nong:
+            # sampled / penalized requests take the plain one-token path
+            rows = torch.tensor(
+                [cu[i + 1] - 1 for i, a in enumerate(acts)
+                 if i not in set(greedy_idx)],
+                dtype=torch.int64, device=logits.device,
+            )
+            for a in nong:
+                a.length += 1
+            self._sample_and_emit(nong, logits[rows])
+        self.spec_stats["steps"] += 1
+        self._dec_seqs = None  # device decode-state caches are stale
+        done_acts = [a for a in acts if a.req.done_ts is not None]
+        for a in done_acts:
+            self.kv.free_seq(a.seq_id)
+        if done_acts:
+            self._active = [a for a in acts if a.req.done_ts is None]
 
     def _sample_and_emit(self, acts: List[_Active], logits: torch.Tensor) -> None:
         # group rows by sampling params so each group is one sample() call

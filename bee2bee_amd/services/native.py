@@ -46,6 +46,8 @@ class NativeEngineService(BaseService):
 
     def load_sync(self) -> None:
         try:
+            import os
+
             from ..engine.engine import InferenceEngine
 
             self.engine = InferenceEngine(
@@ -54,6 +56,9 @@ class NativeEngineService(BaseService):
                 model_path=self.model_path,
                 max_batch=self.max_batch,
                 max_seq_len=self.max_seq_len,
+                # speculative decoding (prompt-lookup + exact verification):
+                # serving opt-in; greedy outputs are provably unchanged
+                spec_decode=os.environ.get("BEE2BEE_SPEC_DECODE") == "1",
             )
             self.engine.start()
         except Exception as e:
