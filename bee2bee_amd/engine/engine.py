@@ -362,7 +362,10 @@ class InferenceEngine:
     def _admit(self) -> List[_Active]:
         admitted: List[_Active] = []
         admit_tokens = 0
-        while len(self._active) + len(admitted) < self.max_batch:
+        # count in-flight chunked prefills too: they join _active later and
+        # must not push it past max_batch (graph buckets size to it)
+        while (len(self._active) + len(self._prefilling) + len(admitted)
+               < self.max_batch):
             if admitted and admit_tokens >= self.max_prefill_tokens:
                 break
             try:

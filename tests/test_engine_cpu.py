@@ -231,3 +231,30 @@ def test_chunked_prefill_interleaves_with_decode():
         assert rl.output_ids == solo_long.output_ids
     finally:
         eng.shutdown()
+
+
+def test_active_set_never_exceeds_max_batch():
+    """Chunked prefills joining the active set must not push it past
+    max_batch (regression: _admit did not count _prefilling, so the
+    decode batch overflowed the graph bucket at high concurrency)."""
+    eng = InferenceEngine("tiny", device="cpu", max_batch=3, max_seq_len=128,
+                          seed=7, max_prefill_tokens=8)
+    try:
+        reqs = [
+            eng.submit(_greedy_req([(i * 13 + j) % 500 for j in range(30)], 6))
+            for i in range(8)
+        ]
+        peak = 0
+        import time as _t
+
+        for _ in range(400):
+            peak = max(peak, len(eng._active))
+            assert len(eng._active) <= 3, f"active set overflowed: {len(eng._active)}"
+            if all(r.done_ts is not None for r in reqs):
+                break
+            _t.sleep(0.01)
+        for r in reqs:
+            _drain(r)
+            assert len(r.output_ids) == 6
+    finally:
+        eng.shutdown()
