@@ -359,9 +359,19 @@ class InferenceEngine:
             return True
         return worked
 
+    def _need_blocks(self, prompt_len: int, max_new: int) -> int:
+        need_len = min(prompt_len + max_new, self.max_seq_len)
+        return -(-need_len // self.kv.block_size)
+
     def _admit(self) -> List[_Active]:
         admitted: List[_Active] = []
         admit_tokens = 0
+        # blocks the current residents will still grow into: admission must
+        # leave room for them or a later extend_seq blows the pool mid-step
+        reserved = 0
+        for a in self._active + self._prefilling:
+            need = self._need_blocks(len(a.req.prompt_ids), a.req.max_new_tokens)
+            reserved += max(0, need - self.kv.seq_n_blocks(a.seq_id))
         # count in-flight chunked prefills too: they join _active later and
         # must not push it past max_batch (graph buckets size to it)
         while (len(self._active) + len(self._prefilling) + len(admitted)
@@ -372,14 +382,14 @@ class InferenceEngine:
                 req = self._pending.get_nowait()
             except queue.Empty:
                 break
-            need_len = min(
-                len(req.prompt_ids) + req.max_new_tokens, self.max_seq_len
+            blocks_needed = self._need_blocks(
+                len(req.prompt_ids), req.max_new_tokens
             )
-            blocks_needed = -(-need_len // self.kv.block_size)
-            if self.kv.free_blocks < blocks_needed:
+            if self.kv.free_blocks - reserved < blocks_needed:
                 # out of KV memory: push back and wait for retirements
                 self._pending.put(req)
                 break
+            reserved += blocks_needed
             seq_id = self._next_seq
             self._next_seq += 1
             self.kv.new_seq(seq_id)
@@ -407,11 +417,10 @@ class InferenceEngine:
             batch.append((a, a.prefilled, take))
             budget -= take
             n_consumed += 1
-        self._prefilling = [
-            a
-            for a, start, take in batch
-            if start + take < len(a.req.prompt_ids)
-        ] + [t for t in queue_[n_consumed:]]
+        # keep EVERYTHING in _prefilling until this step succeeds, so a
+        # mid-prefill failure reaches the error path (which fails + frees
+        # _active and _prefilling) instead of orphaning admitted requests
+        self._prefilling = list(queue_)
 
         dev = self.device
         ids_list, pos_list, slot_list, cu = [], [], [], [0]
@@ -451,6 +460,11 @@ class InferenceEngine:
                 input_ids, positions, slots, cu_seqlens, max_len,
                 block_table=bt, seq_lens=seq_lens, query_lens=query_lens,
             )
+        self._prefilling = [
+            a
+            for a, start, take in batch
+            if start + take < len(a.req.prompt_ids)
+        ] + [t for t in queue_[n_consumed:]]
         # sample only for requests whose whole prompt is now in the cache
         done_idx = [
             i

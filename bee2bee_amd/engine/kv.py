@@ -67,6 +67,9 @@ class PagedKV:
     def seq_len(self, seq_id: int) -> int:
         return self._seq_len.get(seq_id, 0)
 
+    def seq_n_blocks(self, seq_id: int) -> int:
+        return len(self._seq_blocks.get(seq_id, ()))
+
     def extend_seq(self, seq_id: int, new_len: int) -> None:
         """Grow a sequence to new_len tokens, allocating blocks as needed."""
         blocks = self._seq_blocks[seq_id]

@@ -98,3 +98,32 @@ def test_malformed_wire_frames_do_not_crash_node():
         await node.stop()
 
     asyncio.run(run())
+
+
+def test_kv_pool_exhaustion_backpressure():
+    """More concurrent demand than KV blocks: admission must defer (push
+    back to pending) rather than crash, and every request must finish."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    # pool sized for ~2 sequences; submit 6 long ones
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=64,
+                          seed=3, kv_margin_blocks=0)
+    try:
+        reqs = [
+            eng.submit(GenerationRequest(
+                prompt_ids=[(i * 7 + j) % 500 for j in range(40)],
+                max_new_tokens=10,
+                sampling=SamplingParams(greedy=True),
+            ))
+            for i in range(6)
+        ]
+        for r in reqs:
+            while True:
+                x = r.out_queue.get(timeout=120)
+                if not isinstance(x, int):
+                    break
+            assert r.error is None, r.error
+            assert len(r.output_ids) == 10
+    finally:
+        eng.shutdown()
