@@ -108,6 +108,7 @@ class InferenceEngine:
         spec_decode: bool = False,
         spec_k: int = 4,
         spec_ngram: int = 2,
+        kv_dtype: str = "native",
     ) -> None:
         # cap on prompt tokens prefill-batched per step: bounds time-to-first
         # -token for requests behind a burst (they decode while later
@@ -155,7 +156,14 @@ class InferenceEngine:
 
         blocks_per_seq = -(-self.max_seq_len // 32)
         n_blocks = max_batch * blocks_per_seq + kv_margin_blocks
-        self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks)
+        # "fp8": OCP e4m3 KV pool — half the decode-attention bytes; the
+        # chunked-prefill and speculative paths read the cache as bf16 and
+        # are not fp8-enabled yet, so they are rejected up front
+        self.kv_dtype = kv_dtype
+        if kv_dtype == "fp8" and spec_decode:
+            raise ValueError("fp8 KV does not support spec_decode yet")
+        self.kv = PagedKV(self.spec, self.device, dtype, n_blocks=n_blocks,
+                          kv_dtype=kv_dtype)
         self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
 
         # MoE captures only in the dense all-experts range (static shapes);

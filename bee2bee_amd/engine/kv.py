@@ -27,20 +27,28 @@ class PagedKV:
         n_blocks: int,
         block_size: int = BLOCK_SIZE,
         layer_range: Optional[Tuple[int, int]] = None,
+        kv_dtype: str = "native",
     ) -> None:
+        """kv_dtype: "native" stores the compute dtype; "fp8" stores OCP
+        e4m3 bytes (uint8 pool, HALF the KV bytes — the decode kernels read
+        fp8 directly; opt-in, reported separately from the bf16 headline)."""
         self.spec = spec
         self.device = device
         self.dtype = dtype
+        self.kv_dtype = kv_dtype
         self.block_size = block_size
         self.n_blocks = n_blocks
         lo, hi = layer_range or (0, spec.n_layers)
         self.layer_lo = lo
+        store_dtype = torch.uint8 if kv_dtype == "fp8" else dtype
         shape = (n_blocks, spec.n_kv_heads, block_size, spec.head_dim)
         self.k_cache = [
-            torch.zeros(shape, device=device, dtype=dtype) for _ in range(lo, hi)
+            torch.zeros(shape, device=device, dtype=store_dtype)
+            for _ in range(lo, hi)
         ]
         self.v_cache = [
-            torch.zeros(shape, device=device, dtype=dtype) for _ in range(lo, hi)
+            torch.zeros(shape, device=device, dtype=store_dtype)
+            for _ in range(lo, hi)
         ]
         self._free: List[int] = list(range(n_blocks - 1, -1, -1))
         self._seq_blocks: Dict[int, List[int]] = {}

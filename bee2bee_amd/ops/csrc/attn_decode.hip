@@ -21,18 +21,61 @@
 #define DEC_WAVES 4
 #define DEC_CHUNK 256  // keys per workgroup (DEC_WAVES x 64)
 
+// KV element loaders, templated on the cache storage type:
+//   unsigned short = bf16 (2 B/elem), unsigned char = OCP fp8 e4m3 (1 B).
+template <typename KVT>
+__device__ __forceinline__ bf16x8 load_kv_bf16x8(const KVT* p);
+template <>
+__device__ __forceinline__ bf16x8 load_kv_bf16x8(const unsigned short* p) {
+    return *reinterpret_cast<const bf16x8*>(p);
+}
+template <>
+__device__ __forceinline__ bf16x8 load_kv_bf16x8(const unsigned char* p) {
+    float f[8];
+    load_fp8x8(p, f);
+    bf16x8 r;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) r[e] = (__bf16)f[e];
+    return r;
+}
+// raw 2-element cache pair (batched loads keep the raw bytes in flight,
+// converting only when consumed)
+template <typename KVT> struct kv_pair_t;
+template <> struct kv_pair_t<unsigned short> { short2v v; };
+template <> struct kv_pair_t<unsigned char> { u8x2 v; };
+__device__ __forceinline__ float2v kv_pair_f32(kv_pair_t<unsigned short> p) {
+    return float2v{bf2f((unsigned short)p.v[0]), bf2f((unsigned short)p.v[1])};
+}
+__device__ __forceinline__ float2v kv_pair_f32(kv_pair_t<unsigned char> p) {
+    unsigned short w = (unsigned short)p.v[0] |
+                       ((unsigned short)p.v[1] << 8);
+    return fp8x2_2f(w);
+}
+
+template <typename KVT>
+__device__ __forceinline__ float2v load_kv_f32x2(const KVT* p);
+template <>
+__device__ __forceinline__ float2v load_kv_f32x2(const unsigned short* p) {
+    const short2v v = *reinterpret_cast<const short2v*>(p);
+    return float2v{bf2f((unsigned short)v[0]), bf2f((unsigned short)v[1])};
+}
+template <>
+__device__ __forceinline__ float2v load_kv_f32x2(const unsigned char* p) {
+    return fp8x2_2f(*reinterpret_cast<const unsigned short*>(p));
+}
+
 // ------------------------------------------------------------ stage 1: S
 // Whole K row staged into registers (HD/8 independent 16 B loads in one
 // burst), then G dot products against the LDS-held q. Compile-time HD keeps
 // the row array in registers (a runtime bound would spill it to scratch).
-template <int G, int HD>
-__device__ __forceinline__ void score_row(const unsigned short* kr,
+template <int G, int HD, typename KVT>
+__device__ __forceinline__ void score_row(const KVT* kr,
                                           const unsigned short* q_s,
                                           float* s) {
-    short8 krow[HD / 8];
+    bf16x8 krow[HD / 8];
 #pragma unroll
     for (int ii = 0; ii < HD / 8; ++ii)
-        krow[ii] = *reinterpret_cast<const short8*>(kr + ii * 8);
+        krow[ii] = load_kv_bf16x8<KVT>(kr + ii * 8);
 #pragma unroll
     for (int g = 0; g < G; ++g) {
 #pragma unroll
@@ -41,7 +84,7 @@ __device__ __forceinline__ void score_row(const unsigned short* kr,
                 *reinterpret_cast<const short8*>(q_s + g * HD + ii * 8);
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-                s[g] = fmaf(bf2f((unsigned short)krow[ii][j]),
+                s[g] = fmaf((float)krow[ii][j],
                             bf2f((unsigned short)qraw[j]), s[g]);
         }
         // stop the scheduler hoisting every g's LDS q-reads to the top
@@ -50,10 +93,10 @@ __device__ __forceinline__ void score_row(const unsigned short* kr,
     }
 }
 
-template <int G>
+template <int G, typename KVT>
 __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     const unsigned short* __restrict__ q,        // [B, nq, hd] (strided)
-    const unsigned short* __restrict__ k_cache,  // [nb, nkv, bs, hd]
+    const KVT* __restrict__ k_cache,             // [nb, nkv, bs, hd]
     const int* __restrict__ block_table,         // [B, W]
     const int* __restrict__ seq_lens,            // [B]
     unsigned short* __restrict__ p_out,          // [B, nkv, C, CHUNK, G] bf16
@@ -89,24 +132,24 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     for (int g = 0; g < G; ++g) s[g] = -1e30f;
     if (valid) {
         const int page = bt[key / bs];
-        const unsigned short* kr =
+        const KVT* kr =
             k_cache + (((long)page * nkv + kvh) * bs + key % bs) * hd;
 #pragma unroll
         for (int g = 0; g < G; ++g) s[g] = 0.f;
         if (hd == 128) {
-            score_row<G, 128>(kr, q_s, s);
+            score_row<G, 128, KVT>(kr, q_s, s);
         } else if (hd == 64) {
-            score_row<G, 64>(kr, q_s, s);
+            score_row<G, 64, KVT>(kr, q_s, s);
         } else {
             for (int d = 0; d < hd; d += 8) {  // tail models (hd 16/24/...)
-                const short8 kraw = *reinterpret_cast<const short8*>(kr + d);
+                const bf16x8 kraw = load_kv_bf16x8<KVT>(kr + d);
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
                     const short8 qraw =
                         *reinterpret_cast<const short8*>(q_s + g * hd + d);
 #pragma unroll
                     for (int j = 0; j < 8; ++j)
-                        s[g] = fmaf(bf2f((unsigned short)kraw[j]),
+                        s[g] = fmaf((float)kraw[j],
                                     bf2f((unsigned short)qraw[j]), s[g]);
                 }
             }
@@ -175,10 +218,10 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
 //   A (16x32): lane = row (l&15), k = (l>>4)*8+e
 //   B (32x16): lane = col (l&15), k = (l>>4)*8+e
 //   C (16x16): lane = col (l&15), row = (l>>4)*4+r
-template <int G, int HD>
+template <int G, int HD, typename KVT>
 __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     const unsigned short* __restrict__ q,
-    const unsigned short* __restrict__ k_cache,
+    const KVT* __restrict__ k_cache,
     const int* __restrict__ block_table,
     const int* __restrict__ seq_lens,
     unsigned short* __restrict__ p_out,  // bf16
@@ -235,7 +278,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
 #pragma unroll
     for (int t = 0; t < TILES; ++t) {
         const int key = wave_key0 + t * 16 + li;  // A row = li
-        const unsigned short* kr;
+        const KVT* kr;
         if (key < L) {
             const int page = bt[key / bs];
             kr = k_cache + (((long)page * nkv + kvh) * bs + key % bs) * HD;
@@ -245,8 +288,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
         }
 #pragma unroll
         for (int ks = 0; ks < KSTEPS; ++ks)
-            afr[t][ks] =
-                *reinterpret_cast<const bf16x8*>(kr + ks * 32 + lg * 8);
+            afr[t][ks] = load_kv_bf16x8<KVT>(kr + ks * 32 + lg * 8);
     }
 #pragma unroll
     for (int t = 0; t < TILES; ++t) {
@@ -335,9 +377,9 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
 // sizes where B*nkv >= 4096), the kernel carries the running flash-decoding
 // merge (M, l, acc) across chunks in registers and writes the normalized
 // bf16 output directly: no part_o round-trip, no combine kernel.
-template <int G, bool FOLD>
+template <int G, bool FOLD, typename KVT>
 __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
-    const unsigned short* __restrict__ v_cache,  // [nb, nkv, bs, hd]
+    const KVT* __restrict__ v_cache,             // [nb, nkv, bs, hd]
     const int* __restrict__ block_table,
     const int* __restrict__ seq_lens,
     const unsigned short* __restrict__ p_in,     // [B,nkv,C,CHUNK,G] bf16
@@ -408,34 +450,33 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
         const long* voffs = voff_s + wave_start;
         const float* pw = p_s + wave_start * G;
         int t = 0;
-        for (; t + 32 <= nkeys; t += 32) {  // 32 V loads in flight (8 KB/wave)
-            short2v vv[32];
+        for (; t + 32 <= nkeys; t += 32) {  // 32 V loads in flight
+            kv_pair_t<KVT> vv[32];
 #pragma unroll
             for (int j = 0; j < 32; ++j)
-                vv[j] = *reinterpret_cast<const short2v*>(
+                vv[j] = *reinterpret_cast<const kv_pair_t<KVT>*>(
                     v_cache + voffs[t + j] + d0);
 #pragma unroll
             for (int j = 0; j < 32; ++j) {
-                const float v0 = bf2f((unsigned short)vv[j][0]);
-                const float v1 = bf2f((unsigned short)vv[j][1]);
+                const float2v vf = kv_pair_f32(vv[j]);
                 const float* pt = pw + (t + j) * G;
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
-                    o0[g] = fmaf(pt[g], v0, o0[g]);
-                    o1[g] = fmaf(pt[g], v1, o1[g]);
+                    o0[g] = fmaf(pt[g], vf.x, o0[g]);
+                    o1[g] = fmaf(pt[g], vf.y, o1[g]);
                 }
             }
         }
         for (; t < nkeys; ++t) {
-            const short2v vv =
-                *reinterpret_cast<const short2v*>(v_cache + voffs[t] + d0);
-            const float v0 = bf2f((unsigned short)vv[0]);
-            const float v1 = bf2f((unsigned short)vv[1]);
+            const kv_pair_t<KVT> vv =
+                *reinterpret_cast<const kv_pair_t<KVT>*>(
+                    v_cache + voffs[t] + d0);
+            const float2v vf = kv_pair_f32(vv);
             const float* pt = pw + t * G;
 #pragma unroll
             for (int g = 0; g < G; ++g) {
-                o0[g] = fmaf(pt[g], v0, o0[g]);
-                o1[g] = fmaf(pt[g], v1, o1[g]);
+                o0[g] = fmaf(pt[g], vf.x, o0[g]);
+                o1[g] = fmaf(pt[g], vf.y, o1[g]);
             }
         }
     }
@@ -683,11 +724,11 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
 }
 
 extern "C" void launch_attn_decode(
-    const unsigned short* q, const unsigned short* k_cache,
-    const unsigned short* v_cache, const int* block_table,
-    const int* seq_lens, unsigned short* p_buf, float* part_o, float* part_ml,
-    unsigned short* out, int B, int nkv, int G, int W, int bs, int hd, int C,
-    long q_stride, float scale, hipStream_t stream) {
+    const unsigned short* q, const void* k_cache, const void* v_cache,
+    const int* block_table, const int* seq_lens, unsigned short* p_buf,
+    float* part_o, float* part_ml, unsigned short* out, int B, int nkv,
+    int G, int W, int bs, int hd, int C, long q_stride, float scale,
+    int fp8, hipStream_t stream) {
     // enough blocks to fill the chip, but chunks loop within a block so the
     // cold-start/tail phases amortize over multiple 64 KB K/V bursts
     int Z = (4096 + B * nkv - 1) / (B * nkv);
@@ -710,45 +751,54 @@ extern "C" void launch_attn_decode(
         const char* e = getenv("BEE2BEE_PV_MFMA");
         pv_mfma = (e != nullptr && e[0] == '1') ? 1 : 0;
     }
-#define LAUNCH(GG)                                                             \
+#define LAUNCH_T(GG, KVT)                                                      \
     do {                                                                       \
+        const KVT* kc = reinterpret_cast<const KVT*>(k_cache);                 \
+        const KVT* vc = reinterpret_cast<const KVT*>(v_cache);                 \
         if (hd == 128)                                                         \
-            hipLaunchKernelGGL((attn_scores_mfma_kernel<GG, 128>), grid,       \
-                               dim3(DEC_BLOCK), smem_s, stream, q, k_cache,    \
+            hipLaunchKernelGGL((attn_scores_mfma_kernel<GG, 128, KVT>), grid,  \
+                               dim3(DEC_BLOCK), smem_s, stream, q, kc,         \
                                block_table, seq_lens, p_buf, part_ml, nkv, W,  \
                                bs, C, q_stride, scale);                        \
         else if (hd == 64)                                                     \
-            hipLaunchKernelGGL((attn_scores_mfma_kernel<GG, 64>), grid,        \
-                               dim3(DEC_BLOCK), smem_s, stream, q, k_cache,    \
+            hipLaunchKernelGGL((attn_scores_mfma_kernel<GG, 64, KVT>), grid,   \
+                               dim3(DEC_BLOCK), smem_s, stream, q, kc,         \
                                block_table, seq_lens, p_buf, part_ml, nkv, W,  \
                                bs, C, q_stride, scale);                        \
         else                                                                   \
-            hipLaunchKernelGGL(attn_scores_kernel<GG>, grid, dim3(DEC_BLOCK),  \
-                               smem_s, stream, q, k_cache, block_table,        \
-                               seq_lens, p_buf, part_ml, nkv, W, bs, hd, C,    \
-                               q_stride, scale);                               \
-        if (hd == 128 && pv_mfma) {                                            \
+            hipLaunchKernelGGL((attn_scores_kernel<GG, KVT>), grid,            \
+                               dim3(DEC_BLOCK), smem_s, stream, q, kc,         \
+                               block_table, seq_lens, p_buf, part_ml, nkv, W,  \
+                               bs, hd, C, q_stride, scale);                    \
+        if (hd == 128 && pv_mfma && !fp8) {                                    \
             hipLaunchKernelGGL(attn_pv_mfma_kernel<GG>, grid,                  \
-                               dim3(DEC_BLOCK), smem_pvm, stream, v_cache,     \
+                               dim3(DEC_BLOCK), smem_pvm, stream,              \
+                               reinterpret_cast<const unsigned short*>(        \
+                                   v_cache),                                   \
                                block_table, seq_lens, p_buf, part_o, nkv, W,   \
                                bs, C);                                         \
             hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,          \
                                dim3(DEC_BLOCK), 0, stream, part_o, part_ml,    \
                                seq_lens, out, nkv, hd, C);                     \
         } else if (Z == 1) {                                                   \
-            hipLaunchKernelGGL((attn_pv_kernel<GG, true>), grid,               \
-                               dim3(DEC_BLOCK), smem_pv, stream, v_cache,      \
+            hipLaunchKernelGGL((attn_pv_kernel<GG, true, KVT>), grid,          \
+                               dim3(DEC_BLOCK), smem_pv, stream, vc,           \
                                block_table, seq_lens, p_buf, part_o, part_ml,  \
                                out, nkv, W, bs, hd, C);                        \
         } else {                                                               \
-            hipLaunchKernelGGL((attn_pv_kernel<GG, false>), grid,              \
-                               dim3(DEC_BLOCK), smem_pv, stream, v_cache,      \
+            hipLaunchKernelGGL((attn_pv_kernel<GG, false, KVT>), grid,         \
+                               dim3(DEC_BLOCK), smem_pv, stream, vc,           \
                                block_table, seq_lens, p_buf, part_o,           \
                                part_ml, out, nkv, W, bs, hd, C);               \
             hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,          \
                                dim3(DEC_BLOCK), 0, stream, part_o, part_ml,    \
                                seq_lens, out, nkv, hd, C);                     \
         }                                                                      \
+    } while (0)
+#define LAUNCH(GG)                                                             \
+    do {                                                                       \
+        if (fp8) LAUNCH_T(GG, unsigned char);                                  \
+        else LAUNCH_T(GG, unsigned short);                                     \
     } while (0)
     switch (G) {
         case 1: LAUNCH(1); break;
@@ -758,4 +808,5 @@ extern "C" void launch_attn_decode(
         default: LAUNCH(16); break;
     }
 #undef LAUNCH
+#undef LAUNCH_T
 }

@@ -14,6 +14,31 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(2))) short short2v;
 typedef __attribute__((ext_vector_type(2))) float float2v;
+typedef __attribute__((ext_vector_type(8))) unsigned char uchar8;
+typedef __attribute__((ext_vector_type(2))) unsigned char u8x2;
+
+// --- OCP fp8 (e4m3) helpers: gfx950's v_cvt_* work on packed pairs.
+// pack two floats into the low word of an int (2 x fp8 bytes)
+__device__ __forceinline__ unsigned short f2fp8x2(float a, float b) {
+    return (unsigned short)(__builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false) &
+                            0xffff);
+}
+// unpack two fp8 bytes (in the low word) to two floats
+__device__ __forceinline__ float2v fp8x2_2f(unsigned short w) {
+    return __builtin_amdgcn_cvt_pk_f32_fp8((int)w, false);
+}
+// 8 fp8 bytes -> 8 floats
+__device__ __forceinline__ void load_fp8x8(const unsigned char* p,
+                                           float* out) {
+    const unsigned int* pi = reinterpret_cast<const unsigned int*>(p);
+    const unsigned int w0 = pi[0], w1 = pi[1];
+    const float2v a = __builtin_amdgcn_cvt_pk_f32_fp8((int)w0, false);
+    const float2v b = __builtin_amdgcn_cvt_pk_f32_fp8((int)w0, true);
+    const float2v c = __builtin_amdgcn_cvt_pk_f32_fp8((int)w1, false);
+    const float2v d = __builtin_amdgcn_cvt_pk_f32_fp8((int)w1, true);
+    out[0] = a.x; out[1] = a.y; out[2] = b.x; out[3] = b.y;
+    out[4] = c.x; out[5] = c.y; out[6] = d.x; out[7] = d.y;
+}
 
 // bf16 (as ushort bits) <-> float
 __device__ __forceinline__ float bf2f(unsigned short u) {

@@ -124,6 +124,36 @@ __global__ void kv_store_kernel(
     if (hd % 2) { /* head dims are even for all supported models */ }
 }
 
+// fp8 (OCP e4m3) KV cache variant: quantize at store time (RNE pack), so
+// the decode kernels read HALF the bytes. Scale is 1.0 (e4m3 covers +-448;
+// post-RMSNorm K/V magnitudes sit well inside — opt-in, documented).
+__global__ void kv_store_fp8_kernel(
+    const unsigned short* __restrict__ k,  // [T, nkv, hd] bf16
+    const unsigned short* __restrict__ v,
+    unsigned char* __restrict__ k_cache,   // [nb, nkv, bs, hd] fp8
+    unsigned char* __restrict__ v_cache,
+    const int* __restrict__ slots,
+    int T, int nkv, int hd, int bs, long sk, long sv) {
+    const int glob_wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (glob_wave >= T * nkv) return;
+    const int t = glob_wave / nkv;
+    const int h = glob_wave % nkv;
+    const int slot = slots[t];
+    const int blk = slot / bs, off = slot % bs;
+    const long dst = (((long)blk * nkv + h) * bs + off) * hd;
+    const unsigned short* ks = k + (long)t * sk + (long)h * hd;
+    const unsigned short* vs = v + (long)t * sv + (long)h * hd;
+    for (int d = lane * 2; d + 1 < hd; d += WAVE * 2) {
+        const short2v kk = *reinterpret_cast<const short2v*>(ks + d);
+        const short2v vv = *reinterpret_cast<const short2v*>(vs + d);
+        *reinterpret_cast<unsigned short*>(k_cache + dst + d) =
+            f2fp8x2(bf2f((unsigned short)kk[0]), bf2f((unsigned short)kk[1]));
+        *reinterpret_cast<unsigned short*>(v_cache + dst + d) =
+            f2fp8x2(bf2f((unsigned short)vv[0]), bf2f((unsigned short)vv[1]));
+    }
+}
+
 // ----------------------------------------------------------------- swiglu
 // out[t, i] = silu(gu[t, i]) * gu[t, I + i]; grid-stride, short8 loads.
 __global__ void swiglu_kernel(
@@ -185,6 +215,17 @@ void launch_kv_store(const unsigned short* k, const unsigned short* v,
     const long blocks = (waves + 3) / 4;
     hipLaunchKernelGGL(kv_store_kernel, dim3(blocks), dim3(256), 0, stream, k,
                        v, k_cache, v_cache, slots, T, nkv, hd, bs, sk, sv);
+}
+
+void launch_kv_store_fp8(const unsigned short* k, const unsigned short* v,
+                         unsigned char* k_cache, unsigned char* v_cache,
+                         const int* slots, int T, int nkv, int hd, int bs,
+                         long sk, long sv, hipStream_t stream) {
+    const long waves = (long)T * nkv;
+    const long blocks = (waves + 3) / 4;
+    hipLaunchKernelGGL(kv_store_fp8_kernel, dim3(blocks), dim3(256), 0,
+                       stream, k, v, k_cache, v_cache, slots, T, nkv, hd, bs,
+                       sk, sv);
 }
 
 void launch_swiglu(const unsigned short* gu, unsigned short* out, long T,

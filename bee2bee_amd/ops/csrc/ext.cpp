@@ -20,13 +20,15 @@ void launch_rope(unsigned short*, unsigned short*, const int*, const float*,
 void launch_kv_store(const unsigned short*, const unsigned short*,
                      unsigned short*, unsigned short*, const int*, int, int,
                      int, int, long, long, hipStream_t);
+void launch_kv_store_fp8(const unsigned short*, const unsigned short*,
+                         unsigned char*, unsigned char*, const int*, int,
+                         int, int, int, long, long, hipStream_t);
 void launch_swiglu(const unsigned short*, unsigned short*, long, long,
                    hipStream_t);
-void launch_attn_decode(const unsigned short*, const unsigned short*,
-                        const unsigned short*, const int*, const int*,
-                        unsigned short*, float*, float*, unsigned short*,
-                        int, int, int, int, int, int, int, long, float,
-                        hipStream_t);
+void launch_attn_decode(const unsigned short*, const void*, const void*,
+                        const int*, const int*, unsigned short*, float*,
+                        float*, unsigned short*, int, int, int, int, int,
+                        int, int, long, float, int, hipStream_t);
 void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          const unsigned short*, const int*, unsigned short*,
                          int, int, int, int, long, long, long, int, float,
@@ -54,6 +56,8 @@ namespace {
 inline hipStream_t stream() {
     return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
+
+inline bool hd_fp8_ok(long hd) { return hd % 8 == 0; }
 
 inline const unsigned short* bf16p(const Tensor& t) {
     return reinterpret_cast<const unsigned short*>(t.data_ptr());
@@ -119,7 +123,6 @@ void rope_inplace(Tensor& q, Tensor& k, const Tensor& pos, const Tensor& cos_t,
 void kv_cache_store(const Tensor& k, const Tensor& v, Tensor& k_cache,
                     Tensor& v_cache, const Tensor& slots) {
     check_bf16(k, "k");
-    check_bf16(k_cache, "k_cache");
     TORCH_CHECK(slots.scalar_type() == torch::kInt32);
     const int T = k.size(0), nkv = k.size(1), hd = k.size(2);
     const int bs = k_cache.size(2);
@@ -128,6 +131,16 @@ void kv_cache_store(const Tensor& k, const Tensor& v, Tensor& k_cache,
     TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == hd);
     TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == hd);
     TORCH_CHECK(hd % 2 == 0);
+    if (k_cache.scalar_type() == torch::kUInt8) {
+        // fp8 (OCP e4m3) cache: quantize at store time
+        launch_kv_store_fp8(bf16p(k), bf16p(v),
+                            k_cache.data_ptr<unsigned char>(),
+                            v_cache.data_ptr<unsigned char>(),
+                            slots.data_ptr<int>(), T, nkv, hd, bs,
+                            k.stride(0), v.stride(0), stream());
+        return;
+    }
+    check_bf16(k_cache, "k_cache");
     launch_kv_store(bf16p(k), bf16p(v), bf16p_mut(k_cache), bf16p_mut(v_cache),
                     slots.data_ptr<int>(), T, nkv, hd, bs, k.stride(0),
                     v.stride(0), stream());
@@ -148,7 +161,10 @@ Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
                    const Tensor& v_cache, const Tensor& block_table,
                    const Tensor& seq_lens, double scale) {
     check_bf16(q, "q");
-    check_bf16(k_cache, "k_cache");
+    const bool fp8 = k_cache.scalar_type() == torch::kUInt8;
+    if (!fp8) check_bf16(k_cache, "k_cache");
+    TORCH_CHECK(fp8 ? hd_fp8_ok(q.size(2)) : true,
+                "fp8 KV decode needs head_dim % 8 == 0");
     TORCH_CHECK(block_table.scalar_type() == torch::kInt32 &&
                 seq_lens.scalar_type() == torch::kInt32);
     TORCH_CHECK(block_table.is_contiguous());
@@ -170,12 +186,12 @@ Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
     Tensor p_buf = torch::empty({B, nkv, C, kDecChunk, G}, q.options());
     Tensor part_o = torch::empty({B, nkv, C, G, hd}, fopt);
     Tensor part_ml = torch::empty({B, nkv, C, G, 2}, fopt);
-    launch_attn_decode(bf16p(q), bf16p(k_cache), bf16p(v_cache),
+    launch_attn_decode(bf16p(q), k_cache.data_ptr(), v_cache.data_ptr(),
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                        bf16p_mut(p_buf), part_o.data_ptr<float>(),
                        part_ml.data_ptr<float>(),
                        bf16p_mut(out), B, nkv, G, W, bs, hd, C, q.stride(0),
-                       (float)scale, stream());
+                       (float)scale, fp8 ? 1 : 0, stream());
     return out;
 }
 

@@ -66,6 +66,20 @@ def rope_inplace(
         t[..., hd // 2 :] = (x2 * c + x1 * s).to(t.dtype)
 
 
+def _kv_deq(cache: torch.Tensor) -> torch.Tensor:
+    """Dequantize an fp8 (OCP e4m3, uint8 storage) cache to fp32; bf16/fp32
+    caches pass through as fp32."""
+    if cache.dtype == torch.uint8:
+        return cache.view(torch.float8_e4m3fn).float()
+    return cache.float()
+
+
+def _kv_quant_like(x: torch.Tensor, cache: torch.Tensor) -> torch.Tensor:
+    if cache.dtype == torch.uint8:
+        return x.float().to(torch.float8_e4m3fn).view(torch.uint8)
+    return x.to(cache.dtype)
+
+
 def kv_cache_store(
     k: torch.Tensor,
     v: torch.Tensor,
@@ -76,8 +90,8 @@ def kv_cache_store(
     n_blocks, n_kv, block_size, hd = k_cache.shape
     blk = torch.div(slot_mapping, block_size, rounding_mode="floor").long()
     off = (slot_mapping % block_size).long()
-    k_cache[blk, :, off, :] = k.to(k_cache.dtype)
-    v_cache[blk, :, off, :] = v.to(v_cache.dtype)
+    k_cache[blk, :, off, :] = _kv_quant_like(k, k_cache)
+    v_cache[blk, :, off, :] = _kv_quant_like(v, v_cache)
 
 
 def attn_decode(
@@ -97,14 +111,14 @@ def attn_decode(
         L = int(seq_lens[b])
         nblk = (L + block_size - 1) // block_size
         blocks = block_table[b, :nblk].long()
-        keys = k_cache[blocks]  # [nblk, n_kv, bs, hd]
-        vals = v_cache[blocks]
+        keys = _kv_deq(k_cache[blocks])  # [nblk, n_kv, bs, hd]
+        vals = _kv_deq(v_cache[blocks])
         keys = keys.permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
         vals = vals.permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
         qb = q[b].float().view(n_kv, group, hd)
-        scores = torch.einsum("kgd,kld->kgl", qb, keys.float()) * scale
+        scores = torch.einsum("kgd,kld->kgl", qb, keys) * scale
         probs = torch.softmax(scores, dim=-1)
-        ob = torch.einsum("kgl,kld->kgd", probs, vals.float())
+        ob = torch.einsum("kgl,kld->kgd", probs, vals)
         out[b] = ob.reshape(nq, hd).to(q.dtype)
     return out
 
@@ -166,19 +180,21 @@ def attn_decode_with_history(
         nblk = (L + block_size - 1) // block_size
         blocks = block_table[b, :nblk].long()
         keys = (
-            k_cache[blocks].permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
+            _kv_deq(k_cache[blocks])
+            .permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
         )
         vals = (
-            v_cache[blocks].permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
+            _kv_deq(v_cache[blocks])
+            .permute(1, 0, 2, 3).reshape(n_kv, nblk * block_size, hd)[:, :L]
         )
         qb = q[t0 : t0 + QL].float().view(QL, n_kv, group, hd)
-        scores = torch.einsum("qkgd,kld->kgql", qb, keys.float()) * scale
+        scores = torch.einsum("qkgd,kld->kgql", qb, keys) * scale
         # causal within the chunk: query j (absolute pos L-QL+j) sees keys <= pos
         qpos = torch.arange(L - QL, L, device=q.device).view(1, 1, QL, 1)
         kpos = torch.arange(L, device=q.device).view(1, 1, 1, L)
         scores.masked_fill_(kpos > qpos, float("-inf"))
         probs = torch.softmax(scores, dim=-1)
-        ob = torch.einsum("kgql,kld->qkgd", probs, vals.float())
+        ob = torch.einsum("kgql,kld->qkgd", probs, vals)
         out[t0 : t0 + QL] = ob.reshape(QL, nq, hd).to(q.dtype)
         t0 += QL
     return out

@@ -33,6 +33,9 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--batch", type=int, default=1024)
+    ap.add_argument("--kv-dtype", default="native", choices=["native", "fp8"],
+                    help="fp8 = OCP e4m3 KV cache (reported separately from "
+                         "the bf16 headline; never the default)")
     ap.add_argument("--prompt-len", type=int, default=1024)
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--no-tunableop", action="store_true",
@@ -93,6 +96,7 @@ def main() -> None:
         max_seq_len=args.prompt_len + budget + 1,
         use_graphs=(not args.no_graphs) and on_gpu,
         seed=1234 + rank,
+        kv_dtype=args.kv_dtype,
     )
     spec = engine.spec
 
@@ -139,7 +143,7 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": "bf16" if on_gpu else "fp32-cpu-fallback",
+            "dtype": ("bf16+fp8kv" if args.kv_dtype == "fp8" else "bf16") if on_gpu else "fp32-cpu-fallback",
             "data": "synthetic",
             "config": {
                 "model": spec.name,

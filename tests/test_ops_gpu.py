@@ -279,3 +279,47 @@ def test_attn_prefill_paged(hist_q, nq, nkv, hd):
     ref = R.attn_decode_with_history(q, kc, vc, bt, lens_dev, qlens, scale)
     assert_close(got, ref, atol=3e-2, rtol=3e-2,
                  msg=f"prefill_paged {hist_q} hd{hd}")
+
+
+def test_fp8_kv_store_matches_reference():
+    """HIP fp8 quantize-at-store must produce the same e4m3 bytes as the
+    torch reference (both RNE)."""
+    torch.manual_seed(13)
+    T, nkv, hd, bs, nb = 20, 4, 128, 32, 4
+    k = (torch.randn(T, nkv, hd, device=DEV) * 2).bfloat16()
+    v = (torch.randn(T, nkv, hd, device=DEV) * 2).bfloat16()
+    slots = torch.randperm(nb * bs, device=DEV)[:T].to(torch.int32)
+    kc = torch.zeros(nb, nkv, bs, hd, dtype=torch.uint8, device=DEV)
+    vc = torch.zeros_like(kc)
+    kc_ref, vc_ref = kc.clone(), vc.clone()
+    ops.kv_cache_store(k, v, kc, vc, slots)
+    R.kv_cache_store(k, v, kc_ref, vc_ref, slots)
+    assert torch.equal(kc, kc_ref)
+    assert torch.equal(vc, vc_ref)
+
+
+@pytest.mark.parametrize("B,nkv,G,hd,maxlen", [(3, 8, 4, 128, 500),
+                                               (2, 8, 8, 128, 300),
+                                               (2, 8, 4, 64, 129)])
+def test_fp8_kv_attn_decode(B, nkv, G, hd, maxlen):
+    """fp8-KV decode attention (HIP) vs the reference on the SAME quantized
+    cache — exactness of the pipeline, not of the quantization."""
+    torch.manual_seed(14)
+    bs = 32
+    nq = nkv * G
+    lens = torch.randint(1, maxlen + 1, (B,), dtype=torch.int32)
+    lens[0] = maxlen
+    W = (maxlen + bs - 1) // bs
+    nb = B * W + 1
+    perm = torch.randperm(nb - 1) + 1
+    bt = perm[: B * W].reshape(B, W).to(torch.int32).to(DEV)
+    kc = torch.randint(0, 255, (nb, nkv, bs, hd), dtype=torch.uint8, device=DEV)
+    vc = torch.randint(0, 255, (nb, nkv, bs, hd), dtype=torch.uint8, device=DEV)
+    # avoid fp8 NaN encodings (0x7f/0xff)
+    kc[kc == 0x7F] = 0; kc[kc == 0xFF] = 0
+    vc[vc == 0x7F] = 0; vc[vc == 0xFF] = 0
+    q = (torch.randn(B, nq, hd, device=DEV) * 0.2).bfloat16()
+    lens_dev = lens.to(DEV)
+    got = ops.attn_decode(q, kc, vc, bt, lens_dev, hd**-0.5)
+    ref = R.attn_decode(q, kc, vc, bt, lens_dev, hd**-0.5)
+    assert_close(got, ref, atol=3e-2, rtol=3e-2, msg=f"fp8 decode G{G} hd{hd}")
