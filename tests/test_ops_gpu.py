@@ -313,11 +313,15 @@ def test_fp8_kv_attn_decode(B, nkv, G, hd, maxlen):
     nb = B * W + 1
     perm = torch.randperm(nb - 1) + 1
     bt = perm[: B * W].reshape(B, W).to(torch.int32).to(DEV)
-    kc = torch.randint(0, 255, (nb, nkv, bs, hd), dtype=torch.uint8, device=DEV)
-    vc = torch.randint(0, 255, (nb, nkv, bs, hd), dtype=torch.uint8, device=DEV)
-    # avoid fp8 NaN encodings (0x7f/0xff)
-    kc[kc == 0x7F] = 0; kc[kc == 0xFF] = 0
-    vc[vc == 0x7F] = 0; vc[vc == 0xFF] = 0
+    # realistic magnitudes: quantize randn K/V through the fp8 store path
+    # (raw random bytes decode to +-448 outliers where bf16-p rounding noise
+    # crosses any absolute tolerance)
+    kc = torch.zeros(nb, nkv, bs, hd, dtype=torch.uint8, device=DEV)
+    vc = torch.zeros_like(kc)
+    kf = (torch.randn(nb * bs, nkv, hd, device=DEV) * 0.7).bfloat16()
+    vf = (torch.randn(nb * bs, nkv, hd, device=DEV) * 0.7).bfloat16()
+    all_slots = torch.arange(nb * bs, dtype=torch.int32, device=DEV)
+    ops.kv_cache_store(kf, vf, kc, vc, all_slots)
     q = (torch.randn(B, nq, hd, device=DEV) * 0.2).bfloat16()
     lens_dev = lens.to(DEV)
     got = ops.attn_decode(q, kc, vc, bt, lens_dev, hd**-0.5)
