@@ -21,12 +21,31 @@
 // Numerics reference: ops/reference.py attn_prefill.
 #include "common.h"
 
+// KV element loaders shared by the bf16 and fp8 (OCP e4m3) cache paths
+__device__ __forceinline__ float kv_elem_f32(unsigned short x) {
+    return bf2f(x);
+}
+__device__ __forceinline__ float kv_elem_f32(unsigned char x) {
+    return fp8x2_2f((unsigned short)x).x;
+}
+__device__ __forceinline__ bf16x8 load_kv8_bf16(const unsigned short* p) {
+    return *reinterpret_cast<const bf16x8*>(p);
+}
+__device__ __forceinline__ bf16x8 load_kv8_bf16(const unsigned char* p) {
+    float f[8];
+    load_fp8x8(p, f);
+    bf16x8 r;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) r[e] = (__bf16)f[e];
+    return r;
+}
+
 // ------------------------------------------------------------------ basic
-template <bool PAGED>
+template <bool PAGED, typename KVT>
 __global__ __launch_bounds__(64) void attn_prefill_basic(
     const unsigned short* __restrict__ q,  // [T, nq, hd]
-    const unsigned short* __restrict__ k,  // [T,nkv,hd] | PAGED: [nb,nkv,bs,hd]
-    const unsigned short* __restrict__ v,
+    const KVT* __restrict__ k,  // [T,nkv,hd] | PAGED: [nb,nkv,bs,hd]
+    const KVT* __restrict__ v,
     const int* __restrict__ cu,            // [nseq+1] (query tokens)
     const int* __restrict__ block_table,   // PAGED only: [nseq, W]
     const int* __restrict__ seq_lens,      // PAGED only: total len incl chunk
@@ -59,13 +78,14 @@ __global__ __launch_bounds__(64) void attn_prefill_basic(
             float s = -1e30f;
             if (key < kmax) {
                 s = 0.f;
-                const unsigned short* kr;
+                const KVT* kr;
                 if (PAGED)
                     kr = k + (((long)bt[key / bs] * nkv + kvh) * bs +
                               key % bs) * hd;
                 else
                     kr = k + (long)(s0 + key) * s_k + (long)kvh * hd;
-                for (int d = 0; d < hd; ++d) s = fmaf(bf2f(kr[d]), q_s[d], s);
+                for (int d = 0; d < hd; ++d)
+                    s = fmaf(kv_elem_f32(kr[d]), q_s[d], s);
             }
             float cmax = wave_max(s);
             float mn = fmaxf(m, cmax);
@@ -80,14 +100,14 @@ __global__ __launch_bounds__(64) void attn_prefill_basic(
             if (d0 < hd) {
                 for (int t = 0; t < nk; ++t) {
                     const int vk = base + t;
-                    const unsigned short* vr;
+                    const KVT* vr;
                     if (PAGED)
                         vr = v + (((long)bt[vk / bs] * nkv + kvh) * bs +
                                   vk % bs) * hd;
                     else
                         vr = v + (long)(s0 + vk) * s_v + (long)kvh * hd;
-                    o0 = fmaf(p_s[t], bf2f(vr[d0]), o0);
-                    o1 = fmaf(p_s[t], bf2f(vr[d0 + 1]), o1);
+                    o0 = fmaf(p_s[t], kv_elem_f32(vr[d0]), o0);
+                    o1 = fmaf(p_s[t], kv_elem_f32(vr[d0 + 1]), o1);
                 }
             }
         }
@@ -113,11 +133,11 @@ __global__ __launch_bounds__(64) void attn_prefill_basic(
 #define PF_TN 64                     // kv tile (4 x 16-key fragments)
 #define PF_PAD 8                     // LDS row padding (bf16 elems)
 
-template <int HD, bool PAGED>
+template <int HD, bool PAGED, typename KVT>
 __global__ __launch_bounds__(256) void attn_prefill_mfma(
     const unsigned short* __restrict__ q,
-    const unsigned short* __restrict__ k,  // PAGED: k_cache [nb,nkv,bs,hd]
-    const unsigned short* __restrict__ v,
+    const KVT* __restrict__ k,  // PAGED: k_cache [nb,nkv,bs,hd]
+    const KVT* __restrict__ v,
     const int* __restrict__ cu,            // [nseq+1] query tokens
     const int* __restrict__ block_table,   // PAGED only
     const int* __restrict__ seq_lens,      // PAGED only
@@ -198,24 +218,25 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma(
                 const int key = i / THREADS_PER_ROW;
                 const int d = (i % THREADS_PER_ROW) * 8;
                 const int gk = kb + key;
-                short8 kraw{}, vraw{};
+                bf16x8 kraw{}, vraw{};
                 if (gk < kv_end) {
                     if (PAGED) {
                         const long koff = (((long)bt[gk / bs] * nkv + kvh) *
                                            bs + gk % bs) * HD + d;
-                        kraw = *reinterpret_cast<const short8*>(k + koff);
-                        vraw = *reinterpret_cast<const short8*>(v + koff);
+                        kraw = load_kv8_bf16(k + koff);
+                        vraw = load_kv8_bf16(v + koff);
                     } else {
-                        kraw = *reinterpret_cast<const short8*>(
+                        kraw = load_kv8_bf16(
                             k + (long)(s0 + gk) * s_k + (long)kvh * HD + d);
-                        vraw = *reinterpret_cast<const short8*>(
+                        vraw = load_kv8_bf16(
                             v + (long)(s0 + gk) * s_v + (long)kvh * HD + d);
                     }
                 }
-                *reinterpret_cast<short8*>(k_s + key * (HD + PF_PAD) + d) = kraw;
+                *reinterpret_cast<bf16x8*>(k_s + key * (HD + PF_PAD) + d) = kraw;
 #pragma unroll
                 for (int e = 0; e < 8; ++e)
-                    vt_s[(d + e) * VT_STRIDE + key] = (unsigned short)vraw[e];
+                    vt_s[(d + e) * VT_STRIDE + key] =
+                        ((const unsigned short*)&vraw)[e];
             }
         }
         __syncthreads();
@@ -349,51 +370,60 @@ extern "C" void launch_attn_prefill(
         dim3 grid(tiles, nq, nseq);
         const int smem = prefill_mfma_smem(hd);
         if (hd == 128)
-            hipLaunchKernelGGL((attn_prefill_mfma<128, false>), grid,
-                               dim3(256), smem, stream, q, k, v, cu, nullptr,
-                               nullptr, out, nq, nkv, s_q, s_k, s_v, 0, 0,
-                               scale);
+            hipLaunchKernelGGL((attn_prefill_mfma<128, false, unsigned short>),
+                               grid, dim3(256), smem, stream, q, k, v, cu,
+                               nullptr, nullptr, out, nq, nkv, s_q, s_k, s_v,
+                               0, 0, scale);
         else
-            hipLaunchKernelGGL((attn_prefill_mfma<64, false>), grid, dim3(256),
-                               smem, stream, q, k, v, cu, nullptr, nullptr,
-                               out, nq, nkv, s_q, s_k, s_v, 0, 0, scale);
+            hipLaunchKernelGGL((attn_prefill_mfma<64, false, unsigned short>),
+                               grid, dim3(256), smem, stream, q, k, v, cu,
+                               nullptr, nullptr, out, nq, nkv, s_q, s_k, s_v,
+                               0, 0, scale);
         return;
     }
     dim3 grid(nq, nseq);
     const int smem = (hd + WAVE) * 4;
-    hipLaunchKernelGGL(attn_prefill_basic<false>, grid, dim3(WAVE), smem,
-                       stream, q, k, v, cu, nullptr, nullptr, out, nq, nkv,
-                       hd, s_q, s_k, s_v, 0, 0, scale, causal);
+    hipLaunchKernelGGL((attn_prefill_basic<false, unsigned short>), grid,
+                       dim3(WAVE), smem, stream, q, k, v, cu, nullptr,
+                       nullptr, out, nq, nkv, hd, s_q, s_k, s_v, 0, 0, scale,
+                       causal);
 }
 
 // Chunked prefill: query chunks (packed varlen, cu) attend to the full
 // paged history; the chunk's K/V are already in the cache (kv_store ran
 // first), so K/V come exclusively from the paged pool.
 extern "C" void launch_attn_prefill_paged(
-    const unsigned short* q, const unsigned short* k_cache,
-    const unsigned short* v_cache, const int* cu, const int* block_table,
-    const int* seq_lens, unsigned short* out, int nseq, int nq, int nkv,
-    int hd, long s_q, int W, int bs, int max_qlen, float scale,
-    hipStream_t stream) {
-    if (hd == 64 || hd == 128) {
-        const int tiles = (max_qlen + PF_TM - 1) / PF_TM;
-        dim3 grid(tiles, nq, nseq);
-        const int smem = prefill_mfma_smem(hd);
-        if (hd == 128)
-            hipLaunchKernelGGL((attn_prefill_mfma<128, true>), grid,
-                               dim3(256), smem, stream, q, k_cache, v_cache,
-                               cu, block_table, seq_lens, out, nq, nkv, s_q,
-                               0, 0, W, bs, scale);
-        else
-            hipLaunchKernelGGL((attn_prefill_mfma<64, true>), grid, dim3(256),
-                               smem, stream, q, k_cache, v_cache, cu,
-                               block_table, seq_lens, out, nq, nkv, s_q, 0, 0,
-                               W, bs, scale);
-        return;
-    }
-    dim3 grid(nq, nseq);
-    const int smem = (hd + WAVE) * 4;
-    hipLaunchKernelGGL(attn_prefill_basic<true>, grid, dim3(WAVE), smem,
-                       stream, q, k_cache, v_cache, cu, block_table, seq_lens,
-                       out, nq, nkv, hd, s_q, 0, 0, W, bs, scale, 1);
+    const unsigned short* q, const void* k_cache, const void* v_cache,
+    const int* cu, const int* block_table, const int* seq_lens,
+    unsigned short* out, int nseq, int nq, int nkv, int hd, long s_q, int W,
+    int bs, int max_qlen, float scale, int fp8, hipStream_t stream) {
+#define PFP_LAUNCH(KVT)                                                        \
+    do {                                                                       \
+        const KVT* kc = reinterpret_cast<const KVT*>(k_cache);                 \
+        const KVT* vc = reinterpret_cast<const KVT*>(v_cache);                 \
+        if (hd == 64 || hd == 128) {                                           \
+            const int tiles = (max_qlen + PF_TM - 1) / PF_TM;                  \
+            dim3 grid(tiles, nq, nseq);                                        \
+            const int smem = prefill_mfma_smem(hd);                            \
+            if (hd == 128)                                                     \
+                hipLaunchKernelGGL((attn_prefill_mfma<128, true, KVT>), grid,  \
+                                   dim3(256), smem, stream, q, kc, vc, cu,     \
+                                   block_table, seq_lens, out, nq, nkv, s_q,   \
+                                   0, 0, W, bs, scale);                        \
+            else                                                               \
+                hipLaunchKernelGGL((attn_prefill_mfma<64, true, KVT>), grid,   \
+                                   dim3(256), smem, stream, q, kc, vc, cu,     \
+                                   block_table, seq_lens, out, nq, nkv, s_q,   \
+                                   0, 0, W, bs, scale);                        \
+            return;                                                            \
+        }                                                                      \
+        dim3 grid(nq, nseq);                                                   \
+        const int smem = (hd + WAVE) * 4;                                      \
+        hipLaunchKernelGGL((attn_prefill_basic<true, KVT>), grid, dim3(WAVE),  \
+                           smem, stream, q, kc, vc, cu, block_table, seq_lens, \
+                           out, nq, nkv, hd, s_q, 0, 0, W, bs, scale, 1);      \
+    } while (0)
+    if (fp8) PFP_LAUNCH(unsigned char);
+    else PFP_LAUNCH(unsigned short);
+#undef PFP_LAUNCH
 }

@@ -33,10 +33,11 @@ void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          const unsigned short*, const int*, unsigned short*,
                          int, int, int, int, long, long, long, int, float,
                          int, hipStream_t);
-void launch_attn_prefill_paged(const unsigned short*, const unsigned short*,
-                               const unsigned short*, const int*, const int*,
+void launch_attn_prefill_paged(const unsigned short*, const void*,
+                               const void*, const int*, const int*,
                                const int*, unsigned short*, int, int, int,
-                               int, long, int, int, int, float, hipStream_t);
+                               int, long, int, int, int, float, int,
+                               hipStream_t);
 void launch_mfma_probe(const unsigned short*, const unsigned short*, float*,
                        hipStream_t);
 void launch_grouped_gemm(const unsigned short*, const unsigned short*,
@@ -223,7 +224,8 @@ Tensor attn_prefill_paged(const Tensor& q, const Tensor& k_cache,
                           const Tensor& seq_lens, const Tensor& cu_q,
                           long max_qlen, double scale) {
     check_bf16(q, "q");
-    check_bf16(k_cache, "k_cache");
+    const bool fp8 = k_cache.scalar_type() == torch::kUInt8;
+    if (!fp8) check_bf16(k_cache, "k_cache");
     TORCH_CHECK(block_table.scalar_type() == torch::kInt32 &&
                 seq_lens.scalar_type() == torch::kInt32 &&
                 cu_q.scalar_type() == torch::kInt32);
@@ -238,12 +240,12 @@ Tensor attn_prefill_paged(const Tensor& q, const Tensor& k_cache,
     TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
     TORCH_CHECK(block_table.size(0) == nseq && seq_lens.size(0) == nseq);
     Tensor out = torch::empty({T, nq, hd}, q.options());
-    launch_attn_prefill_paged(bf16p(q), bf16p(k_cache), bf16p(v_cache),
-                              cu_q.data_ptr<int>(),
+    launch_attn_prefill_paged(bf16p(q), k_cache.data_ptr(),
+                              v_cache.data_ptr(), cu_q.data_ptr<int>(),
                               block_table.data_ptr<int>(),
                               seq_lens.data_ptr<int>(), bf16p_mut(out), nseq,
                               nq, nkv, hd, q.stride(0), W, bs, (int)max_qlen,
-                              (float)scale, stream());
+                              (float)scale, fp8 ? 1 : 0, stream());
     return out;
 }
 

@@ -327,3 +327,38 @@ def test_fp8_kv_attn_decode(B, nkv, G, hd, maxlen):
     got = ops.attn_decode(q, kc, vc, bt, lens_dev, hd**-0.5)
     ref = R.attn_decode(q, kc, vc, bt, lens_dev, hd**-0.5)
     assert_close(got, ref, atol=3e-2, rtol=3e-2, msg=f"fp8 decode G{G} hd{hd}")
+
+
+def test_fp8_kv_attn_prefill_paged():
+    """Chunked prefill reading an fp8 cache: HIP vs reference on the SAME
+    store-quantized cache (enables chunked prefill under fp8 KV serving)."""
+    torch.manual_seed(15)
+    nq, nkv, hd, bs = 32, 8, 128, 32
+    hist_q = [(200, 70), (0, 64)]
+    B = len(hist_q)
+    seq_lens = [h + ql for h, ql in hist_q]
+    W = (max(seq_lens) + bs - 1) // bs
+    nb = B * W + 1
+    perm = torch.randperm(nb - 1) + 1
+    bt = perm[: B * W].reshape(B, W).to(torch.int32).to(DEV)
+    kc = torch.zeros(nb, nkv, bs, hd, dtype=torch.uint8, device=DEV)
+    vc = torch.zeros_like(kc)
+    kf = (torch.randn(nb * bs, nkv, hd, device=DEV) * 0.6).bfloat16()
+    vf = (torch.randn(nb * bs, nkv, hd, device=DEV) * 0.6).bfloat16()
+    all_slots = torch.arange(nb * bs, dtype=torch.int32, device=DEV)
+    ops.kv_cache_store(kf, vf, kc, vc, all_slots)
+    T = sum(ql for _, ql in hist_q)
+    q = (torch.randn(T, nq, hd, device=DEV) * 0.4).bfloat16()
+    cu_list = [0]
+    for _, ql in hist_q:
+        cu_list.append(cu_list[-1] + ql)
+    cu = torch.tensor(cu_list, dtype=torch.int32, device=DEV)
+    lens_dev = torch.tensor(seq_lens, dtype=torch.int32, device=DEV)
+    qlens = torch.tensor([ql for _, ql in hist_q], dtype=torch.int32,
+                         device=DEV)
+    scale = hd**-0.5
+    got = ops.attn_prefill_paged(
+        q, kc, vc, bt, lens_dev, cu, max(ql for _, ql in hist_q), scale
+    )
+    ref = R.attn_decode_with_history(q, kc, vc, bt, lens_dev, qlens, scale)
+    assert_close(got, ref, atol=3e-2, rtol=3e-2, msg="fp8 prefill_paged")
