@@ -59,6 +59,10 @@ class NativeEngineService(BaseService):
                 # speculative decoding (prompt-lookup + exact verification):
                 # serving opt-in; greedy outputs are provably unchanged
                 spec_decode=os.environ.get("BEE2BEE_SPEC_DECODE") == "1",
+                # fp8 (OCP e4m3) KV pool: halves decode-attention bytes at a
+                # small quantization cost; serving opt-in
+                kv_dtype="fp8" if os.environ.get("BEE2BEE_KV_FP8") == "1"
+                else "native",
             )
             self.engine.start()
         except Exception as e:
