@@ -42,6 +42,10 @@ class Runner:
         # optional expert parallelism (parallel/ep.py); when set, MoE layers
         # dispatch tokens over RCCL all-to-all to the expert owners
         self.ep = None
+        # optional tensor parallelism (parallel/tp.py): weights are a
+        # head/intermediate shard and the two row-parallel projections
+        # all-reduce their partial outputs over RCCL
+        self.tp_group = None
         cos, sin = ops.rope_tables(
             spec.max_seq_len, spec.head_dim, spec.rope_theta, device
         )
@@ -83,7 +87,7 @@ class Runner:
                 hidden, residual, lw.attn_norm, s.rms_eps
             )
 
-        qkv = F.linear(normed, lw.wqkv)
+        qkv = F.linear(normed, lw.wqkv)  # column-parallel under TP
         T = qkv.shape[0]
         q, k, v = qkv.split([s.q_size, s.kv_size, s.kv_size], dim=-1)
         q = q.view(T, s.n_heads, s.head_dim)
@@ -94,6 +98,11 @@ class Runner:
         ops.kv_cache_store(k, v, k_cache, v_cache, slot_mapping)
         attn_out = attn_fn(layer_idx, q, k, v, k_cache, v_cache)
         attn_out = F.linear(attn_out.reshape(T, s.q_size), lw.wo)
+        if self.tp_group is not None:
+            # row-parallel o-projection: sum the per-shard partials
+            import torch.distributed as dist
+
+            dist.all_reduce(attn_out, group=self.tp_group)
 
         normed, residual = ops.fused_add_rmsnorm(
             attn_out, residual, lw.mlp_norm, s.rms_eps
@@ -103,6 +112,11 @@ class Runner:
         else:
             gate_up = F.linear(normed, lw.w_gate_up)
             mlp_out = F.linear(ops.swiglu(gate_up), lw.w_down)
+            if self.tp_group is not None:
+                # row-parallel down-projection
+                import torch.distributed as dist
+
+                dist.all_reduce(mlp_out, group=self.tp_group)
         return mlp_out, residual
 
     # below this many tokens the MoE layer runs all experts densely: at
