@@ -1,5 +1,9 @@
-"""Pipeline-parallel benchmark (BASELINE config 4): Llama-3-70B layer-
-sharded across N GPUs over RCCL/xGMI point-to-point hops.
+"""Model-parallel benchmark (BASELINE config 4): Llama-3-70B sharded
+across N GPUs over RCCL/xGMI.
+
+  --mode pp  (default): layer shards, point-to-point hidden-state hops
+  --mode tp:            Megatron head/intermediate shards, two all-reduces
+                        per layer (lower latency per token at small batch)
 
 Launch (one rank per GPU):
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
@@ -27,6 +31,7 @@ def main() -> None:
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--steps", type=int, default=16)
     ap.add_argument("--warmup", type=int, default=4)
+    ap.add_argument("--mode", default="pp", choices=["pp", "tp"])
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -40,9 +45,12 @@ def main() -> None:
     if on_gpu:
         torch.cuda.set_device(local_rank)
 
-    from bee2bee_amd.parallel.pp import PipelineEngine
+    if args.mode == "tp":
+        from bee2bee_amd.parallel.tp import TPEngine as Engine
+    else:
+        from bee2bee_amd.parallel.pp import PipelineEngine as Engine
 
-    eng = PipelineEngine(
+    eng = Engine(
         args.model,
         device=f"cuda:{local_rank}" if on_gpu else "cpu",
         max_batch=args.batch,
@@ -76,7 +84,8 @@ def main() -> None:
     if rank == 0:
         print(json.dumps({
             "metric": f"output tokens/sec ({eng.spec.name} bf16 greedy decode, "
-                      f"pipeline-parallel pp{world})",
+                      f"{'tensor' if args.mode == 'tp' else 'pipeline'}"
+                      f"-parallel {args.mode}{world})",
             "value": round(args.batch * args.steps / elapsed, 1),
             "unit": "tokens/s",
             "n_gpus": world,
@@ -92,7 +101,7 @@ def main() -> None:
                 "model": eng.spec.name,
                 "global_batch": args.batch,
                 "seq_len": args.prompt_len,
-                "parallelism": f"pp{world}",
+                "parallelism": f"{args.mode}{world}",
             },
         }), flush=True)
     dist.destroy_process_group()
