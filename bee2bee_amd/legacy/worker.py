@@ -92,6 +92,93 @@ class LegacyWorker:
             if eng is not None:
                 eng.shutdown()
             return {"unloaded": payload["model"]}
+        if kind == protocol.TASK_ONNX_LOAD:
+            # gated on onnxruntime like the reference (bee2bee/node.py:210-220)
+            try:
+                import onnxruntime as ort  # noqa: F401
+            except Exception:
+                raise RuntimeError("onnx_support_missing")
+            model_id = payload.get("model_id") or new_id("onnx")
+            sess = ort.InferenceSession(payload["path"])
+            self._engines[model_id] = {"type": "onnx", "session": sess}
+            return {"model_id": model_id}
+        if kind == protocol.TASK_ONNX_INFER:
+            m = self._engines.get(payload.get("model_id"))
+            if not isinstance(m, dict) or m.get("type") != "onnx":
+                raise RuntimeError("onnx_model_not_loaded")
+            inputs = payload.get("inputs") or {}
+            out = m["session"].run(
+                None, {k: np.array(v) for k, v in inputs.items()}
+            )
+            return {"outputs": [o.tolist() if hasattr(o, "tolist") else o
+                                for o in out]}
+        if kind == protocol.TASK_ONNX_UNLOAD:
+            self._engines.pop(payload.get("model_id"), None)
+            return {"ok": True}
+        if kind == protocol.TASK_HF_PART_LOAD:
+            # layer-range partial of OUR transformer (the reference's
+            # DistilBERT partial, bee2bee/node.py:236-249, rebuilt on the
+            # native stack: same math as a pipeline stage)
+            import torch
+
+            from ..engine.kv import PagedKV
+            from ..engine.runner import Runner
+            from ..models.spec import resolve_spec
+            from ..models.tokenizer import load_tokenizer
+            from ..models.weights import ModelWeights
+
+            name = payload.get("model_name", "tiny")
+            start = int(payload.get("start", 0))
+            end = int(payload.get("end", 2))
+            model_id = payload.get("model_id") or new_id("hfpart")
+            spec = resolve_spec(name, payload.get("model_path"))
+            end = min(end, spec.n_layers)
+            dev = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+            dtype = torch.bfloat16 if dev.type == "cuda" else torch.float32
+            w = ModelWeights(spec, dev, dtype)
+            if payload.get("model_path"):
+                w.load_hf(payload["model_path"], layer_range=(start, end))
+            else:
+                w.random_init(seed=int(payload.get("seed", 0)),
+                              layer_range=(start, end))
+            kv = PagedKV(spec, dev, dtype, n_blocks=32,
+                         layer_range=(start, end))
+            runner = Runner(spec, w, kv, dev, dtype,
+                            layer_range=(start, end))
+            tok = load_tokenizer(payload.get("model_path"), spec.vocab_size,
+                                 spec.bos_token_id, spec.eos_token_id)
+            self._engines[model_id] = {
+                "type": "hf_part", "runner": runner, "kv": kv, "tok": tok,
+                "start": start, "end": end, "seq": 0,
+            }
+            return {"model_id": model_id, "start": start, "end": end}
+        if kind == protocol.TASK_HF_PART_FORWARD:
+            import torch
+
+            m = self._engines.get(payload.get("model_id"))
+            if not isinstance(m, dict) or m.get("type") != "hf_part":
+                raise RuntimeError("model_not_loaded")
+            runner, kv = m["runner"], m["kv"]
+            dev = runner.device
+            if payload.get("text") is not None:
+                ids = m["tok"].encode(payload["text"])
+                inp = torch.tensor(ids, dtype=torch.int64, device=dev)
+            else:
+                hs = np.array(payload["hidden"], dtype=np.float32)
+                inp = torch.from_numpy(hs).to(dev, runner.dtype)
+            T = inp.shape[0]
+            sid = m["seq"]
+            m["seq"] += 1
+            kv.new_seq(sid)
+            kv.extend_seq(sid, T)
+            slots = torch.tensor(kv.slot_mapping(sid, range(T)),
+                                 dtype=torch.int32, device=dev)
+            pos = torch.arange(T, dtype=torch.int32, device=dev)
+            cu = torch.tensor([0, T], dtype=torch.int32, device=dev)
+            with torch.no_grad():
+                hid = runner.forward_prefill(inp, pos, slots, cu, T)
+            kv.free_seq(sid)
+            return {"hidden": hid.float().cpu().numpy().tolist()}
         raise ValueError(f"unsupported task kind: {kind}")
 
     # ------------------------------------------------------------ transport

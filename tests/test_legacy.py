@@ -96,3 +96,72 @@ def test_worker_unknown_kind():
     w = LegacyWorker()
     with pytest.raises(ValueError):
         w.execute_task("bogus", {})
+
+
+def test_hf_part_load_forward_chain():
+    """Two layer-range partials chained (text -> stage0 hidden -> stage1
+    hidden) must equal the full-stack forward — the reference's DistilBERT
+    partial workflow (bee2bee/node.py:236-277) on our native stack."""
+    import numpy as np
+    import torch
+
+    from bee2bee_amd.engine.kv import PagedKV
+    from bee2bee_amd.engine.runner import Runner
+    from bee2bee_amd.legacy.worker import LegacyWorker
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.tokenizer import load_tokenizer
+    from bee2bee_amd.models.weights import ModelWeights
+
+    w = LegacyWorker()
+    r0 = w.execute_task("hf_part_load", {
+        "model_name": "tiny", "start": 0, "end": 1, "seed": 5,
+    })
+    r1 = w.execute_task("hf_part_load", {
+        "model_name": "tiny", "start": 1, "end": 2, "seed": 5,
+    })
+    text = "partial forward"
+    h0 = w.execute_task("hf_part_forward", {
+        "model_id": r0["model_id"], "text": text,
+    })["hidden"]
+    h1 = w.execute_task("hf_part_forward", {
+        "model_id": r1["model_id"], "hidden": h0,
+    })["hidden"]
+
+    spec = PRESETS["tiny"]
+    dev = torch.device("cpu")
+    full = ModelWeights(spec, dev, torch.float32).random_init(5)
+    kv = PagedKV(spec, dev, torch.float32, n_blocks=16)
+    runner = Runner(spec, full, kv, dev, torch.float32)
+    tok = load_tokenizer(None, spec.vocab_size, spec.bos_token_id,
+                         spec.eos_token_id)
+    ids = torch.tensor(tok.encode(text), dtype=torch.int64)
+    T = ids.shape[0]
+    kv.new_seq(0)
+    kv.extend_seq(0, T)
+    slots = torch.tensor(kv.slot_mapping(0, range(T)), dtype=torch.int32)
+    pos = torch.arange(T, dtype=torch.int32)
+    cu = torch.tensor([0, T], dtype=torch.int32)
+    ref = runner.forward_prefill(ids, pos, slots, cu, T)
+    got = np.array(h1, dtype=np.float32)
+    assert np.allclose(got, ref.numpy(), atol=1e-4), np.abs(got - ref.numpy()).max()
+
+
+def test_onnx_tasks_gated():
+    """onnx_* tasks behave like the reference: a clean error when
+    onnxruntime is absent, full function when present."""
+    import pytest as _pytest
+
+    from bee2bee_amd.legacy.worker import LegacyWorker
+
+    w = LegacyWorker()
+    try:
+        import onnxruntime  # noqa: F401
+
+        has_ort = True
+    except Exception:
+        has_ort = False
+    if not has_ort:
+        with _pytest.raises(RuntimeError, match="onnx_support_missing"):
+            w.execute_task("onnx_load", {"path": "/nonexistent.onnx"})
+    # unload of an unknown model is a no-op either way
+    assert w.execute_task("onnx_unload", {"model_id": "nope"}) == {"ok": True}
