@@ -71,6 +71,7 @@ class GenerationRequest:
     # filled by the engine
     out_queue: "queue.Queue" = field(default_factory=queue.Queue)
     submit_ts: float = field(default_factory=time.time)
+    admitted_ts: Optional[float] = None   # left the queue, prefill began
     first_token_ts: Optional[float] = None
     done_ts: Optional[float] = None
     error: Optional[str] = None
@@ -322,11 +323,21 @@ class InferenceEngine:
                 text = text[:cut]
         if on_text is not None and decoded_upto < len(text):
             on_text(text[decoded_upto:])
+        now = time.time()
+        ft = req.first_token_ts or now
+        adm = req.admitted_ts or req.submit_ts
         return {
             "text": text,
             "tokens": len(req.output_ids),
-            "latency_ms": int((time.time() - t0) * 1000),
-            "ttft_ms": int(((req.first_token_ts or time.time()) - t0) * 1000),
+            "latency_ms": int((now - t0) * 1000),
+            "ttft_ms": int((ft - t0) * 1000),
+            # per-stage request trace (the reference reports one opaque
+            # latency_ms; SURVEY §5 calls for stage timestamps)
+            "timing": {
+                "queue_ms": int((adm - req.submit_ts) * 1000),
+                "prefill_ms": int((ft - adm) * 1000),
+                "decode_ms": int(((req.done_ts or now) - ft) * 1000),
+            },
         }
 
     # ---------------------------------------------------------- engine loop
@@ -414,10 +425,13 @@ class InferenceEngine:
         budget = self.max_prefill_tokens
         batch: List[tuple] = []  # (active, start, take)
         n_consumed = 0
+        now_admit = time.time()
         for a in queue_:
             if budget <= 0:
                 break
             if a.prefilled == 0:
+                if a.req.admitted_ts is None:
+                    a.req.admitted_ts = now_admit
                 p = a.req.prompt_ids or [self.spec.bos_token_id]
                 p = p[: self.max_seq_len - a.req.max_new_tokens - 1] or p[:1]
                 a.req.prompt_ids = p

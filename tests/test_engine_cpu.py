@@ -258,3 +258,11 @@ def test_active_set_never_exceeds_max_batch():
             assert len(r.output_ids) == 6
     finally:
         eng.shutdown()
+
+
+def test_request_stage_timing(engine):
+    res = engine.generate_text("trace me", max_new_tokens=4, temperature=0.0)
+    t = res["timing"]
+    assert set(t) == {"queue_ms", "prefill_ms", "decode_ms"}
+    assert all(v >= 0 for v in t.values())
+    assert t["queue_ms"] + t["prefill_ms"] + t["decode_ms"] <= res["latency_ms"] + 50
