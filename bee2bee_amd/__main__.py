@@ -56,8 +56,21 @@ def serve_hf(model, model_path, port, region, api_port, device):
 @click.option("--region", default="Auto")
 @click.option("--api-port", default=8000, type=int)
 @click.option("--device", default=None)
-def serve_native(model, model_path, port, region, api_port, device):
+@click.option("--spec-decode", is_flag=True,
+              help="speculative decoding (prompt-lookup + exact greedy "
+                   "verification; output-invariant)")
+@click.option("--kv-fp8", is_flag=True,
+              help="fp8 (OCP e4m3) KV cache: half the decode-attention "
+                   "bytes at a small quantization cost")
+def serve_native(model, model_path, port, region, api_port, device,
+                 spec_decode, kv_fp8):
     """Serve a model on the native MI355X HIP engine (alias of serve-hf)."""
+    import os
+
+    if spec_decode:
+        os.environ["BEE2BEE_SPEC_DECODE"] = "1"
+    if kv_fp8:
+        os.environ["BEE2BEE_KV_FP8"] = "1"
     _serve(
         "native",
         model_name=model,
