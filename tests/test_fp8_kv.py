@@ -79,3 +79,24 @@ def test_fp8_kv_engine_runs_cpu():
 def test_fp8_rejects_spec_decode():
     with pytest.raises(ValueError):
         InferenceEngine("tiny", device="cpu", kv_dtype="fp8", spec_decode=True)
+
+
+def test_fp8_kv_chunked_prefill_cpu():
+    """fp8 KV + chunked prefill together (the serving combination): the
+    engine must complete through the paged-history reference path with a
+    quantized cache."""
+    prompt = [(i * 7 + 3) % 500 for i in range(50)]
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=128,
+                          seed=7, kv_dtype="fp8", max_prefill_tokens=16)
+    try:
+        r = GenerationRequest(prompt_ids=list(prompt), max_new_tokens=6,
+                              sampling=SamplingParams(greedy=True))
+        eng.submit(r)
+        while True:
+            x = r.out_queue.get(timeout=60)
+            if not isinstance(x, int):
+                break
+        assert r.error is None, r.error
+        assert len(r.output_ids) == 6
+    finally:
+        eng.shutdown()
