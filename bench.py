@@ -60,6 +60,21 @@ def main() -> None:
             )
             os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
             os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+            # multi-GPU runs: TunableOp keys files by device ordinal, but
+            # the tuned algorithms are shape-keyed and identical across the
+            # node's MI355Xs — seed ranks 1..7 from rank 0's table so the
+            # scale run's warmup replays instead of re-tuning per GPU
+            base = os.path.join(tune_dir, "tunableop_0.csv")
+            if os.path.exists(base):
+                import shutil
+
+                for d in range(1, 8):
+                    dst = os.path.join(tune_dir, f"tunableop_{d}.csv")
+                    if not os.path.exists(dst):
+                        try:
+                            shutil.copyfile(base, dst)
+                        except OSError:
+                            pass
         except OSError:
             pass  # read-only checkout: run untuned
 
