@@ -803,8 +803,15 @@ class InferenceEngine:
     # ------------------------------------------------ direct batch API (bench)
 
     @torch.no_grad()
-    def bench_prefill(self, batch: int, prompt_len: int, seed: int = 1234) -> List[int]:
-        """Fill the KV cache with synthetic prompts; returns seq_ids."""
+    def bench_prefill(self, batch: int, prompt_len: int, seed: int = 1234,
+                      max_new: Optional[int] = None) -> List[int]:
+        """Fill the KV cache with synthetic prompts; returns seq_ids.
+        max_new must leave room for the prompt under max_seq_len or the
+        admission truncation rule will cut the prompts short (a 10**9
+        sentinel used to truncate every bench prompt to one token)."""
+        if max_new is None:
+            max_new = max(1, self.max_seq_len - prompt_len - 1)
+        assert prompt_len + max_new < self.max_seq_len + 1, "prompt would truncate"
         gen = torch.Generator().manual_seed(seed)
         seq_ids = []
         for b in range(batch):
@@ -818,21 +825,31 @@ class InferenceEngine:
         acts = []
         for b, seq_id in enumerate(seq_ids):
             req = GenerationRequest(
-                prompt_ids=ids[b].tolist(), max_new_tokens=10**9,
+                prompt_ids=ids[b].tolist(), max_new_tokens=max_new,
                 sampling=SamplingParams(greedy=True),
             )
             acts.append(_Active(req, seq_id, 0))
         self._prefill_no_admit(acts)
         self._bench_acts = acts
+        for a in acts:
+            assert a.prefilled == prompt_len, "bench prompt truncated"
         return seq_ids
 
     @torch.no_grad()
     def _prefill_no_admit(self, acts: List[_Active]) -> None:
-        # identical to _prefill but without touching the serving active set
-        saved = self._active
+        """_prefill for the bench path, without touching the serving active
+        set — looping until EVERY prompt is fully in the cache (_prefill
+        consumes at most max_prefill_tokens per call, which also bounds the
+        transient activation memory to ~max_prefill_tokens rows)."""
+        saved_active = self._active
+        saved_prefilling = self._prefilling
         self._active = []
-        self._prefill(acts)
-        self._active = saved
+        self._prefilling = []
+        self._prefill(list(acts))
+        while self._prefilling:
+            self._prefill(list(self._prefilling))
+        self._active = saved_active
+        self._prefilling = saved_prefilling
 
     @torch.no_grad()
     def bench_setup(self, batch: int, prompt_len: int, steps_budget: int,

@@ -8,7 +8,7 @@ untimed warmup steps, times exactly K steps bracketed by barrier +
 torch.cuda.synchronize() on both sides, takes the MAX elapsed over ranks,
 and rank 0 prints ONE JSON line.
 
-A step = one decode iteration of a fixed per-GPU batch (default 2048 seqs at
+A step = one decode iteration of a fixed per-GPU batch (default 1536 seqs at
 prompt length 1024): full layer stack on the HIP kernels via hipGraph
 replay + greedy sampling + paged-KV append. Weights are random-init of the
 real architecture; prompts synthetic (no network for checkpoints).
@@ -32,7 +32,7 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--batch", type=int, default=2048)
+    ap.add_argument("--batch", type=int, default=1536)
     ap.add_argument("--kv-dtype", default="native", choices=["native", "fp8"],
                     help="fp8 = OCP e4m3 KV cache (reported separately from "
                          "the bf16 headline; never the default)")

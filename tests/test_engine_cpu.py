@@ -266,3 +266,20 @@ def test_request_stage_timing(engine):
     assert set(t) == {"queue_ms", "prefill_ms", "decode_ms"}
     assert all(v >= 0 for v in t.values())
     assert t["queue_ms"] + t["prefill_ms"] + t["decode_ms"] <= res["latency_ms"] + 50
+
+
+def test_bench_prefill_fills_every_prompt():
+    """bench_setup must put EVERY prompt token in the cache even when the
+    batch exceeds one prefill budget (regression: chunked prefill left all
+    but the first max_prefill_tokens prompts empty)."""
+    eng = InferenceEngine("tiny", device="cpu", max_batch=8, max_seq_len=64,
+                          seed=7, max_prefill_tokens=32)
+    try:
+        eng.bench_setup(8, 16, steps_budget=4)
+        for a in eng._bench_acts:
+            assert a.prefilled == 16
+            assert len(a.req.output_ids) == 1  # first sampled token emitted
+            assert eng.kv.seq_len(a.seq_id) >= 16
+        assert eng._prefilling == []
+    finally:
+        eng.shutdown()
