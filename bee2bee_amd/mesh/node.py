@@ -175,6 +175,12 @@ class MeshNode:
     async def stop(self) -> None:
         self._running = False
         self._monitor_active = False
+        # fail in-flight requests immediately instead of letting callers
+        # wait out the 300 s request timeout
+        for fut in list(self._pending.values()):
+            if not fut.done():
+                fut.set_exception(RuntimeError("node stopped"))
+        self._pending.clear()
         for t in self._tasks:
             t.cancel()
         async with self._lock:
@@ -243,6 +249,9 @@ class MeshNode:
         async with self._lock:
             self.peers[pid] = Peer(pid, ws, addr)
         await self._send(ws, self._make_hello())
+        # prune finished reader tasks so long-lived nodes with many
+        # reconnects don't accumulate task objects forever
+        self._tasks = [t for t in self._tasks if not t.done()]
         self._tasks.append(asyncio.create_task(self._reader(ws)))
 
     async def _reader(self, ws: Any) -> None:
