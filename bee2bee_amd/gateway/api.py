@@ -228,6 +228,7 @@ async def chat(req: ChatRequest):
                     "service": svc_name,
                     "latency_ms": result.get("latency_ms"),
                     "tokens": result.get("tokens"),
+                    "timing": result.get("timing"),
                 },
             }
 

@@ -107,6 +107,7 @@ class NativeEngineService(BaseService):
                 "tokens": tokens,
                 "latency_ms": latency_ms,
                 "ttft_ms": res.get("ttft_ms"),
+                "timing": res.get("timing"),  # queue/prefill/decode stages
                 "price_per_token": self.price_per_token,
                 "cost": self.price_per_token * tokens,
                 "backend": "bee2bee-amd-native",
