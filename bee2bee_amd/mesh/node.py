@@ -95,6 +95,11 @@ class MeshNode:
 
         # state
         self.peers: Dict[str, Peer] = {}
+        # mesh-replicated KV (rendezvous records etc.): local DHT store whose
+        # writes broadcast one hop over the control plane (wire.DHT_SET)
+        from .dht import DHTNode
+
+        self.dht = DHTNode()
         self.local_services: Dict[str, Any] = {}  # name -> BaseService
         self.providers: Dict[str, Dict[str, Any]] = {}  # pid -> {svc: meta, _latency}
         self.pieces: Dict[str, Dict[str, Any]] = {}
@@ -132,6 +137,7 @@ class MeshNode:
             self.port = server.sockets[0].getsockname()[1]
 
         self._session = aiohttp.ClientSession()
+        await self.dht.start()
 
         # announce address resolution: explicit > NAT/STUN discovery > LAN IP
         display_host = self.announce_host
@@ -329,6 +335,7 @@ class MeshNode:
             wire.GEN_CHUNK: self._handle_gen_chunk,
             wire.PIECE_REQUEST: self._handle_piece_request,
             wire.PIECE_DATA: self._handle_piece_data,
+            wire.DHT_SET: self._handle_dht_set,
         }.get(mtype)
         if handler is None:
             logger.warning("unknown message type: %s", mtype)
@@ -520,6 +527,31 @@ class MeshNode:
                 cb(data.get("text", ""))
             except Exception:
                 logger.exception("chunk callback failed")
+
+    async def _handle_dht_set(self, ws: Any, data: Dict[str, Any]) -> None:
+        key = data.get("key")
+        if not isinstance(key, str):
+            return
+        value = data.get("value")
+        # dict values merge (rendezvous maps peer-id -> record: two peers
+        # announcing concurrently must not clobber each other's entries)
+        if isinstance(value, dict):
+            cur = await self.dht.get(key)
+            if isinstance(cur, dict):
+                value = {**cur, **value}
+        await self.dht.set(key, value)
+
+    async def dht_set(self, key: str, value: Any) -> None:
+        """Store locally AND replicate one hop to every connected peer —
+        enough for same-mesh RCCL rendezvous (parallel/rendezvous.py) without
+        the optional kademlia dependency. With kademlia present the DHT
+        itself replicates and the broadcast is a harmless no-op overlay."""
+        await self.dht.set(key, value)
+        async with self._lock:
+            targets = [p.ws for p in self.peers.values()]
+        for ws in targets:
+            await self._send(ws, {"type": wire.DHT_SET, "key": key,
+                                  "value": value})
 
     async def _handle_piece_request(self, ws: Any, data: Dict[str, Any]) -> None:
         """Serve a locally-held content piece (reference left this a stub,

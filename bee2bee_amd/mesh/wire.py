@@ -35,6 +35,10 @@ GEN_ERROR = "gen_error"
 GEN_RESULT = "gen_result"
 PIECE_REQUEST = "piece_request"
 PIECE_DATA = "piece_data"
+# extension beyond the reference wire set: one-hop DHT record replication so
+# mesh-connected peers can rendezvous into RCCL groups without kademlia
+# (parallel/rendezvous.py); unknown types are ignored by reference peers
+DHT_SET = "dht_set"
 
 # terminal message types for a pending request id (Q1 fix: all three)
 TERMINAL_TYPES = (GEN_RESULT, GEN_SUCCESS, GEN_ERROR)

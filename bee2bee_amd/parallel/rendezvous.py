@@ -63,6 +63,43 @@ async def join_group(
         await asyncio.sleep(poll_s)
 
 
+async def join_group_mesh(
+    node: Any,
+    group: str,
+    peer_id: str,
+    host: str,
+    port: int,
+    world_size: int,
+    gpu: int = 0,
+    poll_s: float = 0.2,
+    timeout_s: float = 120.0,
+) -> Tuple[int, str]:
+    """join_group over a MeshNode's replicated DHT: the announce broadcasts
+    one hop over the WS control plane (wire.DHT_SET), so mesh-connected
+    peers rendezvous without the optional kademlia dependency."""
+    import asyncio
+
+    key = f"rccl:{group}"
+    cur = await node.dht.get(key) or {}
+    cur = dict(cur)
+    cur[peer_id] = {"host": host, "port": port, "gpu": gpu}
+    await node.dht_set(key, cur)
+    deadline = asyncio.get_event_loop().time() + timeout_s
+    while True:
+        records = await find_ranks(node.dht, group)
+        if len(records) >= world_size:
+            peers, master = rank_order(records)
+            return peers.index(peer_id), master
+        if asyncio.get_event_loop().time() > deadline:
+            raise TimeoutError(
+                f"rendezvous {group}: {len(records)}/{world_size} peers"
+            )
+        # re-broadcast our record: a peer that connected after our announce
+        # missed the one-hop replication
+        await node.dht_set(key, {peer_id: cur[peer_id]})
+        await asyncio.sleep(poll_s)
+
+
 def init_distributed(
     rank: int,
     world_size: int,

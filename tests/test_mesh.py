@@ -330,3 +330,36 @@ def test_reference_quirk_messages_handled():
         await node.stop()
 
     asyncio.run(run())
+
+
+def test_mesh_dht_rendezvous():
+    """Two loopback-connected nodes rendezvous into a (virtual) RCCL group
+    through the mesh-replicated DHT: both must converge on the same rank
+    order and master endpoint without kademlia."""
+    import asyncio
+
+    from bee2bee_amd.mesh.node import MeshNode
+    from bee2bee_amd.parallel.rendezvous import join_group_mesh
+
+    async def run():
+        a = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        b = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await a.start()
+        await b.start()
+        try:
+            await b.connect_bootstrap(a.addr)
+            await asyncio.sleep(0.3)  # hello handshake settles
+            ra, rb = await asyncio.gather(
+                join_group_mesh(a, "pp-test", a.peer_id, "127.0.0.1", 29900,
+                                world_size=2, timeout_s=10),
+                join_group_mesh(b, "pp-test", b.peer_id, "127.0.0.1", 29901,
+                                world_size=2, timeout_s=10),
+            )
+            (rank_a, master_a), (rank_b, master_b) = ra, rb
+            assert {rank_a, rank_b} == {0, 1}
+            assert master_a == master_b  # both agree on the TCP store
+        finally:
+            await a.stop()
+            await b.stop()
+
+    asyncio.run(run())
