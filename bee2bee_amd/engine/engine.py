@@ -790,6 +790,28 @@ class InferenceEngine:
                     r.out_queue.put(_STREAM_END)
         self._note_throughput(n_emitted, now)
 
+    def stats(self) -> Dict[str, Any]:
+        """Live engine observability (served at the API home endpoint)."""
+        now = time.time()
+        span = (now - self._tok_window[0][0]) if len(self._tok_window) > 1 else 1.0
+        tps = sum(x[1] for x in self._tok_window) / max(span, 1e-3)
+        out = {
+            "model": self.spec.name,
+            "device": str(self.device),
+            "kv_dtype": self.kv_dtype,
+            "active_requests": len(self._active),
+            "prefilling_requests": len(self._prefilling),
+            "queued_requests": self._pending.qsize(),
+            "kv_free_blocks": self.kv.free_blocks,
+            "kv_total_blocks": self.kv.n_blocks,
+            "tokens_total": self.total_tokens,
+            "tokens_per_sec_10s": round(tps, 1),
+            "decode_graphs": self.graphs is not None,
+        }
+        if self.spec_decode:
+            out["spec_decode"] = dict(self.spec_stats)
+        return out
+
     def _note_throughput(self, n: int, now: float) -> None:
         self.total_tokens += n
         self._tok_window.append((now, n))

@@ -124,6 +124,17 @@ def home():
             "pool_size": len(node.peers),
             "status": "active",
         },
+        # live engine state (queue depth, KV pool, tok/s) — real numbers,
+        # unlike the reference's simulated throughput (bee2bee/utils.py:129)
+        "engine": next(
+            (
+                svc.engine.stats()
+                for svc in node.local_services.values()
+                if getattr(svc, "engine", None) is not None
+                and hasattr(svc.engine, "stats")
+            ),
+            None,
+        ),
     }
 
 

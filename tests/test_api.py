@@ -155,3 +155,31 @@ def test_chat_p2p_fallback_to_remote_provider(client):
         assert body["metadata"]["engine"] == "bee2bee-amd-p2p"
     finally:
         asyncio.run_coroutine_threadsafe(provider.stop(), loop).result(timeout=15)
+
+
+def test_home_exposes_engine_stats(client):
+    """With a native service attached, '/' exposes live engine stats
+    (real numbers — the reference fabricated throughput)."""
+    import asyncio
+
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.services.native import NativeEngineService
+
+    node = gateway_api.node
+    svc = NativeEngineService("tiny", max_batch=2, max_seq_len=64,
+                              device="cpu")
+    svc.load_sync()
+    try:
+        loop = node._tasks[0].get_loop()
+        fut = asyncio.run_coroutine_threadsafe(node.add_service(svc), loop)
+        fut.result(timeout=10)
+        r = client.get("/")
+        eng = r.json().get("engine")
+        assert eng is not None
+        assert eng["model"] == "tiny"
+        assert eng["active_requests"] >= 0
+        assert eng["kv_free_blocks"] <= eng["kv_total_blocks"]
+        assert "tokens_per_sec_10s" in eng
+    finally:
+        if svc.engine is not None:
+            svc.engine.shutdown()
