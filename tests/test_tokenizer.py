@@ -1,0 +1,50 @@
+"""Tokenizer stack: byte tokenizer totality + HF tokenizer.json loading
+(reference loads AutoTokenizer; we load tokenizer.json offline via the
+`tokenizers` package)."""
+import os
+
+import pytest
+
+from bee2bee_amd.models.tokenizer import ByteTokenizer, load_tokenizer
+
+
+def test_byte_tokenizer_total_roundtrip():
+    tok = ByteTokenizer(vocab_size=512, bos_id=1, eos_id=2)
+    text = "héllo wörld ✓"
+    ids = tok.encode(text)
+    assert tok.decode(ids) == text
+    # totality: EVERY id decodes to something (no dropped tokens in streams)
+    assert isinstance(tok.decode(list(range(0, 300))), str)
+
+
+def _build_tokenizer_json(path):
+    tokenizers = pytest.importorskip("tokenizers")
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+
+    tk = Tokenizer(models.BPE(unk_token="<unk>"))
+    tk.pre_tokenizer = pre_tokenizers.Whitespace()
+    trainer = trainers.BpeTrainer(
+        vocab_size=200,
+        special_tokens=["<unk>", "<s>", "</s>"],
+    )
+    tk.train_from_iterator(
+        ["the mesh serves tokens", "tokens stream over the mesh"] * 20,
+        trainer,
+    )
+    tk.save(path)
+
+
+def test_hf_tokenizer_json_loaded(tmp_path):
+    tj = str(tmp_path / "tokenizer.json")
+    _build_tokenizer_json(tj)
+    tok = load_tokenizer(str(tmp_path), 512, 1, 2)
+    assert type(tok).__name__ == "HFTokenizer"
+    ids = tok.encode("the mesh serves tokens")
+    assert ids and ids[0] == tok.bos_token_id  # "<s>" resolved
+    text = tok.decode([i for i in ids if i != tok.bos_token_id])
+    assert "mesh" in text
+
+
+def test_missing_tokenizer_json_falls_back(tmp_path):
+    tok = load_tokenizer(str(tmp_path), 512, 1, 2)
+    assert isinstance(tok, ByteTokenizer)
