@@ -52,6 +52,54 @@ __device__ __forceinline__ float2v kv_pair_f32(kv_pair_t<unsigned char> p) {
     return fp8x2_2f(w);
 }
 
+// MFMA fragment plumbing for the scores kernel: bf16 caches feed the bf16
+// MFMA; fp8 caches feed v_mfma_f32_16x16x32_fp8_fp8 DIRECTLY (same lane
+// map, hardware-verified by test_mfma_fp8_fragment_map) — K bytes go
+// straight from the load into the matrix op, no convert VALU. q is
+// quantized to e4m3 once at fragment setup (the fp8 cache mode is already
+// approximate; documented).
+template <typename KVT> struct kfrag_of { using type = bf16x8; };
+template <> struct kfrag_of<unsigned char> { using type = long; };
+template <typename KVT>
+__device__ __forceinline__ typename kfrag_of<KVT>::type load_kfrag(
+    const KVT* p) {
+    return *reinterpret_cast<const typename kfrag_of<KVT>::type*>(p);
+}
+__device__ __forceinline__ bf16x8 qfrag_from_lds_bf16(
+    const unsigned short* p) {
+    return *reinterpret_cast<const bf16x8*>(p);
+}
+__device__ __forceinline__ long qfrag_from_lds_fp8(const unsigned short* p) {
+    unsigned short w[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        w[i] = f2fp8x2(bf2f(p[2 * i]), bf2f(p[2 * i + 1]));
+    long r;
+    unsigned short* rp = reinterpret_cast<unsigned short*>(&r);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) rp[i] = w[i];
+    return r;
+}
+template <typename KVT>
+__device__ __forceinline__ typename kfrag_of<KVT>::type qfrag_from_lds(
+    const unsigned short* p);
+template <>
+__device__ __forceinline__ bf16x8 qfrag_from_lds<unsigned short>(
+    const unsigned short* p) {
+    return qfrag_from_lds_bf16(p);
+}
+template <>
+__device__ __forceinline__ long qfrag_from_lds<unsigned char>(
+    const unsigned short* p) {
+    return qfrag_from_lds_fp8(p);
+}
+__device__ __forceinline__ f32x4 mfma_kq(bf16x8 a, bf16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+__device__ __forceinline__ f32x4 mfma_kq(long a, long b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, c, 0, 0, 0);
+}
+
 template <typename KVT>
 __device__ __forceinline__ float2v load_kv_f32x2(const KVT* p);
 template <>
@@ -251,15 +299,14 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     __syncthreads();
 
     // B-fragments: col = q head (li), k = dim lg*8+e within each kstep
-    bf16x8 bq[KSTEPS];
+    using kfrag_t = typename kfrag_of<KVT>::type;
+    kfrag_t bq[KSTEPS];
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
         if (li < G) {
-            bq[ks] = *reinterpret_cast<const bf16x8*>(
-                q_s + li * HD + ks * 32 + lg * 8);
+            bq[ks] = qfrag_from_lds<KVT>(q_s + li * HD + ks * 32 + lg * 8);
         } else {
-#pragma unroll
-            for (int e = 0; e < 8; ++e) bq[ks][e] = (__bf16)0.f;
+            bq[ks] = kfrag_t{};
         }
     }
 
@@ -274,7 +321,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
     // scores: all TILES*KSTEPS A-fragments issue before any MFMA (probe-
     // style load batching: 16 independent 16 B loads in flight per wave)
     f32x4 acc[TILES];
-    bf16x8 afr[TILES][KSTEPS];
+    kfrag_t afr[TILES][KSTEPS];
 #pragma unroll
     for (int t = 0; t < TILES; ++t) {
         const int key = wave_key0 + t * 16 + li;  // A row = li
@@ -288,15 +335,14 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_mfma_kernel(
         }
 #pragma unroll
         for (int ks = 0; ks < KSTEPS; ++ks)
-            afr[t][ks] = load_kv_bf16x8<KVT>(kr + ks * 32 + lg * 8);
+            afr[t][ks] = load_kfrag<KVT>(kr + ks * 32 + lg * 8);
     }
 #pragma unroll
     for (int t = 0; t < TILES; ++t) {
         f32x4 c{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ks = 0; ks < KSTEPS; ++ks)
-            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[t][ks], bq[ks], c,
-                                                        0, 0, 0);
+            c = mfma_kq(afr[t][ks], bq[ks], c);
         acc[t] = c;
     }
 

@@ -40,6 +40,8 @@ void launch_attn_prefill_paged(const unsigned short*, const void*,
                                hipStream_t);
 void launch_mfma_probe(const unsigned short*, const unsigned short*, float*,
                        hipStream_t);
+void launch_mfma_probe_fp8(const unsigned char*, const unsigned char*,
+                           float*, hipStream_t);
 void launch_grouped_gemm(const unsigned short*, const unsigned short*,
                          const int*, unsigned short*, int, int, int, int,
                          hipStream_t);
@@ -267,6 +269,19 @@ Tensor grouped_gemm(const Tensor& x, const Tensor& w, const Tensor& offs) {
     return out;
 }
 
+Tensor mfma_probe_fp8(const Tensor& A, const Tensor& B) {
+    TORCH_CHECK(A.scalar_type() == torch::kUInt8 &&
+                B.scalar_type() == torch::kUInt8);
+    TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}));
+    TORCH_CHECK(B.sizes() == torch::IntArrayRef({32, 16}));
+    TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+    Tensor C = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+    launch_mfma_probe_fp8(A.data_ptr<unsigned char>(),
+                          B.data_ptr<unsigned char>(), C.data_ptr<float>(),
+                          stream());
+    return C;
+}
+
 Tensor mfma_probe(const Tensor& A, const Tensor& B) {
     check_bf16(A, "A");
     check_bf16(B, "B");
@@ -316,6 +331,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_prefill_paged", &attn_prefill_paged,
           "chunked prefill vs paged history");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+    m.def("mfma_probe_fp8", &mfma_probe_fp8,
+          "16x16x32 fp8 MFMA layout probe");
     m.def("grouped_gemm", &grouped_gemm, "per-expert segment GEMM (MoE)");
     m.def("bw_probe", &bw_probe, "bandwidth pattern probe");
     m.def("bw_probe_paged", &bw_probe_paged, "paged row-per-lane probe");

@@ -27,6 +27,36 @@ extern "C" void launch_mfma_probe(const unsigned short* A,
     hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream, A, B, C);
 }
 
+// fp8 (OCP e4m3) variant of the layout probe: assumes the SAME index map
+// as the bf16 op (lane -> row/col, k = (l>>4)*8 + e over 8 fp8 BYTES) —
+// tests/test_ops_gpu.py::test_mfma_fp8_fragment_map verifies it on HW
+// before any fp8-MFMA kernel relies on it.
+__global__ void mfma_probe_fp8_kernel(const unsigned char* A,  // [16][32]
+                                      const unsigned char* B,  // [32][16]
+                                      float* C) {              // [16][16]
+    const int l = threadIdx.x;
+    uchar8 a, b;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        const int kk = (l >> 4) * 8 + e;
+        a[e] = A[(l & 15) * 32 + kk];
+        b[e] = B[kk * 16 + (l & 15)];
+    }
+    f32x4 c{0.f, 0.f, 0.f, 0.f};
+    c = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+        *reinterpret_cast<const long*>(&a),
+        *reinterpret_cast<const long*>(&b), c, 0, 0, 0);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) C[((l >> 4) * 4 + r) * 16 + (l & 15)] = c[r];
+}
+
+extern "C" void launch_mfma_probe_fp8(const unsigned char* A,
+                                      const unsigned char* B, float* C,
+                                      hipStream_t stream) {
+    hipLaunchKernelGGL(mfma_probe_fp8_kernel, dim3(1), dim3(64), 0, stream,
+                       A, B, C);
+}
+
 // ---------------------------------------------------------------------------
 // Bandwidth probes over a bf16 pool — used to locate the decode-attention
 // bandwidth ceiling (scripts/bench_attn.py --probe). Each accumulates a

@@ -25,6 +25,8 @@ def main():
     ap.add_argument("--iters", type=int, default=50)
     ap.add_argument("--prefill", action="store_true",
                     help="bench the MFMA prefill kernel instead")
+    ap.add_argument("--fp8", action="store_true",
+                    help="fp8 (e4m3) KV cache variant of the decode bench")
     args = ap.parse_args()
 
     from bee2bee_amd import ops
@@ -60,8 +62,14 @@ def main():
     W = (L + bs - 1) // bs
     nb = B * W + 1
     bt = torch.arange(1, B * W + 1, dtype=torch.int32).reshape(B, W).to(dev)
-    kc = torch.randn(nb, nkv, bs, hd, device=dev).bfloat16()
-    vc = torch.randn(nb, nkv, bs, hd, device=dev).bfloat16()
+    if args.fp8:
+        kc = torch.randint(0, 127, (nb, nkv, bs, hd), dtype=torch.uint8,
+                           device=dev)
+        vc = torch.randint(0, 127, (nb, nkv, bs, hd), dtype=torch.uint8,
+                           device=dev)
+    else:
+        kc = torch.randn(nb, nkv, bs, hd, device=dev).bfloat16()
+        vc = torch.randn(nb, nkv, bs, hd, device=dev).bfloat16()
     q = torch.randn(B, nq, hd, device=dev).bfloat16()
     lens = torch.full((B,), L, dtype=torch.int32, device=dev)
     scale = hd**-0.5
@@ -74,9 +82,10 @@ def main():
         out = ops.attn_decode(q, kc, vc, bt, lens, scale)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.iters
-    kv_bytes = B * L * 2 * nkv * hd * 2
+    kv_bytes = B * L * 2 * nkv * hd * (1 if args.fp8 else 2)
     print(
-        f"decode B{B} L{L} nkv{nkv} G{G} hd{hd}: {dt * 1e6:.1f} us  "
+        f"decode{' fp8' if args.fp8 else ''} "
+        f"B{B} L{L} nkv{nkv} G{G} hd{hd}: {dt * 1e6:.1f} us  "
         f"KV {kv_bytes / 1e6:.0f} MB  {kv_bytes / dt / 1e12:.2f} TB/s"
     )
     _ = out

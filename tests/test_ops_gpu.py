@@ -362,3 +362,16 @@ def test_fp8_kv_attn_prefill_paged():
     )
     ref = R.attn_decode_with_history(q, kc, vc, bt, lens_dev, qlens, scale)
     assert_close(got, ref, atol=3e-2, rtol=3e-2, msg="fp8 prefill_paged")
+
+
+def test_mfma_fp8_fragment_map():
+    """v_mfma_f32_16x16x32_fp8_fp8 must use the same lane->fragment map as
+    the bf16 op (verified before any fp8-MFMA kernel relies on it)."""
+    torch.manual_seed(16)
+    A = (torch.randn(16, 32, device=DEV) * 0.5).to(torch.float8_e4m3fn)
+    B = (torch.randn(32, 16, device=DEV) * 0.5).to(torch.float8_e4m3fn)
+    C = ops.require_hip().mfma_probe_fp8(
+        A.view(torch.uint8), B.view(torch.uint8)
+    )
+    ref = A.float() @ B.float()
+    assert torch.allclose(C, ref, atol=1e-3), (C - ref).abs().max()
