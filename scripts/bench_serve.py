@@ -30,6 +30,9 @@ async def run() -> None:
     ap.add_argument("--requests", type=int, default=64)
     ap.add_argument("--max-new", type=int, default=64)
     ap.add_argument("--prompt-len", type=int, default=512, help="chars of prompt")
+    ap.add_argument("--repetitive", action="store_true",
+                    help="self-similar prompts (speculative-decoding-friendly "
+                         "workload: code/template-like text)")
     ap.add_argument("--max-batch", type=int, default=32)
     ap.add_argument("--api-port", type=int, default=18321)
     args = ap.parse_args()
@@ -66,6 +69,11 @@ async def run() -> None:
     words = ["alpha", "beta", "gamma", "delta", "mesh", "gpu", "tensor", "ring"]
 
     def mk_prompt() -> str:
+        if args.repetitive:
+            # one short phrase repeated (template/code-like self-similarity)
+            phrase = " ".join(rng.choice(words) for _ in range(4))
+            reps = max(1, args.prompt_len // (len(phrase) + 1))
+            return " ".join([phrase] * reps)
         out = []
         while sum(len(w) + 1 for w in out) < args.prompt_len:
             out.append(rng.choice(words))
@@ -138,7 +146,8 @@ async def run() -> None:
         "tokens_per_sec": round(args.requests * args.max_new / wall, 1),
         "wall_s": round(wall, 2),
         "model_load_s": round(load_s, 1),
-        "data": "synthetic prompts, random-init weights",
+        "data": ("repetitive" if args.repetitive else "random")
+                + " synthetic prompts, random-init weights",
     }
     print(json.dumps(result), flush=True)
     svc.engine.shutdown()
