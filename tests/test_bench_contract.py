@@ -1,0 +1,59 @@
+"""Driver contract tests for bench.py: single-process JSON line shape and
+the torch.distributed.run multi-process launch (gloo on CPU; the GPU box
+takes the same path over RCCL)."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _last_json(out: str):
+    for line in reversed(out.strip().splitlines()):
+        if line.startswith("{"):
+            return json.loads(line)
+    raise AssertionError(f"no JSON line in output:\n{out[-2000:]}")
+
+
+REQUIRED = {
+    "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+    "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config",
+}
+
+
+@pytest.mark.timeout(300)
+def test_bench_single_process_contract():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--model", "tiny", "--batch", "2",
+         "--prompt-len", "8", "--steps", "2", "--warmup", "1"],
+        cwd=REPO, capture_output=True, text=True, timeout=240,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    res = _last_json(out.stdout)
+    assert REQUIRED <= set(res)
+    assert res["n_gpus"] == 1 and res["steps"] == 2 and res["warmup"] == 1
+    assert res["scaling"] == "weak" and res["higher_is_better"] is True
+    assert res["config"]["global_batch"] == 2
+    assert res["value"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_bench_multi_process_contract():
+    """The driver's N>1 launch shape: torch.distributed.run, one JSON line
+    from rank 0, whole-job aggregate value, MAX-over-ranks timing."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29753", "bench.py", "--gpus", "2", "--model",
+         "tiny", "--batch", "2", "--prompt-len", "8", "--steps", "2",
+         "--warmup", "1"],
+        cwd=REPO, capture_output=True, text=True, timeout=240,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    res = _last_json(out.stdout)
+    assert res["n_gpus"] == 2
+    assert res["config"]["parallelism"] == "replica-dp2"
+    assert res["config"]["global_batch"] == 4  # whole-job aggregate
