@@ -127,3 +127,45 @@ def test_kv_pool_exhaustion_backpressure():
             assert len(r.output_ids) == 10
     finally:
         eng.shutdown()
+
+
+def test_concurrent_submit_thread_safety():
+    """submit() from many threads concurrently: the queue-based admission
+    must keep every request isolated and complete."""
+    import threading
+
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=4, max_seq_len=64,
+                          seed=3)
+    reqs = []
+    lock = threading.Lock()
+
+    def submit_some(base):
+        for i in range(3):
+            r = GenerationRequest(
+                prompt_ids=[base + i, base, 7], max_new_tokens=4,
+                sampling=SamplingParams(greedy=True),
+            )
+            eng.submit(r)
+            with lock:
+                reqs.append(r)
+
+    try:
+        threads = [threading.Thread(target=submit_some, args=(b,))
+                   for b in (10, 50, 90, 130)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert len(reqs) == 12
+        for r in reqs:
+            while True:
+                x = r.out_queue.get(timeout=120)
+                if not isinstance(x, int):
+                    break
+            assert r.error is None
+            assert len(r.output_ids) == 4
+    finally:
+        eng.shutdown()

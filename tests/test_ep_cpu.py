@@ -131,3 +131,24 @@ def test_moe_engine_matches_single(tmp_path):
                 assert req.output_ids == ep_out, (r, prompt)
     finally:
         single.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_ep4_one_expert_per_rank(tmp_path):
+    """EP at world=4 on 4 experts: the degenerate one-expert-per-rank shard
+    (local_e == 1) must still match the single-process reference."""
+    ref, _x = _single_reference()
+    out_path = str(tmp_path / "ep4_out.pkl")
+    ctx = mp.get_context("spawn")
+    procs = [
+        ctx.Process(target=_ep_worker, args=(r, 4, 29619, out_path))
+        for r in range(4)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    with open(out_path, "rb") as f:
+        ep_out = pickle.load(f)
+    assert torch.allclose(ep_out, ref, atol=1e-5), (ep_out - ref).abs().max()
