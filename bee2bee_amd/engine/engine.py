@@ -121,7 +121,8 @@ class InferenceEngine:
         self.spec_decode = spec_decode
         self.spec_k = spec_k
         self.spec_ngram = spec_ngram
-        self.spec_stats = {"steps": 0, "proposed": 0, "accepted": 0}
+        self.spec_stats = {"steps": 0, "proposed": 0, "accepted": 0,
+                           "delegated": 0}
         self.spec = model if isinstance(model, ModelSpec) else resolve_spec(model, model_path)
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
@@ -169,7 +170,10 @@ class InferenceEngine:
 
         # MoE captures only in the dense all-experts range (static shapes);
         # the padded-bmm path above it has a data-dependent buffer size
-        graphs_ok = on_gpu and not spec_decode and (
+        # spec mode keeps graphs: steps where NO sequence has a proposal
+        # delegate to the captured decode path (neutral cost on workloads
+        # with nothing to speculate on)
+        graphs_ok = on_gpu and (
             not self.spec.is_moe or max_batch <= Runner.MOE_DENSE_MAX_TOKENS
         )
         self.use_graphs = graphs_ok if use_graphs is None else (use_graphs and graphs_ok)
@@ -629,19 +633,29 @@ class InferenceEngine:
         what plain greedy decode would emit (verification is exact)."""
         dev = self.device
         acts = self._active
-        ids_list, pos_list, slot_list, cu = [], [], [], [0]
+        # propose first: when nothing is speculatable this step, take the
+        # plain (hipGraph-captured) decode path instead of the slower
+        # verify-forward — spec mode then costs ~nothing on workloads
+        # without self-similarity
         props: List[List[int]] = []
         for a in acts:
             r = a.req
-            ctx = r.prompt_ids + r.output_ids
             prop: List[int] = []
             if r.sampling.greedy and r.sampling.repetition_penalty == 1.0:
                 room = self.max_seq_len - (a.length + 1) - 1
                 rem = r.max_new_tokens - len(r.output_ids) - 1
                 kcap = min(self.spec_k, room, rem)
-                prop = self._propose(ctx, kcap)
-            toks = [ctx[-1]] + prop
+                prop = self._propose(r.prompt_ids + r.output_ids, kcap)
             props.append(prop)
+        if not any(props):
+            self.spec_stats["delegated"] += 1
+            self._decode_once()
+            return
+        ids_list, pos_list, slot_list, cu = [], [], [], [0]
+        for a, prop in zip(acts, props):
+            r = a.req
+            ctx = r.prompt_ids + r.output_ids
+            toks = [ctx[-1]] + prop
             self.kv.extend_seq(a.seq_id, a.length + len(toks))
             ids_list.extend(toks)
             pos_list.extend(range(a.length, a.length + len(toks)))

@@ -98,3 +98,13 @@ def test_spec_decode_max_new_tokens_respected():
     prompt = [7, 99, 23] * 6
     out, _, _ = _run(True, prompt, 7)
     assert len(out) == 7
+
+
+def test_spec_decode_delegates_without_proposals():
+    """A prompt with no self-similarity must delegate to the plain decode
+    path (spec mode costs ~nothing when there is nothing to speculate)."""
+    prompt = [3, 17, 91, 204, 55]  # no repeated bigram
+    base, _, _ = _run(False, prompt, 3)
+    spec, stats, _ = _run(True, prompt, 3)
+    assert spec == base
+    assert stats["delegated"] >= 1
