@@ -52,7 +52,8 @@ async def seed_checkpoint(
         hashes = piece_hashes(pieces)
         content_hash = sha256_hex_bytes(data)
         node.share_pieces(content_hash, pieces)
-        await dht_mod.announce_piece(dht, content_hash, node.addr)
+        await _publish(node, dht, f"piece:{content_hash}",
+                       lambda cur: sorted(set((cur or []) + [node.addr])))
         files.append(
             {
                 "name": fname,
@@ -63,8 +64,21 @@ async def seed_checkpoint(
             }
         )
     manifest = {"name": name, "files": files}
-    await dht.set(manifest_key(name), json.dumps(manifest))
+    await _publish(node, dht, manifest_key(name),
+                   lambda cur: json.dumps(manifest))
     return manifest
+
+
+async def _publish(node, dht, key: str, update) -> None:
+    """Write a DHT record; when publishing through the node's OWN store,
+    use the mesh-replicated write (node.dht_set) so connected peers can
+    fetch without a shared/kademlia DHT."""
+    cur = await dht.get(key)
+    value = update(cur)
+    if getattr(node, "dht", None) is dht and hasattr(node, "dht_set"):
+        await node.dht_set(key, value)
+    else:
+        await dht.set(key, value)
 
 
 async def fetch_checkpoint(

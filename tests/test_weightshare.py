@@ -62,3 +62,62 @@ def test_fetch_missing_manifest(tmp_path):
         await node.stop()
 
     asyncio.run(run())
+
+
+def test_fetch_over_mesh_dht_then_serve(tmp_path):
+    """Full distribution chain with NO shared DHT object: the seeder
+    publishes through its own node.dht (mesh-replicated dht_set), the
+    leech discovers + fetches through ITS own store, and the fetched
+    checkpoint serves through the engine with outputs identical to the
+    source weights."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    spec = PRESETS["tiny"]
+    src_dir = tmp_path / "src"
+    dst_dir = tmp_path / "dst"
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(9)
+    save_hf(w, str(src_dir))
+
+    async def run():
+        seeder = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        leech = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await seeder.start()
+        await leech.start()
+        try:
+            await leech.connect_bootstrap(seeder.addr)
+            for _ in range(200):
+                if seeder.peer_id in leech.peers and leech.peer_id in seeder.peers:
+                    break
+                await asyncio.sleep(0.02)
+            # publish AFTER the link is up (one-hop replication)
+            await seed_checkpoint(seeder, seeder.dht, "tiny-ckpt2",
+                                  str(src_dir), piece_size=4096)
+            for _ in range(100):
+                if await leech.dht.get("manifest:tiny-ckpt2"):
+                    break
+                await asyncio.sleep(0.02)
+            await fetch_checkpoint(leech, leech.dht, "tiny-ckpt2", str(dst_dir))
+        finally:
+            await leech.stop()
+            await seeder.stop()
+
+    asyncio.run(run())
+
+    def greedy(model_path=None, seed=9):
+        eng = InferenceEngine("tiny", device="cpu", model_path=model_path,
+                              max_batch=2, max_seq_len=64, seed=seed)
+        try:
+            r = GenerationRequest(prompt_ids=[4, 5, 6], max_new_tokens=5,
+                                  sampling=SamplingParams(greedy=True))
+            eng.submit(r)
+            while True:
+                x = r.out_queue.get(timeout=60)
+                if not isinstance(x, int):
+                    break
+            assert r.error is None, r.error
+            return r.output_ids
+        finally:
+            eng.shutdown()
+
+    assert greedy(model_path=str(dst_dir)) == greedy(seed=9)
