@@ -183,3 +183,41 @@ def test_home_exposes_engine_stats(client):
     finally:
         if svc.engine is not None:
             svc.engine.shutdown()
+
+
+def test_connect_endpoint_joins_peer(client):
+    """GET /connect dials a live peer and it appears in /peers (reference
+    api.py:170 semantics)."""
+    import asyncio
+    import time as _t
+
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.mesh.node import MeshNode
+
+    node = gateway_api.node
+    loop = node._tasks[0].get_loop()
+
+    async def mk_peer():
+        p = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await p.start()
+        return p
+
+    peer = asyncio.run_coroutine_threadsafe(mk_peer(), loop).result(timeout=15)
+    try:
+        r = client.get(
+            f"/connect?addr={peer.addr}", headers={"X-API-KEY": "secret-key"}
+        )
+        assert r.status_code == 200, r.text
+        deadline = _t.time() + 5
+        found = False
+        while _t.time() < deadline:
+            peers = client.get(
+                "/peers", headers={"X-API-KEY": "secret-key"}
+            ).json()
+            if any(p["peer_id"] == peer.peer_id for p in peers):
+                found = True
+                break
+            _t.sleep(0.05)
+        assert found, peers
+    finally:
+        asyncio.run_coroutine_threadsafe(peer.stop(), loop).result(timeout=15)
