@@ -88,6 +88,8 @@ class Runner:
             )
 
         qkv = F.linear(normed, lw.wqkv)  # column-parallel under TP
+        if lw.wqkv_bias is not None:  # Qwen2-style attention bias
+            qkv = qkv + lw.wqkv_bias
         T = qkv.shape[0]
         q, k, v = qkv.split([s.q_size, s.kv_size, s.kv_size], dim=-1)
         q = q.view(T, s.n_heads, s.head_dim)

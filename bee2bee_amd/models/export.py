@@ -47,7 +47,10 @@ class ExportableModel(torch.nn.Module):
         freqs = torch.outer(t, inv)
         self.register_buffer("rope_cos", freqs.cos())
         self.register_buffer("rope_sin", freqs.sin())
+        self.qkv_bias = spec.qkv_bias
         for i, lw in enumerate(weights.layers):
+            if lw.wqkv_bias is not None:
+                self.register_buffer(f"wqkv_bias_{i}", lw.wqkv_bias.float())
             self.register_buffer(f"attn_norm_{i}", lw.attn_norm.float())
             self.register_buffer(f"mlp_norm_{i}", lw.mlp_norm.float())
             self.register_buffer(f"wqkv_{i}", lw.wqkv.float())
@@ -79,6 +82,8 @@ class ExportableModel(torch.nn.Module):
         for i in range(self.n_layers):
             normed = self._rms(h, getattr(self, f"attn_norm_{i}"), self.eps)
             qkv = F.linear(normed, getattr(self, f"wqkv_{i}"))
+            if self.qkv_bias:
+                qkv = qkv + getattr(self, f"wqkv_bias_{i}")
             q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], -1)
             q = self._rope(q.view(B, T, self.n_heads, self.hd), T)
             k = self._rope(k.view(B, T, self.n_kv, self.hd), T)

@@ -31,6 +31,8 @@ class ModelSpec:
     # MoE (0 experts = dense)
     n_experts: int = 0
     top_k_experts: int = 2
+    # Qwen2-style attention bias on the fused QKV projection
+    qkv_bias: bool = False
     # vocab specials (byte-tokenizer defaults; overridden by a real tokenizer)
     bos_token_id: int = 1
     eos_token_id: int = 2
@@ -172,6 +174,22 @@ PRESETS: Dict[str, ModelSpec] = {
         n_kv_heads=8,
         head_dim=128,
     ),
+    # Qwen2.5-7B-Instruct class: llama-shaped + QKV bias, large vocab
+    "qwen2.5-7b": ModelSpec(
+        name="qwen2.5-7b",
+        vocab_size=152064,
+        hidden_size=3584,
+        intermediate_size=18944,
+        n_layers=28,
+        n_heads=28,
+        n_kv_heads=4,
+        head_dim=128,
+        rope_theta=1000000.0,
+        rms_eps=1e-6,
+        max_seq_len=32768,
+        tie_embeddings=False,
+        qkv_bias=True,
+    ),
     # Zephyr-7B = Mistral-7B architecture (BASELINE config 3)
     "zephyr-7b": ModelSpec(
         name="zephyr-7b",
@@ -211,6 +229,8 @@ _ALIASES = {
     "llama-3.2-1b": "llama3.2-1b",
     "llama3.2": "llama3.2-1b",
     "huggingfaceh4/zephyr-7b-beta": "zephyr-7b",
+    "qwen/qwen2.5-7b-instruct": "qwen2.5-7b",
+    "qwen2.5": "qwen2.5-7b",
     "zephyr": "zephyr-7b",
     "zephyr-7b-beta": "zephyr-7b",
     "mistral-7b": "zephyr-7b",
@@ -262,6 +282,11 @@ def spec_from_hf_config(model_path: str, name: Optional[str] = None) -> ModelSpe
         tie_embeddings=cfg.get("tie_word_embeddings", False),
         n_experts=cfg.get("num_local_experts", 0),
         top_k_experts=cfg.get("num_experts_per_tok", 2),
+        # Qwen2 sets attention_bias (or ships q/k/v bias tensors implicitly)
+        qkv_bias=bool(
+            cfg.get("attention_bias", False)
+            or cfg.get("model_type") == "qwen2"
+        ),
         bos_token_id=cfg.get("bos_token_id", 1) or 1,
         eos_token_id=(
             cfg.get("eos_token_id")[0]

@@ -31,7 +31,7 @@ from .spec import ModelSpec
 
 class LayerWeights:
     __slots__ = (
-        "attn_norm", "wqkv", "wo", "mlp_norm",
+        "attn_norm", "wqkv", "wqkv_bias", "wo", "mlp_norm",
         "w_gate_up", "w_down",  # dense mlp
         "moe_gate", "moe_w_gate_up", "moe_w_down",  # moe
     )
@@ -39,6 +39,7 @@ class LayerWeights:
     def __init__(self) -> None:
         self.attn_norm = None
         self.wqkv = None
+        self.wqkv_bias = None  # Qwen2-style fused [q+2kv] bias (optional)
         self.wo = None
         self.mlp_norm = None
         self.w_gate_up = None
@@ -117,6 +118,10 @@ class ModelWeights:
             lw.attn_norm = torch.ones(s.hidden_size, device=self.device, dtype=self.dtype)
             lw.mlp_norm = torch.ones(s.hidden_size, device=self.device, dtype=self.dtype)
             lw.wqkv = rnd(f"l{i}.wqkv", s.q_size + 2 * s.kv_size, s.hidden_size, std=std)
+            if s.qkv_bias:
+                lw.wqkv_bias = rnd(
+                    f"l{i}.wqkv_b", s.q_size + 2 * s.kv_size, std=std
+                )
             lw.wo = rnd(f"l{i}.wo", s.hidden_size, s.q_size, std=proj_std)
             if s.is_moe:
                 lw.moe_gate = rnd(f"l{i}.gate", s.n_experts, s.hidden_size, std=std)
@@ -165,6 +170,10 @@ class ModelWeights:
             if p and len(p) == 3 and lw.wqkv is None:
                 lw.wqkv = torch.cat([p["q"], p["k"], p["v"]], dim=0)
                 pending.pop(f"attn.{i}")
+            p = pending.get(f"attnb.{i}")
+            if p and len(p) == 3 and lw.wqkv_bias is None:
+                lw.wqkv_bias = torch.cat([p["q"], p["k"], p["v"]], dim=0)
+                pending.pop(f"attnb.{i}")
             p = pending.get(f"mlp.{i}")
             if p and len(p) == 2 and lw.w_gate_up is None:
                 lw.w_gate_up = torch.cat([p["gate"], p["up"]], dim=0)
@@ -203,6 +212,15 @@ class ModelWeights:
                 try_fuse(i)
             elif sub == "self_attn.v_proj.weight":
                 pending.setdefault(f"attn.{i}", {})["v"] = to_dev(tensor)
+                try_fuse(i)
+            elif sub == "self_attn.q_proj.bias":
+                pending.setdefault(f"attnb.{i}", {})["q"] = to_dev(tensor)
+                try_fuse(i)
+            elif sub == "self_attn.k_proj.bias":
+                pending.setdefault(f"attnb.{i}", {})["k"] = to_dev(tensor)
+                try_fuse(i)
+            elif sub == "self_attn.v_proj.bias":
+                pending.setdefault(f"attnb.{i}", {})["v"] = to_dev(tensor)
                 try_fuse(i)
             elif sub == "self_attn.o_proj.weight":
                 lw.wo = to_dev(tensor)
@@ -291,6 +309,13 @@ def save_hf(weights: ModelWeights, out_dir: str) -> None:
         tensors[f"{pfx}.self_attn.q_proj.weight"] = cpu(q)
         tensors[f"{pfx}.self_attn.k_proj.weight"] = cpu(k)
         tensors[f"{pfx}.self_attn.v_proj.weight"] = cpu(v)
+        if lw.wqkv_bias is not None:
+            qb, kb, vb = torch.split(
+                lw.wqkv_bias, [s.q_size, s.kv_size, s.kv_size], dim=0
+            )
+            tensors[f"{pfx}.self_attn.q_proj.bias"] = cpu(qb)
+            tensors[f"{pfx}.self_attn.k_proj.bias"] = cpu(kb)
+            tensors[f"{pfx}.self_attn.v_proj.bias"] = cpu(vb)
         tensors[f"{pfx}.self_attn.o_proj.weight"] = cpu(lw.wo)
         tensors[f"{pfx}.input_layernorm.weight"] = cpu(lw.attn_norm)
         tensors[f"{pfx}.post_attention_layernorm.weight"] = cpu(lw.mlp_norm)
@@ -324,6 +349,7 @@ def save_hf(weights: ModelWeights, out_dir: str) -> None:
         "num_key_value_heads": s.n_kv_heads,
         "head_dim": s.head_dim,
         "rope_theta": s.rope_theta,
+        "attention_bias": s.qkv_bias,
         "rms_norm_eps": s.rms_eps,
         "max_position_embeddings": s.max_seq_len,
         "tie_word_embeddings": s.tie_embeddings,

@@ -83,6 +83,13 @@ def shard_weights(
         v = lw.wqkv[spec.q_size + spec.kv_size + kv_lo :
                     spec.q_size + spec.kv_size + kv_hi]
         ol.wqkv = torch.cat([q, k, v], dim=0).contiguous()
+        if lw.wqkv_bias is not None:
+            ol.wqkv_bias = torch.cat([
+                lw.wqkv_bias[q_lo:q_hi],
+                lw.wqkv_bias[spec.q_size + kv_lo : spec.q_size + kv_hi],
+                lw.wqkv_bias[spec.q_size + spec.kv_size + kv_lo :
+                             spec.q_size + spec.kv_size + kv_hi],
+            ], dim=0).contiguous()
         ol.wo = lw.wo[:, q_lo:q_hi].contiguous()
         ol.w_gate_up = torch.cat(
             [lw.w_gate_up[i_lo:i_hi], lw.w_gate_up[I + i_lo : I + i_hi]],

@@ -66,3 +66,16 @@ def test_onnx_gated():
     if not has:
         with pytest.raises(RuntimeError, match="onnx_support_missing"):
             export_onnx(spec, w, "/tmp/x.onnx")
+
+
+def test_export_with_qkv_bias_matches_engine():
+    import dataclasses
+
+    spec = dataclasses.replace(PRESETS["tiny"], name="tiny-qb", qkv_bias=True)
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(9)
+    ids = [5, 9, 100]
+    ref = _engine_logits(spec, w, ids)
+    model = ExportableModel(spec, w).eval()
+    with torch.no_grad():
+        got = model(torch.tensor([ids], dtype=torch.int64))[0]
+    assert torch.allclose(got, ref, atol=1e-4), (got - ref).abs().max()

@@ -52,3 +52,46 @@ def test_hf_config_parsing(tmp_path):
     assert spec.n_experts == 8 and spec.head_dim == 128
     spec2 = resolve_spec("whatever", model_path=str(tmp_path))
     assert spec2.n_experts == 8
+
+
+def test_qwen2_preset_and_bias_roundtrip(tmp_path):
+    """Qwen2 family: QKV-bias weights roundtrip through HF save/load and
+    the engine decodes with the bias applied."""
+    import dataclasses
+
+    import torch
+
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+    from bee2bee_amd.models.spec import PRESETS, resolve_spec
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+    assert PRESETS["qwen2.5-7b"].qkv_bias
+    assert resolve_spec("Qwen/Qwen2.5-7B-Instruct").name == "qwen2.5-7b"
+
+    tiny_q = dataclasses.replace(PRESETS["tiny"], name="tiny-qwen", qkv_bias=True)
+    w = ModelWeights(tiny_q, torch.device("cpu"), torch.float32).random_init(4)
+    assert w.layers[0].wqkv_bias is not None
+    save_hf(w, str(tmp_path))
+    w2 = ModelWeights(tiny_q, torch.device("cpu"), torch.float32).load_hf(str(tmp_path))
+    assert torch.allclose(w.layers[1].wqkv_bias, w2.layers[1].wqkv_bias)
+
+    # bias must change the output vs the same weights without bias
+    def gen(spec, seed=4):
+        eng = InferenceEngine(spec, device="cpu", max_batch=2,
+                              max_seq_len=64, seed=seed)
+        try:
+            r = GenerationRequest(prompt_ids=[4, 5, 6], max_new_tokens=5,
+                                  sampling=SamplingParams(greedy=True))
+            eng.submit(r)
+            while True:
+                x = r.out_queue.get(timeout=60)
+                if not isinstance(x, int):
+                    break
+            assert r.error is None, r.error
+            return r.output_ids
+        finally:
+            eng.shutdown()
+
+    with_bias = gen(tiny_q)
+    assert len(with_bias) == 5
