@@ -57,3 +57,24 @@ def test_bench_multi_process_contract():
     assert res["n_gpus"] == 2
     assert res["config"]["parallelism"] == "replica-dp2"
     assert res["config"]["global_batch"] == 4  # whole-job aggregate
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("mode,model", [("pp", "tiny"), ("tp", "tiny")])
+def test_bench_pp_contract(mode, model):
+    """scripts/bench_pp.py (the 8-GPU model-parallel bench) launches under
+    torch.distributed.run and emits the JSON contract (gloo dry run)."""
+    port = "29771" if mode == "pp" else "29772"
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", port, "scripts/bench_pp.py", "--mode", mode,
+         "--model", model, "--batch", "2", "--prompt-len", "8",
+         "--steps", "2", "--warmup", "1"],
+        cwd=REPO, capture_output=True, text=True, timeout=240,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    res = _last_json(out.stdout)
+    assert res["scaling"] == "strong"
+    assert res["config"]["parallelism"] == f"{mode}2"
+    assert res["value"] > 0

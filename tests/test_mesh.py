@@ -363,3 +363,35 @@ def test_mesh_dht_rendezvous():
             await b.stop()
 
     asyncio.run(run())
+
+
+def test_dht_set_merge_on_receive():
+    """Concurrent rendezvous announces from two peers must MERGE dict
+    records, not clobber (the one-hop replication handler)."""
+    import asyncio
+
+    from bee2bee_amd.mesh.node import MeshNode
+
+    async def run():
+        a = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        b = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await a.start()
+        await b.start()
+        try:
+            await b.connect_bootstrap(a.addr)
+            await asyncio.sleep(0.3)
+            await a.dht_set("rccl:g", {"peer-a": {"host": "1.1.1.1"}})
+            await b.dht_set("rccl:g", {"peer-b": {"host": "2.2.2.2"}})
+            for _ in range(100):
+                ra = await a.dht.get("rccl:g") or {}
+                rb = await b.dht.get("rccl:g") or {}
+                if set(ra) == {"peer-a", "peer-b"} == set(rb):
+                    break
+                await asyncio.sleep(0.02)
+            assert set(ra) == {"peer-a", "peer-b"}, ra
+            assert set(rb) == {"peer-a", "peer-b"}, rb
+        finally:
+            await a.stop()
+            await b.stop()
+
+    asyncio.run(run())
