@@ -133,6 +133,15 @@ class InferenceEngine:
         self.dtype = dtype
         self.max_batch = max_batch
         self.max_seq_len = min(max_seq_len or 4096, self.spec.max_seq_len)
+        # the decode combine kernel stages <=128 chunk (m,l) pairs in LDS
+        # (attn_decode.hip ml_s): clamp at construction instead of raising
+        # mid-step and tearing down every active request
+        DECODE_MAX_CTX = 128 * 256
+        if self.max_seq_len > DECODE_MAX_CTX:
+            logger.warning(
+                "max_seq_len %d exceeds the decode kernel's %d-token context"
+                " limit; clamping", self.max_seq_len, DECODE_MAX_CTX)
+            self.max_seq_len = DECODE_MAX_CTX
         self.tokenizer = load_tokenizer(
             model_path, self.spec.vocab_size, self.spec.bos_token_id, self.spec.eos_token_id
         )
@@ -233,12 +242,16 @@ class InferenceEngine:
         temperature: Optional[float] = 0.7,
         stop_token_ids: tuple = (),
         on_token=None,
+        top_p: Optional[float] = None,
+        top_k: Optional[int] = None,
+        repetition_penalty: Optional[float] = None,
     ) -> GenerationRequest:
         """Blocking convenience wrapper: submit + drain the stream."""
         req = GenerationRequest(
             prompt_ids=list(prompt_ids),
             max_new_tokens=max_new_tokens,
-            sampling=SamplingParams.from_request(temperature),
+            sampling=SamplingParams.from_request(
+                temperature, top_p, top_k, repetition_penalty),
             stop_token_ids=tuple(stop_token_ids),
         )
         self.submit(req)
@@ -259,7 +272,9 @@ class InferenceEngine:
         temperature: Optional[float] = 0.7,
         on_text=None,
         stop: Optional[List[str]] = None,
-        repetition_penalty: float = 1.0,
+        repetition_penalty: Optional[float] = None,
+        top_p: Optional[float] = None,
+        top_k: Optional[int] = None,
     ) -> Dict[str, Any]:
         """Text-level wrapper used by the mesh service. `stop` strings
         truncate the output at the first occurrence (reference stop-word
@@ -301,8 +316,8 @@ class InferenceEngine:
         eos = getattr(self.tokenizer, "eos_token_id", None)
         if eos is not None:
             stop_ids = (eos,)
-        sp = SamplingParams.from_request(temperature)
-        sp.repetition_penalty = repetition_penalty
+        sp = SamplingParams.from_request(temperature, top_p, top_k,
+                                         repetition_penalty)
         req = GenerationRequest(
             prompt_ids=ids,
             max_new_tokens=max_new_tokens,
@@ -751,7 +766,8 @@ class InferenceEngine:
         for i, a in enumerate(acts):
             sp = a.req.sampling
             groups.setdefault(
-                (sp.greedy, sp.temperature, sp.top_p, sp.top_k), []
+                (sp.greedy, sp.temperature, sp.top_p, sp.top_k,
+                 sp.repetition_penalty), []
             ).append(i)
         next_dev = torch.empty(
             len(acts), dtype=torch.int64, device=logits.device

@@ -20,29 +20,50 @@ class SamplingParams:
     greedy: bool = False
 
     @classmethod
-    def from_request(cls, temperature: Optional[float]) -> "SamplingParams":
+    def from_request(
+        cls,
+        temperature: Optional[float],
+        top_p: Optional[float] = None,
+        top_k: Optional[int] = None,
+        repetition_penalty: Optional[float] = None,
+    ) -> "SamplingParams":
+        """Wire/API request knobs -> params, with the REFERENCE's generation
+        defaults (bee2bee/hf.py:94-103): temperature 0.7, top_p 0.95,
+        repetition_penalty 1.15. temperature <= 1e-4 selects greedy (the
+        penalty still applies — matching do_sample=False + penalty in HF)."""
         t = 0.7 if temperature is None else float(temperature)
-        if t <= 1e-4:
-            return cls(temperature=1.0, greedy=True)
-        return cls(temperature=t)
+        p = cls(temperature=max(t, 1e-4), greedy=t <= 1e-4)
+        if top_p is not None:
+            p.top_p = float(top_p)
+        if top_k is not None:
+            p.top_k = int(top_k)
+        p.repetition_penalty = (
+            1.15 if repetition_penalty is None else float(repetition_penalty)
+        )
+        return p
 
 
 def apply_repetition_penalty(
     logits: torch.Tensor, prev_ids: torch.Tensor, penalty: float
 ) -> torch.Tensor:
-    """prev_ids [B, L] (pad with -1); penalize already-emitted tokens."""
-    if penalty == 1.0:
+    """prev_ids [B, L] (pad with -1); penalize already-emitted tokens.
+
+    Fully batched: one scatter builds a seen-token mask (pad indices land in
+    a sacrificial extra column), one where applies the penalty — no host
+    loop, so a large decode batch pays O(1) kernel launches."""
+    if penalty == 1.0 or prev_ids.numel() == 0:
         return logits
-    B = logits.shape[0]
-    for b in range(B):
-        ids = prev_ids[b]
-        ids = ids[ids >= 0]
-        if ids.numel() == 0:
-            continue
-        row = logits[b]
-        vals = row[ids]
-        row[ids] = torch.where(vals > 0, vals / penalty, vals * penalty)
-    return logits
+    B, V = logits.shape
+    valid = prev_ids >= 0
+    idx = torch.where(valid, prev_ids, torch.full_like(prev_ids, V))
+    seen = torch.zeros((B, V + 1), dtype=torch.bool, device=logits.device)
+    seen.scatter_(1, idx.long(), True)
+    seen = seen[:, :V]
+    return torch.where(
+        seen,
+        torch.where(logits > 0, logits / penalty, logits * penalty),
+        logits,
+    )
 
 
 def sample(

@@ -27,9 +27,59 @@ from .pieces import piece_hashes, split_pieces, verify_and_reassemble
 
 DEFAULT_PIECE = 8 * 1024 * 1024
 
+# Hard bounds on peer-supplied manifests. A manifest arrives over the
+# replicated DHT (any connected peer can write it), so every field in it is
+# attacker-controlled: file names, counts and sizes must all be validated
+# before they touch the filesystem.
+MAX_MANIFEST_FILES = 4096
+MAX_FILE_BYTES = 512 * 1024 * 1024 * 1024  # 512 GiB — a 70B bf16 shard set fits
+MAX_PIECES_PER_FILE = 1 << 20
+
 
 def manifest_key(name: str) -> str:
     return f"manifest:{name}"
+
+
+def _safe_dest(out_dir: str, name: str) -> str:
+    """Resolve a peer-supplied file name to a path strictly inside out_dir.
+
+    Names come from an untrusted DHT manifest; anything that is not a plain
+    file name (path separators, '..', absolute paths, empty) is rejected, and
+    the resolved real path is checked to stay under out_dir even in the
+    presence of symlinks."""
+    if (
+        not name
+        or name != os.path.basename(name)
+        or name in (".", "..")
+        or "/" in name
+        or "\\" in name
+        or "\x00" in name
+    ):
+        raise ValueError(f"unsafe file name in manifest: {name!r}")
+    dest = os.path.join(out_dir, name)
+    root = os.path.realpath(out_dir)
+    resolved = os.path.realpath(dest)
+    if resolved != root and not resolved.startswith(root + os.sep):
+        raise ValueError(f"manifest file escapes output dir: {name!r}")
+    return dest
+
+
+def _validate_manifest(manifest) -> None:
+    """Structural + bounds validation of an untrusted manifest."""
+    if not isinstance(manifest, dict) or not isinstance(manifest.get("files"), list):
+        raise ValueError("malformed manifest")
+    files = manifest["files"]
+    if len(files) > MAX_MANIFEST_FILES:
+        raise ValueError(f"manifest lists {len(files)} files (max {MAX_MANIFEST_FILES})")
+    for entry in files:
+        if not isinstance(entry, dict):
+            raise ValueError("malformed manifest entry")
+        hashes = entry.get("piece_hashes")
+        if not isinstance(hashes, list) or len(hashes) > MAX_PIECES_PER_FILE:
+            raise ValueError("manifest entry piece list missing or oversized")
+        nbytes = entry.get("bytes")
+        if not isinstance(nbytes, int) or nbytes < 0 or nbytes > MAX_FILE_BYTES:
+            raise ValueError("manifest entry byte count out of bounds")
 
 
 async def seed_checkpoint(
@@ -96,8 +146,10 @@ async def fetch_checkpoint(
     if not raw:
         raise FileNotFoundError(f"no manifest for '{name}' in DHT")
     manifest = json.loads(raw)
+    _validate_manifest(manifest)
     os.makedirs(out_dir, exist_ok=True)
     for entry in manifest["files"]:
+        dest = _safe_dest(out_dir, entry["name"])
         pid = provider_peer_id
         if pid is None:
             providers = await dht_mod.find_providers(dht, entry["content_hash"])
@@ -119,6 +171,8 @@ async def fetch_checkpoint(
         data = verify_and_reassemble(pieces, entry["piece_hashes"])
         if sha256_hex_bytes(data) != entry["content_hash"]:
             raise ValueError(f"content hash mismatch for {entry['name']}")
-        with open(os.path.join(out_dir, entry["name"]), "wb") as f:
+        if len(data) != entry["bytes"]:
+            raise ValueError(f"size mismatch for {entry['name']}")
+        with open(dest, "wb") as f:
             f.write(data)
     return out_dir

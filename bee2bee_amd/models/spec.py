@@ -254,9 +254,16 @@ def resolve_spec(name: str, model_path: Optional[str] = None) -> ModelSpec:
         return PRESETS[key]
     if key in _ALIASES:
         return PRESETS[_ALIASES[key]]
-    for alias, preset in _ALIASES.items():
-        if alias in key or key in alias:
-            return PRESETS[preset]
+    # Prefix match only, and only across a variant-suffix boundary ("-instruct",
+    # "_chat", ...). A bare substring fallback mapped e.g. "llama3.2-1b" to the
+    # llama3-8b preset: a remainder that continues with digits is a
+    # different model size, not a variant, so it must fall through to the demo
+    # spec rather than silently serve the wrong architecture.
+    for alias in sorted(_ALIASES, key=len, reverse=True):
+        if key.startswith(alias):
+            rest = key[len(alias):]
+            if rest and rest[0] in "-_/." and len(rest) > 1 and rest[1].isalpha():
+                return PRESETS[_ALIASES[alias]]
     spec = PRESETS["demo-125m"]
     return ModelSpec(**{**spec.__dict__, "name": name})
 

@@ -152,10 +152,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     int nkv, int W, int bs, int hd, int C, long q_stride, float scale) {
     const int b = blockIdx.x;
     const int kvh = blockIdx.y;
-    const int chunk = blockIdx.z;
     const int L = seq_lens[b];
-    const int start = chunk * DEC_CHUNK;
-    if (start >= L) return;
+    if ((int)blockIdx.z * DEC_CHUNK >= L) return;
     const int lane = threadIdx.x % WAVE;
     const int wid = threadIdx.x / WAVE;
 
@@ -172,6 +170,12 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     __syncthreads();
 
     const int* bt = block_table + (long)b * W;
+    // loop chunks with stride gridDim.z — the PV and combine stages walk
+    // ALL chunks of (b, kvh), so every chunk's p/(m,l) must be produced even
+    // when the launcher picks Z < ceil(L/DEC_CHUNK) (a single-
+    // chunk kernel left chunks >= Z uninitialized for hd not in {64,128})
+    for (int chunk = blockIdx.z; chunk * DEC_CHUNK < L; chunk += gridDim.z) {
+    const int start = chunk * DEC_CHUNK;
     const int key = start + threadIdx.x;  // one key per thread
     const bool valid = key < L;
 
@@ -254,6 +258,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
             pml[g * 2 + 1] = lsum;
         }
     }
+    __syncthreads();  // red reused by the next chunk iteration
+    }  // chunk loop
 }
 
 
@@ -748,7 +754,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
     const long base = ((long)b * nkv + kvh) * C;
 
     // stage this head's (m, l) pairs once
-    __shared__ float ml_s[64][2];  // C is far below 64 in practice
+    __shared__ float ml_s[128][2];  // C <= 128 (32768-token contexts)
     for (int c = threadIdx.x; c < nc; c += DEC_BLOCK) {
         ml_s[c][0] = part_ml[((base + c) * G + g) * 2];
         ml_s[c][1] = part_ml[((base + c) * G + g) * 2 + 1];

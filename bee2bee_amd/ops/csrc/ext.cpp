@@ -184,7 +184,7 @@ Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
     // flash-decoding split: one partial per 256-key chunk, then combine
     const int kDecChunk = 256;  // keep in sync with DEC_CHUNK
     const int C = (W * bs + kDecChunk - 1) / kDecChunk;
-    TORCH_CHECK(C <= 64, "decode supports up to 16384-token contexts for now");
+    TORCH_CHECK(C <= 128, "decode supports up to 32768-token contexts for now");
     auto fopt = q.options().dtype(torch::kFloat32);
     Tensor p_buf = torch::empty({B, nkv, C, kDecChunk, G}, q.options());
     Tensor part_o = torch::empty({B, nkv, C, G, hd}, fopt);

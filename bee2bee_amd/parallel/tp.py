@@ -148,6 +148,11 @@ class TPEngine:
         )
         self.runner = Runner(self.lspec, self.weights, self.kv, self.device, dtype)
         self.runner.tp_group = group if group is not None else dist.group.WORLD
+        # non-greedy sampling must draw IDENTICAL tokens on every rank or the
+        # KV caches silently diverge: a dedicated generator seeded
+        # the same ranks-wide, never the global per-process RNG
+        self.sample_gen = torch.Generator(device=self.device)
+        self.sample_gen.manual_seed(seed + 0x5EED)
         self._seqs: List[int] = []
         self._lens: List[int] = []
         logger.info(
@@ -182,7 +187,8 @@ class TPEngine:
             max(self._lens),
         )
         last = torch.tensor([c - 1 for c in cu[1:]], dtype=torch.int64, device=dev)
-        return sample(self.runner.lm_head(hidden[last]), sampling).cpu()
+        return sample(self.runner.lm_head(hidden[last]), sampling,
+                      generator=self.sample_gen).cpu()
 
     @torch.no_grad()
     def decode_step(
@@ -202,7 +208,8 @@ class TPEngine:
         hidden = self.runner.forward_decode(
             ids.to(dev), positions, slots, bt, lens_t
         )
-        return sample(self.runner.lm_head(hidden), sampling).cpu()
+        return sample(self.runner.lm_head(hidden), sampling,
+                      generator=self.sample_gen).cpu()
 
     @torch.no_grad()
     def generate(

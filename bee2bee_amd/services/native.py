@@ -89,9 +89,16 @@ class NativeEngineService(BaseService):
         max_new = int(params.get("max_new_tokens", self.max_new_tokens))
         max_new = max(1, min(max_new, self.max_new_tokens))
         temperature = float(params.get("temperature", 0.7))
+        # None -> SamplingParams.from_request applies the reference's
+        # generation defaults (top_p 0.95, repetition_penalty 1.15)
+        rp = params.get("repetition_penalty")
+        tp = params.get("top_p")
+        tk = params.get("top_k")
         extra = {
             "stop": params.get("stop"),
-            "repetition_penalty": float(params.get("repetition_penalty", 1.0)),
+            "repetition_penalty": None if rp is None else float(rp),
+            "top_p": None if tp is None else float(tp),
+            "top_k": None if tk is None else int(tk),
         }
         return prompt, max_new, temperature, extra
 
@@ -159,8 +166,9 @@ class NativeEngineService(BaseService):
         eng = self.engine
         ids = eng.tokenizer.encode(prompt)
         ids = ids[-(eng.max_seq_len - max_new - 1):]
-        sp = SamplingParams.from_request(temperature)
-        sp.repetition_penalty = extra.get("repetition_penalty", 1.0)
+        sp = SamplingParams.from_request(
+            temperature, extra.get("top_p"), extra.get("top_k"),
+            extra.get("repetition_penalty"))
         stop_ids = ()
         eos = getattr(eng.tokenizer, "eos_token_id", None)
         if eos is not None:

@@ -121,3 +121,57 @@ def test_fetch_over_mesh_dht_then_serve(tmp_path):
             eng.shutdown()
 
     assert greedy(model_path=str(dst_dir)) == greedy(seed=9)
+
+
+def test_manifest_path_traversal_rejected(tmp_path):
+    """A peer-poisoned manifest with '../' or absolute names must never
+    reach the filesystem."""
+    import json
+
+    import pytest
+
+    from bee2bee_amd.mesh.weightshare import (
+        MAX_MANIFEST_FILES,
+        _safe_dest,
+        _validate_manifest,
+        manifest_key,
+    )
+
+    out = tmp_path / "out"
+    out.mkdir()
+    for bad in ("../evil", "/etc/passwd", "a/b", "..", "", "a\\b", "x\x00y"):
+        with pytest.raises(ValueError):
+            _safe_dest(str(out), bad)
+    # a symlink inside out_dir pointing outside must also be rejected
+    (out / "link").symlink_to(tmp_path)
+    with pytest.raises(ValueError):
+        _safe_dest(str(out), "link")
+    assert _safe_dest(str(out), "model.safetensors").startswith(str(out))
+
+    with pytest.raises(ValueError):
+        _validate_manifest({"files": [{}] * (MAX_MANIFEST_FILES + 1)})
+    with pytest.raises(ValueError):
+        _validate_manifest({"files": [{"piece_hashes": [], "bytes": -1}]})
+    with pytest.raises(ValueError):
+        _validate_manifest({"files": "nope"})
+
+    async def run():
+        dht = DHTNode()
+        await dht.start()
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        evil = {"name": "x", "files": [{
+            "name": "../../pwned", "bytes": 4,
+            "content_hash": "00" * 32, "piece_hashes": ["00" * 32],
+            "piece_size": 4,
+        }]}
+        await dht.set(manifest_key("x"), json.dumps(evil))
+        try:
+            with pytest.raises(ValueError):
+                await fetch_checkpoint(node, dht, "x", str(tmp_path / "fetch"))
+        finally:
+            await node.stop()
+
+    asyncio.run(run())
+    assert not (tmp_path.parent / "pwned").exists()
+    assert not (tmp_path / "pwned").exists()
