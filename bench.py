@@ -40,7 +40,19 @@ def main() -> None:
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--no-tunableop", action="store_true",
                     help="skip hipBLASLt algorithm tuning")
+    ap.add_argument("--preflight", action="store_true",
+                    help="run the scale-run preflight checks (RCCL self-"
+                         "test, master endpoint, env, rank wiring) and exit")
     args = ap.parse_args()
+
+    if args.preflight:
+        from bee2bee_amd.parallel.preflight import run_preflight
+
+        tune_dir = os.path.join(
+            os.path.dirname(os.path.abspath(__file__)), "gpurun_out")
+        report = run_preflight(args.gpus, tune_dir)
+        print(json.dumps(report), flush=True)
+        sys.exit(0 if report["preflight"] == "ok" else 1)
 
     from bee2bee_amd.models.spec import resolve_spec
 

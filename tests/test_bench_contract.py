@@ -78,3 +78,21 @@ def test_bench_pp_contract(mode, model):
     assert res["scaling"] == "strong"
     assert res["config"]["parallelism"] == f"{mode}2"
     assert res["value"] > 0
+
+
+@pytest.mark.timeout(240)
+def test_preflight_cpu():
+    """`bench.py --preflight` must pass on a GPU-less box (rccl/device
+    checks degrade to warn; master/env/gloo wiring are hard checks)."""
+    import json
+    import subprocess
+
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--preflight", "--gpus", "8"],
+        cwd=REPO, capture_output=True, text=True, timeout=220,
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
+    report = json.loads(r.stdout.strip().splitlines()[-1])
+    assert report["preflight"] == "ok"
+    assert report["checks"]["gloo_wiring"]["status"] == "pass"
+    assert report["checks"]["master"]["status"] == "pass"
