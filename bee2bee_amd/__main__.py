@@ -102,6 +102,34 @@ def serve_ollama(model, host, port, public_host, region, api_port):
     )
 
 
+@cli.command("serve-web")
+@click.option("--host", default="0.0.0.0", help="Bind host")
+@click.option("--port", default=8080, type=int, help="HTTP port")
+@click.option("--seeds", default=None,
+              help="Comma-separated seed node WS addrs (else BEE2BEE_SEEDS)")
+def serve_web(host, port, seeds):
+    """Serve the browser gateway + chat dashboard (L6 web layer):
+    /api/p2p/{register,generate,status,global_metrics} over a mesh bridge."""
+    import uvicorn
+
+    from .web.bridge import MeshBridge
+    from .web.gateway import create_app
+
+    seed_list = [s.strip() for s in seeds.split(",")] if seeds else None
+
+    async def _run() -> None:
+        bridge = MeshBridge(seeds=seed_list)
+        await bridge.start()
+        try:
+            config = uvicorn.Config(create_app(bridge), host=host, port=port,
+                                    log_level="info")
+            await uvicorn.Server(config).serve()
+        finally:
+            await bridge.stop()
+
+    asyncio.run(_run())
+
+
 @cli.command("serve-hf-remote")
 @click.option("--model", default="meta-llama/Llama-2-7b-hf", help="HF model name")
 @click.option("--token", required=True, help="HF API token")

@@ -1,0 +1,474 @@
+"""Mesh bridge: the gateway's client side of the wire protocol.
+
+Reimplements the reference's WS bridge (app/api/bridge.js) natively:
+  * a node pool fed by seeds (BEE2BEE_SEEDS) and the public directory
+    (active_nodes pull, bridge.js:133-161), with failed nodes dropped and
+    the pool rotated until one connects (:46-92)
+  * hello/gen_chunk/gen_success/gen_error/ping handling (:163-223); this
+    implementation also treats `gen_result` as terminal (wire.py Q1 fix)
+  * request routing: direct HTTP to the target node's /generate first
+    (fast-fail for firewall probing), then the WS tunnel with a 90 s
+    timeout that resolves with partial chunks if any arrived (:259-349)
+  * join-link registration: priority node + directory push (:372-422)
+
+Differences kept deliberate: no fabricated latencies (the reference
+randomizes dashboard latency, bridge.js:254 — we report measured ping RTT
+or nothing), and directory pushes go through the same typed row builder as
+the node's own registry sync (mesh/registry.py NodeRow).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import os
+import time
+from typing import Any, Callable, Dict, List, Optional
+
+import aiohttp
+
+from ..mesh import wire
+from ..mesh.links import parse_join_link
+from ..utils import new_id
+from .store import WebStore
+
+logger = logging.getLogger("bee2bee_amd.web")
+
+CONNECT_TIMEOUT_S = 5.0
+DIRECT_HTTP_TIMEOUT_S = 5.0
+REQUEST_TIMEOUT_S = 90.0  # bridge-side (reference bridge.js:334), not the
+                          # node's 300 s wire.REQUEST_TIMEOUT
+RECONNECT_DELAY_S = 5.0
+MESH_SYNC_INTERVAL_S = 30.0
+
+
+def _ws_addr(target: str) -> str:
+    """`host:port` / `http://host:port` -> `ws://host:port`."""
+    if target.startswith(("ws://", "wss://")):
+        return target
+    if target.startswith("https://"):
+        return "wss://" + target[len("https://"):]
+    if target.startswith("http://"):
+        return "ws://" + target[len("http://"):]
+    return f"ws://{target}"
+
+
+def _http_addr(target: str) -> str:
+    if target.startswith(("http://", "https://")):
+        return target
+    if target.startswith("wss://"):
+        return "https://" + target[len("wss://"):]
+    if target.startswith("ws://"):
+        return "http://" + target[len("ws://"):]
+    return f"http://{target}"
+
+
+class _Pending:
+    __slots__ = ("future", "on_chunk", "chunks", "start")
+
+    def __init__(self, on_chunk: Optional[Callable[[str], None]]) -> None:
+        self.future: asyncio.Future = asyncio.get_event_loop().create_future()
+        self.on_chunk = on_chunk
+        self.chunks: List[str] = []
+        self.start = time.time()
+
+
+class MeshBridge:
+    """One active WS link into the mesh + a rotating candidate pool."""
+
+    def __init__(
+        self,
+        seeds: Optional[List[str]] = None,
+        store: Optional[WebStore] = None,
+        auto_reconnect: bool = True,
+    ) -> None:
+        if seeds is None:
+            env = os.getenv("BEE2BEE_SEEDS", "")
+            seeds = [s.strip() for s in env.split(",") if s.strip()]
+        self.pool: List[str] = list(dict.fromkeys(seeds))
+        self.store = store if store is not None else WebStore()
+        self.auto_reconnect = auto_reconnect
+        self.registered_node: Optional[str] = None
+        self.peer_meta: Dict[str, Dict[str, Any]] = {}
+        self.started_at = time.time()
+        self._ws: Optional[aiohttp.ClientWebSocketResponse] = None
+        self._url: Optional[str] = None
+        self._session: Optional[aiohttp.ClientSession] = None
+        self._pending: Dict[str, _Pending] = {}
+        self._reader_task: Optional[asyncio.Task] = None
+        self._sync_task: Optional[asyncio.Task] = None
+        self._last_rtt_ms: Optional[float] = None
+        self._closed = False
+
+    # ------------------------------------------------------------ lifecycle
+
+    async def start(self) -> None:
+        self._session = aiohttp.ClientSession()
+        if MESH_SYNC_INTERVAL_S > 0:
+            self._sync_task = asyncio.create_task(self._sync_loop())
+
+    async def stop(self) -> None:
+        self._closed = True
+        for task in (self._sync_task, self._reader_task):
+            if task:
+                task.cancel()
+        if self._ws is not None:
+            await self._ws.close()
+            self._ws = None
+        if self._session is not None:
+            await self._session.close()
+            self._session = None
+        for pending in self._pending.values():
+            if not pending.future.done():
+                pending.future.cancel()
+        self._pending.clear()
+
+    async def _sync_loop(self) -> None:
+        while not self._closed:
+            await asyncio.sleep(MESH_SYNC_INTERVAL_S)
+            try:
+                await self.sync_global_mesh()
+            except Exception:  # noqa: BLE001
+                pass
+
+    # ------------------------------------------------------------ discovery
+
+    async def sync_global_mesh(self) -> None:
+        """Refresh the candidate pool + metadata from the directory."""
+        for row in await self.store.active_nodes():
+            addr = row.get("addr")
+            if not addr:
+                continue
+            if addr not in self.pool:
+                self.pool.append(addr)
+            self._merge_meta(addr, row)
+
+    @property
+    def connected(self) -> bool:
+        return self._ws is not None and not self._ws.closed
+
+    async def connect(self) -> bool:
+        """Try the registered node first, then the pool; prune failures."""
+        if self.connected:
+            return True
+        await self.sync_global_mesh()
+        candidates = list(dict.fromkeys(
+            ([self.registered_node] if self.registered_node else [])
+            + self.pool))
+        for addr in candidates:
+            if await self._dial(addr):
+                return True
+            if addr in self.pool:
+                self.pool.remove(addr)
+            if self.registered_node == addr:
+                self.registered_node = None
+        return False
+
+    async def _dial(self, addr: str) -> bool:
+        assert self._session is not None, "bridge not started"
+        url = _ws_addr(addr)
+        try:
+            ws = await asyncio.wait_for(
+                self._session.ws_connect(url, max_msg_size=wire.MAX_FRAME),
+                timeout=CONNECT_TIMEOUT_S,
+            )
+        except Exception as e:  # noqa: BLE001
+            logger.info("bridge dial failed %s: %s", url, e)
+            return False
+        self._ws, self._url = ws, url
+        self._reader_task = asyncio.create_task(self._reader(ws))
+        self._merge_meta(url, {"status": "active"})
+        await self._push_to_directory(url)
+        logger.info("bridge connected to %s", url)
+        return True
+
+    async def connect_to_peer(self, addr: str) -> None:
+        """Dynamic discovery (status POST action=discover_peer)."""
+        if addr and addr not in self.pool:
+            self.pool.insert(0, addr)
+        if not self.connected:
+            await self.connect()
+
+    # ------------------------------------------------------------- wire I/O
+
+    async def _reader(self, ws: aiohttp.ClientWebSocketResponse) -> None:
+        try:
+            async for msg in ws:
+                if msg.type != aiohttp.WSMsgType.TEXT:
+                    if msg.type in (aiohttp.WSMsgType.ERROR,
+                                    aiohttp.WSMsgType.CLOSE):
+                        break
+                    continue
+                try:
+                    frame = json.loads(msg.data)
+                except json.JSONDecodeError:
+                    continue
+                await self._on_frame(ws, frame)
+        except Exception as e:  # noqa: BLE001
+            logger.debug("bridge reader ended: %s", e)
+        finally:
+            if self._ws is ws:
+                self._ws = None
+                self._url = None
+            if self.auto_reconnect and not self._closed:
+                await asyncio.sleep(RECONNECT_DELAY_S)
+                if not self._closed:
+                    asyncio.ensure_future(self.connect())
+
+    async def _on_frame(self, ws, frame: Dict[str, Any]) -> None:
+        ftype = frame.get("type")
+        rid = frame.get("task_id") or frame.get("rid")
+        pending = self._pending.get(rid) if rid else None
+
+        if ftype == wire.HELLO:
+            if self._url:
+                self._merge_meta(self._url, frame)
+                await self._push_to_directory(self._url)
+            return
+        if ftype == wire.GEN_CHUNK:
+            if pending is not None:
+                text = frame.get("text", "")
+                pending.chunks.append(text)
+                if pending.on_chunk:
+                    pending.on_chunk(text)
+            return
+        if ftype in (wire.GEN_SUCCESS, wire.GEN_RESULT):
+            if pending is not None and not pending.future.done():
+                self._pending.pop(rid, None)
+                err = frame.get("error")
+                if err:
+                    pending.future.set_exception(RuntimeError(str(err)))
+                else:
+                    pending.future.set_result({
+                        "text": frame.get("text") or "".join(pending.chunks),
+                        "rid": rid,
+                        "metadata": {
+                            "backend": frame.get("backend"),
+                            "tokens": frame.get("tokens"),
+                            "latency_ms":
+                                (time.time() - pending.start) * 1000.0,
+                        },
+                    })
+            return
+        if ftype == wire.GEN_ERROR:
+            if pending is not None and not pending.future.done():
+                self._pending.pop(rid, None)
+                pending.future.set_exception(
+                    RuntimeError(frame.get("error") or "node_failure"))
+            return
+        if ftype == wire.PING:
+            await ws.send_str(json.dumps(
+                {"type": wire.PONG, "ts": frame.get("ts")}))
+            return
+        if ftype == wire.PONG:
+            ts = frame.get("ts")
+            if isinstance(ts, (int, float)):
+                self._last_rtt_ms = max(0.0, time.time() * 1000.0 - ts)
+
+    # ------------------------------------------------------------- requests
+
+    async def request(
+        self,
+        payload: Dict[str, Any],
+        on_chunk: Optional[Callable[[str], None]] = None,
+        target_node: Optional[str] = None,
+    ) -> Dict[str, Any]:
+        """Direct-HTTP-first, WS-tunnel-fallback generation request."""
+        if target_node:
+            target_ws = _ws_addr(target_node)
+            if self._url == target_ws and self.connected:
+                logger.info("using established tunnel for %s", target_node)
+            else:
+                result = await self._direct_http(payload, on_chunk, target_node)
+                if result is not None:
+                    return result
+                # direct HTTP failed: swap the tunnel onto the target
+                if self._url != target_ws:
+                    if self._ws is not None:
+                        await self._ws.close()
+                        self._ws = None
+                    self.registered_node = target_ws
+                    if target_ws not in self.pool:
+                        self.pool.insert(0, target_ws)
+
+        if not self.connected:
+            await self.connect()
+        if not self.connected:
+            raise ConnectionError(
+                "mesh unreachable: no node accepted a connection")
+        return await self._tunnel(payload, on_chunk)
+
+    async def _direct_http(
+        self, payload: Dict[str, Any],
+        on_chunk: Optional[Callable[[str], None]],
+        target_node: str,
+    ) -> Optional[Dict[str, Any]]:
+        """POST {node api}/generate and line-parse the JSON stream; None on
+        any failure so the caller falls back to the WS tunnel."""
+        assert self._session is not None, "bridge not started"
+        url = f"{_http_addr(target_node)}/generate"
+        body = {
+            "prompt": payload.get("prompt"),
+            "model": payload.get("model"),
+            "stream": True,
+        }
+        if payload.get("max_tokens") is not None:
+            body["max_tokens"] = payload["max_tokens"]
+        if payload.get("temperature") is not None:
+            body["temperature"] = payload["temperature"]
+        try:
+            text_parts: List[str] = []
+            timeout = aiohttp.ClientTimeout(connect=DIRECT_HTTP_TIMEOUT_S,
+                                            total=REQUEST_TIMEOUT_S)
+            async with self._session.post(url, json=body,
+                                          timeout=timeout) as resp:
+                if resp.status != 200:
+                    return None
+                async for raw_line in resp.content:
+                    line = raw_line.decode("utf-8", "replace").strip()
+                    if not line:
+                        continue
+                    try:
+                        obj = json.loads(line)
+                    except json.JSONDecodeError:
+                        text_parts.append(line)
+                        if on_chunk:
+                            on_chunk(line)
+                        continue
+                    if obj.get("status") == "error":
+                        raise RuntimeError(obj.get("message", "node error"))
+                    delta = obj.get("text") or obj.get("response") or ""
+                    if delta:
+                        text_parts.append(delta)
+                        if on_chunk:
+                            on_chunk(delta)
+            return {"text": "".join(text_parts), "rid": None,
+                    "metadata": {"transport": "direct-http"}}
+        except RuntimeError:
+            raise
+        except Exception as e:  # noqa: BLE001
+            logger.info("direct HTTP to %s failed (%s); trying tunnel",
+                        url, e)
+            return None
+
+    async def _tunnel(
+        self, payload: Dict[str, Any],
+        on_chunk: Optional[Callable[[str], None]],
+    ) -> Dict[str, Any]:
+        task_id = new_id("task")
+        pending = _Pending(on_chunk)
+        self._pending[task_id] = pending
+        frame = {
+            "type": wire.GEN_REQUEST,
+            "task_id": task_id,
+            "rid": task_id,
+            "model": payload.get("model"),
+            "prompt": payload.get("prompt"),
+            "stream": True,
+        }
+        for key in ("max_tokens", "temperature", "svc"):
+            if payload.get(key) is not None:
+                frame[key] = payload[key]
+        assert self._ws is not None
+        await self._ws.send_str(json.dumps(frame))
+        try:
+            return await asyncio.wait_for(pending.future, REQUEST_TIMEOUT_S)
+        except asyncio.TimeoutError:
+            self._pending.pop(task_id, None)
+            if pending.chunks:
+                # partial stream beats a hard error (reference bridge.js:338)
+                return {"text": "".join(pending.chunks), "rid": task_id,
+                        "metadata": {"partial": True}}
+            raise TimeoutError(
+                "node timeout: no response within "
+                f"{REQUEST_TIMEOUT_S:.0f}s") from None
+
+    # ---------------------------------------------------------- registration
+
+    async def register_join_link(self, link: str) -> Dict[str, Any]:
+        """Parse a `coithub.org://join?...` link, make its node the priority
+        target, push it to the directory, and connect."""
+        try:
+            info = parse_join_link(link)
+        except ValueError as e:
+            return {"success": False, "error": str(e)}
+        bootstrap = info.get("bootstrap") or []
+        if not bootstrap:
+            return {"success": False, "error": "missing peer address in link"}
+        node_url = _ws_addr(bootstrap[0])
+        self.registered_node = node_url
+        if node_url not in self.pool:
+            self.pool.insert(0, node_url)
+        self._merge_meta(node_url, {
+            "models": [info["model"]] if info.get("model") else [],
+            "metrics": {"join_link": link, "status": "active"},
+        })
+        await self._push_to_directory(node_url)
+        # force the tunnel onto the registered node
+        if self._ws is not None:
+            await self._ws.close()
+            self._ws = None
+        ok = await self.connect()
+        return {"success": True, "node": node_url, "connected": ok}
+
+    # -------------------------------------------------------------- metadata
+
+    def _merge_meta(self, addr: str, frame: Dict[str, Any]) -> None:
+        meta = self.peer_meta.setdefault(addr, {"addr": addr})
+        if frame.get("peer_id"):
+            meta["peer_id"] = frame["peer_id"]
+        if frame.get("region"):
+            meta["region"] = frame["region"]
+        if frame.get("metrics"):
+            meta["metrics"] = frame["metrics"]
+        models = frame.get("models")
+        if not models and isinstance(frame.get("services"), dict):
+            models = sorted({m for s in frame["services"].values()
+                             for m in (s or {}).get("models", [])})
+        if models:
+            meta["models"] = list(models)
+        for key in ("api_port", "api_host", "public_ip"):
+            if frame.get(key) is not None:
+                meta[key] = frame[key]
+        meta["status"] = "active"
+        meta["last_seen"] = time.time()
+
+    async def _push_to_directory(self, addr: str) -> None:
+        meta = self.peer_meta.get(addr, {})
+        await self.store.upsert_node({
+            "peer_id": meta.get("peer_id") or addr.split("://")[-1],
+            "addr": addr,
+            "region": meta.get("region", "Global"),
+            "models": meta.get("models", []),
+            "metrics": meta.get("metrics") or {"status": "active"},
+            "last_seen": _utc_iso(),
+        })
+
+    # ----------------------------------------------------------------- stats
+
+    def get_stats(self) -> Dict[str, Any]:
+        return {
+            "uptime_s": round(time.time() - self.started_at, 1),
+            "connected": self.connected,
+            "activeNode": self._url,
+            "poolSize": len(self.peer_meta),
+            "totalPeers": len(self.peer_meta),
+            "peers": list(self.peer_meta.values()),
+        }
+
+    def get_regional_mesh(self) -> Dict[str, List[Dict[str, Any]]]:
+        """Peers grouped by region; latency is the MEASURED ping RTT of the
+        active link when known (never fabricated)."""
+        mesh: Dict[str, List[Dict[str, Any]]] = {}
+        for addr, meta in self.peer_meta.items():
+            entry = dict(meta)
+            if addr == self._url and self._last_rtt_ms is not None:
+                entry["latency"] = round(self._last_rtt_ms, 1)
+            mesh.setdefault(meta.get("region", "Global"), []).append(entry)
+        return mesh
+
+
+def _utc_iso() -> str:
+    from datetime import datetime, timezone
+
+    return datetime.now(timezone.utc).isoformat()
