@@ -31,11 +31,15 @@
 
 #define GG_BLOCK 512
 #define GG_WAVES 8
-#define GG_BN 128         // output cols per workgroup (16 per wave)
+#define GG_BN 64          // output cols per workgroup (16 per col-group)
 #define GG_MROWS 128      // rows per pass (8 MFMA row-tiles)
 #define GG_MT (GG_MROWS / 16)
+#define GG_WMT (GG_MT / 2)  // row-tiles per wave (rows split across 2
+                            // wave-halves: halves the per-wave registers ->
+                            // 2 blocks/CU instead of 1; the duplicate W
+                            // fragment loads of a row-half pair hit L2)
 #define GG_BK 32          // K step (one mfma_16x16x32)
-#define GG_KG 128         // K-group per pipeline stage (4 K-steps)
+#define GG_KG 64          // K-group per pipeline stage (2 K-steps)
 #define GG_XPAD 8         // LDS row padding (bf16 elems): 272 B rows keep
                           // b128 reads conflict-free at the 64-dword modulus
 
@@ -56,7 +60,9 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
     const int lg = lane >> 4;
     const int li = lane & 15;
 
-    const unsigned short* wp = w + ((long)e * N + n0 + wid * 16 + li) * K;
+    const int colg = wid & 3;   // wave's 16-col group
+    const int rowh = wid >> 2;  // wave's 64-row half
+    const unsigned short* wp = w + ((long)e * N + n0 + colg * 16 + li) * K;
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     unsigned short* x_s = reinterpret_cast<unsigned short*>(smem_raw);
     constexpr int XS = GG_KG + GG_XPAD;
@@ -64,10 +70,11 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
 
     // x staging ownership: thread t covers GG_NCHUNK of the row×16-dim
     // chunks of a (128 x 128) tile — independent 16 B loads per group/set
-    constexpr int GG_NCHUNK = GG_MROWS * (GG_KG / 16) / GG_BLOCK;
-    constexpr int GG_RSTRIDE = GG_BLOCK / (GG_KG / 16);
-    const int st_row0 = tid >> 3;          // base row; +GG_RSTRIDE per chunk
-    const int st_d = (tid & 7) * 16;       // dim start within the group
+    constexpr int DCH = GG_KG / 16;        // 16-dim chunks per row
+    constexpr int GG_NCHUNK = GG_MROWS * DCH / GG_BLOCK;
+    constexpr int GG_RSTRIDE = GG_BLOCK / DCH;
+    const int st_row0 = tid / DCH;         // base row; +GG_RSTRIDE per chunk
+    const int st_d = (tid % DCH) * 16;     // dim start within the group
     const int ng = K / GG_KG;              // K-groups (K % 128 == 0 checked
                                            // host-side)
 
@@ -75,9 +82,9 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
         const int m_cnt = min(GG_MROWS, m_hi - pass);
         const int n_mt = (m_cnt + 15) / 16;
 
-        f32x4 acc[GG_MT];
+        f32x4 acc[GG_WMT];
 #pragma unroll
-        for (int t = 0; t < GG_MT; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int t = 0; t < GG_WMT; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
         short8 xr[2][GG_NCHUNK][2];   // [set][chunk][2 x 16B]
         bf16x8 wf[2][GG_KG / GG_BK];  // [set][kstep]
@@ -126,12 +133,13 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
         if ((G) + 2 < ng) GG_XLOAD((G) + 2, SET);                              \
         __syncthreads();                                                       \
         const unsigned short* xb_ = x_s + (SET) * GG_MROWS * XS;               \
-        _Pragma("unroll") for (int g = 0; g < GG_KG / GG_BK; ++g) {            \
-            _Pragma("unroll") for (int t = 0; t < GG_MT; ++t) {                \
+        _Pragma("unroll") for (int g_ = 0; g_ < GG_KG / GG_BK; ++g_) {         \
+            _Pragma("unroll") for (int t = 0; t < GG_WMT; ++t) {               \
                 const bf16x8 a_ = *reinterpret_cast<const bf16x8*>(            \
-                    xb_ + (t * 16 + li) * XS + g * GG_BK + lg * 8);            \
+                    xb_ + ((rowh * GG_WMT + t) * 16 + li) * XS +               \
+                    g_ * GG_BK + lg * 8);                                      \
                 acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(              \
-                    a_, wf[SET][g], acc[t], 0, 0, 0);                          \
+                    a_, wf[SET][g_], acc[t], 0, 0, 0);                         \
             }                                                                  \
             __builtin_amdgcn_sched_barrier(0);                                 \
         }                                                                      \
@@ -159,13 +167,13 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
 
         // epilogue: C row = lg*4 + r, col = li
 #pragma unroll
-        for (int t = 0; t < GG_MT; ++t) {
-            if (t >= n_mt) continue;
+        for (int t = 0; t < GG_WMT; ++t) {
+            if (rowh * GG_WMT + t >= n_mt) continue;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int row = t * 16 + lg * 4 + r;
+                const int row = (rowh * GG_WMT + t) * 16 + lg * 4 + r;
                 if (row >= m_cnt) continue;
-                out[(long)(pass + row) * N + n0 + wid * 16 + li] =
+                out[(long)(pass + row) * N + n0 + colg * 16 + li] =
                     f2bf(acc[t][r]);
             }
         }

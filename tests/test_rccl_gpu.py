@@ -138,7 +138,9 @@ def test_preflight_reports_rccl_pass():
         cwd=REPO, capture_output=True, text=True, timeout=280,
     )
     assert r.returncode == 0, r.stdout + r.stderr
-    report = json.loads(r.stdout.strip().splitlines()[-1])
+    # RCCL prints a version banner to stdout: scan for the report line
+    report = next(json.loads(ln) for ln in r.stdout.splitlines()
+                  if ln.startswith("{") and '"preflight"' in ln)
     assert report["checks"]["rccl_self"]["status"] == "pass", report
     assert report["checks"]["env"]["status"] == "pass", report
 
