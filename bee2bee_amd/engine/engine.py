@@ -177,13 +177,16 @@ class InferenceEngine:
                           kv_dtype=kv_dtype)
         self.runner = Runner(self.spec, self.weights, self.kv, self.device, dtype)
 
-        # MoE captures only in the dense all-experts range (static shapes);
-        # the padded-bmm path above it has a data-dependent buffer size
+        # MoE captures while every decode bucket takes a capture-safe path:
+        # dense all-experts bmm, or the grouped-GEMM kernel (device-side
+        # segment sizes, no host sync) — only the padded-bmm fallback above
+        # the grouped range blocks capture.
         # spec mode keeps graphs: steps where NO sequence has a proposal
         # delegate to the captured decode path (neutral cost on workloads
         # with nothing to speculate on)
         graphs_ok = on_gpu and (
-            not self.spec.is_moe or max_batch <= Runner.MOE_DENSE_MAX_TOKENS
+            not self.spec.is_moe
+            or Runner.moe_graph_capturable(self.spec, max_batch)
         )
         self.use_graphs = graphs_ok if use_graphs is None else (use_graphs and graphs_ok)
         self.graphs: Optional[DecodeGraphs] = None

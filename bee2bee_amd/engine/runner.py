@@ -130,8 +130,33 @@ class Runner:
     # above the dense threshold: sort token-slots by expert and run the
     # grouped-GEMM HIP kernel per projection — weights read once per expert,
     # flops proportional to routed tokens, segment sizes stay on device
-    # (static shapes -> graph-capturable, no host sync)
-    MOE_GROUPED_MAX_TOKENS = 100_000
+    # (static shapes -> graph-capturable, no host sync). The kernel
+    # accumulates 128 rows per W pass, so it is the default while the
+    # AVERAGE segment fits one pass (T*k <= E*128); bigger batches re-read
+    # W per extra pass and the padded bmm wins — those sizes only occur on
+    # the (uncaptured) prefill path, where the host sync is harmless.
+    MOE_GROUPED_ROWS = 128
+    MOE_BMM_MAX_TOKENS = 100_000
+
+    @classmethod
+    def moe_grouped_max_tokens(cls, spec) -> int:
+        return cls.MOE_GROUPED_ROWS * spec.n_experts // spec.top_k_experts
+
+    @staticmethod
+    def moe_grouped_aligned(spec) -> bool:
+        """grouped_gemm kernel dim constraints (N%64, K%128) for both
+        projections: gate_up [2I, H] and down [H, I]."""
+        h, i = spec.hidden_size, spec.intermediate_size
+        return h % 128 == 0 and i % 128 == 0
+
+    @classmethod
+    def moe_graph_capturable(cls, spec, max_tokens: int) -> bool:
+        """True when every decode bucket up to max_tokens takes a
+        capture-safe MoE path (dense or grouped — never padded bmm)."""
+        if max_tokens <= cls.MOE_DENSE_MAX_TOKENS:
+            return True
+        return (cls.moe_grouped_aligned(spec)
+                and max_tokens <= cls.moe_grouped_max_tokens(spec))
 
     def _moe_mlp(self, lw, x: torch.Tensor) -> torch.Tensor:
         """Top-k expert MLP: dense all-experts bmm for decode-sized batches,
@@ -161,7 +186,7 @@ class Runner:
             out = torch.einsum("eth,te->th", y.float(), wfull)
             return out.to(x.dtype)
 
-        if T <= self.MOE_GROUPED_MAX_TOKENS:
+        if T <= self.MOE_BMM_MAX_TOKENS:
             # counting sort from one-hot cumsums: every op here (one_hot,
             # cumsum, gather, index_copy) is hipGraph-capture-safe —
             # torch.bincount/argsort are not
@@ -176,10 +201,16 @@ class Runner:
             rank = (oh.cumsum(0) - oh).gather(1, flat_e.unsqueeze(1)).squeeze(1)
             pos = (offs_excl[flat_e] + rank).to(torch.int64)  # dest row
             tok = torch.arange(S, device=x.device, dtype=torch.int64) // k
-            if _os.environ.get("BEE2BEE_MOE_GROUPED") == "1":
-                # in-development grouped-GEMM kernel path (capture-safe,
-                # device-side segment sizes); currently ~1.7 TB/s of W vs
-                # hipBLASLt bmm's ~3.1 — see docs/ROADMAP.md
+            use_grouped = (
+                x.device.type == "cuda"
+                and self.moe_grouped_aligned(s)
+                and T <= self.moe_grouped_max_tokens(s)
+                and _os.environ.get("BEE2BEE_MOE_BMM") != "1"
+            )
+            if use_grouped:
+                # grouped-GEMM HIP kernel (default in this range): no host
+                # sync, static shapes -> the whole MoE decode step hipGraph-
+                # captures; measured 3.8 TB/s of W per projection
                 offsets = torch.zeros(E + 1, dtype=torch.int32, device=x.device)
                 offsets[1:] = counts.cumsum(0).to(torch.int32)
                 x_sorted = torch.empty(S, H, dtype=x.dtype, device=x.device)

@@ -260,7 +260,7 @@ Tensor grouped_gemm(const Tensor& x, const Tensor& w, const Tensor& offs) {
     const int E = w.size(0), N = w.size(1);
     TORCH_CHECK(w.size(2) == K, "K mismatch");
     TORCH_CHECK(offs.size(0) == E + 1);
-    TORCH_CHECK(N % 128 == 0, "N must be a multiple of 128");
+    TORCH_CHECK(N % 64 == 0, "N must be a multiple of 64");
     TORCH_CHECK(K % 128 == 0, "K must be a multiple of 128");
     Tensor out = torch::empty({S, N}, x.options());
     if (S > 0)

@@ -20,15 +20,15 @@ def test_dense_equals_padded_equals_gather():
     x = torch.randn(9, w.spec.hidden_size,
                     generator=torch.Generator().manual_seed(4))
     dense = runner._moe_mlp(w.layers[0], x)  # T=9 <= threshold -> dense
-    saved_d, saved_p = Runner.MOE_DENSE_MAX_TOKENS, Runner.MOE_GROUPED_MAX_TOKENS
+    saved_d, saved_p = Runner.MOE_DENSE_MAX_TOKENS, Runner.MOE_BMM_MAX_TOKENS
     try:
-        Runner.MOE_DENSE_MAX_TOKENS = 0  # force sorted/grouped path
+        Runner.MOE_DENSE_MAX_TOKENS = 0  # force sorted (padded-bmm on CPU)
         padded = runner._moe_mlp(w.layers[0], x)
-        Runner.MOE_GROUPED_MAX_TOKENS = 0  # force per-expert gather path
+        Runner.MOE_BMM_MAX_TOKENS = 0  # force per-expert gather path
         gather = runner._moe_mlp(w.layers[0], x)
     finally:
         Runner.MOE_DENSE_MAX_TOKENS = saved_d
-        Runner.MOE_GROUPED_MAX_TOKENS = saved_p
+        Runner.MOE_BMM_MAX_TOKENS = saved_p
     assert torch.allclose(dense, padded, atol=1e-5), (dense - padded).abs().max()
     assert torch.allclose(dense, gather, atol=1e-5), (dense - gather).abs().max()
 
