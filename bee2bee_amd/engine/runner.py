@@ -186,12 +186,25 @@ class Runner:
             out = torch.einsum("eth,te->th", y.float(), wfull)
             return out.to(x.dtype)
 
-        if T <= self.MOE_BMM_MAX_TOKENS:
+        import os as _os
+
+        use_grouped = (
+            x.device.type == "cuda"
+            and self.moe_grouped_aligned(s)
+            and T <= self.moe_grouped_max_tokens(s)
+            and _os.environ.get("BEE2BEE_MOE_BMM") != "1"
+        )
+        # the padded bmm stays CPU-only (and opt-in for A/B measurement):
+        # hipBLASLt's batched TN bf16 kernel memory-faults on gfx950 once
+        # the padded M reaches ~1K (repro: scripts/debug_moe_t4096.py, C
+        # 1049 AND 1056) — GPU sizes above the grouped range take the
+        # per-expert GEMM loop below, whose plain TN GEMMs are solid
+        use_bmm = (x.device.type != "cuda"
+                   or _os.environ.get("BEE2BEE_MOE_BMM") == "1")
+        if T <= self.MOE_BMM_MAX_TOKENS and (use_grouped or use_bmm):
             # counting sort from one-hot cumsums: every op here (one_hot,
             # cumsum, gather, index_copy) is hipGraph-capture-safe —
             # torch.bincount/argsort are not
-            import os as _os
-
             k = s.top_k_experts
             S = T * k
             flat_e = idx.reshape(-1).to(torch.int64)  # slot s -> expert
@@ -201,12 +214,6 @@ class Runner:
             rank = (oh.cumsum(0) - oh).gather(1, flat_e.unsqueeze(1)).squeeze(1)
             pos = (offs_excl[flat_e] + rank).to(torch.int64)  # dest row
             tok = torch.arange(S, device=x.device, dtype=torch.int64) // k
-            use_grouped = (
-                x.device.type == "cuda"
-                and self.moe_grouped_aligned(s)
-                and T <= self.moe_grouped_max_tokens(s)
-                and _os.environ.get("BEE2BEE_MOE_BMM") != "1"
-            )
             if use_grouped:
                 # grouped-GEMM HIP kernel (default in this range): no host
                 # sync, static shapes -> the whole MoE decode step hipGraph-
@@ -222,8 +229,11 @@ class Runner:
             else:
                 # padded per-expert bmm: weights read once per expert, flops
                 # proportional to max routed count; counts.max() is one host
-                # sync per MoE layer (so no hipGraph above the dense range)
-                C = int(counts.max())
+                # sync per MoE layer (so no hipGraph above the dense range).
+                # M aligns up to 16: odd M (e.g. 1049) faults hipBLASLt's
+                # TN bf16 batched kernel on gfx950 (repro:
+                # scripts/debug_moe_t4096.py; <2% padded flops)
+                C = (int(counts.max()) + 15) & ~15
                 padded = torch.zeros(E, C, H, dtype=x.dtype, device=x.device)
                 padded[flat_e, rank] = x[tok]
                 gu = torch.bmm(padded, lw.moe_w_gate_up.transpose(1, 2))
