@@ -9,9 +9,20 @@ GPU-less plumbing runs).
 
 Kernels (ops/csrc/): fused (add+)RMSNorm, RoPE, paged KV store, paged GQA
 decode attention (flash-decoding style), MFMA flash prefill attention,
-SwiGLU, top-k gating. Plain projection GEMMs go through hipBLASLt via
-torch.nn.functional.linear — library GEMMs are the one place we use a vendor
-library; every fused hot op is hand-written HIP.
+SwiGLU, top-k gating, grouped expert GEMM. Plain projection GEMMs go
+through hipBLASLt via torch.nn.functional.linear — library GEMMs are the
+one place we use a vendor library; every fused hot op is hand-written HIP.
+
+The projection carve-out is MEASURED, not assumed (scripts/bench_skinny.py,
+round 2): across the four Llama-3-8B projection shapes at decode batches
+16/64/256/1536, hipBLASLt beats the hand-written grouped kernel driven as
+a single-segment GEMM on all 16 points (e.g. gate_up at M=64: 6.34 TB/s
+of W vs 3.91 — with one expert segment the grouped grid collapses to
+N/64 workgroups and starves the chip; hipBLASLt's split-K kernels keep
+all 256 CUs fed). The kernel stays the MoE winner because the E-way grid
+and the device-side segment offsets are exactly what the library cannot
+express (its batched path memory-faults at padded M ~1K, and padding an
+uneven routing to bmm shape costs a host sync that blocks hipGraph).
 """
 from __future__ import annotations
 
