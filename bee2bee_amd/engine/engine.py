@@ -84,13 +84,14 @@ class GenerationRequest:
 
 
 class _Active:
-    __slots__ = ("req", "seq_id", "length", "prefilled")
+    __slots__ = ("req", "seq_id", "length", "prefilled", "spec_index")
 
     def __init__(self, req: GenerationRequest, seq_id: int, length: int) -> None:
         self.req = req
         self.seq_id = seq_id
         self.length = length  # tokens currently in KV cache
         self.prefilled = 0  # prompt tokens prefetched so far (chunked prefill)
+        self.spec_index = None  # lazy NGramIndex (speculative proposer)
 
 
 class InferenceEngine:
@@ -626,22 +627,23 @@ class InferenceEngine:
 
     # ------------------------------------------------- speculative decoding
 
-    def _propose(self, ids: List[int], k: int) -> List[int]:
-        """Prompt-lookup proposer: continuation of the most recent earlier
-        occurrence of the trailing n-gram (no draft model needed)."""
-        n = self.spec_ngram
-        if k <= 0 or len(ids) <= n:
+    def _propose(self, a: "_Active", ids: List[int], k: int) -> List[int]:
+        """Longest-match-first prompt-lookup via an incremental per-sequence
+        n-gram index (engine/spec.py) — covers prompt AND output history at
+        O(1) maintenance per token instead of the O(context) rescan of the
+        naive scan; n sizes tried {spec_ngram+2, spec_ngram+1, spec_ngram}."""
+        if k <= 0:
             return []
-        tail = ids[-n:]
-        best: List[int] = []
-        for i in range(len(ids) - n - 1, -1, -1):
-            if ids[i : i + n] == tail:
-                cont = ids[i + n : i + n + k]
-                if len(cont) > len(best):
-                    best = cont
-                if len(best) == k:
-                    break
-        return best
+        idx = getattr(a, "spec_index", None)
+        if idx is None:
+            from .spec import NGramIndex
+
+            n = self.spec_ngram
+            idx = NGramIndex(ids, ns=(n + 2, n + 1, n))
+            a.spec_index = idx
+        else:
+            idx.sync(ids)
+        return idx.propose(k)
 
     @torch.no_grad()
     def _decode_spec_once(self) -> None:
@@ -663,7 +665,7 @@ class InferenceEngine:
                 room = self.max_seq_len - (a.length + 1) - 1
                 rem = r.max_new_tokens - len(r.output_ids) - 1
                 kcap = min(self.spec_k, room, rem)
-                prop = self._propose(r.prompt_ids + r.output_ids, kcap)
+                prop = self._propose(a, r.prompt_ids + r.output_ids, kcap)
             props.append(prop)
         if not any(props):
             self.spec_stats["delegated"] += 1
