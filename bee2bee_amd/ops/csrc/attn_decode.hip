@@ -451,15 +451,20 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     long* voff_s = reinterpret_cast<long*>(p_s + DEC_CHUNK * G);  // [CHUNK]
     float* merge = reinterpret_cast<float*>(voff_s + DEC_CHUNK);  // [Wv][G][hd]
     // FOLD running state lives in LDS so the hot loop stays at ~64 VGPRs
-    // (a register version measured 101 VGPRs -> 5 waves/SIMD). Only wave 0
-    // touches it, in wave-lockstep, so no barriers are needed around it.
+    // (a register version measured 101 VGPRs -> 5 waves/SIMD). Each head
+    // group g is OWNED by wave g%DEC_WAVES: the per-chunk merge + fold and
+    // the final normalize run G-way parallel across waves instead of
+    // serializing on wave 0 (the r1 kernel's measured latency tail), and
+    // per-g slices stay single-writer so no extra barriers appear.
     float* run = merge + DEC_WAVES * G * hd;   // [G][hd] running acc
     float* runml = run + G * hd;               // [G][2] running (m, l)
-    if (FOLD && wid == 0) {
-        for (int i = lane; i < G * hd; i += WAVE) run[i] = 0.f;
-        if (lane < G) {
-            runml[lane * 2] = -1e30f;
-            runml[lane * 2 + 1] = 0.f;
+    if (FOLD) {
+        for (int g = wid; g < G; g += DEC_WAVES) {
+            for (int i = lane; i < hd; i += WAVE) run[g * hd + i] = 0.f;
+            if (lane == 0) {
+                runml[g * 2] = -1e30f;
+                runml[g * 2 + 1] = 0.f;
+            }
         }
     }
 
@@ -543,10 +548,9 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
         }
     }
     __syncthreads();
-    if (wid == 0 && d0 < hd) {
+    if (d0 < hd) {
         const long base = (((long)b * nkv + kvh) * C + chunk) * G;
-#pragma unroll
-        for (int g = 0; g < G; ++g) {
+        for (int g = wid; g < G; g += DEC_WAVES) {
             float a0 = 0.f, a1 = 0.f;
             for (int w2 = 0; w2 < DEC_WAVES; ++w2) {
                 a0 += merge[(w2 * G + g) * hd + d0];
@@ -577,12 +581,11 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     __syncthreads();  // LDS buffers reused by the next chunk iteration
     }  // chunk loop
 
-    if (FOLD && wid == 0) {
+    if (FOLD) {
         const int d0f = lane * 2;
         if (d0f < hd) {
             const int nq = nkv * G;
-#pragma unroll
-            for (int g = 0; g < G; ++g) {
+            for (int g = wid; g < G; g += DEC_WAVES) {
                 const float lr = runml[g * 2 + 1];
                 const float inv = (lr > 0.f) ? 1.f / lr : 0.f;
                 unsigned short* orow =
