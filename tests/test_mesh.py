@@ -395,3 +395,40 @@ def test_dht_set_merge_on_receive():
             await b.stop()
 
     asyncio.run(run())
+
+
+def test_outbound_only_provider_tunnel():
+    """Tunnel-hosting parity (reference notebooks served Colab nodes via
+    ngrok/Bore tunnels): a provider with NO inbound reachability dials OUT
+    to a public relay and serves through that one established connection —
+    requester -> relay -> (inbound ws) -> provider. This is the native
+    equivalent of the reference's tunnel capability: the outbound WS is
+    the tunnel."""
+
+    async def run():
+        relay = await _start_node()          # the only publicly dialable node
+        # behind NAT: announces a black-hole TEST-NET address, so gossiped
+        # dial attempts at it can never succeed — only its outbound link
+        # into the relay can carry traffic
+        provider = await _start_node(announce_host="203.0.113.7",
+                                     announce_port=9)
+        requester = await _start_node()
+        await provider.add_service(EchoService())
+        # provider initiates the only connection it has (outbound)
+        await provider.connect_bootstrap(relay.addr)
+        await _wait_for(lambda: provider.peer_id in relay.providers)
+        await requester.connect_bootstrap(relay.addr)
+        await _wait_for(lambda: relay.peer_id in requester.peers)
+        res = await requester.request_generation(
+            relay.peer_id, "through the tunnel", 8, "echo-model", timeout=10
+        )
+        assert res["text"] == "echo:through the tunnel"
+        # the relay reached the provider over the provider-initiated ws —
+        # no peer holds a live connection AT the provider's (unreachable)
+        # announce addr except via that inbound tunnel
+        assert provider.addr.startswith("ws://203.0.113.7")
+        await requester.stop()
+        await provider.stop()
+        await relay.stop()
+
+    asyncio.run(run())
