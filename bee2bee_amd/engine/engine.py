@@ -24,6 +24,7 @@ from ..models.weights import ModelWeights
 from ..utils import new_id, report_engine_throughput
 from .graphs import DecodeGraphs, decode_slot_mapping
 from .kv import PagedKV
+from . import kv as kv_mod
 from .runner import Runner
 from .sampler import SamplingParams, sample
 
@@ -166,7 +167,7 @@ class InferenceEngine:
         else:
             self.weights.random_init(seed=seed)
 
-        blocks_per_seq = -(-self.max_seq_len // 32)
+        blocks_per_seq = -(-self.max_seq_len // kv_mod.BLOCK_SIZE)
         n_blocks = max_batch * blocks_per_seq + kv_margin_blocks
         # "fp8": OCP e4m3 KV pool — half the decode-attention bytes; the
         # chunked-prefill and speculative paths read the cache as bf16 and

@@ -22,6 +22,7 @@ import torch.distributed as dist
 
 from ..engine.graphs import decode_slot_mapping
 from ..engine.kv import PagedKV
+from ..engine import kv as kv_mod
 from ..engine.runner import Runner
 from ..engine.sampler import SamplingParams, sample
 from ..models.spec import ModelSpec, resolve_spec
@@ -75,7 +76,7 @@ class MoEEngine:
             self.weights.load_hf(model_path, expert_range=expert_range)
         else:
             self.weights.random_init(seed=seed, expert_range=expert_range)
-        blocks_per_seq = -(-self.max_seq_len // 32)
+        blocks_per_seq = -(-self.max_seq_len // kv_mod.BLOCK_SIZE)
         self.kv = PagedKV(
             self.spec, self.device, dtype,
             n_blocks=max_batch * blocks_per_seq + 4,

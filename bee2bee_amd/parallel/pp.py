@@ -25,6 +25,7 @@ import torch.distributed as dist
 
 from ..engine.graphs import decode_slot_mapping
 from ..engine.kv import PagedKV
+from ..engine import kv as kv_mod
 from ..engine.runner import Runner
 from ..engine.sampler import SamplingParams, sample
 from ..models.spec import ModelSpec, resolve_spec
@@ -78,7 +79,7 @@ class PipelineEngine:
             self.weights.load_hf(model_path, layer_range=self.layer_range)
         else:
             self.weights.random_init(seed=seed, layer_range=self.layer_range)
-        blocks_per_seq = -(-self.max_seq_len // 32)
+        blocks_per_seq = -(-self.max_seq_len // kv_mod.BLOCK_SIZE)
         self.kv = PagedKV(
             self.spec, self.device, dtype,
             n_blocks=max_batch * blocks_per_seq + 4,

@@ -33,6 +33,7 @@ import torch.distributed as dist
 
 from ..engine.graphs import decode_slot_mapping
 from ..engine.kv import PagedKV
+from ..engine import kv as kv_mod
 from ..engine.runner import Runner
 from ..engine.sampler import SamplingParams, sample
 from ..models.spec import ModelSpec, resolve_spec
@@ -141,7 +142,7 @@ class TPEngine:
         self.lspec = self.weights.spec
         del full  # transient: only the shard stays resident
 
-        blocks_per_seq = -(-self.max_seq_len // 32)
+        blocks_per_seq = -(-self.max_seq_len // kv_mod.BLOCK_SIZE)
         self.kv = PagedKV(
             self.lspec, self.device, dtype,
             n_blocks=max_batch * blocks_per_seq + 4,
