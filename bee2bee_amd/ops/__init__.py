@@ -139,6 +139,25 @@ def attn_decode(
     return reference.attn_decode(q, k_cache, v_cache, block_table, seq_lens, scale)
 
 
+def attn_decode_lse(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_table: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+):
+    """attn_decode that ALSO returns the per-(seq, q-head) flash merge
+    state (m, l) — what context-parallel ranks exchange (parallel/cp.py)."""
+    if _use_hip(q):
+        out, ml = require_hip().attn_decode_lse(
+            q, k_cache, v_cache, block_table, seq_lens, scale
+        )
+        return out, ml
+    return reference.attn_decode_lse(
+        q, k_cache, v_cache, block_table, seq_lens, scale)
+
+
 def attn_prefill(
     q: torch.Tensor,
     k: torch.Tensor,

@@ -438,6 +438,11 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
     float* __restrict__ part_o,                  // [B, nkv, C, G, hd]
     const float* __restrict__ part_ml,           // [B, nkv, C, G, 2]
     unsigned short* __restrict__ out,            // [B, nq, hd] (FOLD only)
+    float* __restrict__ out_ml,                  // optional [B, nq, 2]:
+                                                 // final (m, l) per q head —
+                                                 // the cross-rank merge
+                                                 // state for context
+                                                 // parallelism (parallel/cp)
     int nkv, int W, int bs, int hd, int C) {
     const int b = blockIdx.x;
     const int kvh = blockIdx.y;
@@ -583,8 +588,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
 
     if (FOLD) {
         const int d0f = lane * 2;
+        const int nq = nkv * G;
         if (d0f < hd) {
-            const int nq = nkv * G;
             for (int g = wid; g < G; g += DEC_WAVES) {
                 const float lr = runml[g * 2 + 1];
                 const float inv = (lr > 0.f) ? 1.f / lr : 0.f;
@@ -592,6 +597,13 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_pv_kernel(
                     out + ((long)b * nq + kvh * G + g) * hd;
                 orow[d0f] = f2bf(run[g * hd + d0f] * inv);
                 orow[d0f + 1] = f2bf(run[g * hd + d0f + 1] * inv);
+            }
+        }
+        if (out_ml != nullptr && lane == 0) {
+            for (int g = wid; g < G; g += DEC_WAVES) {
+                float* mlrow = out_ml + ((long)b * nq + kvh * G + g) * 2;
+                mlrow[0] = runml[g * 2];
+                mlrow[1] = runml[g * 2 + 1];
             }
         }
     }
@@ -747,6 +759,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
     const float* __restrict__ part_ml,  // [B, nkv, C, G, 2]
     const int* __restrict__ seq_lens,
     unsigned short* __restrict__ out,   // [B, nq, hd]
+    float* __restrict__ out_ml,         // optional [B, nq, 2] (see PV)
     int nkv, int hd, int C) {
     const int b = blockIdx.x;
     const int kvh = blockIdx.y;
@@ -768,6 +781,11 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
     float Ltot = 0.f;
     for (int c = 0; c < nc; ++c) Ltot += __expf(ml_s[c][0] - M) * ml_s[c][1];
     const float inv = (Ltot > 0.f) ? 1.f / Ltot : 0.f;
+    if (out_ml != nullptr && threadIdx.x == 0) {
+        float* mlrow = out_ml + ((long)b * nq + kvh * G + g) * 2;
+        mlrow[0] = M;
+        mlrow[1] = Ltot;
+    }
 
     for (int d = threadIdx.x; d < hd; d += DEC_BLOCK) {
         float acc = 0.f;
@@ -781,7 +799,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_decode_combine_kernel(
 extern "C" void launch_attn_decode(
     const unsigned short* q, const void* k_cache, const void* v_cache,
     const int* block_table, const int* seq_lens, unsigned short* p_buf,
-    float* part_o, float* part_ml, unsigned short* out, int B, int nkv,
+    float* part_o, float* part_ml, unsigned short* out, float* out_ml,
+    int B, int nkv,
     int G, int W, int bs, int hd, int C, long q_stride, float scale,
     int fp8, hipStream_t stream) {
     // enough blocks to fill the chip, but chunks loop within a block so the
@@ -834,20 +853,20 @@ extern "C" void launch_attn_decode(
                                bs, C);                                         \
             hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,          \
                                dim3(DEC_BLOCK), 0, stream, part_o, part_ml,    \
-                               seq_lens, out, nkv, hd, C);                     \
+                               seq_lens, out, out_ml, nkv, hd, C);             \
         } else if (Z == 1) {                                                   \
             hipLaunchKernelGGL((attn_pv_kernel<GG, true, KVT>), grid,          \
                                dim3(DEC_BLOCK), smem_pv, stream, vc,           \
                                block_table, seq_lens, p_buf, part_o, part_ml,  \
-                               out, nkv, W, bs, hd, C);                        \
+                               out, out_ml, nkv, W, bs, hd, C);                \
         } else {                                                               \
             hipLaunchKernelGGL((attn_pv_kernel<GG, false, KVT>), grid,         \
                                dim3(DEC_BLOCK), smem_pv, stream, vc,           \
                                block_table, seq_lens, p_buf, part_o,           \
-                               part_ml, out, nkv, W, bs, hd, C);               \
+                               part_ml, out, nullptr, nkv, W, bs, hd, C);      \
             hipLaunchKernelGGL(attn_decode_combine_kernel<GG>, cgrid,          \
                                dim3(DEC_BLOCK), 0, stream, part_o, part_ml,    \
-                               seq_lens, out, nkv, hd, C);                     \
+                               seq_lens, out, out_ml, nkv, hd, C);             \
         }                                                                      \
     } while (0)
 #define LAUNCH(GG)                                                             \

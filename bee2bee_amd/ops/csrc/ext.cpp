@@ -27,8 +27,8 @@ void launch_swiglu(const unsigned short*, unsigned short*, long, long,
                    hipStream_t);
 void launch_attn_decode(const unsigned short*, const void*, const void*,
                         const int*, const int*, unsigned short*, float*,
-                        float*, unsigned short*, int, int, int, int, int,
-                        int, int, long, float, int, hipStream_t);
+                        float*, unsigned short*, float*, int, int, int, int,
+                        int, int, int, long, float, int, hipStream_t);
 void launch_attn_prefill(const unsigned short*, const unsigned short*,
                          const unsigned short*, const int*, unsigned short*,
                          int, int, int, int, long, long, long, int, float,
@@ -160,9 +160,11 @@ Tensor swiglu(const Tensor& gu) {
     return out;
 }
 
-Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
-                   const Tensor& v_cache, const Tensor& block_table,
-                   const Tensor& seq_lens, double scale) {
+std::vector<Tensor> attn_decode_impl(const Tensor& q, const Tensor& k_cache,
+                                     const Tensor& v_cache,
+                                     const Tensor& block_table,
+                                     const Tensor& seq_lens, double scale,
+                                     bool want_lse) {
     check_bf16(q, "q");
     const bool fp8 = k_cache.scalar_type() == torch::kUInt8;
     if (!fp8) check_bf16(k_cache, "k_cache");
@@ -189,13 +191,37 @@ Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
     Tensor p_buf = torch::empty({B, nkv, C, kDecChunk, G}, q.options());
     Tensor part_o = torch::empty({B, nkv, C, G, hd}, fopt);
     Tensor part_ml = torch::empty({B, nkv, C, G, 2}, fopt);
+    Tensor out_ml;
+    float* out_ml_p = nullptr;
+    if (want_lse) {
+        out_ml = torch::empty({B, nq, 2}, fopt);
+        out_ml_p = out_ml.data_ptr<float>();
+    }
     launch_attn_decode(bf16p(q), k_cache.data_ptr(), v_cache.data_ptr(),
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                        bf16p_mut(p_buf), part_o.data_ptr<float>(),
                        part_ml.data_ptr<float>(),
-                       bf16p_mut(out), B, nkv, G, W, bs, hd, C, q.stride(0),
-                       (float)scale, fp8 ? 1 : 0, stream());
-    return out;
+                       bf16p_mut(out), out_ml_p, B, nkv, G, W, bs, hd, C,
+                       q.stride(0), (float)scale, fp8 ? 1 : 0, stream());
+    if (want_lse) return {out, out_ml};
+    return {out};
+}
+
+Tensor attn_decode(const Tensor& q, const Tensor& k_cache,
+                   const Tensor& v_cache, const Tensor& block_table,
+                   const Tensor& seq_lens, double scale) {
+    return attn_decode_impl(q, k_cache, v_cache, block_table, seq_lens,
+                            scale, false)[0];
+}
+
+// out + per-(seq, q-head) (m, l): the flash merge state that context
+// parallelism exchanges across ranks (parallel/cp.py)
+std::vector<Tensor> attn_decode_lse(const Tensor& q, const Tensor& k_cache,
+                                    const Tensor& v_cache,
+                                    const Tensor& block_table,
+                                    const Tensor& seq_lens, double scale) {
+    return attn_decode_impl(q, k_cache, v_cache, block_table, seq_lens,
+                            scale, true);
 }
 
 Tensor attn_prefill(const Tensor& q, const Tensor& k, const Tensor& v,
@@ -327,6 +353,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("kv_cache_store", &kv_cache_store, "paged KV scatter");
     m.def("swiglu", &swiglu, "silu(g) * u");
     m.def("attn_decode", &attn_decode, "paged GQA decode attention");
+    m.def("attn_decode_lse", &attn_decode_lse,
+          "paged GQA decode attention + per-head (m, l) merge state");
     m.def("attn_prefill", &attn_prefill, "varlen causal flash prefill");
     m.def("attn_prefill_paged", &attn_prefill_paged,
           "chunked prefill vs paged history");

@@ -101,8 +101,12 @@ def attn_decode(
     block_table: torch.Tensor,
     seq_lens: torch.Tensor,
     scale: float,
+    out_ml: torch.Tensor = None,
 ) -> torch.Tensor:
-    """Paged single-token attention, GQA. q [B, nq, hd] -> out [B, nq, hd]."""
+    """Paged single-token attention, GQA. q [B, nq, hd] -> out [B, nq, hd].
+    When out_ml [B, nq, 2] is given, also writes the flash merge state
+    (max scaled score m, sum-exp l) per query head — the cross-rank
+    exchange for context parallelism."""
     B, nq, hd = q.shape
     _nb, n_kv, block_size, _ = k_cache.shape
     group = nq // n_kv
@@ -120,6 +124,11 @@ def attn_decode(
         probs = torch.softmax(scores, dim=-1)
         ob = torch.einsum("kgl,kld->kgd", probs, vals)
         out[b] = ob.reshape(nq, hd).to(q.dtype)
+        if out_ml is not None:
+            m = scores.amax(dim=-1)                         # [n_kv, group]
+            l = torch.exp(scores - m.unsqueeze(-1)).sum(-1)  # [n_kv, group]
+            out_ml[b, :, 0] = m.reshape(nq)
+            out_ml[b, :, 1] = l.reshape(nq)
     return out
 
 
@@ -229,3 +238,19 @@ def grouped_gemm(
         if hi > lo:
             out[lo:hi] = (x[lo:hi].float() @ w[e].float().t()).to(x.dtype)
     return out
+
+
+def attn_decode_lse(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_table: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+):
+    """attn_decode + per-head (m, l) merge state (see parallel/cp.py)."""
+    out_ml = torch.empty(q.shape[0], q.shape[1], 2, dtype=torch.float32,
+                         device=q.device)
+    out = attn_decode(q, k_cache, v_cache, block_table, seq_lens, scale,
+                      out_ml=out_ml)
+    return out, out_ml
