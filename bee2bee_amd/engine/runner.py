@@ -46,6 +46,7 @@ class Runner:
         # head/intermediate shard and the two row-parallel projections
         # all-reduce their partial outputs over RCCL
         self.tp_group = None
+        self.decode_attn_fn = None  # CP hook (parallel/cp.py)
         cos, sin = ops.rope_tables(
             spec.max_seq_len, spec.head_dim, spec.rope_theta, device
         )
@@ -319,6 +320,12 @@ class Runner:
         later pipeline stages); returns final hidden [B, H]."""
 
         def attn(layer_idx, q, k, v, k_cache, v_cache):
+            if self.decode_attn_fn is not None:
+                # injected attention (context parallelism: parallel/cp.py
+                # merges per-rank partials over the local KV shard)
+                return self.decode_attn_fn(
+                    q, k_cache, v_cache, block_table, seq_lens, self.scale
+                )
             return ops.attn_decode(
                 q, k_cache, v_cache, block_table, seq_lens, self.scale
             )

@@ -104,3 +104,166 @@ def cp_attn_decode(
         [g[..., :hd] for g in gathered],
         [g[..., hd:] for g in gathered],
     ).to(q.dtype)
+
+
+class CPEngine:
+    """Lockstep context-parallel engine: every rank holds the FULL weights
+    and computes the SAME batch; the KV history is page-sharded round-robin
+    across ranks (global page p lives on rank p % world), so the servable
+    context per sequence scales with world x HBM while activations stay
+    replicated. Prefill compute is replicated (the dense flash prefill
+    needs no cache) and only the owned pages are written; decode runs
+    cp_attn_decode per layer via the Runner's decode_attn_fn hook.
+
+    Outputs are identical on every rank (greedy or shared-seed sampling),
+    mirroring TPEngine's contract. CPU-tested on gloo world 2 against the
+    single-process engine (tests/test_cp_cpu.py)."""
+
+    def __init__(
+        self,
+        model,
+        device: Optional[str] = None,
+        dtype: Optional[torch.dtype] = None,
+        model_path: Optional[str] = None,
+        max_batch: int = 8,
+        max_seq_len: int = 4096,
+        seed: int = 0,
+        group: Optional[dist.ProcessGroup] = None,
+    ) -> None:
+        import dataclasses as _dc
+
+        from ..engine import kv as kv_mod
+        from ..engine.kv import PagedKV
+        from ..engine.runner import Runner
+        from ..models.spec import ModelSpec, resolve_spec
+        from ..models.weights import ModelWeights
+
+        assert dist.is_initialized(), "init the process group first"
+        self.group = group
+        self.rank = dist.get_rank(group)
+        self.world = dist.get_world_size(group)
+        spec = model if isinstance(model, ModelSpec) else resolve_spec(
+            model, model_path)
+        self.spec = spec
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        if dtype is None:
+            dtype = (torch.bfloat16 if self.device.type == "cuda"
+                     else torch.float32)
+        self.dtype = dtype
+        self.page = kv_mod.BLOCK_SIZE
+        self.max_seq_len = min(max_seq_len, spec.max_seq_len)
+
+        self.weights = ModelWeights(spec, self.device, dtype)
+        if model_path:
+            self.weights.load_hf(model_path)
+        else:
+            self.weights.random_init(seed=seed)
+
+        # each rank stores ~1/world of every sequence's pages (+1 slack)
+        pages_total = -(-self.max_seq_len // self.page)
+        local_pages = pages_total // self.world + 1
+        self.kv = PagedKV(spec, self.device, dtype,
+                          n_blocks=max_batch * local_pages + 2)
+        self.runner = Runner(spec, self.weights, self.kv, self.device, dtype)
+        self.runner.decode_attn_fn = self._cp_attn
+        self._seqs: List[int] = []
+        self._lens: List[int] = []          # GLOBAL lengths
+        self._scratch = -7
+        self.kv.new_seq(self._scratch)
+        self.kv.extend_seq(self._scratch, 1)
+
+    # page p of a sequence lives on rank p % world; its LOCAL logical
+    # position for PagedKV is (p // world) * page + offset
+    def _owned(self, pos: int) -> bool:
+        return (pos // self.page) % self.world == self.rank
+
+    def _local_pos(self, pos: int) -> int:
+        return (pos // self.page) // self.world * self.page + pos % self.page
+
+    def _local_len(self, L: int) -> int:
+        """Tokens of a length-L history stored on this rank."""
+        n = 0
+        for p in range(0, -(-L // self.page)):
+            if p % self.world == self.rank:
+                n += min(self.page, L - p * self.page)
+        return n
+
+    def _slots(self, sid: int, positions) -> List[int]:
+        """Slot per global position: owned -> local slot, else scratch."""
+        scratch = self.kv.slot_mapping(self._scratch, [0])[0]
+        out = []
+        for pos in positions:
+            if self._owned(pos):
+                out.append(self.kv.slot_mapping(
+                    sid, [self._local_pos(pos)])[0])
+            else:
+                out.append(scratch)
+        return out
+
+    def _cp_attn(self, q, k_cache, v_cache, block_table, seq_lens, scale):
+        return cp_attn_decode(q, k_cache, v_cache, block_table, seq_lens,
+                              scale, group=self.group)
+
+    def _local_lens_t(self) -> torch.Tensor:
+        return torch.tensor([self._local_len(L) for L in self._lens],
+                            dtype=torch.int32, device=self.device)
+
+    @torch.no_grad()
+    def prefill(self, prompts, sampling=None) -> torch.Tensor:
+        from ..engine.sampler import SamplingParams, sample
+
+        sampling = sampling or SamplingParams(greedy=True)
+        dev = self.device
+        self._seqs = list(range(len(prompts)))
+        self._lens = [len(p) for p in prompts]
+        ids_list, pos_list, slot_list, cu = [], [], [], [0]
+        for sid, p in zip(self._seqs, prompts):
+            self.kv.new_seq(sid)
+            self.kv.extend_seq(sid, self._local_len(len(p)))
+            ids_list.extend(p)
+            pos_list.extend(range(len(p)))
+            slot_list.extend(self._slots(sid, range(len(p))))
+            cu.append(cu[-1] + len(p))
+        hidden = self.runner.forward_prefill(
+            torch.tensor(ids_list, dtype=torch.int64, device=dev),
+            torch.tensor(pos_list, dtype=torch.int32, device=dev),
+            torch.tensor(slot_list, dtype=torch.int32, device=dev),
+            torch.tensor(cu, dtype=torch.int32, device=dev),
+            max(self._lens),
+        )
+        last = torch.tensor([c - 1 for c in cu[1:]], dtype=torch.int64,
+                            device=dev)
+        return sample(self.runner.lm_head(hidden[last]), sampling).cpu()
+
+    @torch.no_grad()
+    def decode_step(self, ids: torch.Tensor, sampling=None) -> torch.Tensor:
+        from ..engine.graphs import decode_slot_mapping  # noqa: F401
+        from ..engine.sampler import SamplingParams, sample
+
+        sampling = sampling or SamplingParams(greedy=True)
+        dev = self.device
+        positions = torch.tensor(self._lens, dtype=torch.int32, device=dev)
+        slot_list = []
+        for i, sid in enumerate(self._seqs):
+            pos = self._lens[i]
+            self.kv.extend_seq(sid, self._local_len(pos + 1))
+            slot_list.append(self._slots(sid, [pos])[0])
+            self._lens[i] = pos + 1
+        lens_t = self._local_lens_t()
+        bt = self.kv.block_table(self._seqs)
+        slots = torch.tensor(slot_list, dtype=torch.int32, device=dev)
+        hidden = self.runner.forward_decode(
+            ids.to(dev), positions, slots, bt, lens_t)
+        return sample(self.runner.lm_head(hidden), sampling).cpu()
+
+    @torch.no_grad()
+    def generate(self, prompts, max_new_tokens: int, sampling=None):
+        ids = self.prefill(prompts, sampling)
+        outs = [[int(t)] for t in ids]
+        for _ in range(max_new_tokens - 1):
+            ids = self.decode_step(ids, sampling)
+            for o, t in zip(outs, ids):
+                o.append(int(t))
+        return outs

@@ -126,3 +126,65 @@ def test_cp2_gloo_matches_single(tmp_path):
     with open(out_path, "rb") as f:
         got = pickle.load(f)
     assert torch.allclose(got.float(), full.float(), atol=1e-5)
+
+
+def _cpengine_worker(rank, world, port, out_path):
+    import torch.distributed as dist
+
+    from bee2bee_amd.parallel.cp import CPEngine
+
+    dist.init_process_group(
+        backend="gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world,
+    )
+    try:
+        eng = CPEngine("tiny", device="cpu", max_batch=2, max_seq_len=512,
+                       seed=21)
+        g = torch.Generator().manual_seed(13)
+        prompts = [torch.randint(4, 500, (300,), generator=g).tolist(),
+                   torch.randint(4, 500, (280,), generator=g).tolist()]
+        outs = eng.generate(prompts, 5)
+        if rank == 0:
+            with open(out_path, "wb") as f:
+                pickle.dump((prompts, outs), f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_cpengine2_matches_single(tmp_path):
+    """CPEngine on gloo world 2 (each rank holding half the 256-token
+    pages) must emit the same greedy tokens as the plain single-process
+    engine on the same weights."""
+    out_path = str(tmp_path / "cpe.pkl")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_cpengine_worker, args=(r, 2, 29873, out_path))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=260)
+        assert p.exitcode == 0
+    with open(out_path, "rb") as f:
+        prompts, cp_outs = pickle.load(f)
+
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=512,
+                          seed=21)
+    try:
+        ref = []
+        for prompt in prompts:
+            req = GenerationRequest(prompt_ids=list(prompt), max_new_tokens=5,
+                                    sampling=SamplingParams(greedy=True))
+            eng.submit(req)
+            while True:
+                item = req.out_queue.get(timeout=120)
+                if not isinstance(item, int):
+                    break
+            assert req.error is None, req.error
+            ref.append(req.output_ids)
+    finally:
+        eng.shutdown()
+    assert cp_outs == ref, (cp_outs, ref)
