@@ -184,6 +184,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void attn_scores_kernel(
     for (int g = 0; g < G; ++g) s[g] = -1e30f;
     if (valid) {
         const int page = bt[key / bs];
+        BB_KASSERT(page >= 0);  // block table row must be fully mapped
         const KVT* kr =
             k_cache + (((long)page * nkv + kvh) * bs + key % bs) * hd;
 #pragma unroll

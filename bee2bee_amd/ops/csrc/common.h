@@ -85,6 +85,26 @@ __device__ __forceinline__ float wave_sum(float v) {
 
 #define DEV_INLINE __device__ __forceinline__
 
+// Debug-assert build (BEE2BEE_DEBUG_KERNELS=1 at build time -> hipcc gets
+// -DBEE2BEE_DEBUG): bounds/invariant checks compile into the kernels and
+// trap with file:line on violation. The release build compiles them away
+// entirely (no branch, no register cost). This is the kernel-side
+// sanitizer story: HIP has no compute-sanitizer equivalent on this stack,
+// so invariants are asserted at the source level instead.
+#ifdef BEE2BEE_DEBUG
+#define BB_KASSERT(cond)                                                      \
+    do {                                                                      \
+        if (!(cond)) {                                                        \
+            printf("BB_KASSERT failed %s:%d: %s (block %d,%d,%d thread %d)\n",\
+                   __FILE__, __LINE__, #cond, (int)blockIdx.x,                \
+                   (int)blockIdx.y, (int)blockIdx.z, (int)threadIdx.x);       \
+            __builtin_trap();                                                 \
+        }                                                                     \
+    } while (0)
+#else
+#define BB_KASSERT(cond) do { } while (0)
+#endif
+
 #define HIP_CHECK_KERNEL()                                                    \
     do {                                                                      \
         hipError_t e_ = hipGetLastError();                                    \

@@ -56,6 +56,7 @@ __global__ __launch_bounds__(GG_BLOCK) void grouped_gemm_kernel(
     const int n0 = blockIdx.x * GG_BN;
     const int m_lo = offs[e];
     const int m_hi = offs[e + 1];
+    BB_KASSERT(m_lo >= 0 && m_lo <= m_hi);  // offsets must be a prefix
     if (m_lo >= m_hi) return;
     const int tid = threadIdx.x;
     const int lane = tid % WAVE;

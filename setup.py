@@ -26,8 +26,10 @@ ext = CUDAExtension(
         os.path.join(CSRC, "grouped_gemm.hip"),
     ],
     extra_compile_args={
-        "cxx": ["-O3", "-std=c++17"],
-        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+        "cxx": ["-O3", "-std=c++17"]
+        + (["-DBEE2BEE_DEBUG"] if os.environ.get("BEE2BEE_DEBUG_KERNELS") == "1" else []),
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"]
+        + (["-DBEE2BEE_DEBUG", "-g"] if os.environ.get("BEE2BEE_DEBUG_KERNELS") == "1" else []),
     },
 )
 
