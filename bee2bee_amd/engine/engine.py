@@ -85,7 +85,8 @@ class GenerationRequest:
 
 
 class _Active:
-    __slots__ = ("req", "seq_id", "length", "prefilled", "spec_index")
+    __slots__ = ("req", "seq_id", "length", "prefilled", "spec_index",
+                 "pen_slot")
 
     def __init__(self, req: GenerationRequest, seq_id: int, length: int) -> None:
         self.req = req
@@ -93,6 +94,7 @@ class _Active:
         self.length = length  # tokens currently in KV cache
         self.prefilled = 0  # prompt tokens prefetched so far (chunked prefill)
         self.spec_index = None  # lazy NGramIndex (speculative proposer)
+        self.pen_slot = None    # repetition-penalty seen-mask pool row
 
 
 class InferenceEngine:
@@ -201,6 +203,15 @@ class InferenceEngine:
 
         self._gen = torch.Generator(device=self.device)
         self._gen.manual_seed(seed)
+        # repetition-penalty seen-token masks: one pooled [slots, V] bool
+        # buffer, maintained INCREMENTALLY (prompt scatter at admission, one
+        # batched scatter of the sampled tokens per step) — the r2 parity
+        # change made penalty 1.15 the wire default, and rebuilding a
+        # [B, context] history tensor on the host every step cost ms-scale
+        # GIL time on the serving path. Lazy: never allocated for
+        # penalty-1.0-only workloads (bench).
+        self._pen_pool: Optional[torch.Tensor] = None
+        self._pen_free: List[int] = []
 
         self._pending: "queue.Queue[GenerationRequest]" = queue.Queue()
         self._active: List[_Active] = []
@@ -382,6 +393,7 @@ class InferenceEngine:
                     else:
                         a.req.out_queue.put(_STREAM_END)
                     self.kv.free_seq(a.seq_id)
+                    self._pen_release(a)
                 self._active.clear()
                 self._prefilling.clear()
                 did_work = True
@@ -436,7 +448,10 @@ class InferenceEngine:
             seq_id = self._next_seq
             self._next_seq += 1
             self.kv.new_seq(seq_id)
-            admitted.append(_Active(req, seq_id, 0))
+            a_new = _Active(req, seq_id, 0)
+            if req.sampling.repetition_penalty != 1.0:
+                self._pen_assign(a_new)
+            admitted.append(a_new)
             admit_tokens += len(req.prompt_ids)
         return admitted
 
@@ -622,6 +637,7 @@ class InferenceEngine:
         done = [a for a in self._active if a.req.done_ts is not None]
         for a in done:
             self.kv.free_seq(a.seq_id)
+            self._pen_release(a)
         if done:
             self._active = [a for a in self._active if a.req.done_ts is None]
             self._dec_seqs = None  # membership changed
@@ -763,8 +779,32 @@ class InferenceEngine:
         done_acts = [a for a in acts if a.req.done_ts is not None]
         for a in done_acts:
             self.kv.free_seq(a.seq_id)
+            self._pen_release(a)
         if done_acts:
             self._active = [a for a in acts if a.req.done_ts is None]
+
+    def _pen_assign(self, a: "_Active") -> None:
+        """Give a penalized request a seen-mask row seeded with its prompt."""
+        V = self.spec.vocab_size
+        if self._pen_pool is None:
+            n = self.max_batch + 8
+            self._pen_pool = torch.zeros(n, V, dtype=torch.bool,
+                                         device=self.device)
+            self._pen_free = list(range(n))
+        if not self._pen_free:
+            return  # degrade: penalty falls back to the host rebuild below
+        slot = self._pen_free.pop()
+        a.pen_slot = slot
+        row = self._pen_pool[slot]
+        row.zero_()
+        ids = torch.tensor(a.req.prompt_ids, dtype=torch.int64,
+                           device=self.device)
+        row[ids] = True
+
+    def _pen_release(self, a: "_Active") -> None:
+        if a.pen_slot is not None:
+            self._pen_free.append(a.pen_slot)
+            a.pen_slot = None
 
     def _sample_and_emit(self, acts: List[_Active], logits: torch.Tensor) -> None:
         # group rows by sampling params so each group is one sample() call
@@ -784,22 +824,47 @@ class InferenceEngine:
             gen = self._gen if self.device.type == "cuda" else None
             grp_logits = logits[idx]
             if sp.repetition_penalty != 1.0:
-                from .sampler import apply_repetition_penalty
+                slots = [acts[r].pen_slot for r in rows]
+                if self._pen_pool is not None and all(
+                        sl is not None for sl in slots):
+                    seen = self._pen_pool[
+                        torch.tensor(slots, dtype=torch.int64,
+                                     device=logits.device)]
+                    p = sp.repetition_penalty
+                    gl = grp_logits.float()
+                    grp_logits = torch.where(
+                        seen, torch.where(gl > 0, gl / p, gl * p), gl)
+                else:
+                    # pool exhausted fallback: host-rebuilt history
+                    from .sampler import apply_repetition_penalty
 
-                width = max(
-                    (len(acts[r].req.output_ids) + len(acts[r].req.prompt_ids))
-                    for r in rows
-                )
-                prev = torch.full((len(rows), width), -1, dtype=torch.int64)
-                for j, r in enumerate(rows):
-                    ids = acts[r].req.prompt_ids + acts[r].req.output_ids
-                    prev[j, : len(ids)] = torch.tensor(ids, dtype=torch.int64)
-                grp_logits = apply_repetition_penalty(
-                    grp_logits.float(), prev.to(grp_logits.device),
-                    sp.repetition_penalty,
-                )
+                    width = max(
+                        (len(acts[r].req.output_ids)
+                         + len(acts[r].req.prompt_ids))
+                        for r in rows
+                    )
+                    prev = torch.full((len(rows), width), -1,
+                                      dtype=torch.int64)
+                    for j, r in enumerate(rows):
+                        ids = acts[r].req.prompt_ids + acts[r].req.output_ids
+                        prev[j, : len(ids)] = torch.tensor(
+                            ids, dtype=torch.int64)
+                    grp_logits = apply_repetition_penalty(
+                        grp_logits.float(), prev.to(grp_logits.device),
+                        sp.repetition_penalty,
+                    )
             toks = sample(grp_logits, sp, generator=gen)
             next_dev[idx] = toks
+        # mark the freshly sampled tokens seen (one batched device scatter)
+        if self._pen_pool is not None:
+            pen_idx = [i for i, a in enumerate(acts) if a.pen_slot is not None]
+            if pen_idx:
+                slot_t = torch.tensor(
+                    [acts[i].pen_slot for i in pen_idx], dtype=torch.int64,
+                    device=logits.device)
+                idx_t = torch.tensor(pen_idx, dtype=torch.int64,
+                                     device=logits.device)
+                self._pen_pool[slot_t, next_dev[idx_t]] = True
         self._last_sampled = next_dev  # feeds the next step without H2D
         next_ids = next_dev.cpu()  # the one host sync per step (emission)
         now = time.time()

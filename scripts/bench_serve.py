@@ -5,6 +5,10 @@ This measures the second half of BASELINE.json's metric ("output tokens/sec
 + p50 e2e request latency"): concurrent clients POST /generate against a
 live node serving the native engine, exactly like an external user.
 
+The server runs in its OWN process (round 2): with client and server in
+one interpreter the benchmark measured its own GIL contention — separate
+processes measure what an external user actually sees.
+
 Usage:
   python scripts/bench_serve.py --model llama3-8b --clients 16 \
       --requests 64 --max-new 64 [--device cuda:0] [--prompt-len 512]
@@ -13,13 +17,41 @@ Prints one JSON line with latency percentiles and aggregate throughput.
 import argparse
 import asyncio
 import json
+import multiprocessing
 import os
 import random
-import statistics
 import sys
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def _server_main(opts: dict) -> None:
+    """Child process: mesh node + native engine + uvicorn gateway."""
+
+    async def serve() -> None:
+        import uvicorn
+
+        from bee2bee_amd.gateway import api as gateway_api
+        from bee2bee_amd.mesh.node import MeshNode
+        from bee2bee_amd.services.native import NativeEngineService
+
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        node.api_port = opts["api_port"]
+        svc = NativeEngineService(
+            opts["model"], device=opts["device"],
+            max_batch=opts["max_batch"], max_seq_len=2048,
+        )
+        loop = asyncio.get_running_loop()
+        await loop.run_in_executor(None, svc.load_sync)
+        await node.add_service(svc)
+        gateway_api.node = node
+        config = uvicorn.Config(gateway_api.app, host="127.0.0.1",
+                                port=opts["api_port"], log_level="error")
+        await uvicorn.Server(config).serve()
+
+    asyncio.run(serve())
 
 
 async def run() -> None:
@@ -38,32 +70,33 @@ async def run() -> None:
     args = ap.parse_args()
 
     import aiohttp
-    import uvicorn
 
-    from bee2bee_amd.gateway import api as gateway_api
-    from bee2bee_amd.mesh.node import MeshNode
-    from bee2bee_amd.services.native import NativeEngineService
-
-    node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
-    await node.start()
-    node.api_port = args.api_port
-    svc = NativeEngineService(
-        args.model, device=args.device, max_batch=args.max_batch,
-        max_seq_len=2048,
-    )
-    loop = asyncio.get_running_loop()
     t0 = time.time()
-    await loop.run_in_executor(None, svc.load_sync)
-    load_s = time.time() - t0
-    await node.add_service(svc)
-
-    gateway_api.node = node
-    config = uvicorn.Config(
-        gateway_api.app, host="127.0.0.1", port=args.api_port, log_level="error"
+    proc = multiprocessing.get_context("spawn").Process(
+        target=_server_main,
+        args=({"model": args.model, "device": args.device,
+               "max_batch": args.max_batch, "api_port": args.api_port},),
+        daemon=True,
     )
-    server = uvicorn.Server(config)
-    asyncio.get_running_loop().create_task(server.serve())
-    await asyncio.sleep(1.0)
+    proc.start()
+
+    base = f"http://127.0.0.1:{args.api_port}"
+    device = "?"
+    async with aiohttp.ClientSession() as session:
+        for _ in range(600):  # engine build + server bind
+            try:
+                async with session.get(f"{base}/", timeout=aiohttp.ClientTimeout(total=2)) as r:
+                    body = await r.json()
+                    if body.get("services"):
+                        device = (body.get("engine") or {}).get("device", "?")
+                        break
+            except Exception:
+                pass
+            await asyncio.sleep(0.5)
+        else:
+            proc.terminate()
+            raise RuntimeError("server did not become ready")
+    load_s = time.time() - t0
 
     rng = random.Random(0)
     words = ["alpha", "beta", "gamma", "delta", "mesh", "gpu", "tensor", "ring"]
@@ -83,7 +116,7 @@ async def run() -> None:
     ttft: list = []
     tokens_total = 0
     sem = asyncio.Semaphore(args.clients)
-    url = f"http://127.0.0.1:{args.api_port}/generate"
+    url = f"{base}/generate"
 
     async def one_request(session) -> None:
         nonlocal tokens_total
@@ -120,14 +153,19 @@ async def run() -> None:
                 ttft.append(first - t_start)
             tokens_total += args.max_new
 
-    t_bench = time.perf_counter()
-    async with aiohttp.ClientSession() as session:
-        # small warmup
-        await one_request(session)
-        lat.clear(); ttft.clear()
-        tokens = tokens_total = 0
-        await asyncio.gather(*(one_request(session) for _ in range(args.requests)))
-    wall = time.perf_counter() - t_bench
+    try:
+        t_bench = time.perf_counter()
+        async with aiohttp.ClientSession() as session:
+            # small warmup
+            await one_request(session)
+            lat.clear(); ttft.clear()
+            tokens_total = 0
+            await asyncio.gather(
+                *(one_request(session) for _ in range(args.requests)))
+        wall = time.perf_counter() - t_bench
+    finally:
+        proc.terminate()
+        proc.join(timeout=10)
 
     lat.sort()
     def pct(xs, p):
@@ -136,7 +174,7 @@ async def run() -> None:
     result = {
         "metric": "e2e request latency + tokens/sec via /generate (streaming)",
         "model": args.model,
-        "device": str(svc.engine.device),
+        "device": device,
         "requests": args.requests,
         "concurrency": args.clients,
         "max_new_tokens": args.max_new,
@@ -150,10 +188,6 @@ async def run() -> None:
                 + " synthetic prompts, random-init weights",
     }
     print(json.dumps(result), flush=True)
-    svc.engine.shutdown()
-    await node.stop()
-    server.should_exit = True
-    await asyncio.sleep(0.2)
 
 
 if __name__ == "__main__":

@@ -283,3 +283,36 @@ def test_bench_prefill_fills_every_prompt():
         assert eng._prefilling == []
     finally:
         eng.shutdown()
+
+
+def test_repetition_penalty_seen_mask_equals_history_rebuild():
+    """The pooled seen-mask penalty (incremental) must emit exactly the
+    tokens of the host-rebuilt-history fallback path."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    def run(exhaust_pool):
+        eng = InferenceEngine("tiny", device="cpu", max_batch=2,
+                              max_seq_len=96, seed=5)
+        try:
+            if exhaust_pool:
+                # force the fallback branch by leaving no free slots
+                eng._pen_assign  # noqa: B018 — ensure attr exists
+                eng._pen_pool = None
+                eng._pen_free = []
+                eng._pen_assign = lambda a: None
+            req = GenerationRequest(
+                prompt_ids=[7, 8, 9, 7, 8], max_new_tokens=12,
+                sampling=SamplingParams(greedy=True, repetition_penalty=1.3),
+            )
+            eng.submit(req)
+            while True:
+                item = req.out_queue.get(timeout=60)
+                if not isinstance(item, int):
+                    break
+            assert req.error is None, req.error
+            return req.output_ids
+        finally:
+            eng.shutdown()
+
+    assert run(False) == run(True)
