@@ -678,7 +678,12 @@ class InferenceEngine:
         for a in acts:
             r = a.req
             prop: List[int] = []
-            if r.sampling.greedy and r.sampling.repetition_penalty == 1.0:
+            # greedy with repetition penalty is ALSO speculatable: the
+            # verify pass applies the penalty exactly (seen-mask + the
+            # in-window proposal prefix) before each argmax
+            if r.sampling.greedy and (
+                    r.sampling.repetition_penalty == 1.0
+                    or a.pen_slot is not None):
                 room = self.max_seq_len - (a.length + 1) - 1
                 rem = r.max_new_tokens - len(r.output_ids) - 1
                 kcap = min(self.spec_k, room, rem)
@@ -720,23 +725,74 @@ class InferenceEngine:
         greedy_idx = [i for i, a in enumerate(acts)
                       if a.req.sampling.greedy
                       and a.req.sampling.repetition_penalty == 1.0]
-        nong = [a for i, a in enumerate(acts) if i not in set(greedy_idx)]
+        pen_idx = [i for i, a in enumerate(acts)
+                   if a.req.sampling.greedy
+                   and a.req.sampling.repetition_penalty != 1.0
+                   and a.pen_slot is not None]
+        handled = set(greedy_idx) | set(pen_idx)
+        nong = [a for i, a in enumerate(acts) if i not in handled]
         if greedy_idx:
             argmax = logits.argmax(dim=-1).cpu()
+        pen_argmax = {}
+        if pen_idx:
+            # penalized argmax under the BASE seen mask for every verify
+            # row in one batch + one sync; in-window corrections happen on
+            # the host walk below
+            row_ids, row_act, row_p = [], [], []
+            for i in pen_idx:
+                q = cu[i + 1] - cu[i]
+                row_ids.extend(range(cu[i], cu[i] + q))
+                row_act.extend([i] * q)
+                row_p.extend([acts[i].req.sampling.repetition_penalty] * q)
+            rid = torch.tensor(row_ids, dtype=torch.int64, device=dev)
+            slot_t = torch.tensor(
+                [acts[i].pen_slot for i in row_act], dtype=torch.int64,
+                device=dev)
+            p_t = torch.tensor(row_p, dtype=torch.float32,
+                               device=dev).unsqueeze(1)
+            gl = logits[rid].float()
+            seen = self._pen_pool[slot_t]
+            pen_rows = torch.where(
+                seen, torch.where(gl > 0, gl / p_t, gl * p_t), gl)
+            pa = pen_rows.argmax(dim=-1).cpu()
+            for k, row in enumerate(row_ids):
+                pen_argmax[row] = int(pa[k])
         now = time.time()
         n_emitted = 0
-        for i in greedy_idx:
+        pen_new_slots: List[int] = []
+        pen_new_toks: List[int] = []
+        for i in greedy_idx + pen_idx:
             a, r, prop = acts[i], acts[i].req, props[i]
             self.spec_stats["proposed"] += len(prop)
             emitted: List[int] = []
             j = 0
-            while True:
-                tok = int(argmax[cu[i] + j])
-                emitted.append(tok)
-                if j < len(prop) and tok == prop[j] and tok not in r.stop_token_ids:
-                    j += 1
-                    continue
-                break
+            if a.req.sampling.repetition_penalty == 1.0:
+                while True:
+                    tok = int(argmax[cu[i] + j])
+                    emitted.append(tok)
+                    if (j < len(prop) and tok == prop[j]
+                            and tok not in r.stop_token_ids):
+                        j += 1
+                        continue
+                    break
+            else:
+                # exact penalized walk: row j's base-mask argmax is exact
+                # unless it collides with a token proposed EARLIER in this
+                # window (whose in-window penalty could demote it) — at a
+                # collision, truncate acceptance (still exact: every
+                # emitted token equals what plain penalized decode emits)
+                window: set = set()
+                while True:
+                    tok = pen_argmax[cu[i] + j]
+                    if j > 0 and tok in window:
+                        break  # ambiguous under in-window penalty: stop
+                    emitted.append(tok)
+                    if (j < len(prop) and tok == prop[j]
+                            and tok not in r.stop_token_ids):
+                        window.add(tok)
+                        j += 1
+                        continue
+                    break
             self.spec_stats["accepted"] += j
             # rewind the KV length past rejected proposals (their stored
             # keys get overwritten; attention never reads past seq_len)
@@ -748,6 +804,9 @@ class InferenceEngine:
                 if r.first_token_ts is None:
                     r.first_token_ts = now
                 r.output_ids.append(tok)
+                if a.pen_slot is not None:
+                    pen_new_slots.append(a.pen_slot)
+                    pen_new_toks.append(tok)
                 n_emitted += 1
                 done = (r.cancelled or tok in r.stop_token_ids
                         or len(r.output_ids) >= r.max_new_tokens)
@@ -764,11 +823,17 @@ class InferenceEngine:
                         r.out_queue.put(_STREAM_END)
         if n_emitted:
             self._note_throughput(n_emitted, now)
+        # mark penalized-greedy emissions seen (one batched scatter)
+        if pen_new_slots:
+            self._pen_pool[
+                torch.tensor(pen_new_slots, dtype=torch.int64, device=dev),
+                torch.tensor(pen_new_toks, dtype=torch.int64, device=dev),
+            ] = True
         if nong:
-            # sampled / penalized requests take the plain one-token path
+            # non-greedy requests take the plain one-token path
             rows = torch.tensor(
                 [cu[i + 1] - 1 for i, a in enumerate(acts)
-                 if i not in set(greedy_idx)],
+                 if i not in handled],
                 dtype=torch.int64, device=logits.device,
             )
             for a in nong:

@@ -175,3 +175,38 @@ def test_ngram_index_detects_output_cycle():
     for _ in range(3):
         idx.extend([5, 6, 4])            # model falls into a cycle
     assert idx.propose(3) == [5, 6, 4]
+
+
+def test_spec_output_invariant_with_repetition_penalty():
+    """Penalized-greedy speculative verification must be EXACT: identical
+    tokens to the plain engine under the reference-default penalty 1.15,
+    on a self-similar prompt that forces proposals and acceptances."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    prompt = [5, 6, 7, 8, 5, 6, 7, 8, 5, 6, 7, 8, 5, 6]
+
+    def run(spec):
+        eng = InferenceEngine("tiny", device="cpu", max_batch=2,
+                              max_seq_len=128, seed=11, spec_decode=spec)
+        try:
+            req = GenerationRequest(
+                prompt_ids=list(prompt), max_new_tokens=24,
+                sampling=SamplingParams(greedy=True,
+                                        repetition_penalty=1.15))
+            eng.submit(req)
+            while True:
+                item = req.out_queue.get(timeout=120)
+                if not isinstance(item, int):
+                    break
+            assert req.error is None, req.error
+            stats = dict(eng.spec_stats)
+            return req.output_ids, stats
+        finally:
+            eng.shutdown()
+
+    plain, _ = run(False)
+    spec, stats = run(True)
+    assert spec == plain, (spec, plain)
+    # the penalized-greedy path must actually have speculated
+    assert stats["proposed"] > 0
