@@ -4,6 +4,9 @@ across N GPUs over RCCL/xGMI.
   --mode pp  (default): layer shards, point-to-point hidden-state hops
   --mode tp:            Megatron head/intermediate shards, two all-reduces
                         per layer (lower latency per token at small batch)
+  --mode cp:            context parallelism — KV pages sharded across
+                        ranks, one (m,l,o) all_gather per layer; servable
+                        context scales with world x 288 GB
 
 Launch (one rank per GPU):
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
@@ -31,7 +34,7 @@ def main() -> None:
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--steps", type=int, default=16)
     ap.add_argument("--warmup", type=int, default=4)
-    ap.add_argument("--mode", default="pp", choices=["pp", "tp"])
+    ap.add_argument("--mode", default="pp", choices=["pp", "tp", "cp"])
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -47,6 +50,8 @@ def main() -> None:
 
     if args.mode == "tp":
         from bee2bee_amd.parallel.tp import TPEngine as Engine
+    elif args.mode == "cp":
+        from bee2bee_amd.parallel.cp import CPEngine as Engine
     else:
         from bee2bee_amd.parallel.pp import PipelineEngine as Engine
 
@@ -84,7 +89,7 @@ def main() -> None:
     if rank == 0:
         print(json.dumps({
             "metric": f"output tokens/sec ({eng.spec.name} bf16 greedy decode, "
-                      f"{'tensor' if args.mode == 'tp' else 'pipeline'}"
+                      f"{dict(tp='tensor', cp='context', pp='pipeline')[args.mode]}"
                       f"-parallel {args.mode}{world})",
             "value": round(args.batch * args.steps / elapsed, 1),
             "unit": "tokens/s",

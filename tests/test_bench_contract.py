@@ -60,11 +60,11 @@ def test_bench_multi_process_contract():
 
 
 @pytest.mark.timeout(300)
-@pytest.mark.parametrize("mode,model", [("pp", "tiny"), ("tp", "tiny")])
+@pytest.mark.parametrize("mode,model", [("pp", "tiny"), ("tp", "tiny"), ("cp", "tiny")])
 def test_bench_pp_contract(mode, model):
     """scripts/bench_pp.py (the 8-GPU model-parallel bench) launches under
     torch.distributed.run and emits the JSON contract (gloo dry run)."""
-    port = "29771" if mode == "pp" else "29772"
+    port = {"pp": "29771", "tp": "29772", "cp": "29773"}[mode]
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
