@@ -113,6 +113,14 @@ def attn_decode(
     out = torch.empty_like(q)
     for b in range(B):
         L = int(seq_lens[b])
+        if L == 0:
+            # empty context (a CP rank owning no pages of this sequence):
+            # zero output with (m, l) = (-inf, 0) merge state
+            out[b] = 0
+            if out_ml is not None:
+                out_ml[b, :, 0] = -1e30
+                out_ml[b, :, 1] = 0.0
+            continue
         nblk = (L + block_size - 1) // block_size
         blocks = block_table[b, :nblk].long()
         keys = _kv_deq(k_cache[blocks])  # [nblk, n_kv, bs, hd]
