@@ -27,7 +27,28 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def _server_main(opts: dict) -> None:
-    """Child process: mesh node + native engine + uvicorn gateway."""
+    """Child process: mesh node + native engine + uvicorn gateway.
+    BEE2BEE_PROFILE_SERVER=1 cProfiles this process and dumps the top
+    cumulative entries to gpurun_out/serve_profile.txt on SIGTERM."""
+    prof = None
+    if os.environ.get("BEE2BEE_PROFILE_SERVER") == "1":
+        import cProfile
+        import pstats
+        import signal
+
+        prof = cProfile.Profile()
+        prof.enable()
+
+        def _dump(_sig, _frm):
+            prof.disable()
+            os.makedirs("gpurun_out", exist_ok=True)
+            with open("gpurun_out/serve_profile.txt", "w") as f:
+                st = pstats.Stats(prof, stream=f)
+                st.sort_stats("cumulative").print_stats(50)
+                st.sort_stats("tottime").print_stats(50)
+            os._exit(0)
+
+        signal.signal(signal.SIGTERM, _dump)
 
     async def serve() -> None:
         import uvicorn
