@@ -213,6 +213,9 @@ class InferenceEngine:
         self._pen_pool: Optional[torch.Tensor] = None
         self._pen_free: List[int] = []
 
+        self._busy_s = 0.0       # engine-thread time inside working steps
+        self._busy_steps = 0
+        self._born = time.time()
         self._pending: "queue.Queue[GenerationRequest]" = queue.Queue()
         self._active: List[_Active] = []
         # admitted requests whose prompt is longer than one prefill budget:
@@ -379,6 +382,7 @@ class InferenceEngine:
 
     def _loop(self) -> None:
         while not self._stop:
+            t0 = time.perf_counter()
             try:
                 did_work = self._step()
             except Exception as e:  # engine errors fail all active requests
@@ -397,6 +401,9 @@ class InferenceEngine:
                 self._active.clear()
                 self._prefilling.clear()
                 did_work = True
+            if did_work:
+                self._busy_s += time.perf_counter() - t0
+                self._busy_steps += 1
             if not did_work:
                 self._wake.wait(timeout=0.05)
                 self._wake.clear()
@@ -973,6 +980,10 @@ class InferenceEngine:
             "tokens_total": self.total_tokens,
             "tokens_per_sec_10s": round(tps, 1),
             "decode_graphs": self.graphs is not None,
+            "engine_busy_s": round(self._busy_s, 2),
+            "engine_steps": self._busy_steps,
+            "engine_ms_per_step": round(
+                1e3 * self._busy_s / max(self._busy_steps, 1), 2),
         }
         if self.spec_decode:
             out["spec_decode"] = dict(self.spec_stats)

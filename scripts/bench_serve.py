@@ -184,6 +184,13 @@ async def run() -> None:
             await asyncio.gather(
                 *(one_request(session) for _ in range(args.requests)))
         wall = time.perf_counter() - t_bench
+        engine_stats = {}
+        try:
+            async with aiohttp.ClientSession() as session:
+                async with session.get(f"{base}/", timeout=aiohttp.ClientTimeout(total=3)) as r:
+                    engine_stats = (await r.json()).get("engine") or {}
+        except Exception:
+            pass
     finally:
         proc.terminate()
         proc.join(timeout=10)
@@ -207,6 +214,9 @@ async def run() -> None:
         "model_load_s": round(load_s, 1),
         "data": ("repetitive" if args.repetitive else "random")
                 + " synthetic prompts, random-init weights",
+        "engine_busy_s": engine_stats.get("engine_busy_s"),
+        "engine_steps": engine_stats.get("engine_steps"),
+        "engine_ms_per_step": engine_stats.get("engine_ms_per_step"),
     }
     print(json.dumps(result), flush=True)
 
