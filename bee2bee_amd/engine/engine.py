@@ -10,6 +10,7 @@ retirement, so the 288 GB HBM pool is shared across the whole request mix.
 from __future__ import annotations
 
 import logging
+import os
 import queue
 import threading
 import time
@@ -381,6 +382,12 @@ class InferenceEngine:
     # ---------------------------------------------------------- engine loop
 
     def _loop(self) -> None:
+        prof = None
+        if os.environ.get("BEE2BEE_PROFILE_ENGINE") == "1":
+            import cProfile
+
+            prof = cProfile.Profile()
+            prof.enable()
         while not self._stop:
             t0 = time.perf_counter()
             try:
@@ -407,6 +414,15 @@ class InferenceEngine:
             if not did_work:
                 self._wake.wait(timeout=0.05)
                 self._wake.clear()
+        if prof is not None:
+            import pstats
+
+            prof.disable()
+            os.makedirs("gpurun_out", exist_ok=True)
+            with open("gpurun_out/engine_profile.txt", "w") as f:
+                st = pstats.Stats(prof, stream=f)
+                st.sort_stats("tottime").print_stats(40)
+                st.sort_stats("cumulative").print_stats(40)
 
     def _step(self) -> bool:
         admitted = self._admit()

@@ -49,6 +49,22 @@ def _server_main(opts: dict) -> None:
             os._exit(0)
 
         signal.signal(signal.SIGTERM, _dump)
+    elif os.environ.get("BEE2BEE_PROFILE_ENGINE") == "1":
+        import signal
+
+        def _stop_engine(_sig, _frm):
+            # shut the engine thread down so its profiler dump runs
+            from bee2bee_amd.gateway import api as gateway_api
+
+            node = gateway_api.node
+            if node is not None:
+                for svc in list(getattr(node, "local_services", {}).values()):
+                    eng = getattr(svc, "engine", None)
+                    if eng is not None:
+                        eng.shutdown()
+            os._exit(0)
+
+        signal.signal(signal.SIGTERM, _stop_engine)
 
     async def serve() -> None:
         import uvicorn
