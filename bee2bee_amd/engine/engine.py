@@ -518,10 +518,10 @@ class InferenceEngine:
                 self.kv.slot_mapping(a.seq_id, range(start, start + take))
             )
             cu.append(cu[-1] + take)
-        input_ids = torch.tensor(ids_list, dtype=torch.int64, device=dev)
-        positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
-        slots = torch.tensor(slot_list, dtype=torch.int32, device=dev)
-        cu_seqlens = torch.tensor(cu, dtype=torch.int32, device=dev)
+        input_ids = torch.tensor(ids_list, dtype=torch.int64).to(dev, non_blocking=True)
+        positions = torch.tensor(pos_list, dtype=torch.int32).to(dev, non_blocking=True)
+        slots = torch.tensor(slot_list, dtype=torch.int32).to(dev, non_blocking=True)
+        cu_seqlens = torch.tensor(cu, dtype=torch.int32).to(dev, non_blocking=True)
         max_len = max(take for _, _, take in batch)
         whole = all(
             start == 0 and take == len(a.req.prompt_ids)
@@ -533,13 +533,11 @@ class InferenceEngine:
             )
         else:
             bt = self.kv.block_table([a.seq_id for a, _, _ in batch])
-            seq_lens = torch.tensor(
-                [start + take for _, start, take in batch],
+            seq_lens = torch.tensor([start + take for _, start, take in batch],
                 dtype=torch.int32, device=dev,
             )
             query_lens = torch.tensor(
-                [take for _, _, take in batch], dtype=torch.int32, device=dev
-            )
+                [take for _, _, take in batch], dtype=torch.int32).to(dev, non_blocking=True)
             hidden = self.runner.forward_prefill(
                 input_ids, positions, slots, cu_seqlens, max_len,
                 block_table=bt, seq_lens=seq_lens, query_lens=query_lens,
@@ -558,9 +556,7 @@ class InferenceEngine:
         if not done_idx:
             return
         completed = [batch[i][0] for i in done_idx]
-        last_rows = torch.tensor(
-            [cu[i + 1] - 1 for i in done_idx], dtype=torch.int64, device=dev
-        )
+        last_rows = torch.tensor([cu[i + 1] - 1 for i in done_idx], dtype=torch.int64).to(dev, non_blocking=True)
         logits = self.runner.lm_head(hidden[last_rows])
         self._sample_and_emit(completed, logits)
         self._active.extend([a for a in completed if a.req.done_ts is None])
@@ -639,11 +635,9 @@ class InferenceEngine:
                     for a in self._active
                 ]
                 st = {
-                    "ids": torch.tensor(last_ids, dtype=torch.int64, device=dev),
-                    "pos": torch.tensor([a.length for a in self._active],
-                                        dtype=torch.int32, device=dev),
-                    "lens": torch.tensor([a.length + 1 for a in self._active],
-                                         dtype=torch.int32, device=dev),
+                    "ids": torch.tensor(last_ids, dtype=torch.int64).to(dev, non_blocking=True),
+                    "pos": torch.tensor([a.length for a in self._active], dtype=torch.int32).to(dev, non_blocking=True),
+                    "lens": torch.tensor([a.length + 1 for a in self._active], dtype=torch.int32).to(dev, non_blocking=True),
                     "bt": self.kv.block_table(seqs),
                 }
                 self._eager_state = st
@@ -728,17 +722,16 @@ class InferenceEngine:
                 self.kv.slot_mapping(a.seq_id, range(a.length, a.length + len(toks)))
             )
             cu.append(cu[-1] + len(toks))
-        input_ids = torch.tensor(ids_list, dtype=torch.int64, device=dev)
-        positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
-        slots = torch.tensor(slot_list, dtype=torch.int32, device=dev)
-        cu_t = torch.tensor(cu, dtype=torch.int32, device=dev)
+        input_ids = torch.tensor(ids_list, dtype=torch.int64).to(dev, non_blocking=True)
+        positions = torch.tensor(pos_list, dtype=torch.int32).to(dev, non_blocking=True)
+        slots = torch.tensor(slot_list, dtype=torch.int32).to(dev, non_blocking=True)
+        cu_t = torch.tensor(cu, dtype=torch.int32).to(dev, non_blocking=True)
         qlens = [cu[i + 1] - cu[i] for i in range(len(acts))]
         bt = self.kv.block_table([a.seq_id for a in acts])
-        seq_lens = torch.tensor(
-            [a.length + q for a, q in zip(acts, qlens)],
+        seq_lens = torch.tensor([a.length + q for a, q in zip(acts, qlens)],
             dtype=torch.int32, device=dev,
         )
-        qlens_t = torch.tensor(qlens, dtype=torch.int32, device=dev)
+        qlens_t = torch.tensor(qlens, dtype=torch.int32).to(dev, non_blocking=True)
         hidden = self.runner.forward_prefill(
             input_ids, positions, slots, cu_t, max(qlens),
             block_table=bt, seq_lens=seq_lens, query_lens=qlens_t,
@@ -767,12 +760,9 @@ class InferenceEngine:
                 row_ids.extend(range(cu[i], cu[i] + q))
                 row_act.extend([i] * q)
                 row_p.extend([acts[i].req.sampling.repetition_penalty] * q)
-            rid = torch.tensor(row_ids, dtype=torch.int64, device=dev)
-            slot_t = torch.tensor(
-                [acts[i].pen_slot for i in row_act], dtype=torch.int64,
-                device=dev)
-            p_t = torch.tensor(row_p, dtype=torch.float32,
-                               device=dev).unsqueeze(1)
+            rid = torch.tensor(row_ids, dtype=torch.int64).to(dev, non_blocking=True)
+            slot_t = torch.tensor([acts[i].pen_slot for i in row_act], dtype=torch.int64).to(dev, non_blocking=True)
+            p_t = torch.tensor(row_p, dtype=torch.float32).to(dev, non_blocking=True).unsqueeze(1)
             gl = logits[rid].float()
             seen = self._pen_pool[slot_t]
             pen_rows = torch.where(
@@ -849,16 +839,16 @@ class InferenceEngine:
         # mark penalized-greedy emissions seen (one batched scatter)
         if pen_new_slots:
             self._pen_pool[
-                torch.tensor(pen_new_slots, dtype=torch.int64, device=dev),
-                torch.tensor(pen_new_toks, dtype=torch.int64, device=dev),
+                torch.tensor(pen_new_slots, dtype=torch.int64).to(dev, non_blocking=True),
+                torch.tensor(pen_new_toks, dtype=torch.int64).to(dev, non_blocking=True),
             ] = True
         if nong:
             # non-greedy requests take the plain one-token path
             rows = torch.tensor(
                 [cu[i + 1] - 1 for i, a in enumerate(acts)
                  if i not in handled],
-                dtype=torch.int64, device=logits.device,
-            )
+                dtype=torch.int64,
+            ).to(logits.device, non_blocking=True)
             for a in nong:
                 a.length += 1
             self._sample_and_emit(nong, logits[rows])
@@ -1104,9 +1094,7 @@ class InferenceEngine:
             self.kv.extend_seq(sid, prompt_len + steps_budget + 1)
         dev = self.device
         acts = self._bench_acts
-        first = torch.tensor(
-            [a.req.output_ids[-1] for a in acts], dtype=torch.int64, device=dev
-        )
+        first = torch.tensor([a.req.output_ids[-1] for a in acts], dtype=torch.int64).to(dev, non_blocking=True)
         bt = self.kv.block_table([a.seq_id for a in acts])
         if self.graphs is not None:
             g = self.graphs
