@@ -188,3 +188,61 @@ def test_cpengine2_matches_single(tmp_path):
     finally:
         eng.shutdown()
     assert cp_outs == ref, (cp_outs, ref)
+
+
+def _cpengine_short_worker(rank, world, port, out_path):
+    import torch.distributed as dist
+
+    from bee2bee_amd.parallel.cp import CPEngine
+
+    dist.init_process_group(
+        backend="gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world,
+    )
+    try:
+        eng = CPEngine("tiny", device="cpu", max_batch=2, max_seq_len=512,
+                       seed=23)
+        prompts = [[7, 8, 9, 10, 11]]  # < one page: rank 1 owns NOTHING
+        outs = eng.generate(prompts, 4)
+        if rank == 0:
+            with open(out_path, "wb") as f:
+                pickle.dump(outs, f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_cpengine_short_prompt_empty_shard(tmp_path):
+    """A prompt shorter than one 256-token page leaves rank 1 with an
+    EMPTY KV shard; the merge must still be exact."""
+    out_path = str(tmp_path / "cps.pkl")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_cpengine_short_worker,
+                         args=(r, 2, 29877, out_path)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=200)
+        assert p.exitcode == 0
+    with open(out_path, "rb") as f:
+        cp_outs = pickle.load(f)
+
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=512,
+                          seed=23)
+    try:
+        req = GenerationRequest(prompt_ids=[7, 8, 9, 10, 11],
+                                max_new_tokens=4,
+                                sampling=SamplingParams(greedy=True))
+        eng.submit(req)
+        while True:
+            item = req.out_queue.get(timeout=120)
+            if not isinstance(item, int):
+                break
+        assert req.error is None, req.error
+        ref = req.output_ids
+    finally:
+        eng.shutdown()
+    assert cp_outs == [ref], (cp_outs, ref)
