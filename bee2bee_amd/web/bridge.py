@@ -86,6 +86,9 @@ class MeshBridge:
             env = os.getenv("BEE2BEE_SEEDS", "")
             seeds = [s.strip() for s in env.split(",") if s.strip()]
         self.pool: List[str] = list(dict.fromkeys(seeds))
+        self._seeds0 = list(self.pool)  # recovery set: a dead seed is
+        # pruned from the pool, but when EVERYTHING is gone the original
+        # seeds are retried (a node may come back at the same address)
         self.store = store if store is not None else WebStore()
         self.auto_reconnect = auto_reconnect
         self.registered_node: Optional[str] = None
@@ -162,6 +165,8 @@ class MeshBridge:
                 self.pool.remove(addr)
             if self.registered_node == addr:
                 self.registered_node = None
+        if not self.pool:
+            self.pool = list(self._seeds0)
         return False
 
     async def _dial(self, addr: str) -> bool:

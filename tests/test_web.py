@@ -258,3 +258,48 @@ def test_bridge_pool_rotation_drops_dead_seed():
             await node.stop()
 
     asyncio.run(run())
+
+
+@pytest.mark.timeout(120)
+def test_bridge_reconnects_after_node_restart():
+    """The bridge must survive its active node dying and reattach when a
+    node at the same address returns (reference bridge.js:217-222)."""
+    async def run():
+        from bee2bee_amd.web import bridge as bridge_mod
+
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        port = node.port
+        node.local_services["hf"] = EchoService(model="echo-model")
+        store = WebStore(base_url=None, key=None)
+        bridge = MeshBridge(seeds=[f"ws://127.0.0.1:{port}"], store=store,
+                            auto_reconnect=True)
+        # fast reconnect for the test
+        orig_delay = bridge_mod.RECONNECT_DELAY_S
+        bridge_mod.RECONNECT_DELAY_S = 0.2
+        await bridge.start()
+        try:
+            assert await bridge.connect()
+            r1 = await bridge.request({"prompt": "a", "model": "echo-model"})
+            assert "echo:a" in r1["text"]
+            await node.stop()
+            await asyncio.sleep(0.5)
+            assert not bridge.connected
+            # node comes back on the SAME port
+            node2 = MeshNode(host="127.0.0.1", port=port, enable_nat=False)
+            await node2.start()
+            node2.local_services["hf"] = EchoService(model="echo-model")
+            for _ in range(100):
+                if bridge.connected:
+                    break
+                await asyncio.sleep(0.1)
+            # even if the auto-reconnect timer hasn't fired, a request
+            # must re-dial and succeed
+            r2 = await bridge.request({"prompt": "b", "model": "echo-model"})
+            assert "echo:b" in r2["text"]
+            await node2.stop()
+        finally:
+            bridge_mod.RECONNECT_DELAY_S = orig_delay
+            await bridge.stop()
+
+    asyncio.run(run())
