@@ -239,7 +239,9 @@ DASHBOARD_HTML = """<!doctype html>
  .stat b{color:#e6edf3}
 </style></head><body><main>
 <h1>bee2bee-amd &mdash; decentralized inference mesh</h1>
-<section><h3>Network</h3><div id="stats">loading&hellip;</div></section>
+<section><h3>Network</h3><div id="stats">loading&hellip;</div>
+<table id="mesh" style="width:100%;margin-top:12px;border-collapse:collapse">
+</table></section>
 <section><h3>Register a node</h3>
 <input id="link" placeholder="coithub.org://join?..."/>
 <button onclick="registerNode()">Register</button>
@@ -259,6 +261,17 @@ async function refresh(){
    `<span class=stat>peers <b>${s.poolSize}</b></span>`+
    `<span class=stat>total tokens <b>${m.tokens||0}</b></span>`+
    `<span class=stat>chats <b>${m.chats||0}</b></span>`;
+  const rows=[['region','node','models','latency','status']];
+  for(const [region,peers] of Object.entries(s.mesh||{}))
+   for(const p of peers)
+    rows.push([region,p.peer_id||p.addr||'-',
+     (p.models||[]).join(', ')||'-',
+     p.latency!=null?p.latency+' ms':'-',p.status||'-']);
+  document.getElementById('mesh').innerHTML=rows.map((r,i)=>
+   `<tr style="border-bottom:1px solid #30363d">`+r.map(c=>
+    i?`<td style="padding:4px 8px">${c}</td>`
+     :`<th style="text-align:left;padding:4px 8px;color:#7d8590">${c}</th>`)
+   .join('')+`</tr>`).join('');
  }catch(e){}
 }
 async function registerNode(){
