@@ -62,11 +62,26 @@ class Runner:
 
     # ------------------------------------------------------------- forwards
 
-    def embed(self, input_ids: torch.Tensor) -> torch.Tensor:
-        return F.embedding(input_ids, self.weights.embed)
+    def embed(self, input_ids: torch.Tensor,
+              positions: Optional[torch.Tensor] = None) -> torch.Tensor:
+        h = F.embedding(input_ids, self.weights.embed)
+        if self.spec.pos_type == "learned" and positions is not None:
+            h = h + F.embedding(positions.long(), self.weights.pos_embed)
+        return h
+
+    def _norm(self, x, w, b):
+        if self.spec.norm_type == "layernorm":
+            return ops.layernorm(x, w, b, self.spec.rms_eps)
+        return ops.rmsnorm(x, w, self.spec.rms_eps)
+
+    def _fused_add_norm(self, x, resid, w, b):
+        if self.spec.norm_type == "layernorm":
+            return ops.fused_add_layernorm(x, resid, w, b, self.spec.rms_eps)
+        return ops.fused_add_rmsnorm(x, resid, w, self.spec.rms_eps)
 
     def lm_head(self, hidden: torch.Tensor) -> torch.Tensor:
-        normed = ops.rmsnorm(hidden, self.weights.final_norm, self.spec.rms_eps)
+        normed = self._norm(hidden, self.weights.final_norm,
+                            self.weights.final_norm_bias)
         return F.linear(normed, self.weights.lm_head)
 
     def _layer(
@@ -82,10 +97,10 @@ class Runner:
         lw = self.weights.layers[layer_idx]
         if residual is None:
             residual = hidden
-            normed = ops.rmsnorm(hidden, lw.attn_norm, s.rms_eps)
+            normed = self._norm(hidden, lw.attn_norm, lw.attn_norm_bias)
         else:
-            normed, residual = ops.fused_add_rmsnorm(
-                hidden, residual, lw.attn_norm, s.rms_eps
+            normed, residual = self._fused_add_norm(
+                hidden, residual, lw.attn_norm, lw.attn_norm_bias
             )
 
         qkv = F.linear(normed, lw.wqkv)  # column-parallel under TP
@@ -96,22 +111,26 @@ class Runner:
         q = q.view(T, s.n_heads, s.head_dim)
         k = k.view(T, s.n_kv_heads, s.head_dim)
         v = v.view(T, s.n_kv_heads, s.head_dim)
-        ops.rope_inplace(q, k, positions, self.rope_cos, self.rope_sin)
+        if s.pos_type == "rope":
+            ops.rope_inplace(q, k, positions, self.rope_cos, self.rope_sin)
         k_cache, v_cache = self.kv.layer(layer_idx)
         ops.kv_cache_store(k, v, k_cache, v_cache, slot_mapping)
         attn_out = attn_fn(layer_idx, q, k, v, k_cache, v_cache)
-        attn_out = F.linear(attn_out.reshape(T, s.q_size), lw.wo)
+        attn_out = F.linear(attn_out.reshape(T, s.q_size), lw.wo, lw.wo_bias)
         if self.tp_group is not None:
             # row-parallel o-projection: sum the per-shard partials
             import torch.distributed as dist
 
             dist.all_reduce(attn_out, group=self.tp_group)
 
-        normed, residual = ops.fused_add_rmsnorm(
-            attn_out, residual, lw.mlp_norm, s.rms_eps
+        normed, residual = self._fused_add_norm(
+            attn_out, residual, lw.mlp_norm, lw.mlp_norm_bias
         )
         if s.is_moe:
             mlp_out = self._moe_mlp(lw, normed)
+        elif s.act_type == "gelu":
+            fc = F.linear(normed, lw.w_gate_up, lw.w_gate_up_bias)
+            mlp_out = F.linear(ops.gelu(fc), lw.w_down, lw.w_down_bias)
         else:
             gate_up = F.linear(normed, lw.w_gate_up)
             mlp_out = F.linear(ops.swiglu(gate_up), lw.w_down)
@@ -302,7 +321,7 @@ class Runner:
             return ops.attn_prefill(q, k, v, cu_seqlens, max_seqlen, self.scale)
 
         hidden = (
-            self.embed(input_ids_or_hidden)
+            self.embed(input_ids_or_hidden, positions)
             if self.is_first_stage
             else input_ids_or_hidden
         )
@@ -331,7 +350,7 @@ class Runner:
             )
 
         hidden = (
-            self.embed(input_ids_or_hidden)
+            self.embed(input_ids_or_hidden, positions)
             if self.is_first_stage
             else input_ids_or_hidden
         )

@@ -33,6 +33,14 @@ class ModelSpec:
     top_k_experts: int = 2
     # Qwen2-style attention bias on the fused QKV projection
     qkv_bias: bool = False
+    # architecture family knobs (gpt2: layernorm + learned positions +
+    # gelu MLP + biases everywhere + tied embeddings)
+    arch: str = "llama"          # "llama" | "gpt2" (HF weight mapping)
+    norm_type: str = "rmsnorm"   # "rmsnorm" | "layernorm"
+    act_type: str = "swiglu"     # "swiglu" | "gelu" (gelu: fc -> act -> proj)
+    pos_type: str = "rope"       # "rope" | "learned"
+    attn_out_bias: bool = False
+    mlp_bias: bool = False
     # vocab specials (byte-tokenizer defaults; overridden by a real tokenizer)
     bos_token_id: int = 1
     eos_token_id: int = 2
@@ -55,6 +63,8 @@ class ModelSpec:
         attn = h * (self.q_size + 2 * self.kv_size) + self.q_size * h
         if self.is_moe:
             mlp = self.n_experts * 3 * h * i + h * self.n_experts
+        elif self.act_type == "gelu":
+            mlp = 2 * h * i
         else:
             mlp = 3 * h * i
         norms = 2 * h
@@ -219,7 +229,42 @@ PRESETS: Dict[str, ModelSpec] = {
     ),
 }
 
+def _gpt2(name: str, n_layers: int, hidden: int, n_heads: int) -> ModelSpec:
+    return ModelSpec(
+        name=name,
+        vocab_size=50257,
+        hidden_size=hidden,
+        intermediate_size=4 * hidden,
+        n_layers=n_layers,
+        n_heads=n_heads,
+        n_kv_heads=n_heads,
+        head_dim=hidden // n_heads,
+        rms_eps=1e-5,
+        max_seq_len=1024,
+        tie_embeddings=True,
+        qkv_bias=True,
+        arch="gpt2",
+        norm_type="layernorm",
+        act_type="gelu",
+        pos_type="learned",
+        attn_out_bias=True,
+        mlp_bias=True,
+        bos_token_id=50256,
+        eos_token_id=50256,
+    )
+
+
+PRESETS["distilgpt2"] = _gpt2("distilgpt2", 6, 768, 12)
+PRESETS["gpt2"] = _gpt2("gpt2", 12, 768, 12)
+PRESETS["gpt2-medium"] = _gpt2("gpt2-medium", 24, 1024, 16)
+PRESETS["tiny-gpt2"] = _gpt2("tiny-gpt2", 2, 64, 4)
+PRESETS["tiny-gpt2"].vocab_size = 512
+PRESETS["tiny-gpt2"].max_seq_len = 512
+
 _ALIASES = {
+    "distilbert/distilgpt2": "distilgpt2",
+    "openai-community/gpt2": "gpt2",
+    "openai-community/gpt2-medium": "gpt2-medium",
     "meta-llama/meta-llama-3-8b": "llama3-8b",
     "meta-llama/llama-3-8b": "llama3-8b",
     "llama-3-8b": "llama3-8b",
@@ -269,9 +314,19 @@ def resolve_spec(name: str, model_path: Optional[str] = None) -> ModelSpec:
 
 
 def spec_from_hf_config(model_path: str, name: Optional[str] = None) -> ModelSpec:
-    """Parse a HuggingFace config.json (Llama/Mistral/Mixtral style keys)."""
+    """Parse a HuggingFace config.json (Llama/Mistral/Mixtral/GPT-2 keys)."""
     with open(os.path.join(model_path, "config.json")) as f:
         cfg = json.load(f)
+    if cfg.get("model_type") == "gpt2" or "n_embd" in cfg:
+        spec = _gpt2(
+            name or cfg.get("_name_or_path", os.path.basename(model_path)),
+            cfg.get("n_layer", 12), cfg.get("n_embd", 768),
+            cfg.get("n_head", 12),
+        )
+        spec.vocab_size = cfg.get("vocab_size", 50257)
+        spec.max_seq_len = cfg.get("n_positions", 1024)
+        spec.rms_eps = cfg.get("layer_norm_epsilon", 1e-5)
+        return spec
     n_heads = cfg.get("num_attention_heads", 32)
     hidden = cfg.get("hidden_size", 4096)
     return ModelSpec(

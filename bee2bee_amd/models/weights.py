@@ -31,19 +31,25 @@ from .spec import ModelSpec
 
 class LayerWeights:
     __slots__ = (
-        "attn_norm", "wqkv", "wqkv_bias", "wo", "mlp_norm",
-        "w_gate_up", "w_down",  # dense mlp
+        "attn_norm", "attn_norm_bias", "wqkv", "wqkv_bias", "wo", "wo_bias",
+        "mlp_norm", "mlp_norm_bias",
+        "w_gate_up", "w_gate_up_bias", "w_down", "w_down_bias",  # dense mlp
         "moe_gate", "moe_w_gate_up", "moe_w_down",  # moe
     )
 
     def __init__(self) -> None:
         self.attn_norm = None
+        self.attn_norm_bias = None  # layernorm arch (gpt2)
         self.wqkv = None
-        self.wqkv_bias = None  # Qwen2-style fused [q+2kv] bias (optional)
+        self.wqkv_bias = None  # Qwen2/gpt2 fused [q+2kv] bias (optional)
         self.wo = None
+        self.wo_bias = None
         self.mlp_norm = None
-        self.w_gate_up = None
+        self.mlp_norm_bias = None
+        self.w_gate_up = None       # gelu arch: the single fc [I, H]
+        self.w_gate_up_bias = None
         self.w_down = None
+        self.w_down_bias = None
         self.moe_gate = None
         self.moe_w_gate_up = None
         self.moe_w_down = None
@@ -55,7 +61,9 @@ class ModelWeights:
         self.device = device
         self.dtype = dtype
         self.embed: Optional[torch.Tensor] = None  # [vocab, hidden]
+        self.pos_embed: Optional[torch.Tensor] = None  # [max_seq, hidden]
         self.final_norm: Optional[torch.Tensor] = None  # [hidden]
+        self.final_norm_bias: Optional[torch.Tensor] = None
         self.lm_head: Optional[torch.Tensor] = None  # [vocab, hidden]
         self.layers = [LayerWeights() for _ in range(spec.n_layers)]
         # expert-parallel shard: when set, MoE expert tensors hold only
@@ -99,12 +107,19 @@ class ModelWeights:
 
         std = 0.02
         proj_std = std / max(1.0, (2 * s.n_layers) ** 0.5)
+        zeros = lambda *shape: torch.zeros(  # noqa: E731
+            *shape, device=self.device, dtype=self.dtype)
         if lo == 0:
             self.embed = rnd("embed", s.vocab_size, s.hidden_size, std=std)
+            if s.pos_type == "learned":
+                self.pos_embed = rnd("pos_embed", s.max_seq_len,
+                                     s.hidden_size, std=0.01)
         if hi == s.n_layers:
             self.final_norm = torch.ones(
                 s.hidden_size, device=self.device, dtype=self.dtype
             )
+            if s.norm_type == "layernorm":
+                self.final_norm_bias = zeros(s.hidden_size)
             if s.tie_embeddings:
                 self.lm_head = (
                     self.embed
@@ -122,7 +137,12 @@ class ModelWeights:
                 lw.wqkv_bias = rnd(
                     f"l{i}.wqkv_b", s.q_size + 2 * s.kv_size, std=std
                 )
+            if s.norm_type == "layernorm":
+                lw.attn_norm_bias = zeros(s.hidden_size)
+                lw.mlp_norm_bias = zeros(s.hidden_size)
             lw.wo = rnd(f"l{i}.wo", s.hidden_size, s.q_size, std=proj_std)
+            if s.attn_out_bias:
+                lw.wo_bias = zeros(s.hidden_size)
             if s.is_moe:
                 lw.moe_gate = rnd(f"l{i}.gate", s.n_experts, s.hidden_size, std=std)
                 lw.moe_w_gate_up = torch.stack([
@@ -135,6 +155,14 @@ class ModelWeights:
                         s.intermediate_size, std=proj_std)
                     for e in range(e_lo, e_hi)
                 ])
+            elif s.act_type == "gelu":
+                lw.w_gate_up = rnd(f"l{i}.w_fc", s.intermediate_size,
+                                   s.hidden_size, std=std)
+                lw.w_down = rnd(f"l{i}.w_dn", s.hidden_size,
+                                s.intermediate_size, std=proj_std)
+                if s.mlp_bias:
+                    lw.w_gate_up_bias = zeros(s.intermediate_size)
+                    lw.w_down_bias = zeros(s.hidden_size)
             else:
                 lw.w_gate_up = rnd(f"l{i}.w_gu", 2 * s.intermediate_size, s.hidden_size, std=std)
                 lw.w_down = rnd(f"l{i}.w_dn", s.hidden_size, s.intermediate_size, std=proj_std)
@@ -152,6 +180,8 @@ class ModelWeights:
         """Load HF safetensors shards, fusing qkv / gate|up on the fly.
         expert_range: keep only that expert shard on device (EP ranks)."""
         s = self.spec
+        if s.arch == "gpt2":
+            return self._load_gpt2(model_path, layer_range)
         lo, hi = layer_range or (0, s.n_layers)
         self.expert_range = expert_range
         e_lo, e_hi = expert_range or (0, getattr(s, "n_experts", 0) or 0)
@@ -270,6 +300,122 @@ class ModelWeights:
         return self
 
 
+    @torch.no_grad()
+    def _load_gpt2(self, model_path: str,
+                   layer_range: Optional[Tuple[int, int]] = None
+                   ) -> "ModelWeights":
+        """GPT-2 family safetensors (wte/wpe/h.N.*): Conv1D weights are
+        stored transposed ([in, out]) and qkv ships pre-fused as c_attn."""
+        s = self.spec
+        lo, hi = layer_range or (0, s.n_layers)
+
+        def to_dev(t: torch.Tensor) -> torch.Tensor:
+            return t.to(self.device, self.dtype, non_blocking=True)
+
+        for name, tensor in iter_safetensors(model_path):
+            name = name[len("transformer."):] if name.startswith(
+                "transformer.") else name
+            if name == "wte.weight":
+                if lo == 0:
+                    self.embed = to_dev(tensor)
+                continue
+            if name == "wpe.weight":
+                if lo == 0:
+                    self.pos_embed = to_dev(tensor)
+                continue
+            if name == "ln_f.weight":
+                if hi == s.n_layers:
+                    self.final_norm = to_dev(tensor)
+                continue
+            if name == "ln_f.bias":
+                if hi == s.n_layers:
+                    self.final_norm_bias = to_dev(tensor)
+                continue
+            parts = name.split(".")
+            if parts[0] != "h":
+                continue
+            i = int(parts[1])
+            if not (lo <= i < hi):
+                continue
+            lw = self.layers[i]
+            rest = ".".join(parts[2:])
+            if rest == "ln_1.weight":
+                lw.attn_norm = to_dev(tensor)
+            elif rest == "ln_1.bias":
+                lw.attn_norm_bias = to_dev(tensor)
+            elif rest == "ln_2.weight":
+                lw.mlp_norm = to_dev(tensor)
+            elif rest == "ln_2.bias":
+                lw.mlp_norm_bias = to_dev(tensor)
+            elif rest == "attn.c_attn.weight":
+                lw.wqkv = to_dev(tensor.t().contiguous())  # Conv1D -> [3H, H]
+            elif rest == "attn.c_attn.bias":
+                lw.wqkv_bias = to_dev(tensor)
+            elif rest == "attn.c_proj.weight":
+                lw.wo = to_dev(tensor.t().contiguous())
+            elif rest == "attn.c_proj.bias":
+                lw.wo_bias = to_dev(tensor)
+            elif rest == "mlp.c_fc.weight":
+                lw.w_gate_up = to_dev(tensor.t().contiguous())  # [I, H]
+            elif rest == "mlp.c_fc.bias":
+                lw.w_gate_up_bias = to_dev(tensor)
+            elif rest == "mlp.c_proj.weight":
+                lw.w_down = to_dev(tensor.t().contiguous())  # [H, I]
+            elif rest == "mlp.c_proj.bias":
+                lw.w_down_bias = to_dev(tensor)
+        if hi == s.n_layers and self.lm_head is None and self.embed is not None:
+            self.lm_head = self.embed  # tied
+        return self
+
+
+
+def _save_gpt2(weights: "ModelWeights", out_dir: str, tensors, cpu) -> None:
+    """GPT-2-format shard + config (Conv1D tensors re-transposed)."""
+    from safetensors.torch import save_file
+
+    s = weights.spec
+    if weights.embed is not None:
+        tensors["wte.weight"] = cpu(weights.embed)
+    if weights.pos_embed is not None:
+        tensors["wpe.weight"] = cpu(weights.pos_embed)
+    if weights.final_norm is not None:
+        tensors["ln_f.weight"] = cpu(weights.final_norm)
+    if weights.final_norm_bias is not None:
+        tensors["ln_f.bias"] = cpu(weights.final_norm_bias)
+    for i, lw in enumerate(weights.layers):
+        if lw.wqkv is None:
+            continue
+        p = f"h.{i}"
+        tensors[f"{p}.ln_1.weight"] = cpu(lw.attn_norm)
+        tensors[f"{p}.ln_1.bias"] = cpu(lw.attn_norm_bias)
+        tensors[f"{p}.ln_2.weight"] = cpu(lw.mlp_norm)
+        tensors[f"{p}.ln_2.bias"] = cpu(lw.mlp_norm_bias)
+        tensors[f"{p}.attn.c_attn.weight"] = cpu(lw.wqkv.t())
+        tensors[f"{p}.attn.c_attn.bias"] = cpu(lw.wqkv_bias)
+        tensors[f"{p}.attn.c_proj.weight"] = cpu(lw.wo.t())
+        tensors[f"{p}.attn.c_proj.bias"] = cpu(lw.wo_bias)
+        tensors[f"{p}.mlp.c_fc.weight"] = cpu(lw.w_gate_up.t())
+        tensors[f"{p}.mlp.c_fc.bias"] = cpu(lw.w_gate_up_bias)
+        tensors[f"{p}.mlp.c_proj.weight"] = cpu(lw.w_down.t())
+        tensors[f"{p}.mlp.c_proj.bias"] = cpu(lw.w_down_bias)
+    save_file(tensors, os.path.join(out_dir, "model.safetensors"))
+    cfg = {
+        "architectures": ["GPT2LMHeadModel"],
+        "model_type": "gpt2",
+        "vocab_size": s.vocab_size,
+        "n_embd": s.hidden_size,
+        "n_layer": s.n_layers,
+        "n_head": s.n_heads,
+        "n_positions": s.max_seq_len,
+        "layer_norm_epsilon": s.rms_eps,
+        "tie_word_embeddings": True,
+        "bos_token_id": s.bos_token_id,
+        "eos_token_id": s.eos_token_id,
+    }
+    with open(os.path.join(out_dir, "config.json"), "w") as f:
+        json.dump(cfg, f, indent=1)
+
+
 def iter_safetensors(model_path: str) -> Iterator[Tuple[str, torch.Tensor]]:
     """Yield (name, tensor) from every *.safetensors shard in a directory."""
     from safetensors import safe_open
@@ -294,6 +440,10 @@ def save_hf(weights: ModelWeights, out_dir: str) -> None:
 
     def cpu(t: torch.Tensor) -> torch.Tensor:
         return t.detach().to("cpu").contiguous()
+
+    if s.arch == "gpt2":
+        _save_gpt2(weights, out_dir, tensors, cpu)
+        return
 
     if weights.embed is not None:
         tensors["model.embed_tokens.weight"] = cpu(weights.embed)

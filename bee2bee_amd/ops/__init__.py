@@ -196,6 +196,26 @@ def attn_prefill_paged(
     )
 
 
+def layernorm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+              eps: float) -> torch.Tensor:
+    if _use_hip(x):
+        return require_hip().layernorm(x, weight, bias, eps)
+    return reference.layernorm(x, weight, bias, eps)
+
+
+def fused_add_layernorm(x, residual, weight, bias, eps: float):
+    if _use_hip(x):
+        return tuple(require_hip().fused_add_layernorm(
+            x, residual, weight, bias, eps))
+    return reference.fused_add_layernorm(x, residual, weight, bias, eps)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    if _use_hip(x):
+        return require_hip().gelu(x)
+    return reference.gelu(x)
+
+
 def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
     if _use_hip(gate_up):
         return require_hip().swiglu(gate_up)

@@ -65,6 +65,94 @@ __global__ void rmsnorm_kernel(
     }
 }
 
+// -------------------------------------------------------------- layernorm
+// GPT-2-family LayerNorm (mean/variance, weight + bias), same one-HBM-read
+// structure as rmsnorm_kernel: row cached fp32 in LDS between passes.
+template <bool FUSED_ADD>
+__global__ void layernorm_kernel(
+    const unsigned short* __restrict__ x,      // [T, H]
+    const unsigned short* __restrict__ resid,  // [T, H] or null
+    const unsigned short* __restrict__ w,      // [H]
+    const unsigned short* __restrict__ b,      // [H]
+    unsigned short* __restrict__ y,            // [T, H]
+    unsigned short* __restrict__ resid_out,    // [T, H] or null
+    int H, float eps) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    float* row = reinterpret_cast<float*>(smem_raw);  // [H]
+    __shared__ float red[8], red2[8];
+
+    const long t = blockIdx.x;
+    const unsigned short* xr = x + t * (long)H;
+    const unsigned short* rr = FUSED_ADD ? resid + t * (long)H : nullptr;
+    unsigned short* ro = FUSED_ADD ? resid_out + t * (long)H : nullptr;
+
+    float lsum = 0.f, lsumsq = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        float v[8];
+        load_bf16x8(xr + i, v);
+        if (FUSED_ADD) {
+            float r[8];
+            load_bf16x8(rr + i, r);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) v[j] += r[j];
+            store_bf16x8(ro + i, v);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            row[i + j] = v[j];
+            lsum += v[j];
+            lsumsq += v[j] * v[j];
+        }
+    }
+    lsum = wave_sum(lsum);
+    lsumsq = wave_sum(lsumsq);
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (lane == 0) { red[wid] = lsum; red2[wid] = lsumsq; }
+    __syncthreads();
+    const int nw = blockDim.x / WAVE;
+    float tot = 0.f, totsq = 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+        tot += (i < nw) ? red[i] : 0.f;
+        totsq += (i < nw) ? red2[i] : 0.f;
+    }
+    const float mean = tot / (float)H;
+    const float var = totsq / (float)H - mean * mean;
+    const float inv = rsqrtf(var + eps);
+
+    unsigned short* yr = y + t * (long)H;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        float wv[8], bv[8], o[8];
+        load_bf16x8(w + i, wv);
+        load_bf16x8(b + i, bv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            o[j] = (row[i + j] - mean) * inv * wv[j] + bv[j];
+        store_bf16x8(yr + i, o);
+    }
+}
+
+// ------------------------------------------------------------------ gelu
+// HF gelu_new (tanh approximation) — matches transformers GPT-2 exactly.
+__global__ void gelu_kernel(
+    const unsigned short* __restrict__ x,  // [N] (flattened, 8-aligned)
+    unsigned short* __restrict__ y,
+    long n8) {
+    for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < n8;
+         idx += (long)gridDim.x * blockDim.x) {
+        float v[8], o[8];
+        load_bf16x8(x + idx * 8, v);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const float u = v[j];
+            const float c = 0.7978845608028654f * (u + 0.044715f * u * u * u);
+            o[j] = 0.5f * u * (1.f + tanhf(c));
+        }
+        store_bf16x8(y + idx * 8, o);
+    }
+}
+
 // ------------------------------------------------------------------- rope
 // Llama rotate-half RoPE, in place on strided q/k views of the fused qkv
 // projection. One wave per (token, head); lane owns pair (d, d+hd/2).
@@ -186,6 +274,35 @@ void launch_rmsnorm(const unsigned short* x, const unsigned short* w,
     const int smem = H * 4;
     hipLaunchKernelGGL((rmsnorm_kernel<false>), dim3(T), dim3(256), smem,
                        stream, x, nullptr, w, y, nullptr, H, eps);
+}
+
+void launch_layernorm(const unsigned short* x, const unsigned short* w,
+                      const unsigned short* b, unsigned short* y, long T,
+                      int H, float eps, hipStream_t stream) {
+    const int smem = H * 4;
+    hipLaunchKernelGGL((layernorm_kernel<false>), dim3(T), dim3(256), smem,
+                       stream, x, nullptr, w, b, y, nullptr, H, eps);
+}
+
+void launch_fused_add_layernorm(const unsigned short* x,
+                                const unsigned short* resid,
+                                const unsigned short* w,
+                                const unsigned short* b, unsigned short* y,
+                                unsigned short* resid_out, long T, int H,
+                                float eps, hipStream_t stream) {
+    const int smem = H * 4;
+    hipLaunchKernelGGL((layernorm_kernel<true>), dim3(T), dim3(256), smem,
+                       stream, x, resid, w, b, y, resid_out, H, eps);
+}
+
+void launch_gelu(const unsigned short* x, unsigned short* y, long n,
+                 hipStream_t stream) {
+    const long n8 = n / 8;
+    long blocks = (n8 + 255) / 256;
+    if (blocks > 2048) blocks = 2048;  // grid-stride
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(gelu_kernel, dim3(blocks), dim3(256), 0, stream, x, y,
+                       n8);
 }
 
 void launch_fused_add_rmsnorm(const unsigned short* x,

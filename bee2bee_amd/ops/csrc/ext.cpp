@@ -15,6 +15,14 @@ void launch_rmsnorm(const unsigned short*, const unsigned short*,
 void launch_fused_add_rmsnorm(const unsigned short*, const unsigned short*,
                               const unsigned short*, unsigned short*,
                               unsigned short*, long, int, float, hipStream_t);
+void launch_layernorm(const unsigned short*, const unsigned short*,
+                      const unsigned short*, unsigned short*, long, int,
+                      float, hipStream_t);
+void launch_fused_add_layernorm(const unsigned short*, const unsigned short*,
+                                const unsigned short*, const unsigned short*,
+                                unsigned short*, unsigned short*, long, int,
+                                float, hipStream_t);
+void launch_gelu(const unsigned short*, unsigned short*, long, hipStream_t);
 void launch_rope(unsigned short*, unsigned short*, const int*, const float*,
                  const float*, int, int, int, int, long, long, hipStream_t);
 void launch_kv_store(const unsigned short*, const unsigned short*,
@@ -147,6 +155,48 @@ void kv_cache_store(const Tensor& k, const Tensor& v, Tensor& k_cache,
     launch_kv_store(bf16p(k), bf16p(v), bf16p_mut(k_cache), bf16p_mut(v_cache),
                     slots.data_ptr<int>(), T, nkv, hd, bs, k.stride(0),
                     v.stride(0), stream());
+}
+
+Tensor layernorm(const Tensor& x, const Tensor& w, const Tensor& b,
+                 double eps) {
+    check_bf16(x, "x");
+    check_bf16(w, "w");
+    check_bf16(b, "b");
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && b.is_contiguous());
+    const long T = x.numel() / x.size(-1);
+    const int H = x.size(-1);
+    TORCH_CHECK(H % 8 == 0);
+    Tensor y = torch::empty_like(x);
+    launch_layernorm(bf16p(x), bf16p(w), bf16p(b), bf16p_mut(y), T, H,
+                     (float)eps, stream());
+    return y;
+}
+
+std::vector<Tensor> fused_add_layernorm(const Tensor& x, const Tensor& resid,
+                                        const Tensor& w, const Tensor& b,
+                                        double eps) {
+    check_bf16(x, "x");
+    check_bf16(resid, "resid");
+    TORCH_CHECK(x.is_contiguous() && resid.is_contiguous());
+    TORCH_CHECK(x.sizes() == resid.sizes());
+    const long T = x.numel() / x.size(-1);
+    const int H = x.size(-1);
+    TORCH_CHECK(H % 8 == 0);
+    Tensor y = torch::empty_like(x);
+    Tensor resid_out = torch::empty_like(x);
+    launch_fused_add_layernorm(bf16p(x), bf16p(resid), bf16p(w), bf16p(b),
+                               bf16p_mut(y), bf16p_mut(resid_out), T, H,
+                               (float)eps, stream());
+    return {y, resid_out};
+}
+
+Tensor gelu(const Tensor& x) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.is_contiguous());
+    TORCH_CHECK(x.numel() % 8 == 0);
+    Tensor y = torch::empty_like(x);
+    launch_gelu(bf16p(x), bf16p_mut(y), x.numel(), stream());
+    return y;
 }
 
 Tensor swiglu(const Tensor& gu) {
@@ -352,6 +402,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rope_inplace", &rope_inplace, "llama RoPE in place");
     m.def("kv_cache_store", &kv_cache_store, "paged KV scatter");
     m.def("swiglu", &swiglu, "silu(g) * u");
+    m.def("layernorm", &layernorm, "LayerNorm (bf16, weight+bias)");
+    m.def("fused_add_layernorm", &fused_add_layernorm,
+          "residual add + LayerNorm");
+    m.def("gelu", &gelu, "gelu_new (tanh approximation)");
     m.def("attn_decode", &attn_decode, "paged GQA decode attention");
     m.def("attn_decode_lse", &attn_decode_lse,
           "paged GQA decode attention + per-head (m, l) merge state");

@@ -36,6 +36,30 @@ def fused_add_rmsnorm(
     return rmsnorm(new_residual, weight, eps), new_residual
 
 
+def layernorm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+              eps: float) -> torch.Tensor:
+    xf = x.float()
+    mean = xf.mean(-1, keepdim=True)
+    var = xf.var(-1, unbiased=False, keepdim=True)
+    out = (xf - mean) * torch.rsqrt(var + eps)
+    return (out * weight.float() + bias.float()).to(x.dtype)
+
+
+def fused_add_layernorm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor,
+    bias: torch.Tensor, eps: float,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    new_residual = (x.float() + residual.float()).to(x.dtype)
+    return layernorm(new_residual, weight, bias, eps), new_residual
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    """HF gelu_new (tanh approximation) — GPT-2's activation."""
+    xf = x.float()
+    c = 0.7978845608028654 * (xf + 0.044715 * xf.pow(3))
+    return (0.5 * xf * (1.0 + torch.tanh(c))).to(x.dtype)
+
+
 def rope_tables(
     max_len: int, head_dim: int, theta: float, device, dtype=torch.float32
 ) -> Tuple[torch.Tensor, torch.Tensor]:

@@ -1,0 +1,106 @@
+"""GPT-2 family (round 2): BASELINE config 1 names distilgpt2, and the
+engine now executes the REAL architecture (LayerNorm + learned positions +
+gelu MLP + biases + tied embeddings) instead of a llama-shaped stand-in.
+
+The ground-truth anchor: a checkpoint saved by this framework loads into
+transformers' GPT2LMHeadModel and produces the SAME logits as our runner."""
+import numpy as np
+import pytest
+import torch
+
+from bee2bee_amd.models.spec import PRESETS, resolve_spec
+from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+
+def test_gpt2_roundtrip(tmp_path):
+    spec = PRESETS["tiny-gpt2"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(3)
+    save_hf(w, str(tmp_path))
+    w2 = ModelWeights(spec, torch.device("cpu"), torch.float32).load_hf(
+        str(tmp_path))
+    assert torch.equal(w.embed, w2.embed)
+    assert torch.equal(w.pos_embed, w2.pos_embed)
+    assert torch.equal(w.layers[0].wqkv, w2.layers[0].wqkv)
+    assert torch.equal(w.layers[1].w_down_bias, w2.layers[1].w_down_bias)
+    assert torch.equal(w2.lm_head, w2.embed)  # tied
+    # the config round-trips to the same spec
+    spec2 = resolve_spec("x", model_path=str(tmp_path))
+    assert (spec2.arch, spec2.n_layers, spec2.n_heads) == ("gpt2", 2, 4)
+
+
+def test_gpt2_logits_match_transformers(tmp_path):
+    """Byte-for-byte architecture check: our runner's logits vs
+    transformers' GPT2LMHeadModel on the SAME saved checkpoint."""
+    transformers = pytest.importorskip("transformers")
+
+    spec = PRESETS["tiny-gpt2"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(7)
+    save_hf(w, str(tmp_path))
+
+    hf = transformers.GPT2LMHeadModel.from_pretrained(
+        str(tmp_path), torch_dtype=torch.float32)
+    hf.eval()
+
+    ids = torch.tensor([[5, 9, 17, 3, 250, 44, 8]])
+    with torch.no_grad():
+        hf_logits = hf(ids).logits[0]  # [T, V]
+
+    from bee2bee_amd.engine.kv import PagedKV
+    from bee2bee_amd.engine.runner import Runner
+
+    kv = PagedKV(spec, torch.device("cpu"), torch.float32, n_blocks=8)
+    runner = Runner(spec, w, kv, torch.device("cpu"), torch.float32)
+    T = ids.shape[1]
+    kv.new_seq(0)
+    kv.extend_seq(0, T)
+    slots = torch.tensor(kv.slot_mapping(0, range(T)), dtype=torch.int32)
+    hidden = runner.forward_prefill(
+        ids[0], torch.arange(T, dtype=torch.int32), slots,
+        torch.tensor([0, T], dtype=torch.int32), T,
+    )
+    ours = runner.lm_head(hidden)
+    err = (ours - hf_logits).abs().max().item()
+    assert err < 2e-3, f"logits diverge from transformers: {err}"
+
+
+def test_gpt2_engine_greedy_decode():
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    def run():
+        eng = InferenceEngine("tiny-gpt2", device="cpu", max_batch=2,
+                              max_seq_len=128, seed=11)
+        try:
+            req = GenerationRequest(prompt_ids=[5, 6, 7, 8],
+                                    max_new_tokens=8,
+                                    sampling=SamplingParams(greedy=True))
+            eng.submit(req)
+            while True:
+                item = req.out_queue.get(timeout=60)
+                if not isinstance(item, int):
+                    break
+            assert req.error is None, req.error
+            return req.output_ids
+    # determinism across engine instances
+        finally:
+            eng.shutdown()
+
+    a, b = run(), run()
+    assert len(a) == 8 and a == b
+
+
+def test_distilgpt2_served_by_name():
+    """Config 1 shape: serving 'distilgpt2' by name builds the real
+    6-layer GPT-2 architecture (random weights offline)."""
+    from bee2bee_amd.engine.engine import InferenceEngine
+
+    eng = InferenceEngine("distilgpt2", device="cpu", max_batch=1,
+                          max_seq_len=64, seed=1)
+    try:
+        assert eng.spec.arch == "gpt2" and eng.spec.n_layers == 6
+        assert eng.weights.pos_embed is not None
+        req = eng.generate([50, 51, 52], max_new_tokens=3, temperature=0.0,
+                           repetition_penalty=1.0)
+        assert len(req.output_ids) == 3
+    finally:
+        eng.shutdown()

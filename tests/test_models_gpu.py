@@ -43,3 +43,31 @@ def test_mixtral_8x7b_decodes():
     # MoE path must be deterministic under greedy
     toks2 = _generate("mixtral-8x7b")
     assert toks == toks2
+
+
+@pytest.mark.timeout(300)
+def test_distilgpt2_decodes_on_gpu():
+    """The real GPT-2 architecture (layernorm/gelu/learned-pos kernels) on
+    the HIP path, deterministic greedy."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("distilgpt2", device="cuda:0", max_batch=4,
+                          max_seq_len=256, seed=3)
+    try:
+        assert eng.spec.arch == "gpt2"
+        outs = []
+        for _ in range(2):
+            req = GenerationRequest(prompt_ids=[50, 51, 52, 99],
+                                    max_new_tokens=8,
+                                    sampling=SamplingParams(greedy=True))
+            eng.submit(req)
+            while True:
+                item = req.out_queue.get(timeout=120)
+                if not isinstance(item, int):
+                    break
+            assert req.error is None, req.error
+            outs.append(req.output_ids)
+        assert outs[0] == outs[1] and len(outs[0]) == 8
+    finally:
+        eng.shutdown()

@@ -437,3 +437,25 @@ def test_attn_decode_lse_merge(B, nkv, G, hd, fold):
     merged = merge_partials(outs, mls)
     assert_close(merged, full, atol=3e-2, rtol=3e-2,
                  msg=f"cp merge B{B} G{G} hd{hd} fold={fold}")
+
+
+def test_layernorm_matches_reference():
+    torch.manual_seed(4)
+    for T, H in [(7, 768), (256, 1024), (3, 64)]:
+        x = torch.randn(T, H, device=DEV).bfloat16()
+        w = torch.randn(H, device=DEV).bfloat16()
+        b = torch.randn(H, device=DEV).bfloat16()
+        got = ops.layernorm(x, w, b, 1e-5)
+        ref = R.layernorm(x, w, b, 1e-5)
+        assert_close(got, ref, msg=f"layernorm {T}x{H}")
+        resid = torch.randn(T, H, device=DEV).bfloat16()
+        g2, r2 = ops.fused_add_layernorm(x, resid, w, b, 1e-5)
+        e2, er2 = R.fused_add_layernorm(x, resid, w, b, 1e-5)
+        assert_close(g2, e2, msg="fused_add_layernorm y")
+        assert_close(r2, er2, msg="fused_add_layernorm resid")
+
+
+def test_gelu_matches_reference():
+    torch.manual_seed(5)
+    x = (torch.randn(33, 768, device=DEV) * 3).bfloat16()
+    assert_close(ops.gelu(x), R.gelu(x), msg="gelu")

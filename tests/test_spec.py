@@ -28,9 +28,19 @@ def test_alias_resolution():
 
 
 def test_unknown_falls_back_to_demo():
-    spec = resolve_spec("distilgpt2")
+    spec = resolve_spec("some-model-nobody-knows")
     assert spec.hidden_size == 768  # demo spec under the requested name
-    assert spec.name == "distilgpt2"
+    assert spec.name == "some-model-nobody-knows"
+
+
+def test_distilgpt2_is_a_real_gpt2_spec():
+    # round 2: distilgpt2 (BASELINE config 1) is a REAL architecture now,
+    # not the llama-shaped demo spec under a borrowed name
+    spec = resolve_spec("distilgpt2")
+    assert (spec.arch, spec.norm_type, spec.act_type, spec.pos_type) == (
+        "gpt2", "layernorm", "gelu", "learned")
+    assert spec.n_layers == 6 and spec.n_heads == 12
+    assert resolve_spec("distilbert/distilgpt2").name == "distilgpt2"
 
 
 def test_hf_config_parsing(tmp_path):
