@@ -46,6 +46,12 @@ def shard_spec(spec: ModelSpec, world: int) -> ModelSpec:
     assert spec.n_kv_heads % world == 0, "world must divide n_kv_heads"
     assert spec.intermediate_size % world == 0
     assert not spec.is_moe, "use expert parallelism (ep.py) for MoE models"
+    # row-parallel projections all-reduce partial outputs; an output-side
+    # bias (gpt2 family) would be added once PER RANK — scope TP to the
+    # swiglu/no-out-bias families until bias handling lands
+    assert spec.act_type == "swiglu" and not spec.attn_out_bias, (
+        "TP supports the llama/mistral/qwen families; gpt2-family biases "
+        "need rank-0-only bias handling")
     return dataclasses.replace(
         spec,
         n_heads=spec.n_heads // world,
