@@ -559,7 +559,14 @@ class InferenceEngine:
         last_rows = torch.tensor([cu[i + 1] - 1 for i in done_idx], dtype=torch.int64).to(dev, non_blocking=True)
         logits = self.runner.lm_head(hidden[last_rows])
         self._sample_and_emit(completed, logits)
-        self._active.extend([a for a in completed if a.req.done_ts is None])
+        for a in completed:
+            if a.req.done_ts is not None:
+                # finished AT prefill (max_new=1 / instant stop token):
+                # retire here or the KV blocks leak until shutdown
+                self.kv.free_seq(a.seq_id)
+                self._pen_release(a)
+            else:
+                self._active.append(a)
 
     @torch.no_grad()
     def _decode_once(self) -> None:

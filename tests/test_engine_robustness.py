@@ -169,3 +169,35 @@ def test_concurrent_submit_thread_safety():
             assert len(r.output_ids) == 4
     finally:
         eng.shutdown()
+
+
+def test_kv_freed_for_requests_finishing_at_prefill():
+    """max_new_tokens=1 requests finish at prefill completion and must
+    release their KV blocks (regression: they skipped retirement and the
+    pool drained)."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=96,
+                          seed=2)
+    try:
+        base = eng.kv.free_blocks
+        for round_ in range(6):  # 6 rounds x 2 blocks would exhaust a
+            req = GenerationRequest(  # leaky pool long before this
+                prompt_ids=list(range(4, 40)), max_new_tokens=1,
+                sampling=SamplingParams(greedy=True))
+            eng.submit(req)
+            while True:
+                item = req.out_queue.get(timeout=60)
+                if not isinstance(item, int):
+                    break
+            assert req.error is None, req.error
+            assert len(req.output_ids) == 1
+        for _ in range(100):
+            if eng.kv.free_blocks == base:
+                break
+            import time as _t
+            _t.sleep(0.02)
+        assert eng.kv.free_blocks == base, (eng.kv.free_blocks, base)
+    finally:
+        eng.shutdown()
