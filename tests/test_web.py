@@ -303,3 +303,51 @@ def test_bridge_reconnects_after_node_restart():
             await bridge.stop()
 
     asyncio.run(run())
+
+
+@pytest.mark.timeout(180)
+def test_bridge_direct_http_success_path():
+    """targetNode with a LIVE FastAPI /generate: the bridge must stream via
+    direct HTTP and never open a tunnel (reference bridge.js:271-289)."""
+    import threading
+
+    import uvicorn
+
+    async def run():
+        from bee2bee_amd.gateway import api as gateway_api
+
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        node.local_services["hf"] = EchoService(model="echo-model")
+        gateway_api.node = node
+        config = uvicorn.Config(gateway_api.app, host="127.0.0.1",
+                                port=18461, log_level="error")
+        server = uvicorn.Server(config)
+        task = asyncio.ensure_future(server.serve())
+        for _ in range(100):
+            if server.started:
+                break
+            await asyncio.sleep(0.05)
+
+        store = WebStore(base_url=None, key=None)
+        bridge = MeshBridge(seeds=[], store=store, auto_reconnect=False)
+        await bridge.start()
+        try:
+            chunks = []
+            result = await bridge.request(
+                {"prompt": "direct hit", "model": "echo-model"},
+                on_chunk=chunks.append,
+                target_node="127.0.0.1:18461",
+            )
+            assert "echo:direct" in result["text"]
+            assert result["metadata"]["transport"] == "direct-http"
+            assert not bridge.connected  # no tunnel was needed
+            assert "".join(chunks) == result["text"]
+        finally:
+            await bridge.stop()
+            server.should_exit = True
+            await asyncio.sleep(0.2)
+            task.cancel()
+            await node.stop()
+
+    asyncio.run(run())
