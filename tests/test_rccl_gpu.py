@@ -183,3 +183,35 @@ def test_rccl_two_rank(kind):
     import torch.multiprocessing as mp
 
     mp.spawn(_nccl2_worker, args=(2, _free_port(), kind), nprocs=2, join=True)
+
+
+@pytest.mark.timeout(300)
+def test_cpengine_world1_nccl_matches_single(nccl_world1):
+    """CPEngine on a real RCCL group (world 1 => the all_gather is an
+    identity through RCCL) must match the plain engine token-for-token."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+    from bee2bee_amd.parallel.cp import CPEngine
+
+    g = torch.Generator().manual_seed(17)
+    prompts = [torch.randint(4, 500, (300,), generator=g).tolist()]
+    eng = CPEngine("tiny", device="cuda:0", max_batch=2, max_seq_len=512,
+                   seed=29)
+    cp_outs = eng.generate(prompts, 5)
+
+    ref_eng = InferenceEngine("tiny", device="cuda:0", max_batch=2,
+                              max_seq_len=512, seed=29, use_graphs=False)
+    try:
+        req = GenerationRequest(prompt_ids=list(prompts[0]),
+                                max_new_tokens=5,
+                                sampling=SamplingParams(greedy=True))
+        ref_eng.submit(req)
+        while True:
+            item = req.out_queue.get(timeout=120)
+            if not isinstance(item, int):
+                break
+        assert req.error is None, req.error
+        ref = req.output_ids
+    finally:
+        ref_eng.shutdown()
+    assert cp_outs == [ref], (cp_outs, ref)
