@@ -168,6 +168,10 @@ class CPEngine:
                           n_blocks=max_batch * local_pages + 2)
         self.runner = Runner(spec, self.weights, self.kv, self.device, dtype)
         self.runner.decode_attn_fn = self._cp_attn
+        # non-greedy sampling must draw IDENTICAL tokens on every rank
+        # (same contract and fix as TPEngine)
+        self.sample_gen = torch.Generator(device=self.device)
+        self.sample_gen.manual_seed(seed + 0x5EED)
         self._seqs: List[int] = []
         self._lens: List[int] = []          # GLOBAL lengths
         self._scratch = -7
@@ -235,7 +239,8 @@ class CPEngine:
         )
         last = torch.tensor([c - 1 for c in cu[1:]], dtype=torch.int64,
                             device=dev)
-        return sample(self.runner.lm_head(hidden[last]), sampling).cpu()
+        return sample(self.runner.lm_head(hidden[last]), sampling,
+                      generator=self.sample_gen).cpu()
 
     @torch.no_grad()
     def decode_step(self, ids: torch.Tensor, sampling=None) -> torch.Tensor:
@@ -256,7 +261,8 @@ class CPEngine:
         slots = torch.tensor(slot_list, dtype=torch.int32, device=dev)
         hidden = self.runner.forward_decode(
             ids.to(dev), positions, slots, bt, lens_t)
-        return sample(self.runner.lm_head(hidden), sampling).cpu()
+        return sample(self.runner.lm_head(hidden), sampling,
+                      generator=self.sample_gen).cpu()
 
     @torch.no_grad()
     def generate(self, prompts, max_new_tokens: int, sampling=None):

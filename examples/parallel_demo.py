@@ -6,7 +6,8 @@ Run (one rank per GPU; gloo on CPU works for a dry run):
     python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
         --master-addr 127.0.0.1 examples/parallel_demo.py --mode tp
 
-Modes: tp (tensor parallel), pp (pipeline parallel), ep (Mixtral expert
+Modes: tp (tensor parallel), pp (pipeline parallel), cp (context
+parallel: KV pages sharded across ranks), ep (Mixtral expert
 parallel). All three engines share the lockstep generate() API and are
 exact-match tested against the single-process engine (tests/test_*_cpu.py).
 """
@@ -22,7 +23,7 @@ import torch.distributed as dist
 
 def main() -> None:
     ap = argparse.ArgumentParser()
-    ap.add_argument("--mode", default="tp", choices=["tp", "pp", "ep"])
+    ap.add_argument("--mode", default="tp", choices=["tp", "pp", "ep", "cp"])
     ap.add_argument("--model", default=None,
                     help="preset (defaults: tp/pp=tiny or llama3-70b on GPU, "
                          "ep=tiny-moe or mixtral-8x7b on GPU)")
@@ -44,6 +45,10 @@ def main() -> None:
         from bee2bee_amd.parallel.moe_engine import MoEEngine as Engine
 
         model = args.model or ("mixtral-8x7b" if on_gpu else "tiny-moe")
+    elif args.mode == "cp":
+        from bee2bee_amd.parallel.cp import CPEngine as Engine
+
+        model = args.model or ("llama3-8b" if on_gpu else "tiny")
     elif args.mode == "pp":
         from bee2bee_amd.parallel.pp import PipelineEngine as Engine
 
