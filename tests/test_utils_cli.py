@@ -56,3 +56,43 @@ def test_cli_config_set(tmp_path, monkeypatch):
     assert res.exit_code == 0
     cfg = json.loads((tmp_path / "config.json").read_text())
     assert cfg["bootstrap_url"] == "ws://1.1.1.1:9"
+
+
+def test_serve_web_cli_end_to_end(tmp_path):
+    """`python -m bee2bee_amd serve-web` boots the browser gateway; the
+    dashboard and the api/p2p endpoints answer."""
+    import json
+    import socket
+    import subprocess
+    import sys
+    import time
+    import urllib.request
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "bee2bee_amd", "serve-web",
+         "--host", "127.0.0.1", "--port", str(port)],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+    )
+    try:
+        base = f"http://127.0.0.1:{port}"
+        for _ in range(100):
+            try:
+                html = urllib.request.urlopen(base + "/", timeout=1).read()
+                break
+            except Exception:
+                time.sleep(0.2)
+        else:
+            raise AssertionError("serve-web did not come up")
+        assert b"bee2bee" in html
+        status = json.load(urllib.request.urlopen(
+            base + "/api/p2p/status", timeout=5))
+        assert status["mode"] == "fusion-serverless"
+        metrics = json.load(urllib.request.urlopen(
+            base + "/api/p2p/global_metrics", timeout=5))
+        assert metrics.get("tokens", 0) == 0  # offline: zeros
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
