@@ -109,7 +109,11 @@ def _single_process_reference(model: str = "tiny"):
 
 @pytest.mark.timeout(300)
 @pytest.mark.parametrize(
-    "world,port,model", [(2, 29611, "tiny"), (3, 29612, "tiny3")]
+    "world,port,model",
+    [(2, 29611, "tiny"), (3, 29612, "tiny3"),
+     # gpt2 family across stages: learned positions live on the first
+     # stage, LayerNorm/gelu/biases run on every stage
+     (2, 29613, "tiny-gpt2")]
 )
 def test_pp_matches_single_process(tmp_path, world, port, model):
     """2- and 3-stage pipeline greedy decode == single-process engine
