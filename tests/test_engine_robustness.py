@@ -201,3 +201,42 @@ def test_kv_freed_for_requests_finishing_at_prefill():
         assert eng.kv.free_blocks == base, (eng.kv.free_blocks, base)
     finally:
         eng.shutdown()
+
+
+def test_cancel_during_chunked_prefill():
+    """Cancelling a request mid-chunked-prefill must free its KV and not
+    disturb other requests."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=480,
+                          seed=4, max_prefill_tokens=32)
+    try:
+        base = eng.kv.free_blocks
+        big = GenerationRequest(prompt_ids=list(range(4, 404)),
+                                max_new_tokens=8,
+                                sampling=SamplingParams(greedy=True))
+        small = GenerationRequest(prompt_ids=[7, 8, 9],
+                                  max_new_tokens=4,
+                                  sampling=SamplingParams(greedy=True))
+        eng.submit(big)
+        eng.submit(small)
+        big.cancelled = True  # cancel while the 400-token prompt chunks in
+        while True:
+            item = small.out_queue.get(timeout=60)
+            if not isinstance(item, int):
+                break
+        assert small.error is None and len(small.output_ids) == 4
+        # big either errored/cancelled or finished; wait for it to settle
+        import time as _t
+        for _ in range(200):
+            if big.done_ts is not None or big.error is not None:
+                break
+            _t.sleep(0.02)
+        for _ in range(200):
+            if eng.kv.free_blocks == base:
+                break
+            _t.sleep(0.02)
+        assert eng.kv.free_blocks == base, (eng.kv.free_blocks, base)
+    finally:
+        eng.shutdown()

@@ -432,3 +432,67 @@ def test_outbound_only_provider_tunnel():
         await relay.stop()
 
     asyncio.run(run())
+
+
+def test_node_survives_adversarial_frames():
+    """Malformed/hostile frames from a peer must never kill the node:
+    non-JSON, wrong-typed fields, unknown types, oversized rids, dht_set
+    garbage — the node stays responsive afterwards."""
+
+    async def run():
+        import aiohttp
+
+        node = await _start_node()
+        await node.add_service(EchoService())
+        session = aiohttp.ClientSession()
+        try:
+            ws = await session.ws_connect(node.addr)
+            evil = [
+                "this is not json",
+                "[]",
+                '"just a string"',
+                json.dumps({"no_type": 1}),
+                json.dumps({"type": 42}),
+                json.dumps({"type": "unknown_fancy_type", "x": [1, 2]}),
+                json.dumps({"type": "ping", "ts": "not-a-number"}),
+                json.dumps({"type": "pong", "ts": {"nested": True}}),
+                json.dumps({"type": "gen_result", "rid": "R" * 100000}),
+                json.dumps({"type": "gen_chunk"}),  # no rid at all
+                json.dumps({"type": "service_announce", "service": None,
+                            "meta": "nope"}),
+                json.dumps({"type": "peer_list", "peers": [123, None, {}]}),
+                json.dumps({"type": "dht_set", "key": None, "value": object
+                            .__class__.__name__}),
+                json.dumps({"type": "piece_request", "hash": True,
+                            "index": "NaN"}),
+                json.dumps({"type": "hello", "peer_id": {"k": 1},
+                            "services": 7}),
+            ]
+            for frame in evil:
+                await ws.send_str(frame)
+            await asyncio.sleep(0.3)
+            # the node is still alive and serves a real request
+            await ws.close()
+            client = await session.ws_connect(node.addr)
+            rid = "probe-1"
+            await client.send_str(json.dumps({
+                "type": "gen_request", "rid": rid, "svc": "hf",
+                "prompt": "still alive", "max_new_tokens": 4,
+            }))
+            text = None
+            for _ in range(50):
+                msg = await asyncio.wait_for(client.receive(), timeout=5)
+                if msg.type != 1:  # TEXT
+                    continue
+                d = json.loads(msg.data)
+                if d.get("type") in ("gen_success", "gen_result") and \
+                        d.get("rid") == rid and d.get("text"):
+                    text = d["text"]
+                    break
+            assert text == "echo:still alive", text
+            await client.close()
+        finally:
+            await session.close()
+            await node.stop()
+
+    asyncio.run(run())
