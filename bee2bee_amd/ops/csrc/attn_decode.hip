@@ -878,7 +878,11 @@ extern "C" void launch_attn_decode(
     switch (G) {
         case 1: LAUNCH(1); break;
         case 2: LAUNCH(2); break;
+        case 3: LAUNCH(3); break;
         case 4: LAUNCH(4); break;
+        case 5: LAUNCH(5); break;
+        case 6: LAUNCH(6); break;
+        case 7: LAUNCH(7); break;  // Qwen2.5-7B: 28 q / 4 kv heads
         case 8: LAUNCH(8); break;
         default: LAUNCH(16); break;
     }

@@ -229,7 +229,7 @@ std::vector<Tensor> attn_decode_impl(const Tensor& q, const Tensor& k_cache,
     TORCH_CHECK(hd <= 128 && hd % 2 == 0, "decode kernel supports hd<=128");
     TORCH_CHECK(nq % nkv == 0);
     const int G = nq / nkv;
-    TORCH_CHECK(G == 1 || G == 2 || G == 4 || G == 8 || G == 16,
+    TORCH_CHECK(G >= 1 && G <= 8 || G == 16,
                 "unsupported GQA group size ", G);
     TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == hd);
     Tensor out = torch::empty({B, nq, hd}, q.options());

@@ -118,6 +118,9 @@ def test_swiglu(T, I):
     [
         (3, 8, 4, 128, 32, 500),   # llama3-8b shape
         (2, 8, 8, 128, 32, 300),   # llama3-70b shape
+        (2, 4, 7, 128, 32, 300),   # qwen2.5-7b shape (G=7, regression:
+                                   # was never instantiated -> GPU decode
+                                   # raised for qwen)
         (2, 8, 4, 64, 32, 129),    # llama3.2-1b shape
         (2, 2, 2, 16, 32, 70),     # tiny
         (1, 1, 1, 128, 32, 33),
