@@ -138,8 +138,9 @@ PRESETS: Dict[str, ModelSpec] = {
         n_experts=4,
         top_k_experts=2,
     ),
-    # small demo model served under names our engine has no checkpoint for
-    # (e.g. "distilgpt2" in BASELINE config 1 — plumbing checks on CPU)
+    # small demo model served under UNKNOWN names (fallback spec; known
+    # families — incl. the real distilgpt2/gpt2 since round 2 — have their
+    # own presets below)
     "demo-125m": ModelSpec(
         name="demo-125m",
         vocab_size=32000,
