@@ -351,3 +351,75 @@ def test_bridge_direct_http_success_path():
             await node.stop()
 
     asyncio.run(run())
+
+
+def test_web_store_offline_guards():
+    """Offline store (no creds): every call degrades to no-op/zeros."""
+    async def run():
+        store = WebStore(base_url=None, key=None)
+        assert not store.enabled
+        assert await store.insert_message("n", 5) is False
+        assert await store.system_stats() == {"visits": 0, "chats": 0,
+                                              "tokens": 0}
+        assert await store.active_nodes() == []
+        assert await store.upsert_node({"peer_id": "x"}) is False
+        # zero tokens never writes even when enabled-looking
+        store2 = WebStore(base_url="http://127.0.0.1:9", key="k")
+        assert await store2.insert_message("n", 0) is False
+
+    asyncio.run(run())
+
+
+@pytest.mark.timeout(180)
+def test_gateway_generate_with_target_node_direct_http():
+    """Full L6 path with targetNode: gateway -> bridge -> DIRECT HTTP to
+    the node's FastAPI /generate (no WS tunnel), tally persisted under the
+    target's id (reference index.js:80: node_id = targetNode)."""
+    import threading
+
+    import uvicorn
+
+    async def run():
+        from bee2bee_amd.gateway import api as gateway_api
+
+        directory = MockDirectory()
+        await directory.start()
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        node.local_services["hf"] = EchoService(model="echo-model")
+        gateway_api.node = node
+        config = uvicorn.Config(gateway_api.app, host="127.0.0.1",
+                                port=18462, log_level="error")
+        server = uvicorn.Server(config)
+        task = asyncio.ensure_future(server.serve())
+        for _ in range(100):
+            if server.started:
+                break
+            await asyncio.sleep(0.05)
+
+        store = WebStore(base_url=directory.url, key="k")
+        app = create_app(store=store)
+        try:
+            async with app.router.lifespan_context(app):
+                transport = httpx.ASGITransport(app=app)
+                async with httpx.AsyncClient(transport=transport,
+                                             base_url="http://gw") as client:
+                    r = await client.post(
+                        "/api/p2p/generate",
+                        json={"task": {"prompt": "hit the target",
+                                       "model": "echo-model",
+                                       "targetNode": "127.0.0.1:18462"}})
+                    assert r.status_code == 200
+                    assert "echo:hit" in r.text
+            tallies = [m for m in directory.messages
+                       if m["content"] == "[Metric Log]"]
+            assert len(tallies) == 1
+            assert tallies[0]["node_id"] == "127.0.0.1:18462"
+        finally:
+            server.should_exit = True
+            await asyncio.sleep(0.2)
+            task.cancel()
+            await node.stop()
+            await directory.stop()
+
+    asyncio.run(run())
