@@ -116,6 +116,13 @@ def main() -> None:
     from bee2bee_amd.engine.engine import InferenceEngine
 
     budget = args.warmup + args.steps + 4
+    # models with short contexts (gpt2 family: 1024) can't hold the default
+    # 1024-token prompt plus the step budget — clamp instead of asserting
+    max_prompt = _spec_probe.max_seq_len - budget - 2
+    if args.prompt_len > max_prompt:
+        print(f"# prompt-len {args.prompt_len} > {_spec_probe.name} context "
+              f"budget; clamping to {max_prompt}", file=sys.stderr)
+        args.prompt_len = max_prompt
     engine = InferenceEngine(
         args.model,
         device=device,
