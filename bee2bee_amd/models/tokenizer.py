@@ -64,9 +64,38 @@ class HFTokenizer:
         return self.tk.decode(ids)
 
 
+class BPETokenizer:
+    """Older GPT-2-style checkpoints ship vocab.json + merges.txt instead
+    of tokenizer.json; build a byte-level BPE from them."""
+
+    def __init__(self, vocab_path: str, merges_path: str,
+                 eos_id: int = 50256) -> None:
+        from tokenizers import Tokenizer, decoders, models, pre_tokenizers
+
+        tk = Tokenizer(models.BPE.from_file(vocab_path, merges_path))
+        tk.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+        tk.decoder = decoders.ByteLevel()
+        self.tk = tk
+        self.vocab_size = tk.get_vocab_size()
+        tid = tk.token_to_id("<|endoftext|>")
+        self.bos_token_id = tid if tid is not None else eos_id
+        self.eos_token_id = self.bos_token_id
+        self.pad_token_id = self.bos_token_id
+
+    def encode(self, text: str, add_bos: bool = False) -> List[int]:
+        return self.tk.encode(text).ids  # GPT-2 adds no BOS
+
+    def decode(self, ids: List[int]) -> str:
+        return self.tk.decode(ids)
+
+
 def load_tokenizer(model_path: Optional[str], vocab_size: int, bos_id: int, eos_id: int):
     if model_path:
         tj = os.path.join(model_path, "tokenizer.json")
         if os.path.isfile(tj):
             return HFTokenizer(tj)
+        vj = os.path.join(model_path, "vocab.json")
+        mg = os.path.join(model_path, "merges.txt")
+        if os.path.isfile(vj) and os.path.isfile(mg):
+            return BPETokenizer(vj, mg, eos_id=eos_id)
     return ByteTokenizer(vocab_size=vocab_size, bos_id=bos_id, eos_id=eos_id)

@@ -48,3 +48,21 @@ def test_hf_tokenizer_json_loaded(tmp_path):
 def test_missing_tokenizer_json_falls_back(tmp_path):
     tok = load_tokenizer(str(tmp_path), 512, 1, 2)
     assert isinstance(tok, ByteTokenizer)
+
+
+def test_bpe_vocab_merges_fallback(tmp_path):
+    """GPT-2-era checkpoints with vocab.json + merges.txt (no
+    tokenizer.json) load as a byte-level BPE."""
+    import json
+
+    vocab = {"<|endoftext|>": 0, "h": 1, "e": 2, "l": 3, "o": 4,
+             "he": 5, "ll": 6, "hell": 7, "hello": 8, "Ġ": 9}
+    json.dump(vocab, open(tmp_path / "vocab.json", "w"))
+    (tmp_path / "merges.txt").write_text(
+        "#version: 0.2\nh e\nl l\nhe ll\nhell o\n")
+    from bee2bee_amd.models.tokenizer import load_tokenizer
+
+    tk = load_tokenizer(str(tmp_path), 512, 1, 2)
+    ids = tk.encode("hello")
+    assert tk.decode(ids) == "hello"
+    assert tk.eos_token_id == 0  # <|endoftext|>
