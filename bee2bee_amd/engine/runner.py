@@ -48,7 +48,8 @@ class Runner:
         self.tp_group = None
         self.decode_attn_fn = None  # CP hook (parallel/cp.py)
         cos, sin = ops.rope_tables(
-            spec.max_seq_len, spec.head_dim, spec.rope_theta, device
+            spec.max_seq_len, spec.head_dim, spec.rope_theta, device,
+            scaling=spec.rope_scaling,
         )
         self.rope_cos, self.rope_sin = cos, sin
 

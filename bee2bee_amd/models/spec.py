@@ -41,6 +41,10 @@ class ModelSpec:
     pos_type: str = "rope"       # "rope" | "learned"
     attn_out_bias: bool = False
     mlp_bias: bool = False
+    # llama3-type RoPE scaling (Llama-3.1/3.2 checkpoints):
+    # {"rope_type": "llama3", "factor", "low_freq_factor",
+    #  "high_freq_factor", "original_max_position_embeddings"}
+    rope_scaling: Optional[Dict[str, Any]] = None
     # vocab specials (byte-tokenizer defaults; overridden by a real tokenizer)
     bos_token_id: int = 1
     eos_token_id: int = 2
@@ -164,6 +168,9 @@ PRESETS: Dict[str, ModelSpec] = {
         n_kv_heads=8,
         head_dim=64,
         tie_embeddings=True,
+        rope_scaling={"rope_type": "llama3", "factor": 32.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
     ),
     "llama3-8b": _llama(
         "llama3-8b",
@@ -343,6 +350,12 @@ def spec_from_hf_config(model_path: str, name: Optional[str] = None) -> ModelSpe
         rms_eps=cfg.get("rms_norm_eps", 1e-5),
         max_seq_len=min(cfg.get("max_position_embeddings", 8192), 131072),
         tie_embeddings=cfg.get("tie_word_embeddings", False),
+        rope_scaling=(
+            cfg.get("rope_scaling")
+            if (cfg.get("rope_scaling") or {}).get(
+                "rope_type", (cfg.get("rope_scaling") or {}).get("type"))
+            == "llama3" else None
+        ),
         n_experts=cfg.get("num_local_experts", 0),
         top_k_experts=cfg.get("num_experts_per_tok", 2),
         # Qwen2 sets attention_bias (or ships q/k/v bias tensors implicitly)

@@ -61,11 +61,31 @@ def gelu(x: torch.Tensor) -> torch.Tensor:
 
 
 def rope_tables(
-    max_len: int, head_dim: int, theta: float, device, dtype=torch.float32
+    max_len: int, head_dim: int, theta: float, device, dtype=torch.float32,
+    scaling: "Optional[dict]" = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Host-precomputed cos/sin tables. `scaling` supports the llama3
+    rope_type (Llama-3.1/3.2 checkpoints): frequencies below the low-freq
+    wavelength divide by `factor`, above high-freq stay, in between
+    interpolate smoothly — matches transformers' llama3 rope init."""
+    import math
+
     inv_freq = 1.0 / (
         theta ** (torch.arange(0, head_dim, 2, dtype=torch.float32) / head_dim)
     )
+    if scaling and scaling.get("rope_type", scaling.get("type")) == "llama3":
+        factor = float(scaling["factor"])
+        lo_f = float(scaling.get("low_freq_factor", 1.0))
+        hi_f = float(scaling.get("high_freq_factor", 4.0))
+        orig = float(scaling.get("original_max_position_embeddings", 8192))
+        low_wl = orig / lo_f
+        high_wl = orig / hi_f
+        wavelen = 2 * math.pi / inv_freq
+        scaled = torch.where(wavelen > low_wl, inv_freq / factor, inv_freq)
+        smooth = (orig / wavelen - lo_f) / (hi_f - lo_f)
+        mid = (1 - smooth) * inv_freq / factor + smooth * inv_freq
+        is_mid = (wavelen <= low_wl) & (wavelen >= high_wl)
+        inv_freq = torch.where(is_mid, mid, scaled)
     t = torch.arange(max_len, dtype=torch.float32)
     freqs = torch.outer(t, inv_freq)  # [max_len, hd/2]
     return freqs.cos().to(device, dtype), freqs.sin().to(device, dtype)

@@ -105,3 +105,49 @@ def test_qwen2_preset_and_bias_roundtrip(tmp_path):
 
     with_bias = gen(tiny_q)
     assert len(with_bias) == 5
+
+
+def test_llama3_rope_scaling_matches_transformers():
+    """rope_tables with the llama3 rope_type must reproduce transformers'
+    ROPE_INIT_FUNCTIONS['llama3'] frequencies exactly (Llama-3.1/3.2
+    checkpoint compatibility)."""
+    import pytest
+    import torch
+
+    transformers = pytest.importorskip("transformers")
+    from transformers.modeling_rope_utils import ROPE_INIT_FUNCTIONS
+
+    from bee2bee_amd.ops.reference import rope_tables
+
+    scaling = {"rope_type": "llama3", "factor": 32.0,
+               "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+               "original_max_position_embeddings": 8192}
+    cfg = transformers.LlamaConfig(
+        hidden_size=2048, num_attention_heads=32, head_dim=64,
+        rope_theta=500000.0, max_position_embeddings=131072,
+        rope_scaling=dict(scaling))
+    hf_inv, _ = ROPE_INIT_FUNCTIONS["llama3"](cfg, "cpu")
+    cos, sin = rope_tables(64, 64, 500000.0, "cpu", scaling=scaling)
+    ours = torch.atan2(sin[1], cos[1])  # inv_freq from position-1 angles
+    assert (ours - hf_inv).abs().max().item() < 1e-6
+
+
+def test_rope_scaling_parsed_from_hf_config(tmp_path):
+    import json
+
+    cfg = {"vocab_size": 1024, "hidden_size": 64, "intermediate_size": 128,
+           "num_hidden_layers": 2, "num_attention_heads": 4,
+           "num_key_value_heads": 2, "head_dim": 16,
+           "rope_theta": 500000.0,
+           "rope_scaling": {"rope_type": "llama3", "factor": 8.0,
+                            "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                            "original_max_position_embeddings": 8192}}
+    (tmp_path / "config.json").write_text(json.dumps(cfg))
+    from bee2bee_amd.models.spec import resolve_spec
+
+    spec = resolve_spec("x", model_path=str(tmp_path))
+    assert spec.rope_scaling["factor"] == 8.0
+    # non-llama3 types are ignored (unsupported) rather than misapplied
+    cfg["rope_scaling"] = {"rope_type": "yarn", "factor": 4.0}
+    (tmp_path / "config.json").write_text(json.dumps(cfg))
+    assert resolve_spec("x", model_path=str(tmp_path)).rope_scaling is None
