@@ -96,3 +96,43 @@ def test_ollama_missing_prompt():
     svc = OllamaService("gemma3", host="http://127.0.0.1:9")
     with pytest.raises(ServiceError):
         svc.execute({})
+
+
+class _FakeHFClient:
+    def __init__(self, fail=False):
+        self.fail = fail
+        self.calls = []
+
+    def text_generation(self, prompt, **kw):
+        self.calls.append((prompt, kw))
+        if self.fail:
+            raise RuntimeError("rate limited")
+        return "remote says: " + prompt
+
+
+def test_hf_remote_execute_and_pricing():
+    from bee2bee_amd.services.hf_remote import HFRemoteService
+
+    svc = HFRemoteService("some/model", token="t", price_per_token=0.01)
+    svc.client = _FakeHFClient()
+    res = svc.execute({"prompt": "ping", "max_new_tokens": 8})
+    assert res["text"] == "remote says: ping"
+    assert res["cost"] == pytest.approx(res["tokens"] * 0.01)
+    assert svc.client.calls[0][1]["max_new_tokens"] == 8
+
+    chunks = [json.loads(x) for x in svc.execute_stream({"prompt": "s"})]
+    assert chunks[0]["text"].startswith("remote says")
+    assert chunks[-1]["done"] is True
+
+
+def test_hf_remote_error_paths():
+    from bee2bee_amd.services.hf_remote import HFRemoteService
+
+    svc = HFRemoteService("m", token="t")
+    with pytest.raises(ServiceError):
+        svc.execute({"prompt": "x"})  # not initialized
+    svc.client = _FakeHFClient(fail=True)
+    with pytest.raises(ServiceError):
+        svc.execute({"prompt": "x"})
+    stream = list(svc.execute_stream({"prompt": "x"}))
+    assert json.loads(stream[0])["status"] == "error"
