@@ -79,3 +79,30 @@ def test_export_with_qkv_bias_matches_engine():
     with torch.no_grad():
         got = model(torch.tensor([ids], dtype=torch.int64))[0]
     assert torch.allclose(got, ref, atol=1e-4), (got - ref).abs().max()
+
+
+def test_gpt2_export_matches_transformers(tmp_path):
+    """TorchScript export of a GPT-2-family model: traced logits equal
+    transformers' GPT2LMHeadModel on the same checkpoint."""
+    import pytest
+    import torch
+
+    transformers = pytest.importorskip("transformers")
+    from bee2bee_amd.models.export import export_torchscript
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+    spec = PRESETS["tiny-gpt2"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(5)
+    path = str(tmp_path / "m.pt")
+    export_torchscript(spec, w, path, example_len=6)
+    traced = torch.jit.load(path)
+
+    save_hf(w, str(tmp_path / "hf"))
+    hf = transformers.GPT2LMHeadModel.from_pretrained(
+        str(tmp_path / "hf"), torch_dtype=torch.float32).eval()
+    ids = torch.tensor([[3, 9, 100, 7, 45, 2]])
+    with torch.no_grad():
+        got = traced(ids)
+        want = hf(ids).logits
+    assert torch.allclose(got, want, atol=2e-3), (got - want).abs().max()
