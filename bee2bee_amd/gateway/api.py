@@ -99,6 +99,11 @@ class ChatRequest(BaseModel):
     model: Optional[str] = None
     max_new_tokens: Optional[int] = None
     temperature: Optional[float] = 0.7
+    # None -> the engine applies the reference generation defaults
+    # (top_p 0.95, repetition_penalty 1.15 — bee2bee/hf.py:94-103)
+    top_p: Optional[float] = None
+    top_k: Optional[int] = None
+    repetition_penalty: Optional[float] = None
     stream: Optional[bool] = False
 
 
@@ -201,6 +206,9 @@ async def chat(req: ChatRequest):
                 "prompt": req.prompt,
                 "max_new_tokens": req.max_new_tokens or 2048,
                 "temperature": req.temperature or 0.7,
+                "top_p": req.top_p,
+                "top_k": req.top_k,
+                "repetition_penalty": req.repetition_penalty,
             }
             if req.stream:
                 if hasattr(svc, "execute_stream_async"):

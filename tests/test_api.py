@@ -221,3 +221,36 @@ def test_connect_endpoint_joins_peer(client):
         assert found, peers
     finally:
         asyncio.run_coroutine_threadsafe(peer.stop(), loop).result(timeout=15)
+
+
+def test_generate_forwards_sampling_knobs(client):
+    """HTTP sampling knobs reach the service params (reference generation
+    surface: temperature/top_p/repetition_penalty)."""
+    from bee2bee_amd.gateway import api as gateway_api
+
+    seen = {}
+
+    class Probe:
+        name = "hf"
+        price_per_token = 0.0
+
+        def get_metadata(self):
+            return {"models": ["probe-model"], "price_per_token": 0.0}
+
+        def execute(self, params):
+            seen.update(params)
+            return {"text": "ok", "tokens": 1, "latency_ms": 1,
+                    "price_per_token": 0.0, "cost": 0.0}
+
+    gateway_api.node.local_services["hf"] = Probe()
+    r = client.post(
+        "/generate",
+        headers={"X-API-KEY": "secret-key"},
+        json={"prompt": "x", "model": "probe-model", "temperature": 0.9,
+              "top_p": 0.5, "top_k": 11, "repetition_penalty": 1.3},
+    )
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+    assert seen["temperature"] == 0.9
+    assert seen["top_p"] == 0.5
+    assert seen["top_k"] == 11
+    assert seen["repetition_penalty"] == 1.3
