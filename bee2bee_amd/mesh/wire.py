@@ -145,13 +145,19 @@ def request_params(data: Dict[str, Any]) -> Dict[str, Any]:
 
     Accepts both `max_new_tokens` and the legacy `max_tokens` key and both
     `rid` and legacy `task_id` (reference :574-586)."""
-    return {
+    out = {
         "prompt": data.get("prompt", ""),
         "max_new_tokens": int(
             data.get("max_new_tokens") or data.get("max_tokens") or 2048
         ),
         "temperature": float(data.get("temperature", 0.7)),
     }
+    # optional sampling knobs pass through when present (None -> the
+    # engine applies the reference generation defaults)
+    for key in ("top_p", "top_k", "repetition_penalty"):
+        if data.get(key) is not None:
+            out[key] = data[key]
+    return out
 
 
 def request_id(data: Dict[str, Any]) -> Optional[str]:

@@ -63,3 +63,13 @@ def test_reference_shaped_messages_parse():
     # hello without services/metrics (early-handshake JS bridge shape)
     bare = {"type": "hello", "peer_id": "p1", "addr": "ws://x:1"}
     assert bare.get("services") is None  # nodes must tolerate absence
+
+
+def test_request_params_optional_sampling_knobs():
+    from bee2bee_amd.mesh import wire
+
+    p = wire.request_params({"prompt": "x", "top_p": 0.4,
+                             "repetition_penalty": 1.2})
+    assert p["top_p"] == 0.4 and p["repetition_penalty"] == 1.2
+    p2 = wire.request_params({"prompt": "x"})
+    assert "top_p" not in p2  # absent stays absent (engine defaults apply)
