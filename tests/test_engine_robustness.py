@@ -240,3 +240,55 @@ def test_cancel_during_chunked_prefill():
         assert eng.kv.free_blocks == base, (eng.kv.free_blocks, base)
     finally:
         eng.shutdown()
+
+
+def test_randomized_workload_soak():
+    """Property test: a burst of randomized requests (lengths, sampling
+    params, penalties, cancels, stop tokens) completes with no errors and
+    the KV pool returns exactly to baseline."""
+    import random
+
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    rng = random.Random(42)
+    eng = InferenceEngine("tiny", device="cpu", max_batch=4, max_seq_len=256,
+                          seed=7, max_prefill_tokens=64)
+    try:
+        base = eng.kv.free_blocks
+        reqs = []
+        for i in range(24):
+            n = rng.randrange(1, 180)
+            sp = rng.choice([
+                SamplingParams(greedy=True),
+                SamplingParams(greedy=True, repetition_penalty=1.15),
+                SamplingParams(temperature=0.8, top_p=0.9, top_k=20),
+                SamplingParams(temperature=1.2, repetition_penalty=1.3),
+            ])
+            req = GenerationRequest(
+                prompt_ids=[rng.randrange(4, 500) for _ in range(n)],
+                max_new_tokens=rng.randrange(1, 12),
+                sampling=sp,
+                stop_token_ids=(rng.randrange(4, 500),) if rng.random() < 0.3
+                else (),
+            )
+            eng.submit(req)
+            if rng.random() < 0.2:
+                req.cancelled = True
+            reqs.append(req)
+        for req in reqs:
+            while True:
+                item = req.out_queue.get(timeout=120)
+                if not isinstance(item, int):
+                    break
+            assert req.error is None, req.error
+            assert len(req.output_ids) <= req.max_new_tokens
+        import time as _t
+        for _ in range(300):
+            if eng.kv.free_blocks == base and not eng._active \
+                    and not eng._prefilling:
+                break
+            _t.sleep(0.02)
+        assert eng.kv.free_blocks == base, (eng.kv.free_blocks, base)
+    finally:
+        eng.shutdown()
