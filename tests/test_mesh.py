@@ -496,3 +496,49 @@ def test_node_survives_adversarial_frames():
             await node.stop()
 
     asyncio.run(run())
+
+
+def test_mesh_churn_requests_keep_flowing():
+    """Peer churn: providers joining and leaving mid-traffic. Requests to
+    live providers keep succeeding; requests to dead ones fail with typed
+    errors, never hang."""
+
+    async def run():
+        hub = await _start_node()
+        requester = await _start_node()
+        await requester.connect_bootstrap(hub.addr)
+        await _wait_for(lambda: hub.peer_id in requester.peers)
+
+        for round_ in range(3):
+            prov = await _start_node()
+            await prov.add_service(EchoService(model=f"churn-{round_}"))
+            await prov.connect_bootstrap(hub.addr)
+            def has_model(name):
+                return any(
+                    name in m.get("models", [])
+                    for svcs in hub.providers.values()
+                    for m in svcs.values() if isinstance(m, dict)
+                )
+
+            await _wait_for(lambda: has_model(f"churn-{round_}"))
+            res = await requester.request_generation(
+                hub.peer_id, f"ping {round_}", 8, f"churn-{round_}",
+                timeout=10)
+            assert res["text"] == f"echo:ping {round_}"
+            await prov.stop()  # churn out
+            await _wait_for(lambda: not has_model(f"churn-{round_}"),
+                            timeout=10)
+            # now the model is gone: typed failure, not a hang
+            try:
+                await requester.request_generation(
+                    hub.peer_id, "x", 4, f"churn-{round_}", timeout=5)
+                raise AssertionError("expected a typed failure")
+            except AssertionError:
+                raise
+            except Exception as e:
+                assert "consensus_deadlock" in str(e) or "no_local" in str(e) \
+                    or "timed_out" in str(e), e
+        await requester.stop()
+        await hub.stop()
+
+    asyncio.run(run())
