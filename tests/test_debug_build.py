@@ -9,16 +9,18 @@ import pytest
 HIPCC = shutil.which("hipcc") or "/opt/rocm/bin/hipcc"
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 SRC = os.path.join(REPO, "bee2bee_amd", "ops", "csrc", "grouped_gemm.hip")
+SRC_ATTN = os.path.join(REPO, "bee2bee_amd", "ops", "csrc", "attn_decode.hip")
 
 
 @pytest.mark.skipif(not os.path.exists(HIPCC), reason="no hipcc")
-@pytest.mark.timeout(300)
-def test_debug_assert_build_compiles(tmp_path):
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("src", [SRC, SRC_ATTN])
+def test_debug_assert_build_compiles(tmp_path, src):
     for extra in ([], ["-DBEE2BEE_DEBUG"]):
         r = subprocess.run(
-            [HIPCC, "-c", SRC, "-o", str(tmp_path / "gg.o"),
+            [HIPCC, "-c", src, "-o", str(tmp_path / "k.o"),
              "--offload-arch=gfx950", "-O2", "-std=c++17", *extra],
-            capture_output=True, text=True, timeout=280,
+            capture_output=True, text=True, timeout=560,
         )
         assert r.returncode == 0, r.stderr[-2000:]
 
