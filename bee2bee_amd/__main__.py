@@ -14,6 +14,8 @@ from .utils import setup_logging
 
 
 @click.group()
+@click.version_option(package_name=None, version=__import__(
+    "bee2bee_amd").__version__, prog_name="bee2bee-amd")
 def cli() -> None:
     """Bee2Bee-AMD: MI355X-native decentralized inference mesh."""
     setup_logging()

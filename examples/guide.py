@@ -107,7 +107,29 @@ async def main() -> None:
     console.print(Panel(result["text"], title="gen_result",
                         border_style="green"))
 
-    console.print("[bold]6. Teardown[/bold]")
+    console.print("[bold]6. Torrent-style weight distribution[/bold] — seed "
+                  "a checkpoint on the provider, fetch it over the wire:")
+    import tempfile
+
+    import torch
+
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+    from bee2bee_amd.mesh.weightshare import fetch_checkpoint, seed_checkpoint
+
+    src = tempfile.mkdtemp()
+    dst = tempfile.mkdtemp()
+    w = ModelWeights(PRESETS["tiny"], torch.device("cpu"),
+                     torch.float32).random_init(1)
+    save_hf(w, src)
+    manifest = await seed_checkpoint(provider, provider.dht, "demo-ckpt",
+                                     src, piece_size=65536)
+    await asyncio.sleep(0.3)  # one-hop DHT replication
+    await fetch_checkpoint(edge, edge.dht, "demo-ckpt", dst)
+    files = ", ".join(f["name"] for f in manifest["files"])
+    console.print(f"   fetched + hash-verified: [green]{files}[/green]")
+
+    console.print("[bold]7. Teardown[/bold]")
     await edge.stop()
     await provider.stop()
     console.print("[green]done[/green] — next steps: serve a real model with "
