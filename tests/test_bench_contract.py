@@ -96,3 +96,21 @@ def test_preflight_cpu():
     assert report["preflight"] == "ok"
     assert report["checks"]["gloo_wiring"]["status"] == "pass"
     assert report["checks"]["master"]["status"] == "pass"
+
+
+def test_preflight_unit_checks():
+    """Individual preflight checks behave sanely off-GPU."""
+    from bee2bee_amd.parallel import preflight as pf
+
+    assert pf._check_env()["status"] in ("pass", "warn")
+    m = pf._check_master()
+    assert m["status"] == "pass", m
+    d = pf._check_devices(8)
+    assert d["status"] in ("pass", "warn")
+    r = pf._check_rccl_self()
+    assert r["status"] in ("pass", "warn")  # warn: no GPU here
+    t = pf._check_tunableop(str(REPO) + "/definitely-missing-dir")
+    assert t["status"] == "warn"
+    report = pf.run_preflight(2, skip=["gloo_wiring"])
+    assert "gloo_wiring" not in report["checks"]
+    assert report["preflight"] in ("ok", "failed")
