@@ -316,3 +316,31 @@ def test_repetition_penalty_seen_mask_equals_history_rebuild():
             eng.shutdown()
 
     assert run(False) == run(True)
+
+
+def test_engine_stats_surface():
+    """stats() exposes the observability keys the API home endpoint and
+    the serve bench rely on (incl. the r2 busy-time counters)."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=96,
+                          seed=1)
+    try:
+        req = GenerationRequest(prompt_ids=[5, 6, 7], max_new_tokens=4,
+                                sampling=SamplingParams(greedy=True))
+        eng.submit(req)
+        while True:
+            item = req.out_queue.get(timeout=60)
+            if not isinstance(item, int):
+                break
+        s = eng.stats()
+        for key in ("model", "device", "kv_free_blocks", "kv_total_blocks",
+                    "tokens_total", "tokens_per_sec_10s", "decode_graphs",
+                    "engine_busy_s", "engine_steps", "engine_ms_per_step"):
+            assert key in s, key
+        assert s["tokens_total"] >= 4
+        assert s["engine_steps"] >= 1
+        assert s["engine_busy_s"] > 0
+    finally:
+        eng.shutdown()

@@ -1,0 +1,57 @@
+"""PagedKV pool mechanics: allocation, growth, recycling, exhaustion."""
+import pytest
+import torch
+
+from bee2bee_amd.engine.kv import BLOCK_SIZE, PagedKV
+from bee2bee_amd.models.spec import PRESETS
+
+
+def _pool(n_blocks=4):
+    return PagedKV(PRESETS["tiny"], torch.device("cpu"), torch.float32,
+                   n_blocks=n_blocks)
+
+
+def test_page_size_is_decode_chunk():
+    # one decode chunk (DEC_CHUNK=256) == one page: the r2 layout contract
+    assert BLOCK_SIZE == 256
+
+
+def test_alloc_grow_free_recycle():
+    kv = _pool(4)
+    kv.new_seq(1)
+    kv.extend_seq(1, 1)
+    assert kv.free_blocks == 3
+    kv.extend_seq(1, BLOCK_SIZE + 1)  # crosses into a second page
+    assert kv.free_blocks == 2
+    kv.extend_seq(1, BLOCK_SIZE)      # shrink request is a no-op
+    assert kv.free_blocks == 2
+    kv.new_seq(2)
+    kv.extend_seq(2, 2 * BLOCK_SIZE)
+    assert kv.free_blocks == 0
+    kv.free_seq(1)
+    assert kv.free_blocks == 2        # pages recycled
+    kv.new_seq(3)
+    kv.extend_seq(3, 2 * BLOCK_SIZE)  # reuses the freed pages
+    assert kv.free_blocks == 0
+
+
+def test_pool_exhaustion_raises():
+    kv = _pool(2)
+    kv.new_seq(1)
+    kv.extend_seq(1, 2 * BLOCK_SIZE)
+    kv.new_seq(2)
+    with pytest.raises(RuntimeError, match="KV pool exhausted"):
+        kv.extend_seq(2, 1)
+
+
+def test_slot_mapping_spans_pages():
+    kv = _pool(4)
+    kv.new_seq(9)
+    kv.extend_seq(9, BLOCK_SIZE + 5)
+    slots = kv.slot_mapping(9, [0, BLOCK_SIZE - 1, BLOCK_SIZE,
+                                BLOCK_SIZE + 4])
+    blocks = kv.block_table([9])[0]
+    assert slots[0] == int(blocks[0]) * BLOCK_SIZE
+    assert slots[1] == int(blocks[0]) * BLOCK_SIZE + BLOCK_SIZE - 1
+    assert slots[2] == int(blocks[1]) * BLOCK_SIZE
+    assert slots[3] == int(blocks[1]) * BLOCK_SIZE + 4
