@@ -104,3 +104,25 @@ def test_distilgpt2_served_by_name():
         assert len(req.output_ids) == 3
     finally:
         eng.shutdown()
+
+
+def test_gpt2_layer_range_load(tmp_path):
+    """PP stages load only their layer slice of a GPT-2 checkpoint (embed/
+    pos on the first stage, ln_f/head on the last)."""
+    spec = PRESETS["tiny-gpt2"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(9)
+    save_hf(w, str(tmp_path))
+
+    first = ModelWeights(spec, torch.device("cpu"), torch.float32).load_hf(
+        str(tmp_path), layer_range=(0, 1))
+    assert first.embed is not None and first.pos_embed is not None
+    assert first.layers[0].wqkv is not None
+    assert first.layers[1].wqkv is None
+    assert first.final_norm is None  # not the last stage
+
+    last = ModelWeights(spec, torch.device("cpu"), torch.float32).load_hf(
+        str(tmp_path), layer_range=(1, 2))
+    assert last.embed is None
+    assert last.layers[0].wqkv is None
+    assert last.layers[1].wqkv is not None
+    assert last.final_norm is not None and last.final_norm_bias is not None
