@@ -214,6 +214,10 @@ class ModelWeights:
             if name == "model.embed_tokens.weight":
                 if want_embed:
                     self.embed = to_dev(tensor)
+                elif want_head and s.tie_embeddings:
+                    # PP LAST stage of a tied checkpoint: the head IS the
+                    # embedding matrix, which only ships as embed_tokens
+                    self.lm_head = to_dev(tensor)
                 continue
             if name == "model.norm.weight":
                 if want_head:
@@ -318,6 +322,8 @@ class ModelWeights:
             if name == "wte.weight":
                 if lo == 0:
                     self.embed = to_dev(tensor)
+                elif hi == s.n_layers:
+                    self.lm_head = to_dev(tensor)  # tied head, PP last stage
                 continue
             if name == "wpe.weight":
                 if lo == 0:

@@ -101,3 +101,25 @@ def test_expert_range_load_hf(tmp_path):
     assert torch.allclose(
         shard.layers[1].moe_w_down, full.layers[1].moe_w_down[1:3]
     )
+
+
+def test_tied_head_loads_on_pp_last_stage(tmp_path):
+    """PP last stages of tied-embedding checkpoints must get lm_head from
+    the embedding tensor (regression: both loaders left it None)."""
+    import torch
+
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+    for preset in ("tiny", "tiny-gpt2"):  # llama-format and gpt2-format
+        spec = PRESETS[preset]
+        assert spec.tie_embeddings
+        w = ModelWeights(spec, torch.device("cpu"),
+                         torch.float32).random_init(6)
+        out = tmp_path / preset
+        save_hf(w, str(out))
+        last = ModelWeights(spec, torch.device("cpu"), torch.float32).load_hf(
+            str(out), layer_range=(1, spec.n_layers))
+        assert last.embed is None, preset
+        assert last.lm_head is not None, preset
+        assert torch.equal(last.lm_head, w.embed), preset
