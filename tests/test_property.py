@@ -1,0 +1,143 @@
+"""Property-based tests (hypothesis) for the round's pure algorithms.
+
+Each test pins an optimized implementation to a brute-force reference:
+- NGramIndex (incremental longest-match proposer) vs an O(L·n) rescan
+- batched repetition penalty (single scatter) vs a per-row Python loop
+- content sharding split/hash/reassemble roundtrip on arbitrary blobs
+
+These run on CPU in CI; random structure (tiny alphabets, short grams)
+forces the collision/repeat cases hand-written examples tend to miss.
+"""
+import pytest
+import torch
+
+hypothesis = pytest.importorskip("hypothesis")
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from bee2bee_amd.engine.sampler import apply_repetition_penalty
+from bee2bee_amd.engine.spec import NGramIndex
+from bee2bee_amd.mesh import pieces as P
+
+token_streams = st.lists(st.integers(min_value=0, max_value=5),
+                         min_size=0, max_size=64)
+
+
+def brute_force_propose(ids, k, ns=(4, 3, 2)):
+    """Reference semantics: continuation of the MOST RECENT strictly-earlier
+    occurrence of the longest matching trailing gram."""
+    L = len(ids)
+    for n in sorted(ns, reverse=True):
+        if L <= n:
+            continue
+        gram = ids[L - n:]
+        pos = next(
+            (p for p in range(L - n - 1, -1, -1) if ids[p:p + n] == gram),
+            None,
+        )
+        if pos is None:
+            continue
+        cont = ids[pos + n: pos + n + k]
+        if cont:
+            return cont
+    return []
+
+
+class TestNGramIndex:
+    @given(ids=token_streams, k=st.integers(min_value=1, max_value=8))
+    @settings(max_examples=300, deadline=None)
+    def test_matches_bruteforce(self, ids, k):
+        idx = NGramIndex(ids)
+        assert idx.propose(k) == brute_force_propose(ids, k)
+
+    @given(ids=token_streams, split=st.integers(min_value=0, max_value=64))
+    @settings(max_examples=100, deadline=None)
+    def test_incremental_equals_batch(self, ids, split):
+        """Building via extend+append+sync must equal building at once."""
+        split = min(split, len(ids))
+        a = NGramIndex(ids)
+        b = NGramIndex(ids[:split])
+        for t in ids[split:]:
+            b.append(t)
+        c = NGramIndex()
+        c.sync(ids)
+        assert a.propose(6) == b.propose(6) == c.propose(6)
+        assert a.maps == b.maps == c.maps
+
+    @given(ids=token_streams)
+    @settings(max_examples=50, deadline=None)
+    def test_propose_zero_k(self, ids):
+        assert NGramIndex(ids).propose(0) == []
+
+
+class TestRepetitionPenalty:
+    @given(
+        b=st.integers(min_value=1, max_value=4),
+        v=st.integers(min_value=2, max_value=24),
+        data=st.data(),
+        penalty=st.sampled_from([1.0, 1.15, 2.0]),
+    )
+    @settings(max_examples=150, deadline=None)
+    def test_matches_row_loop(self, b, v, data, penalty):
+        L = data.draw(st.integers(min_value=1, max_value=12))
+        prev = data.draw(
+            st.lists(
+                st.lists(st.integers(min_value=-1, max_value=v - 1),
+                         min_size=L, max_size=L),
+                min_size=b, max_size=b,
+            )
+        )
+        torch.manual_seed(0)
+        logits = torch.randn(b, v)
+        prev_t = torch.tensor(prev, dtype=torch.int64)
+        got = apply_repetition_penalty(logits.clone(), prev_t, penalty)
+
+        want = logits.clone()
+        for i in range(b):
+            for tok in set(t for t in prev[i] if t >= 0):
+                x = want[i, tok].item()
+                want[i, tok] = x / penalty if x > 0 else x * penalty
+        assert torch.allclose(got, want), (got - want).abs().max()
+
+    def test_pad_only_rows_untouched(self):
+        logits = torch.randn(3, 8)
+        prev = torch.full((3, 5), -1, dtype=torch.int64)
+        out = apply_repetition_penalty(logits.clone(), prev, 1.15)
+        assert torch.equal(out, logits)
+
+
+class TestPieces:
+    @given(
+        blob=st.binary(min_size=0, max_size=4096),
+        piece_size=st.integers(min_value=1, max_value=512),
+    )
+    @settings(max_examples=150, deadline=None)
+    def test_split_verify_reassemble_roundtrip(self, blob, piece_size):
+        ps = P.split_pieces(blob, piece_size)
+        assert sum(len(p) for p in ps) == len(blob)
+        assert all(len(p) == piece_size for p in ps[:-1])
+        hashes = P.piece_hashes(ps)
+        assert P.verify_and_reassemble(ps, hashes) == blob
+
+    @given(blob=st.binary(min_size=2, max_size=512),
+           data=st.data())
+    @settings(max_examples=50, deadline=None)
+    def test_corruption_detected_at_index(self, blob, data):
+        ps = P.split_pieces(blob, 7)
+        hashes = P.piece_hashes(ps)
+        i = data.draw(st.integers(min_value=0, max_value=len(ps) - 1))
+        bad = list(ps)
+        flipped = bytearray(bad[i])
+        flipped[0] ^= 0xFF
+        bad[i] = bytes(flipped)
+        with pytest.raises(ValueError, match=f"hash_mismatch_at_{i}"):
+            P.verify_and_reassemble(bad, hashes)
+
+    @given(total=st.integers(min_value=0, max_value=32),
+           have=st.lists(st.integers(min_value=-4, max_value=36), max_size=40))
+    @settings(max_examples=50, deadline=None)
+    def test_bitfield(self, total, have):
+        bf = P.bitfield_from_pieces(total, have)
+        assert len(bf) == total
+        want = {i for i in have if 0 <= i < total}
+        assert {i for i, v in enumerate(bf) if v} == want
