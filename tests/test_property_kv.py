@@ -1,0 +1,157 @@
+"""Property-based tests for the PagedKV block allocator and DHT announce
+semantics: random alloc/extend/free interleavings must conserve the block
+pool exactly (no leak, no double-hand-out, no cross-sequence slot overlap),
+and piece/rank announcements must be idempotent and order-insensitive.
+
+The allocator invariants back the engine's continuous batching — the soak
+tests check end-state equality; these check EVERY intermediate state on
+randomized schedules.
+"""
+import asyncio
+
+import pytest
+import torch
+
+hypothesis = pytest.importorskip("hypothesis")
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from bee2bee_amd.engine.kv import PagedKV
+from bee2bee_amd.models.spec import PRESETS
+
+N_BLOCKS = 12
+BLOCK = 8
+
+
+def make_kv():
+    return PagedKV(PRESETS["tiny"], torch.device("cpu"), torch.float32,
+                   n_blocks=N_BLOCKS, block_size=BLOCK)
+
+
+# ops: (kind, seq_id, amount) — schedules drawn over a small id space so
+# new/free/extend collide and re-use ids aggressively
+ops = st.lists(
+    st.tuples(st.sampled_from(["new", "extend", "free"]),
+              st.integers(min_value=0, max_value=4),
+              st.integers(min_value=1, max_value=3 * BLOCK)),
+    max_size=60,
+)
+
+
+class TestPagedKVAllocator:
+    @given(schedule=ops)
+    @settings(max_examples=200, deadline=None)
+    def test_pool_conservation_under_random_schedule(self, schedule):
+        kv = make_kv()
+        live = set()
+        for kind, sid, amount in schedule:
+            if kind == "new":
+                kv.new_seq(sid)
+                live.add(sid)
+            elif kind == "free":
+                kv.free_seq(sid)
+                live.discard(sid)
+            elif kind == "extend" and sid in live:
+                new_len = kv.seq_len(sid) + amount
+                need = (new_len + BLOCK - 1) // BLOCK - kv.seq_n_blocks(sid)
+                if need > kv.free_blocks:
+                    with pytest.raises(RuntimeError, match="KV pool exhausted"):
+                        kv.extend_seq(sid, new_len)
+                    continue
+                kv.extend_seq(sid, new_len)
+
+            # invariant 1: every block is exactly one of {free, owned-once}
+            owned = [b for s in kv._seq_blocks.values() for b in s]
+            assert len(owned) == len(set(owned)), "block handed out twice"
+            assert set(owned).isdisjoint(kv._free), "owned block also free"
+            assert len(owned) + kv.free_blocks == N_BLOCKS, "pool leak"
+
+            # invariant 2: block count always matches the sequence length
+            for s in live:
+                assert kv.seq_n_blocks(s) == (kv.seq_len(s) + BLOCK - 1) // BLOCK
+
+        # invariant 3: freeing everything restores the full pool
+        for s in list(live):
+            kv.free_seq(s)
+        assert kv.free_blocks == N_BLOCKS
+
+    @given(lens=st.lists(st.integers(min_value=1, max_value=2 * BLOCK),
+                         min_size=1, max_size=4))
+    @settings(max_examples=100, deadline=None)
+    def test_slot_mappings_disjoint_across_sequences(self, lens):
+        kv = make_kv()
+        slots = {}
+        for sid, ln in enumerate(lens):
+            kv.new_seq(sid)
+            kv.extend_seq(sid, ln)
+            s = kv.slot_mapping(sid, range(ln))
+            assert len(s) == len(set(s)), "duplicate slot within a sequence"
+            slots[sid] = set(s)
+        seen = set()
+        for sid, s in slots.items():
+            assert seen.isdisjoint(s), "slot shared across sequences"
+            seen |= s
+        # every slot is inside the pool
+        assert all(0 <= x < N_BLOCKS * BLOCK for x in seen)
+
+    @given(ln=st.integers(min_value=1, max_value=3 * BLOCK))
+    @settings(max_examples=50, deadline=None)
+    def test_block_table_covers_slot_mapping(self, ln):
+        """slot // block_size must be exactly the sequence's block table —
+        the contract between host allocation and the decode kernel's
+        device-side page walk."""
+        kv = make_kv()
+        kv.new_seq(7)
+        kv.extend_seq(7, ln)
+        slots = kv.slot_mapping(7, range(ln))
+        table = kv.block_table([7])[0].tolist()
+        n = kv.seq_n_blocks(7)
+        for p, s in enumerate(slots):
+            assert s // BLOCK == table[p // BLOCK]
+            assert s % BLOCK == p % BLOCK
+        assert table[:n] == kv._seq_blocks[7]
+
+
+class TestDHTAnnounceProperties:
+    @given(addrs=st.lists(st.sampled_from(["a", "b", "c", "d"]),
+                          min_size=1, max_size=12))
+    @settings(max_examples=50, deadline=None)
+    def test_announce_idempotent_and_order_insensitive(self, addrs):
+        from bee2bee_amd.mesh.dht import DHTNode, announce_piece, find_providers
+
+        async def run(seq):
+            dht = DHTNode()
+            await dht.start()
+            for a in seq:
+                await announce_piece(dht, "h" * 8, a)
+            out = await find_providers(dht, "h" * 8)
+            await dht.stop()
+            return out
+
+        fwd = asyncio.run(run(addrs))
+        rev = asyncio.run(run(list(reversed(addrs))))
+        assert sorted(fwd) == sorted(rev) == sorted(set(addrs))
+
+    @given(ranks=st.lists(st.integers(min_value=0, max_value=3),
+                          min_size=1, max_size=8))
+    @settings(max_examples=30, deadline=None)
+    def test_rank_announce_last_write_wins_per_peer(self, ranks):
+        from bee2bee_amd.mesh.dht import DHTNode, announce_rank, find_ranks
+
+        async def run():
+            dht = DHTNode()
+            await dht.start()
+            for i, r in enumerate(ranks):
+                await announce_rank(
+                    dht, "g", f"peer{r}",
+                    {"rank": i, "endpoint": f"127.0.0.1:{5000 + i}"})
+            out = await find_ranks(dht, "g")
+            await dht.stop()
+            return out
+
+        table = asyncio.run(run())
+        assert set(table) == {f"peer{r}" for r in ranks}
+        # each peer's record is its LAST announcement
+        for r in set(ranks):
+            last = max(i for i, x in enumerate(ranks) if x == r)
+            assert table[f"peer{r}"]["rank"] == last
