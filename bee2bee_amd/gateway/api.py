@@ -77,6 +77,31 @@ app.add_middleware(
     allow_headers=["*"],
 )
 
+# OpenAI-compatible surface (/v1/models, /v1/completions,
+# /v1/chat/completions) behind the same optional API key
+from . import metrics as _metrics  # noqa: E402
+from .openai_compat import router as _openai_router  # noqa: E402
+
+app.include_router(_openai_router, dependencies=[Depends(get_api_key)])
+
+
+@app.middleware("http")
+async def _count_requests(request, call_next):
+    response = await call_next(request)
+    if request.url.path != "/metrics":
+        _metrics.HTTP_REQUESTS.labels(
+            path=request.url.path, method=request.method,
+            status=str(response.status_code)).inc()
+    return response
+
+
+@app.get("/metrics")
+def metrics_endpoint():
+    from fastapi import Response
+
+    _metrics.refresh(node)
+    return Response(_metrics.render(), media_type=_metrics.CONTENT_TYPE)
+
 
 class PeerInfo(BaseModel):
     peer_id: str
@@ -260,7 +285,10 @@ async def chat(req: ChatRequest):
             else:
                 pid = node.peer_id
         res = await node.request_generation(
-            pid, req.prompt, req.max_new_tokens or 2048, req.model
+            pid, req.prompt, req.max_new_tokens or 2048, req.model,
+            temperature=req.temperature or 0.7,
+            sampling={"top_p": req.top_p, "top_k": req.top_k,
+                      "repetition_penalty": req.repetition_penalty},
         )
         return {
             "status": "ok",

@@ -1,0 +1,299 @@
+"""OpenAI-compatible API on the node gateway: /v1/models, /v1/completions,
+/v1/chat/completions (buffered + SSE streaming).
+
+Beyond reference parity (the reference exposes only its own /chat and
+/generate shapes): existing OpenAI-client tooling points at a mesh node
+unchanged. Routing reuses the gateway's local-first-then-P2P semantics
+(api.py /chat); chat messages are flattened to the `user:`/`assistant:`
+transcript the reference's web chat sends as its prompt
+(app/src/App.jsx:993-997, parsed by bee2bee/hf.py:56-81).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from typing import Any, AsyncIterator, Dict, List, Optional
+
+from fastapi import APIRouter, Depends, HTTPException
+from fastapi.responses import StreamingResponse
+from pydantic import BaseModel
+
+from ..utils import new_id
+
+router = APIRouter()
+
+
+class CompletionRequest(BaseModel):
+    model: Optional[str] = None
+    prompt: str = ""
+    max_tokens: Optional[int] = 256
+    temperature: Optional[float] = None
+    top_p: Optional[float] = None
+    stream: Optional[bool] = False
+    n: Optional[int] = 1
+    stop: Optional[Any] = None  # str | List[str]
+
+
+class ChatMessage(BaseModel):
+    role: str
+    content: str
+
+
+class ChatCompletionRequest(BaseModel):
+    model: Optional[str] = None
+    messages: List[ChatMessage]
+    max_tokens: Optional[int] = 256
+    temperature: Optional[float] = None
+    top_p: Optional[float] = None
+    stream: Optional[bool] = False
+    n: Optional[int] = 1
+    stop: Optional[Any] = None
+
+
+def _gateway():
+    """The live gateway module (node + auth are its globals)."""
+    from . import api as gateway_api
+
+    if gateway_api.node is None:
+        raise HTTPException(status_code=503, detail="node not running")
+    return gateway_api
+
+
+def _stop_list(stop: Any) -> List[str]:
+    if stop is None:
+        return []
+    return [stop] if isinstance(stop, str) else [s for s in stop if s]
+
+
+def _truncate_at_stop(text: str, stops: List[str]) -> str:
+    for s in stops:
+        i = text.find(s)
+        if i >= 0:
+            text = text[:i]
+    return text
+
+
+def _est_tokens(text: str) -> int:
+    # the directory tally estimate (web/gateway.py): ceil(len/4)
+    return max(0, -(-len(text) // 4))
+
+
+def _pick_service(node, model: Optional[str]):
+    """Local-first with the gateway's fuzzy matching; None -> P2P."""
+    from .api import _model_matches
+
+    for _name, svc in node.local_services.items():
+        if _model_matches(model, svc.get_metadata().get("models", [])):
+            return svc
+    return None
+
+
+async def _run_buffered(node, model: Optional[str], prompt: str,
+                        req) -> Dict[str, Any]:
+    params = {
+        "prompt": prompt,
+        "max_new_tokens": req.max_tokens or 256,
+        "temperature": 0.7 if req.temperature is None else req.temperature,
+    }
+    if req.top_p is not None:
+        params["top_p"] = req.top_p
+    svc = _pick_service(node, model)
+    loop = asyncio.get_running_loop()
+    if svc is not None:
+        return await loop.run_in_executor(None, svc.execute, params)
+    picked = node.pick_provider(model) if model else None
+    pid = picked[0] if picked else node.peer_id
+    return await node.request_generation(
+        pid, prompt, params["max_new_tokens"], model,
+        temperature=params["temperature"],
+        sampling={"top_p": req.top_p},
+    )
+
+
+def _service_chunks(svc, params) -> List[str]:
+    """Drain a service's JSON-lines stream into text deltas."""
+    out = []
+    for line in svc.execute_stream(params):
+        try:
+            obj = json.loads(line)
+        except (TypeError, json.JSONDecodeError):
+            continue
+        if obj.get("text"):
+            out.append(obj["text"])
+    return out
+
+
+async def _sse(events: AsyncIterator[Dict[str, Any]]) -> AsyncIterator[str]:
+    async for ev in events:
+        yield f"data: {json.dumps(ev)}\n\n"
+    yield "data: [DONE]\n\n"
+
+
+async def _stream_deltas(node, model, prompt, req) -> AsyncIterator[str]:
+    """Text deltas from the local service stream (or one buffered burst
+    when only the P2P path exists)."""
+    svc = _pick_service(node, model)
+    stops = _stop_list(req.stop)
+    emitted = ""
+    if svc is not None:
+        params = {
+            "prompt": prompt,
+            "max_new_tokens": req.max_tokens or 256,
+            "temperature": 0.7 if req.temperature is None else req.temperature,
+        }
+        if req.top_p is not None:
+            params["top_p"] = req.top_p
+        loop = asyncio.get_running_loop()
+        queue: asyncio.Queue = asyncio.Queue()
+
+        def _pump() -> None:
+            try:
+                for line in svc.execute_stream(params):
+                    loop.call_soon_threadsafe(queue.put_nowait, line)
+            finally:
+                loop.call_soon_threadsafe(queue.put_nowait, None)
+
+        loop.run_in_executor(None, _pump)
+        while True:
+            line = await queue.get()
+            if line is None:
+                return
+            try:
+                obj = json.loads(line)
+            except (TypeError, json.JSONDecodeError):
+                continue
+            delta = obj.get("text") or ""
+            if not delta:
+                continue
+            if stops:
+                cut = _truncate_at_stop(emitted + delta, stops)
+                if len(cut) < len(emitted) + len(delta):
+                    if len(cut) > len(emitted):
+                        yield cut[len(emitted):]
+                    return
+            emitted += delta
+            yield delta
+    else:
+        result = await _run_buffered(node, model, prompt, req)
+        text = _truncate_at_stop(result.get("text", ""), stops)
+        if text:
+            yield text
+
+
+def _reject_n(n: Optional[int]) -> None:
+    if n is not None and n != 1:
+        raise HTTPException(status_code=400, detail="only n=1 is supported")
+
+
+@router.get("/v1/models")
+async def list_models():
+    gw = _gateway()
+    node = gw.node
+    ids = set()
+    for svc in node.local_services.values():
+        ids.update(svc.get_metadata().get("models", []))
+    for entry in node.list_providers():
+        ids.update(entry.get("models", []))
+    now = int(time.time())
+    return {
+        "object": "list",
+        "data": [{"id": m, "object": "model", "created": now,
+                  "owned_by": "bee2bee-amd"} for m in sorted(ids)],
+    }
+
+
+@router.post("/v1/completions")
+async def completions(req: CompletionRequest):
+    gw = _gateway()
+    _reject_n(req.n)
+    cid = new_id("cmpl")
+    created = int(time.time())
+    model = req.model
+
+    if req.stream:
+        async def events():
+            async for delta in _stream_deltas(gw.node, model, req.prompt, req):
+                yield {
+                    "id": cid, "object": "text_completion",
+                    "created": created, "model": model or "auto",
+                    "choices": [{"index": 0, "text": delta,
+                                 "finish_reason": None}],
+                }
+
+        return StreamingResponse(_sse(events()),
+                                 media_type="text/event-stream")
+
+    result = await _run_buffered(gw.node, model, req.prompt, req)
+    text = _truncate_at_stop(result.get("text", ""), _stop_list(req.stop))
+    completion_tokens = result.get("tokens") or _est_tokens(text)
+    return {
+        "id": cid,
+        "object": "text_completion",
+        "created": created,
+        "model": model or "auto",
+        "choices": [{"index": 0, "text": text, "logprobs": None,
+                     "finish_reason": "stop"}],
+        "usage": {
+            "prompt_tokens": _est_tokens(req.prompt),
+            "completion_tokens": completion_tokens,
+            "total_tokens": _est_tokens(req.prompt) + completion_tokens,
+        },
+    }
+
+
+def _transcript(messages: List[ChatMessage]) -> str:
+    lines = []
+    for m in messages:
+        role = m.role if m.role in ("user", "assistant", "system") else "user"
+        lines.append(f"{role}: {m.content}")
+    lines.append("assistant:")
+    return "\n".join(lines)
+
+
+@router.post("/v1/chat/completions")
+async def chat_completions(req: ChatCompletionRequest):
+    gw = _gateway()
+    _reject_n(req.n)
+    if not req.messages:
+        raise HTTPException(status_code=400, detail="messages must not be empty")
+    cid = new_id("chatcmpl")
+    created = int(time.time())
+    prompt = _transcript(req.messages)
+    model = req.model
+
+    if req.stream:
+        async def events():
+            first = True
+            async for delta in _stream_deltas(gw.node, model, prompt, req):
+                d: Dict[str, Any] = {"content": delta}
+                if first:
+                    d["role"] = "assistant"
+                    first = False
+                yield {
+                    "id": cid, "object": "chat.completion.chunk",
+                    "created": created, "model": model or "auto",
+                    "choices": [{"index": 0, "delta": d,
+                                 "finish_reason": None}],
+                }
+
+        return StreamingResponse(_sse(events()),
+                                 media_type="text/event-stream")
+
+    result = await _run_buffered(gw.node, model, prompt, req)
+    text = _truncate_at_stop(result.get("text", ""), _stop_list(req.stop))
+    completion_tokens = result.get("tokens") or _est_tokens(text)
+    return {
+        "id": cid,
+        "object": "chat.completion",
+        "created": created,
+        "model": model or "auto",
+        "choices": [{"index": 0,
+                     "message": {"role": "assistant", "content": text},
+                     "finish_reason": "stop"}],
+        "usage": {
+            "prompt_tokens": _est_tokens(prompt),
+            "completion_tokens": completion_tokens,
+            "total_tokens": _est_tokens(prompt) + completion_tokens,
+        },
+    }

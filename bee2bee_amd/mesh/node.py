@@ -691,6 +691,7 @@ class MeshNode:
         stream: bool = False,
         on_chunk: Optional[Callable[[str], None]] = None,
         timeout: float = wire.REQUEST_TIMEOUT,
+        sampling: Optional[Dict[str, Any]] = None,
     ) -> Dict[str, Any]:
         # self-request short-circuits to the local service
         if provider_id in (self.peer_id, "local"):
@@ -702,16 +703,16 @@ class MeshNode:
             if svc is None and self.local_services:
                 svc = next(iter(self.local_services.values()))
             if svc is not None:
+                params = {
+                    "prompt": prompt,
+                    "max_new_tokens": max_new_tokens,
+                    "temperature": temperature,
+                }
+                for key in ("top_p", "top_k", "repetition_penalty"):
+                    if sampling and sampling.get(key) is not None:
+                        params[key] = sampling[key]
                 loop = asyncio.get_running_loop()
-                return await loop.run_in_executor(
-                    None,
-                    svc.execute,
-                    {
-                        "prompt": prompt,
-                        "max_new_tokens": max_new_tokens,
-                        "temperature": temperature,
-                    },
-                )
+                return await loop.run_in_executor(None, svc.execute, params)
             raise RuntimeError(wire.ERR_NO_LOCAL_SERVICE)
 
         peer = self.peers.get(provider_id)
@@ -748,6 +749,7 @@ class MeshNode:
                 max_new_tokens=max_new_tokens,
                 temperature=temperature,
                 stream=stream,
+                sampling=sampling,
             ),
         )
         try:

@@ -105,8 +105,9 @@ def gen_request(
     max_new_tokens: int = 2048,
     temperature: float = 0.7,
     stream: bool = False,
+    sampling: Optional[Dict[str, Any]] = None,
 ) -> Dict[str, Any]:
-    return {
+    frame = {
         "type": GEN_REQUEST,
         "rid": rid,
         "prompt": prompt,
@@ -118,6 +119,12 @@ def gen_request(
         "temperature": temperature,
         "stream": stream,
     }
+    # optional sampling knobs ride alongside (request_params reads the same
+    # keys on the receiving side; absent -> reference generation defaults)
+    for key in ("top_p", "top_k", "repetition_penalty"):
+        if sampling and sampling.get(key) is not None:
+            frame[key] = sampling[key]
+    return frame
 
 
 def gen_chunk(rid: str, text: str) -> Dict[str, Any]:
