@@ -194,6 +194,41 @@ PRESETS: Dict[str, ModelSpec] = {
         n_kv_heads=8,
         head_dim=128,
     ),
+    # Llama-3.1: same dims as Llama-3 at 8B/70B. Real 3.1 checkpoints ship
+    # llama3-type rope_scaling + 128k context in config.json, which the
+    # checkpoint-loading path applies (spec_from_hf_config); presets stay
+    # unscaled per the llama3.2-1b note above.
+    "llama3.1-8b": _llama(
+        "llama3.1-8b",
+        vocab_size=128256,
+        hidden_size=4096,
+        intermediate_size=14336,
+        n_layers=32,
+        n_heads=32,
+        n_kv_heads=8,
+        head_dim=128,
+    ),
+    "llama3.1-70b": _llama(
+        "llama3.1-70b",
+        vocab_size=128256,
+        hidden_size=8192,
+        intermediate_size=28672,
+        n_layers=80,
+        n_heads=64,
+        n_kv_heads=8,
+        head_dim=128,
+    ),
+    "llama3.2-3b": _llama(
+        "llama3.2-3b",
+        vocab_size=128256,
+        hidden_size=3072,
+        intermediate_size=8192,
+        n_layers=28,
+        n_heads=24,
+        n_kv_heads=8,
+        head_dim=128,
+        tie_embeddings=True,
+    ),
     # Qwen2.5-7B-Instruct class: llama-shaped + QKV bias, large vocab
     "qwen2.5-7b": ModelSpec(
         name="qwen2.5-7b",
@@ -203,6 +238,53 @@ PRESETS: Dict[str, ModelSpec] = {
         n_layers=28,
         n_heads=28,
         n_kv_heads=4,
+        head_dim=128,
+        rope_theta=1000000.0,
+        rms_eps=1e-6,
+        max_seq_len=32768,
+        tie_embeddings=False,
+        qkv_bias=True,
+    ),
+    # Qwen2.5 larger sizes (14B/32B/72B): same family, GQA groups 5/5/8 —
+    # all decode-kernel-instantiated group sizes
+    "qwen2.5-14b": ModelSpec(
+        name="qwen2.5-14b",
+        vocab_size=152064,
+        hidden_size=5120,
+        intermediate_size=13824,
+        n_layers=48,
+        n_heads=40,
+        n_kv_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        rms_eps=1e-6,
+        max_seq_len=32768,
+        tie_embeddings=False,
+        qkv_bias=True,
+    ),
+    "qwen2.5-32b": ModelSpec(
+        name="qwen2.5-32b",
+        vocab_size=152064,
+        hidden_size=5120,
+        intermediate_size=27648,
+        n_layers=64,
+        n_heads=40,
+        n_kv_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        rms_eps=1e-6,
+        max_seq_len=32768,
+        tie_embeddings=False,
+        qkv_bias=True,
+    ),
+    "qwen2.5-72b": ModelSpec(
+        name="qwen2.5-72b",
+        vocab_size=152064,
+        hidden_size=8192,
+        intermediate_size=29568,
+        n_layers=80,
+        n_heads=64,
+        n_kv_heads=8,
         head_dim=128,
         rope_theta=1000000.0,
         rms_eps=1e-6,
@@ -282,9 +364,23 @@ _ALIASES = {
     "meta-llama/meta-llama-3-70b": "llama3-70b",
     "llama-3-70b": "llama3-70b",
     "llama-3.2-1b": "llama3.2-1b",
+    "meta-llama/llama-3.2-1b": "llama3.2-1b",
     "llama3.2": "llama3.2-1b",
+    "llama-3.2-3b": "llama3.2-3b",
+    "meta-llama/llama-3.2-3b": "llama3.2-3b",
+    "meta-llama/llama-3.1-8b": "llama3.1-8b",
+    "meta-llama/meta-llama-3.1-8b": "llama3.1-8b",
+    "llama-3.1-8b": "llama3.1-8b",
+    "llama3.1": "llama3.1-8b",
+    "meta-llama/llama-3.1-70b": "llama3.1-70b",
+    "meta-llama/meta-llama-3.1-70b": "llama3.1-70b",
+    "llama-3.1-70b": "llama3.1-70b",
     "huggingfaceh4/zephyr-7b-beta": "zephyr-7b",
     "qwen/qwen2.5-7b-instruct": "qwen2.5-7b",
+    "qwen/qwen2.5-7b": "qwen2.5-7b",
+    "qwen/qwen2.5-14b": "qwen2.5-14b",
+    "qwen/qwen2.5-32b": "qwen2.5-32b",
+    "qwen/qwen2.5-72b": "qwen2.5-72b",
     "qwen2.5": "qwen2.5-7b",
     "zephyr": "zephyr-7b",
     "zephyr-7b-beta": "zephyr-7b",

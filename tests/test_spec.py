@@ -151,3 +151,31 @@ def test_rope_scaling_parsed_from_hf_config(tmp_path):
     cfg["rope_scaling"] = {"rope_type": "yarn", "factor": 4.0}
     (tmp_path / "config.json").write_text(json.dumps(cfg))
     assert resolve_spec("x", model_path=str(tmp_path)).rope_scaling is None
+
+
+def test_llama31_32_and_qwen_size_presets():
+    """HF ids of the Llama-3.1/3.2 and larger Qwen2.5 sizes resolve to the
+    real architectures (previously fell through to the demo spec), with
+    parameter counts matching the published models and GQA group sizes the
+    decode kernel instantiates (1..8)."""
+    from bee2bee_amd.models.spec import resolve_spec
+
+    cases = {
+        "meta-llama/Llama-3.1-8B-Instruct": ("llama3.1-8b", 8.03),
+        "meta-llama/Meta-Llama-3.1-70B": ("llama3.1-70b", 70.55),
+        "meta-llama/Llama-3.2-3B-Instruct": ("llama3.2-3b", 3.21),
+        "meta-llama/Llama-3.2-1B": ("llama3.2-1b", 1.24),
+        "Qwen/Qwen2.5-14B-Instruct": ("qwen2.5-14b", 14.77),
+        "Qwen/Qwen2.5-32B": ("qwen2.5-32b", 32.76),
+        "Qwen/Qwen2.5-72B-Instruct": ("qwen2.5-72b", 72.71),
+    }
+    for hf_id, (name, billions) in cases.items():
+        spec = resolve_spec(hf_id)
+        assert spec.name == name, (hf_id, spec.name)
+        assert abs(spec.n_params() / 1e9 - billions) < 0.05, (hf_id, spec.n_params())
+        assert spec.head_dim in (64, 128)
+        assert 1 <= spec.n_heads // spec.n_kv_heads <= 8
+    # qwen family traits hold at every size
+    for n in ("qwen2.5-14b", "qwen2.5-32b", "qwen2.5-72b"):
+        s = resolve_spec(n)
+        assert s.qkv_bias and not s.tie_embeddings and s.rope_theta == 1e6
