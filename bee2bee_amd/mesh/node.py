@@ -107,6 +107,7 @@ class MeshNode:
         self._lock = asyncio.Lock()
         self._bootstrap_addrs: List[str] = []  # reconnect targets
         self._pending: Dict[str, asyncio.Future] = {}
+        self._pending_ws: Dict[str, Any] = {}  # rid -> ws it was sent over
         self._chunk_cbs: Dict[str, Callable[[str], None]] = {}
         self._running = False
         self._monitor_active = False
@@ -187,6 +188,7 @@ class MeshNode:
             if not fut.done():
                 fut.set_exception(RuntimeError("node stopped"))
         self._pending.clear()
+        self._pending_ws.clear()
         for t in self._tasks:
             t.cancel()
         async with self._lock:
@@ -287,6 +289,15 @@ class MeshNode:
                     self.providers.pop(pid, None)
                     logger.info("peer disconnected: %s", pid)
                     break
+        # fail-fast: in-flight requests routed over this link error NOW with
+        # a typed message instead of waiting out the 300 s request timeout
+        for rid, routed_ws in list(self._pending_ws.items()):
+            if routed_ws is ws:
+                self._pending_ws.pop(rid, None)
+                fut = self._pending.pop(rid, None)
+                self._chunk_cbs.pop(rid, None)
+                if fut is not None and not fut.done():
+                    fut.set_exception(RuntimeError(wire.ERR_NOT_CONNECTED))
 
     async def _send(self, ws: Any, obj: Dict[str, Any]) -> None:
         try:
@@ -506,6 +517,7 @@ class MeshNode:
         """gen_result / gen_success / gen_error all settle a pending rid."""
         rid = data.get("rid")
         fut = self._pending.pop(rid, None) if rid else None
+        self._pending_ws.pop(rid, None)
         self._chunk_cbs.pop(rid, None)
         if fut is None or fut.done():
             return
@@ -589,6 +601,7 @@ class MeshNode:
     async def _handle_piece_data(self, ws: Any, data: Dict[str, Any]) -> None:
         rid = f"piece:{data.get('hash')}:{data.get('index')}"
         fut = self._pending.pop(rid, None)
+        self._pending_ws.pop(rid, None)
         if fut is None or fut.done():
             return
         if data.get("error"):
@@ -605,11 +618,17 @@ class MeshNode:
         rid = f"piece:{content_hash}:{index}"
         fut: asyncio.Future = asyncio.get_running_loop().create_future()
         self._pending[rid] = fut
+        self._pending_ws[rid] = peer.ws
         await self._send(
             peer.ws,
             {"type": wire.PIECE_REQUEST, "hash": content_hash, "index": index},
         )
-        return await asyncio.wait_for(fut, timeout=60.0)
+        try:
+            return await asyncio.wait_for(fut, timeout=60.0)
+        except asyncio.TimeoutError:
+            self._pending.pop(rid, None)
+            self._pending_ws.pop(rid, None)
+            raise RuntimeError(wire.ERR_TIMEOUT) from None
 
     def share_pieces(self, content_hash: str, pieces: List[bytes]) -> None:
         """Register locally-held pieces for serving to the mesh."""
@@ -722,6 +741,7 @@ class MeshNode:
         rid = new_id("req")
         fut: asyncio.Future = asyncio.get_running_loop().create_future()
         self._pending[rid] = fut
+        self._pending_ws[rid] = peer.ws
         if on_chunk is not None:
             self._chunk_cbs[rid] = on_chunk
 
@@ -756,6 +776,7 @@ class MeshNode:
             return await asyncio.wait_for(fut, timeout=timeout)
         except asyncio.TimeoutError:
             self._pending.pop(rid, None)
+            self._pending_ws.pop(rid, None)
             self._chunk_cbs.pop(rid, None)
             raise RuntimeError(wire.ERR_TIMEOUT) from None
 

@@ -542,3 +542,31 @@ def test_mesh_churn_requests_keep_flowing():
         await hub.stop()
 
     asyncio.run(run())
+
+
+def test_provider_death_fails_inflight_request_fast():
+    """A provider dying MID-REQUEST rejects the requester's future
+    immediately with the typed provider_not_connected error — the reference
+    (and round-1) behavior was a silent 300 s timeout wait."""
+
+    async def run():
+        a = await _start_node()
+        b = await _start_node()
+        # slow service: the request is guaranteed in-flight when b dies
+        await b.add_service(EchoService(latency_s=8.0))
+        await a.connect_bootstrap(b.addr)
+        await _wait_for(lambda: b.peer_id in a.providers)
+
+        t0 = asyncio.get_event_loop().time()
+        task = asyncio.create_task(
+            a.request_generation(b.peer_id, "x", 8, "echo-model", timeout=60)
+        )
+        await asyncio.sleep(0.3)  # let the gen_request land on b
+        await b.stop()
+        with pytest.raises(RuntimeError, match="provider_not_connected"):
+            await task
+        assert asyncio.get_event_loop().time() - t0 < 5.0, "waited out timeout"
+        assert not a._pending and not a._pending_ws  # no leaked entries
+        await a.stop()
+
+    asyncio.run(run())
