@@ -215,6 +215,20 @@ class MeshBridge:
             if self._ws is ws:
                 self._ws = None
                 self._url = None
+            # fail-fast: tunnel requests in flight on this link resolve NOW
+            # (with their partial stream if any) instead of waiting out the
+            # 90 s request timeout
+            for rid, pending in list(self._pending.items()):
+                if pending.future.done():
+                    continue
+                self._pending.pop(rid, None)
+                if pending.chunks:
+                    pending.future.set_result({
+                        "text": "".join(pending.chunks), "rid": rid,
+                        "metadata": {"partial": True}})
+                else:
+                    pending.future.set_exception(
+                        ConnectionError("mesh link lost mid-request"))
             if self.auto_reconnect and not self._closed:
                 await asyncio.sleep(RECONNECT_DELAY_S)
                 if not self._closed:

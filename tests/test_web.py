@@ -16,7 +16,7 @@ from bee2bee_amd.web.bridge import MeshBridge, _http_addr, _ws_addr
 from bee2bee_amd.web.gateway import create_app
 from bee2bee_amd.web.store import GLOBAL_METRICS_NODE, WebStore
 
-from tests.test_mesh import EchoService
+from tests.test_mesh import EchoService, _wait_for
 
 
 class MockDirectory:
@@ -421,5 +421,44 @@ def test_gateway_generate_with_target_node_direct_http():
             task.cancel()
             await node.stop()
             await directory.stop()
+
+    asyncio.run(run())
+
+
+def test_bridge_link_loss_fails_inflight_tunnel_fast():
+    """The mesh link dropping mid-tunnel-request resolves the pending
+    future immediately (ConnectionError, or the partial stream if chunks
+    already arrived) instead of waiting out the 90 s request timeout."""
+    class SlowStream(EchoService):
+        def execute_stream(self, params):
+            import time as _t
+
+            _t.sleep(3.0)  # in the provider's executor thread
+            yield from super().execute_stream(params)
+
+    async def run():
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        node.local_services["hf"] = SlowStream(model="echo-model")
+        store = WebStore(base_url=None, key=None)
+        bridge = MeshBridge(seeds=[node.addr], store=store,
+                            auto_reconnect=False)
+        await bridge.start()
+        try:
+            assert await bridge.connect()
+            t0 = asyncio.get_event_loop().time()
+            task = asyncio.create_task(bridge.request(
+                {"prompt": "x", "model": "echo-model"}))
+            await _wait_for(lambda: bridge._pending)  # request in flight
+            # abrupt link loss: close the transport under the WS (no
+            # graceful close handshake — the peer is still mid-request)
+            bridge._ws._writer.transport.close()
+            with pytest.raises(ConnectionError, match="mesh link lost"):
+                await asyncio.wait_for(task, timeout=6.0)
+            assert asyncio.get_event_loop().time() - t0 < 6.0
+            assert not bridge._pending
+        finally:
+            await bridge.stop()
+            await node.stop()
 
     asyncio.run(run())
