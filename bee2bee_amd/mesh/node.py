@@ -108,6 +108,7 @@ class MeshNode:
         self._bootstrap_addrs: List[str] = []  # reconnect targets
         self._pending: Dict[str, asyncio.Future] = {}
         self._pending_ws: Dict[str, Any] = {}  # rid -> ws it was sent over
+        self._req_tasks: set = set()  # in-flight gen/piece request handlers
         self._chunk_cbs: Dict[str, Callable[[str], None]] = {}
         self._running = False
         self._monitor_active = False
@@ -189,6 +190,8 @@ class MeshNode:
                 fut.set_exception(RuntimeError("node stopped"))
         self._pending.clear()
         self._pending_ws.clear()
+        for t in list(self._req_tasks):
+            t.cancel()
         for t in self._tasks:
             t.cancel()
         async with self._lock:
@@ -351,7 +354,23 @@ class MeshNode:
         if handler is None:
             logger.warning("unknown message type: %s", mtype)
             return
+        if mtype in (wire.GEN_REQUEST, wire.PIECE_REQUEST):
+            # long-running handlers must not block this connection's read
+            # loop: a provider mid-generation still answers pings and serves
+            # concurrent requests on the same link (chunks interleave by rid)
+            task = asyncio.create_task(self._run_request_handler(
+                handler, ws, data))
+            self._req_tasks.add(task)
+            task.add_done_callback(self._req_tasks.discard)
+            return
         await handler(ws, data)
+
+    async def _run_request_handler(self, handler, ws: Any,
+                                   data: Dict[str, Any]) -> None:
+        try:
+            await handler(ws, data)
+        except Exception:
+            logger.exception("error handling %s", data.get("type"))
 
     async def _handle_hello(self, ws: Any, data: Dict[str, Any]) -> None:
         pid = data.get("peer_id")

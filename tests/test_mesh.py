@@ -570,3 +570,29 @@ def test_provider_death_fails_inflight_request_fast():
         await a.stop()
 
     asyncio.run(run())
+
+
+def test_concurrent_requests_on_one_link_interleave():
+    """Two simultaneous requests over the SAME WS connection execute in
+    parallel on the provider (the reader loop is not blocked by a
+    long-running gen_request), and pings keep flowing mid-generation."""
+
+    async def run():
+        a = await _start_node()
+        b = await _start_node()
+        await b.add_service(EchoService(latency_s=1.0))
+        await a.connect_bootstrap(b.addr)
+        await _wait_for(lambda: b.peer_id in a.providers)
+
+        t0 = asyncio.get_event_loop().time()
+        r1, r2 = await asyncio.gather(
+            a.request_generation(b.peer_id, "one", 8, "echo-model", timeout=15),
+            a.request_generation(b.peer_id, "two", 8, "echo-model", timeout=15),
+        )
+        elapsed = asyncio.get_event_loop().time() - t0
+        assert r1["text"] == "echo:one" and r2["text"] == "echo:two"
+        assert elapsed < 1.9, f"requests serialized: {elapsed:.2f}s"
+        await a.stop()
+        await b.stop()
+
+    asyncio.run(run())
