@@ -510,25 +510,47 @@ class MeshNode:
                 await self._send(ws, wire.gen_result_error(rid, f"local_error: {e}"))
             return
 
-        # 2. one-hop swarm relay
+        # 2. one-hop swarm relay: failover down the (price, latency) ranking,
+        # never relaying back to the requester; sampling knobs and stream
+        # chunks pass through (the reference relays prompt/max_tokens only
+        # and has no failover — one dead provider fails the request)
         if model_name:
-            provider = self.pick_provider(model_name)
-            if provider:
-                pid, _meta = provider
+            requester_pid = next(
+                (p for p, peer in self.peers.items() if peer.ws is ws), None)
+            last_err: Optional[Exception] = None
+            want_stream = bool(data.get("stream"))
+
+            def _forward_chunk(text: str) -> None:
+                task = asyncio.create_task(
+                    self._send(ws, wire.gen_chunk(rid, text)))
+                self._req_tasks.add(task)
+                task.add_done_callback(self._req_tasks.discard)
+
+            for pid, _meta in self.pick_providers(
+                model_name, limit=3, exclude=(requester_pid,)
+            ):
                 try:
                     result = await self.request_generation(
                         provider_id=pid,
                         prompt=params["prompt"],
                         max_new_tokens=params["max_new_tokens"],
                         model_name=model_name,
+                        temperature=params.get("temperature", 0.7),
+                        sampling={k: params.get(k) for k in
+                                  ("top_p", "top_k", "repetition_penalty")},
+                        stream=want_stream,
+                        on_chunk=_forward_chunk if want_stream else None,
                     )
                     await self._send(ws, wire.gen_result(rid, result))
                     return
-                except Exception as e:
-                    await self._send(
-                        ws, wire.gen_result_error(rid, f"{wire.ERR_RELAY}: {e}")
-                    )
-                    return
+                except Exception as e:  # noqa: BLE001 — try the next provider
+                    logger.info("relay to %s failed (%s); failing over", pid, e)
+                    last_err = e
+            if last_err is not None:
+                await self._send(
+                    ws, wire.gen_result_error(rid, f"{wire.ERR_RELAY}: {last_err}")
+                )
+                return
 
         await self._send(ws, wire.gen_result_error(rid, wire.ERR_NO_NODE))
 
@@ -693,11 +715,19 @@ class MeshNode:
                 )
         return out
 
-    def pick_provider(self, model_name: str) -> Optional[Tuple[str, Dict[str, Any]]]:
-        """Cheapest-then-fastest provider advertising the model
-        (reference sort key :745)."""
+    def pick_providers(
+        self,
+        model_name: str,
+        limit: int = 1,
+        exclude: Tuple[Optional[str], ...] = (),
+    ) -> List[Tuple[str, Dict[str, Any]]]:
+        """Up to `limit` providers advertising the model, cheapest-then-
+        fastest (reference sort key :745) — the ranking the relay's
+        failover walks."""
         candidates = []
         for pid, svcs in self.providers.items():
+            if pid in exclude:
+                continue
             for svc_name, meta in svcs.items():
                 if svc_name.startswith("_") or not isinstance(meta, dict):
                     continue
@@ -711,13 +741,18 @@ class MeshNode:
                         )
                     )
                     break
-        if not candidates:
-            return None
         candidates.sort(key=lambda c: (c[0], c[1]))
-        _price, _lat, pid, svc_name = candidates[0]
-        meta = dict(self.providers[pid][svc_name])
-        meta["_svc_name"] = svc_name
-        return pid, meta
+        out = []
+        for _price, _lat, pid, svc_name in candidates[:limit]:
+            meta = dict(self.providers[pid][svc_name])
+            meta["_svc_name"] = svc_name
+            out.append((pid, meta))
+        return out
+
+    def pick_provider(self, model_name: str) -> Optional[Tuple[str, Dict[str, Any]]]:
+        """Cheapest-then-fastest provider advertising the model."""
+        picked = self.pick_providers(model_name, limit=1)
+        return picked[0] if picked else None
 
     async def request_generation(
         self,

@@ -596,3 +596,65 @@ def test_concurrent_requests_on_one_link_interleave():
         await b.stop()
 
     asyncio.run(run())
+
+
+def test_relay_failover_to_next_provider():
+    """Relay failover: requester asks R (no local service); R knows two
+    providers of the model — the cheaper one is DEAD. The relay must fail
+    over to the live provider instead of erroring (the reference fails the
+    request on the first dead provider)."""
+
+    async def run():
+        relay = await _start_node()
+        dead = await _start_node()
+        live = await _start_node()
+        requester = await _start_node()
+
+        # dead is CHEAPER so the ranking tries it first
+        await dead.add_service(EchoService(model="fo-model", price=0.0001))
+        await live.add_service(EchoService(model="fo-model", price=0.01))
+        await relay.connect_bootstrap(dead.addr)
+        await relay.connect_bootstrap(live.addr)
+        await _wait_for(lambda: len(relay.providers) == 2)
+        await requester.connect_bootstrap(relay.addr)
+        await _wait_for(lambda: relay.peer_id in requester.peers)
+
+        dead_pid = dead.peer_id
+        await dead.stop()
+        await _wait_for(lambda: dead_pid not in relay.peers)
+        # provider table may lag the peer table; the relay's attempt to the
+        # dead pid raises provider_not_connected and fails over regardless
+
+        res = await requester.request_generation(
+            relay.peer_id, "ping", 8, "fo-model", timeout=15)
+        assert res["text"] == "echo:ping"
+        for n in (requester, relay, live):
+            await n.stop()
+
+    asyncio.run(run())
+
+
+def test_relay_streams_chunks_through():
+    """A relayed streaming request forwards gen_chunk frames end-to-end:
+    requester -> relay -> provider, chunks come back with the original rid."""
+
+    async def run():
+        relay = await _start_node()
+        provider = await _start_node()
+        requester = await _start_node()
+        await provider.add_service(EchoService(model="st-model"))
+        await relay.connect_bootstrap(provider.addr)
+        await _wait_for(lambda: provider.peer_id in relay.providers)
+        await requester.connect_bootstrap(relay.addr)
+        await _wait_for(lambda: relay.peer_id in requester.peers)
+
+        chunks = []
+        res = await requester.request_generation(
+            relay.peer_id, "a b c", 8, "st-model", stream=True,
+            on_chunk=chunks.append, timeout=15)
+        assert "".join(chunks).strip() == "echo:a b c"
+        assert isinstance(res, dict)
+        for n in (requester, relay, provider):
+            await n.stop()
+
+    asyncio.run(run())
