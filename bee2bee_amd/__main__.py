@@ -143,6 +143,19 @@ def serve_hf_remote(model, token, region, api_port):
     _serve("hf_remote", model_name=model, region=region, api_port=api_port)
 
 
+@cli.command("doctor")
+@click.option("--port", default=None, type=int,
+              help="Also probe a specific mesh port for bindability")
+def doctor_cmd(port):
+    """Diagnose the node environment (torch/ROCm, GPU, HIP extension,
+    RCCL, IPC env, state dir, ports) — run before first serve."""
+    import sys as _sys
+
+    from .doctor import run_doctor
+
+    _sys.exit(run_doctor(port, echo=click.echo))
+
+
 @cli.command("config")
 @click.argument("key")
 @click.argument("value")

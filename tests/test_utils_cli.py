@@ -96,3 +96,26 @@ def test_serve_web_cli_end_to_end(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_doctor_command_runs_green_on_this_image():
+    """doctor reports every non-GPU check ok on the CI image (GPU check
+    degrades to a warning on CPU-only hosts, never a failure)."""
+    res = CliRunner().invoke(cli, ["doctor"])
+    assert res.exit_code == 0, res.output
+    assert "pytorch-rocm" in res.output
+    assert "hip-extension" in res.output
+    assert "[FAIL]" not in res.output
+    assert "0 failures" in res.output
+
+
+def test_doctor_collect_checks_statuses():
+    from bee2bee_amd.doctor import collect_checks
+
+    checks = {name: (status, detail) for name, status, detail
+              in collect_checks()}
+    assert checks["python"][0] == "ok"
+    assert checks["rccl"][0] == "ok"  # RCCL backend compiled into this torch
+    assert checks["ipc-env"][0] == "ok"  # exported in this image
+    assert checks["mesh-port"][0] == "ok"
+    assert "presets" in checks["model-catalog"][1]
