@@ -1,6 +1,7 @@
-"""CLI: serve-native / serve-hf / serve-ollama / serve-hf-remote / register /
-bench — click group, same command names as the reference (bee2bee/__main__.py
-:30-123) plus native-engine extras.
+"""CLI: serve-native / serve-hf / serve-ollama / serve-hf-remote / serve-web /
+register / config / doctor / export-model / seed-model / bench — click group,
+same command names as the reference (bee2bee/__main__.py:30-123) plus
+native-engine extras (doctor, export-model, seed-model, serve-web).
 """
 from __future__ import annotations
 
