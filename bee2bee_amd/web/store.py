@@ -84,7 +84,7 @@ class WebStore:
 
     async def system_stats(self) -> Dict[str, int]:
         """Aggregate usage via the system_stats view; zeros when offline."""
-        empty = {"visits": 0, "chats": 0, "tokens": 0}
+        empty = {"users": 0, "chats": 0, "tokens": 0}
         if not self.enabled:
             return empty
         try:

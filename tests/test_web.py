@@ -359,7 +359,7 @@ def test_web_store_offline_guards():
         store = WebStore(base_url=None, key=None)
         assert not store.enabled
         assert await store.insert_message("n", 5) is False
-        assert await store.system_stats() == {"visits": 0, "chats": 0,
+        assert await store.system_stats() == {"users": 0, "chats": 0,
                                               "tokens": 0}
         assert await store.active_nodes() == []
         assert await store.upsert_node({"peer_id": "x"}) is False
