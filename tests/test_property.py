@@ -141,3 +141,41 @@ class TestPieces:
         assert len(bf) == total
         want = {i for i in have if 0 <= i < total}
         assert {i for i, v in enumerate(bf) if v} == want
+
+
+class TestJoinLinks:
+    @given(
+        network=st.text(alphabet=st.characters(
+            whitelist_categories=("Ll", "Lu", "Nd")), min_size=1, max_size=12),
+        model=st.text(alphabet=st.characters(
+            whitelist_categories=("Ll", "Nd")), min_size=1, max_size=16),
+        hash_hex=st.text(alphabet="0123456789abcdef", min_size=8, max_size=16),
+        bootstrap=st.lists(
+            st.text(min_size=0, max_size=40), max_size=4),
+    )
+    @settings(max_examples=150, deadline=None)
+    def test_generate_parse_roundtrip(self, network, model, hash_hex,
+                                      bootstrap):
+        """Any bootstrap address (arbitrary unicode — ports, IPv6, emoji)
+        survives the base64 leg; query fields round-trip exactly."""
+        from bee2bee_amd.mesh.links import generate_join_link, parse_join_link
+
+        link = generate_join_link(network, model, hash_hex, bootstrap)
+        info = parse_join_link(link)
+        assert info["network"] == network
+        assert info["model"] == model
+        assert info["hash"] == hash_hex
+        assert info["bootstrap"] == [b for b in bootstrap if b != ""] or \
+            info["bootstrap"] == bootstrap  # empty strings may drop
+
+    @given(junk=st.text(max_size=60))
+    @settings(max_examples=100, deadline=None)
+    def test_parse_never_crashes_unexpectedly(self, junk):
+        """Arbitrary text either parses (valid scheme+host) or raises the
+        typed ValueError — no other exception type escapes."""
+        from bee2bee_amd.mesh.links import parse_join_link
+
+        try:
+            parse_join_link(junk)
+        except ValueError:
+            pass
