@@ -180,3 +180,48 @@ def test_chat_forwards_sampling_to_service(client):
     assert r.json()["status"] == "ok"
     assert svc.last_params["top_p"] == 0.8
     assert svc.last_params["repetition_penalty"] == 1.2
+
+
+def test_openai_completions_p2p_fallback(client):
+    """/v1/completions on a node WITHOUT a local service for the model:
+    the request routes over the mesh to a provider peer and comes back in
+    OpenAI shape."""
+    import asyncio
+    import threading
+    import time
+
+    from bee2bee_amd.mesh.node import MeshNode
+    from tests.test_mesh import EchoService
+
+    # provider node on its own loop/thread (the gateway node lives on the
+    # TestClient app loop; mesh wiring goes through the HTTP /connect)
+    loop = asyncio.new_event_loop()
+    t = threading.Thread(target=loop.run_forever, daemon=True)
+    t.start()
+
+    def on_loop(coro, timeout=15):
+        return asyncio.run_coroutine_threadsafe(coro, loop).result(timeout)
+
+    provider = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+    on_loop(provider.start())
+    on_loop(provider.add_service(EchoService(model="remote-model")))
+    try:
+        r = client.get("/connect", headers=KEY,
+                       params={"addr": provider.addr})
+        assert r.json().get("status") == "connected", r.json()
+        for _ in range(100):  # service_announce propagation
+            rows = client.get("/providers", headers=KEY).json()
+            if any("remote-model" in p["models"] for p in rows):
+                break
+            time.sleep(0.05)
+        else:
+            raise AssertionError(f"provider never advertised: {rows}")
+
+        r = client.post("/v1/completions", headers=KEY,
+                        json={"model": "remote-model", "prompt": "mesh hop"})
+        assert r.status_code == 200, r.text
+        assert r.json()["choices"][0]["text"] == "echo:mesh hop"
+    finally:
+        on_loop(provider.stop())
+        loop.call_soon_threadsafe(loop.stop)
+        t.join(timeout=5)
