@@ -23,7 +23,8 @@ from typing import Dict, List, Optional
 
 from ..utils import sha256_hex_bytes
 from . import dht as dht_mod
-from .pieces import piece_hashes, split_pieces, verify_and_reassemble
+from .pieces import (part_path, piece_hashes, split_pieces,
+                     verify_and_reassemble)
 
 DEFAULT_PIECE = 8 * 1024 * 1024
 
@@ -131,48 +132,114 @@ async def _publish(node, dht, key: str, update) -> None:
         await dht.set(key, value)
 
 
+def _drop_parts(parts_dir: str, content_hash: str, n: int) -> None:
+    for i in range(n):
+        try:
+            os.remove(part_path(parts_dir, content_hash, i))
+        except OSError:
+            pass
+
+
 async def fetch_checkpoint(
     node,
     dht: dht_mod.DHTNode,
     name: str,
     out_dir: str,
     provider_peer_id: Optional[str] = None,
+    keep_parts: bool = False,
 ) -> str:
     """Pull a seeded checkpoint from the mesh into out_dir (hash-verified).
 
-    Providers are discovered per content hash from the DHT; the caller can
-    pin a peer id instead (it must already be connected)."""
+    Torrent semantics:
+      * pieces stripe across EVERY connected provider of a file's content
+        hash, with per-piece failover to the next provider on error or a
+        hash-mismatched (corrupt) piece;
+      * each verified piece persists to `<out_dir>/.parts/` in the
+        reference's `{hash}_{i:08d}.part` naming (bee2bee/pieces.py:24-32),
+        so an interrupted fetch RESUMES — already-verified pieces are never
+        re-requested — and part caches interoperate with reference peers;
+      * files already complete in out_dir (content hash matches) are
+        skipped entirely.
+    The caller can pin one provider peer id instead (must be connected)."""
     raw = await dht.get(manifest_key(name))
     if not raw:
         raise FileNotFoundError(f"no manifest for '{name}' in DHT")
     manifest = json.loads(raw)
     _validate_manifest(manifest)
     os.makedirs(out_dir, exist_ok=True)
+    parts_dir = os.path.join(out_dir, ".parts")
+
     for entry in manifest["files"]:
         dest = _safe_dest(out_dir, entry["name"])
-        pid = provider_peer_id
-        if pid is None:
-            providers = await dht_mod.find_providers(dht, entry["content_hash"])
-            # map announced addrs back to connected peers
-            for p, peer in node.peers.items():
-                if peer.addr in providers:
-                    pid = p
-                    break
-        if pid is None:
+        chash = entry["content_hash"]
+        want_hashes = entry["piece_hashes"]
+        n = len(want_hashes)
+
+        # complete-file short-circuit (idempotent re-fetch)
+        if os.path.isfile(dest):
+            with open(dest, "rb") as f:
+                if sha256_hex_bytes(f.read()) == chash:
+                    if not keep_parts:
+                        _drop_parts(parts_dir, chash, n)
+                    continue
+
+        if provider_peer_id is not None:
+            pids = [provider_peer_id]
+        else:
+            providers = await dht_mod.find_providers(dht, chash)
+            pids = [p for p, peer in node.peers.items()
+                    if peer.addr in providers]
+        if not pids:
             raise RuntimeError(
-                f"no connected provider for {entry['name']} "
-                f"({entry['content_hash'][:12]})"
+                f"no connected provider for {entry['name']} ({chash[:12]})"
             )
-        pieces: List[bytes] = []
-        for i in range(len(entry["piece_hashes"])):
-            pieces.append(
-                await node.request_piece(pid, entry["content_hash"], i)
-            )
-        data = verify_and_reassemble(pieces, entry["piece_hashes"])
-        if sha256_hex_bytes(data) != entry["content_hash"]:
+
+        # resume: verified pieces from the part cache
+        pieces: List[Optional[bytes]] = [None] * n
+        os.makedirs(parts_dir, exist_ok=True)
+        for i in range(n):
+            ppath = part_path(parts_dir, chash, i)
+            if os.path.isfile(ppath):
+                with open(ppath, "rb") as f:
+                    blob = f.read()
+                if sha256_hex_bytes(blob) == want_hashes[i]:
+                    pieces[i] = blob
+
+        for i in range(n):
+            if pieces[i] is not None:
+                continue
+            blob = None
+            last_err: Optional[Exception] = None
+            for k in range(len(pids)):
+                pid = pids[(i + k) % len(pids)]  # stripe + failover
+                try:
+                    cand = await node.request_piece(pid, chash, i)
+                except Exception as e:  # noqa: BLE001 — next provider
+                    last_err = e
+                    continue
+                if sha256_hex_bytes(cand) != want_hashes[i]:
+                    last_err = ValueError(
+                        f"corrupt piece {i} from {pid[:12]}")
+                    continue
+                blob = cand
+                break
+            if blob is None:
+                raise RuntimeError(
+                    f"piece {i} of {entry['name']} unavailable: {last_err}")
+            pieces[i] = blob
+            with open(part_path(parts_dir, chash, i), "wb") as f:
+                f.write(blob)
+
+        data = verify_and_reassemble(pieces, want_hashes)
+        if sha256_hex_bytes(data) != chash:
             raise ValueError(f"content hash mismatch for {entry['name']}")
         if len(data) != entry["bytes"]:
             raise ValueError(f"size mismatch for {entry['name']}")
         with open(dest, "wb") as f:
             f.write(data)
+        if not keep_parts:
+            _drop_parts(parts_dir, chash, n)
+
+    if not keep_parts and os.path.isdir(parts_dir) and not os.listdir(parts_dir):
+        os.rmdir(parts_dir)
     return out_dir

@@ -175,3 +175,151 @@ def test_manifest_path_traversal_rejected(tmp_path):
     asyncio.run(run())
     assert not (tmp_path.parent / "pwned").exists()
     assert not (tmp_path / "pwned").exists()
+
+
+def test_fetch_resumes_from_part_cache(tmp_path):
+    """Interrupted fetch: verified pieces persisted under .parts/ are not
+    re-requested on the retry (torrent resume); corrupt cached parts are
+    re-fetched."""
+    spec = PRESETS["tiny"]
+    src_dir = tmp_path / "src"
+    dst_dir = tmp_path / "dst"
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(3)
+    save_hf(w, str(src_dir))
+
+    async def run():
+        dht = DHTNode()
+        await dht.start()
+        seeder = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        leech = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await seeder.start()
+        await leech.start()
+        await seed_checkpoint(seeder, dht, "ck", str(src_dir),
+                              piece_size=2048)
+        await leech.connect_bootstrap(seeder.addr)
+        for _ in range(200):
+            if seeder.peer_id in leech.peers:
+                break
+            await asyncio.sleep(0.02)
+
+        requested = []
+        real = leech.request_piece
+
+        async def counting(pid, chash, i):
+            requested.append((chash[:8], i))
+            return await real(pid, chash, i)
+
+        leech.request_piece = counting
+
+        # interrupted first fetch: fail after 3 pieces
+        calls = {"n": 0}
+
+        async def failing(pid, chash, i):
+            if calls["n"] >= 3:
+                raise RuntimeError("simulated network drop")
+            calls["n"] += 1
+            return await counting(pid, chash, i)
+
+        leech.request_piece = failing
+        try:
+            await fetch_checkpoint(leech, dht, "ck", str(dst_dir),
+                                   keep_parts=True)
+            raise AssertionError("fetch should have failed")
+        except RuntimeError:
+            pass
+        n_first = len(requested)
+        assert n_first == 3
+
+        # corrupt ONE cached part: it must be re-fetched and replaced
+        import os as _os
+
+        from bee2bee_amd.mesh.pieces import part_path
+
+        parts_dir = str(dst_dir / ".parts")
+        cached = sorted(_os.listdir(parts_dir))
+        assert len(cached) == 3
+        victim = _os.path.join(parts_dir, cached[0])
+        with open(victim, "wb") as f:
+            f.write(b"garbage")
+
+        # resumed fetch completes; the 2 intact parts are NOT re-requested
+        leech.request_piece = counting
+        await fetch_checkpoint(leech, dht, "ck", str(dst_dir))
+        refetched = set(requested[n_first:])
+        assert refetched  # the corrupted piece + the never-fetched rest
+
+        def part_key(fname):
+            stem = fname[:-len(".part")]
+            h, idx = stem.rsplit("_", 1)
+            return (h[:8], int(idx))
+
+        intact_keys = {part_key(c) for c in cached[1:]}
+        assert part_key(cached[0]) in refetched  # corrupt part re-fetched
+        assert not (intact_keys & refetched)  # intact parts NOT re-fetched
+
+        await leech.stop()
+        await seeder.stop()
+
+        # fetched checkpoint is byte-identical
+        got = (dst_dir / "model.safetensors").read_bytes()
+        want = (src_dir / "model.safetensors").read_bytes()
+        assert got == want
+        assert not (dst_dir / ".parts").exists()  # cleaned after success
+
+    asyncio.run(run())
+
+
+def test_fetch_stripes_and_fails_over_providers(tmp_path):
+    """Two seeders: pieces stripe across both; killing one mid-catalog still
+    completes via failover."""
+    spec = PRESETS["tiny"]
+    src_dir = tmp_path / "src"
+    dst_dir = tmp_path / "dst"
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(5)
+    save_hf(w, str(src_dir))
+
+    async def run():
+        dht = DHTNode()
+        await dht.start()
+        s1 = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        s2 = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        leech = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        for n in (s1, s2, leech):
+            await n.start()
+        await seed_checkpoint(s1, dht, "ck2", str(src_dir), piece_size=1024)
+        await seed_checkpoint(s2, dht, "ck2", str(src_dir), piece_size=1024)
+        for s in (s1, s2):
+            await leech.connect_bootstrap(s.addr)
+        for _ in range(200):
+            if s1.peer_id in leech.peers and s2.peer_id in leech.peers:
+                break
+            await asyncio.sleep(0.02)
+
+        served = {s1.peer_id: 0, s2.peer_id: 0}
+        real = leech.request_piece
+
+        async def counting(pid, chash, i):
+            served[pid] += 1
+            return await real(pid, chash, i)
+
+        leech.request_piece = counting
+        await fetch_checkpoint(leech, dht, "ck2", str(dst_dir))
+        assert served[s1.peer_id] > 0 and served[s2.peer_id] > 0, served
+
+        # failover: one seeder dies; a fresh fetch still completes
+        import shutil
+
+        shutil.rmtree(dst_dir)
+        await s1.stop()
+        for _ in range(200):
+            if s1.peer_id not in leech.peers:
+                break
+            await asyncio.sleep(0.02)
+        await fetch_checkpoint(leech, dht, "ck2", str(dst_dir))
+        got = (dst_dir / "model.safetensors").read_bytes()
+        assert got == (src_dir / "model.safetensors").read_bytes()
+
+        await leech.stop()
+        await s2.stop()
+
+    asyncio.run(run())
