@@ -17,6 +17,7 @@ base64 inside the 32 MiB WS limit).
 """
 from __future__ import annotations
 
+import asyncio
 import json
 import os
 from typing import Dict, List, Optional
@@ -205,30 +206,36 @@ async def fetch_checkpoint(
                 if sha256_hex_bytes(blob) == want_hashes[i]:
                     pieces[i] = blob
 
-        for i in range(n):
-            if pieces[i] is not None:
-                continue
-            blob = None
-            last_err: Optional[Exception] = None
-            for k in range(len(pids)):
-                pid = pids[(i + k) % len(pids)]  # stripe + failover
-                try:
-                    cand = await node.request_piece(pid, chash, i)
-                except Exception as e:  # noqa: BLE001 — next provider
-                    last_err = e
-                    continue
-                if sha256_hex_bytes(cand) != want_hashes[i]:
-                    last_err = ValueError(
-                        f"corrupt piece {i} from {pid[:12]}")
-                    continue
-                blob = cand
-                break
-            if blob is None:
+        sem = asyncio.Semaphore(min(8, 2 * len(pids)))
+
+        async def fetch_piece(i: int) -> None:
+            async with sem:
+                last_err: Optional[Exception] = None
+                for k in range(len(pids)):
+                    pid = pids[(i + k) % len(pids)]  # stripe + failover
+                    try:
+                        cand = await node.request_piece(pid, chash, i)
+                    except Exception as e:  # noqa: BLE001 — next provider
+                        last_err = e
+                        continue
+                    if sha256_hex_bytes(cand) != want_hashes[i]:
+                        last_err = ValueError(
+                            f"corrupt piece {i} from {pid[:12]}")
+                        continue
+                    pieces[i] = cand
+                    with open(part_path(parts_dir, chash, i), "wb") as f:
+                        f.write(cand)
+                    return
                 raise RuntimeError(
                     f"piece {i} of {entry['name']} unavailable: {last_err}")
-            pieces[i] = blob
-            with open(part_path(parts_dir, chash, i), "wb") as f:
-                f.write(blob)
+
+        missing = [i for i in range(n) if pieces[i] is None]
+        # up to 8 pieces in flight across the provider set
+        results = await asyncio.gather(
+            *(fetch_piece(i) for i in missing), return_exceptions=True)
+        for r in results:
+            if isinstance(r, BaseException):
+                raise r
 
         data = verify_and_reassemble(pieces, want_hashes)
         if sha256_hex_bytes(data) != chash:
