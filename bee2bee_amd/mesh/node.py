@@ -621,21 +621,47 @@ class MeshNode:
             return
         import base64
 
-        pieces = info.get("pieces", [])
-        if index >= len(pieces):
-            await self._send(
-                ws,
-                {"type": wire.PIECE_DATA, "hash": content_hash, "index": index,
-                 "error": "index_out_of_range"},
-            )
-            return
+        if "path" in info:  # disk-backed share_file registration
+            if index >= info["n"]:
+                await self._send(
+                    ws,
+                    {"type": wire.PIECE_DATA, "hash": content_hash,
+                     "index": index, "error": "index_out_of_range"},
+                )
+                return
+
+            def _read_slice() -> bytes:
+                with open(info["path"], "rb") as f:
+                    f.seek(index * info["piece_size"])
+                    return f.read(info["piece_size"])
+
+            try:
+                blob = await asyncio.get_running_loop().run_in_executor(
+                    None, _read_slice)
+            except OSError as e:
+                await self._send(
+                    ws,
+                    {"type": wire.PIECE_DATA, "hash": content_hash,
+                     "index": index, "error": f"read_failed: {e}"},
+                )
+                return
+        else:
+            pieces = info.get("pieces", [])
+            if index >= len(pieces):
+                await self._send(
+                    ws,
+                    {"type": wire.PIECE_DATA, "hash": content_hash,
+                     "index": index, "error": "index_out_of_range"},
+                )
+                return
+            blob = pieces[index]
         await self._send(
             ws,
             {
                 "type": wire.PIECE_DATA,
                 "hash": content_hash,
                 "index": index,
-                "data": base64.b64encode(pieces[index]).decode(),
+                "data": base64.b64encode(blob).decode(),
             },
         )
 
@@ -672,8 +698,17 @@ class MeshNode:
             raise RuntimeError(wire.ERR_TIMEOUT) from None
 
     def share_pieces(self, content_hash: str, pieces: List[bytes]) -> None:
-        """Register locally-held pieces for serving to the mesh."""
+        """Register in-memory pieces for serving to the mesh."""
         self.pieces[content_hash] = {"pieces": pieces}
+
+    def share_file(self, content_hash: str, path: str, piece_size: int,
+                   n_pieces: int) -> None:
+        """Register a DISK-BACKED file for piece serving: the handler reads
+        the requested slice on demand (executor), so seeding a multi-GB
+        checkpoint holds no payload bytes in host RAM."""
+        self.pieces[content_hash] = {
+            "path": path, "piece_size": int(piece_size), "n": int(n_pieces)
+        }
 
     # -------------------------------------------------------------- services
 

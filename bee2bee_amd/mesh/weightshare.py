@@ -24,8 +24,7 @@ from typing import Dict, List, Optional
 
 from ..utils import sha256_hex_bytes
 from . import dht as dht_mod
-from .pieces import (part_path, piece_hashes, split_pieces,
-                     verify_and_reassemble)
+from .pieces import part_path, verify_and_reassemble
 
 DEFAULT_PIECE = 8 * 1024 * 1024
 
@@ -93,23 +92,36 @@ async def seed_checkpoint(
 ) -> Dict:
     """Split every file of a checkpoint dir into pieces, register them with
     the node for serving, and publish a manifest + provider records."""
+    import hashlib
+
     files = []
     for fname in sorted(os.listdir(directory)):
         path = os.path.join(directory, fname)
         if not os.path.isfile(path):
             continue
+        # stream: per-piece + whole-file hashes in one pass, never holding
+        # the file in memory (safetensors shards run to tens of GB)
+        hashes: List[str] = []
+        whole = hashlib.sha256()
+        total = 0
         with open(path, "rb") as f:
-            data = f.read()
-        pieces = split_pieces(data, piece_size) if data else [b""]
-        hashes = piece_hashes(pieces)
-        content_hash = sha256_hex_bytes(data)
-        node.share_pieces(content_hash, pieces)
+            while True:
+                chunk = f.read(piece_size)
+                if not chunk:
+                    break
+                hashes.append(sha256_hex_bytes(chunk))
+                whole.update(chunk)
+                total += len(chunk)
+        if not hashes:
+            hashes = [sha256_hex_bytes(b"")]  # empty file = one empty piece
+        content_hash = whole.hexdigest()
+        node.share_file(content_hash, path, piece_size, len(hashes))
         await _publish(node, dht, f"piece:{content_hash}",
                        lambda cur: sorted(set((cur or []) + [node.addr])))
         files.append(
             {
                 "name": fname,
-                "bytes": len(data),
+                "bytes": total,
                 "content_hash": content_hash,
                 "piece_hashes": hashes,
                 "piece_size": piece_size,

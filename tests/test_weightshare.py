@@ -323,3 +323,34 @@ def test_fetch_stripes_and_fails_over_providers(tmp_path):
         await s2.stop()
 
     asyncio.run(run())
+
+
+def test_seeding_is_disk_backed(tmp_path):
+    """seed_checkpoint must register files WITHOUT holding payload bytes in
+    RAM (multi-GB shards): the node's piece table stores a path + geometry,
+    and serving reads slices on demand."""
+    spec = PRESETS["tiny"]
+    src_dir = tmp_path / "src"
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(1)
+    save_hf(w, str(src_dir))
+
+    async def run():
+        dht = DHTNode()
+        await dht.start()
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        manifest = await seed_checkpoint(node, dht, "dk", str(src_dir),
+                                         piece_size=4096)
+        for info in node.pieces.values():
+            assert "pieces" not in info, "payload held in memory"
+            assert "path" in info and info["n"] >= 1
+        # bytes in the manifest match the real file sizes
+        import os as _os
+
+        for entry in manifest["files"]:
+            real = _os.path.getsize(_os.path.join(str(src_dir),
+                                                  entry["name"]))
+            assert entry["bytes"] == real
+        await node.stop()
+
+    asyncio.run(run())
