@@ -465,14 +465,26 @@ class MeshBridge:
 
     # ----------------------------------------------------------------- stats
 
+    STALE_AFTER_S = 180.0  # directory rows older than this show "stale"
+
+    def _with_staleness(self, meta: Dict[str, Any]) -> Dict[str, Any]:
+        entry = dict(meta)
+        seen = entry.get("last_seen")
+        if isinstance(seen, (int, float)) and \
+                time.time() - seen > self.STALE_AFTER_S:
+            entry["status"] = "stale"
+        return entry
+
     def get_stats(self) -> Dict[str, Any]:
+        peers = [self._with_staleness(m) for m in self.peer_meta.values()]
+        active = [p for p in peers if p.get("status") == "active"]
         return {
             "uptime_s": round(time.time() - self.started_at, 1),
             "connected": self.connected,
             "activeNode": self._url,
-            "poolSize": len(self.peer_meta),
-            "totalPeers": len(self.peer_meta),
-            "peers": list(self.peer_meta.values()),
+            "poolSize": len(active),
+            "totalPeers": len(peers),
+            "peers": peers,
         }
 
     def get_regional_mesh(self) -> Dict[str, List[Dict[str, Any]]]:
@@ -480,7 +492,7 @@ class MeshBridge:
         active link when known (never fabricated)."""
         mesh: Dict[str, List[Dict[str, Any]]] = {}
         for addr, meta in self.peer_meta.items():
-            entry = dict(meta)
+            entry = self._with_staleness(meta)
             if addr == self._url and self._last_rtt_ms is not None:
                 entry["latency"] = round(self._last_rtt_ms, 1)
             mesh.setdefault(meta.get("region", "Global"), []).append(entry)

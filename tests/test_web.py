@@ -462,3 +462,26 @@ def test_bridge_link_loss_fails_inflight_tunnel_fast():
             await node.stop()
 
     asyncio.run(run())
+
+
+def test_bridge_marks_stale_peers():
+    """Directory rows not refreshed within STALE_AFTER_S surface as
+    status=stale in stats and the regional mesh (the reference shows every
+    directory row as active forever)."""
+    from bee2bee_amd.web.store import WebStore
+
+    bridge = MeshBridge(seeds=[], store=WebStore(base_url=None, key=None),
+                        auto_reconnect=False)
+    bridge._merge_meta("ws://1.2.3.4:1", {"region": "EU"})
+    bridge._merge_meta("ws://5.6.7.8:2", {"region": "EU"})
+    bridge.peer_meta["ws://1.2.3.4:1"]["last_seen"] -= 10_000  # long ago
+
+    stats = bridge.get_stats()
+    by_addr = {p["addr"]: p for p in stats["peers"]}
+    assert by_addr["ws://1.2.3.4:1"]["status"] == "stale"
+    assert by_addr["ws://5.6.7.8:2"]["status"] == "active"
+    assert stats["poolSize"] == 1 and stats["totalPeers"] == 2
+
+    mesh = bridge.get_regional_mesh()
+    statuses = {e["addr"]: e["status"] for e in mesh["EU"]}
+    assert statuses == {"ws://1.2.3.4:1": "stale", "ws://5.6.7.8:2": "active"}
