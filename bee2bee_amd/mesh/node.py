@@ -514,7 +514,11 @@ class MeshNode:
         # never relaying back to the requester; sampling knobs and stream
         # chunks pass through (the reference relays prompt/max_tokens only
         # and has no failover — one dead provider fails the request)
-        if model_name:
+        try:
+            hops = int(data.get("hops") or 0)
+        except (TypeError, ValueError):
+            hops = 0
+        if model_name and hops < 1:
             requester_pid = next(
                 (p for p, peer in self.peers.items() if peer.ws is ws), None)
             last_err: Optional[Exception] = None
@@ -540,6 +544,7 @@ class MeshNode:
                                   ("top_p", "top_k", "repetition_penalty")},
                         stream=want_stream,
                         on_chunk=_forward_chunk if want_stream else None,
+                        hops=hops + 1,
                     )
                     await self._send(ws, wire.gen_result(rid, result))
                     return
@@ -800,6 +805,7 @@ class MeshNode:
         on_chunk: Optional[Callable[[str], None]] = None,
         timeout: float = wire.REQUEST_TIMEOUT,
         sampling: Optional[Dict[str, Any]] = None,
+        hops: int = 0,
     ) -> Dict[str, Any]:
         # self-request short-circuits to the local service
         if provider_id in (self.peer_id, "local"):
@@ -859,6 +865,7 @@ class MeshNode:
                 temperature=temperature,
                 stream=stream,
                 sampling=sampling,
+                hops=hops,
             ),
         )
         try:

@@ -106,6 +106,7 @@ def gen_request(
     temperature: float = 0.7,
     stream: bool = False,
     sampling: Optional[Dict[str, Any]] = None,
+    hops: int = 0,
 ) -> Dict[str, Any]:
     frame = {
         "type": GEN_REQUEST,
@@ -124,6 +125,11 @@ def gen_request(
     for key in ("top_p", "top_k", "repetition_penalty"):
         if sampling and sampling.get(key) is not None:
             frame[key] = sampling[key]
+    if hops:
+        # relay-loop guard: relays forward hops+1; a frame that already
+        # relayed once is never relayed again (one-hop semantics, matching
+        # the reference's design). Extra key is ignored by reference peers.
+        frame["hops"] = int(hops)
     return frame
 
 

@@ -738,3 +738,47 @@ def test_node_survives_seeded_frame_fuzz():
             await node.stop()
 
     asyncio.run(run())
+
+
+def test_relay_hop_limit_prevents_loops():
+    """A gen_request that already relayed once (hops=1) is never relayed
+    again — nodes without a local service answer consensus_deadlock instead
+    of bouncing the request around the mesh forever."""
+
+    async def run():
+        import aiohttp
+
+        a = await _start_node()
+        b = await _start_node()
+        # b advertises a provider (a itself) for the model but has no local
+        # service: a hops=1 frame must NOT be relayed back out
+        await a.connect_bootstrap(b.addr)
+        await _wait_for(lambda: a.peer_id in b.peers)
+        b.providers[a.peer_id] = {"hf": {"models": ["loop-model"],
+                                         "price_per_token": 0.001}}
+
+        session = aiohttp.ClientSession()
+        try:
+            ws = await session.ws_connect(b.addr)
+            await ws.send_str(json.dumps({
+                "type": "gen_request", "rid": "loop-1", "svc": "hf",
+                "model": "loop-model", "prompt": "x", "max_new_tokens": 4,
+                "hops": 1,
+            }))
+            data = None
+            for _ in range(50):
+                msg = await asyncio.wait_for(ws.receive(), timeout=10)
+                if msg.type != aiohttp.WSMsgType.TEXT:
+                    break
+                data = json.loads(msg.data)
+                if data.get("type") in ("gen_result", "gen_error"):
+                    break
+            assert data is not None and "error" in data
+            assert "no_node_available" in data["error"]
+            await ws.close()
+        finally:
+            await session.close()
+            await a.stop()
+            await b.stop()
+
+    asyncio.run(run())
