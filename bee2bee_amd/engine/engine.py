@@ -295,10 +295,13 @@ class InferenceEngine:
         repetition_penalty: Optional[float] = None,
         top_p: Optional[float] = None,
         top_k: Optional[int] = None,
+        cancel=None,
     ) -> Dict[str, Any]:
         """Text-level wrapper used by the mesh service. `stop` strings
         truncate the output at the first occurrence (reference stop-word
-        scan, bee2bee/hf.py:111-136)."""
+        scan, bee2bee/hf.py:111-136). `cancel` (a threading.Event) stops
+        generation at the next emitted token — a disconnected client must
+        not keep burning decode steps."""
         t0 = time.time()
         ids = self.tokenizer.encode(prompt)
         ids = ids[-(self.max_seq_len - max_new_tokens - 1) :]
@@ -350,6 +353,8 @@ class InferenceEngine:
             if item is _STREAM_END:
                 break
             _on_token(item)
+            if cancel is not None and cancel.is_set():
+                req.cancelled = True  # engine stops at the next step
         if req.error:
             raise RuntimeError(req.error)
         text = self.tokenizer.decode(req.output_ids)

@@ -35,17 +35,18 @@ _PROVIDERS = Gauge("bee2bee_mesh_providers", "Known model providers",
 _UPTIME = Gauge("bee2bee_uptime_seconds", "Node uptime", registry=REGISTRY)
 _ENGINE = {}  # stat key -> Gauge, created lazily from engine.stats() keys
 
+# keys exactly as InferenceEngine.stats() emits them
 _ENGINE_STATS = (
-    ("queued", "requests waiting for admission"),
-    ("active", "sequences in the running batch"),
-    ("free_kv_blocks", "KV pool blocks free"),
-    ("total_kv_blocks", "KV pool blocks total"),
-    ("steps", "engine decode steps executed"),
-    ("tokens_generated", "tokens generated since start"),
+    ("queued_requests", "requests waiting for admission"),
+    ("active_requests", "sequences in the running batch"),
+    ("prefilling_requests", "requests in chunked prefill"),
+    ("kv_free_blocks", "KV pool blocks free"),
+    ("kv_total_blocks", "KV pool blocks total"),
+    ("tokens_total", "tokens generated since start"),
+    ("tokens_per_sec_10s", "generation throughput (10s window)"),
     ("engine_busy_s", "engine thread busy seconds"),
     ("engine_steps", "engine busy-step count"),
-    ("ms_per_step", "mean engine step latency (ms)"),
-    ("tokens_per_s", "recent generation throughput"),
+    ("engine_ms_per_step", "mean engine step latency (ms)"),
 )
 
 

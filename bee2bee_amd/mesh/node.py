@@ -475,14 +475,23 @@ class MeshNode:
                     await self._send(ws, wire.gen_success(rid, closing))
                     await self._send(ws, wire.gen_result(rid, closing))  # Q1
                 elif data.get("stream"):
-                    # pump the (sync, blocking) stream generator from a thread
+                    # pump the (sync, blocking) stream generator from a
+                    # thread; a requester that disconnects mid-stream closes
+                    # the generator, which cancels the engine request —
+                    # dead clients must not keep burning decode steps
                     queue: asyncio.Queue = asyncio.Queue()
+                    gone = False
 
                     def _pump() -> None:
+                        gen = svc.execute_stream(params)
                         try:
-                            for chunk_raw in svc.execute_stream(params):
-                                loop.call_soon_threadsafe(queue.put_nowait, chunk_raw)
+                            for chunk_raw in gen:
+                                if gone:
+                                    break
+                                loop.call_soon_threadsafe(
+                                    queue.put_nowait, chunk_raw)
                         finally:
+                            gen.close()
                             loop.call_soon_threadsafe(queue.put_nowait, None)
 
                     pump_fut = loop.run_in_executor(None, _pump)
@@ -490,6 +499,9 @@ class MeshNode:
                         chunk_raw = await queue.get()
                         if chunk_raw is None:
                             break
+                        if getattr(ws, "closed", False):
+                            gone = True
+                            continue  # drain until the pump notices
                         try:
                             text = json.loads(chunk_raw).get("text", "")
                         except Exception:

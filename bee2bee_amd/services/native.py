@@ -128,11 +128,13 @@ class NativeEngineService(BaseService):
         prompt, max_new, temperature, extra = self._check(params)
         chunks: "queue.Queue" = queue.Queue()
         DONE = object()
+        cancel = threading.Event()  # closing the generator stops the engine
 
         def _run() -> None:
             try:
                 self.engine.generate_text(
-                    prompt, max_new, temperature, on_text=chunks.put, **extra
+                    prompt, max_new, temperature, on_text=chunks.put,
+                    cancel=cancel, **extra
                 )
                 chunks.put(DONE)
             except Exception as e:  # noqa: BLE001
@@ -140,17 +142,22 @@ class NativeEngineService(BaseService):
 
         t = threading.Thread(target=_run, daemon=True)
         t.start()
-        while True:
-            item = chunks.get()
-            if item is DONE:
-                break
-            if isinstance(item, Exception):
-                yield json.dumps(
-                    {"status": "error", "message": f"Stream error: {item}"}
-                ) + "\n"
-                return
-            yield json.dumps({"text": item}) + "\n"
-        yield json.dumps({"done": True}) + "\n"
+        try:
+            while True:
+                item = chunks.get()
+                if item is DONE:
+                    break
+                if isinstance(item, Exception):
+                    yield json.dumps(
+                        {"status": "error", "message": f"Stream error: {item}"}
+                    ) + "\n"
+                    return
+                yield json.dumps({"text": item}) + "\n"
+            yield json.dumps({"done": True}) + "\n"
+        finally:
+            # GeneratorExit (consumer gone) or normal end: either way the
+            # engine request must not keep burning decode steps
+            cancel.set()
 
     async def execute_stream_async(
         self, params: Dict[str, Any]
@@ -190,26 +197,32 @@ class NativeEngineService(BaseService):
         eng.submit(req)
         stop_words = extra.get("stop") or []
         emitted = ""
-        while True:
-            delta, done = await aq.get()
-            if delta:
-                cut = -1
-                probe = emitted + delta
-                for w in stop_words:
-                    pos = probe.find(w)
-                    if pos >= 0 and (cut < 0 or pos < cut):
-                        cut = pos
-                if cut >= 0:
-                    tail = probe[:cut][len(emitted):]
-                    if tail:
-                        yield json.dumps({"text": tail}) + "\n"
-                    req.cancelled = True
+        try:
+            while True:
+                delta, done = await aq.get()
+                if delta:
+                    cut = -1
+                    probe = emitted + delta
+                    for w in stop_words:
+                        pos = probe.find(w)
+                        if pos >= 0 and (cut < 0 or pos < cut):
+                            cut = pos
+                    if cut >= 0:
+                        tail = probe[:cut][len(emitted):]
+                        if tail:
+                            yield json.dumps({"text": tail}) + "\n"
+                        req.cancelled = True
+                        break
+                    emitted = probe
+                    yield json.dumps({"text": delta}) + "\n"
+                if done:
                     break
-                emitted = probe
-                yield json.dumps({"text": delta}) + "\n"
-            if done:
-                break
-        if req.error:
-            yield json.dumps({"status": "error", "message": req.error}) + "\n"
-            return
-        yield json.dumps({"done": True}) + "\n"
+            if req.error:
+                yield json.dumps(
+                    {"status": "error", "message": req.error}) + "\n"
+                return
+            yield json.dumps({"done": True}) + "\n"
+        finally:
+            # consumer disconnected (aclose/GeneratorExit) or finished:
+            # make sure the engine request stops either way
+            req.cancelled = True

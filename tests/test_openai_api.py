@@ -225,3 +225,31 @@ def test_openai_completions_p2p_fallback(client):
         on_loop(provider.stop())
         loop.call_soon_threadsafe(loop.stop)
         t.join(timeout=5)
+
+
+def test_metrics_engine_gauges_with_native_service(client):
+    """With a real engine attached, /metrics exports the engine gauges
+    under the exact stats() keys (guards against key drift)."""
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.services.native import NativeEngineService
+
+    svc = NativeEngineService("tiny", device="cpu", max_batch=2,
+                              max_seq_len=128)
+    svc.load_sync()
+    try:
+        gateway_api.node.local_services["hf"] = svc
+        r = client.post("/chat", headers=KEY,
+                        json={"prompt": "hi", "model": "tiny",
+                              "max_new_tokens": 4, "temperature": 0.0})
+        assert r.json()["status"] == "ok"
+        text = client.get("/metrics").text
+        for g in ("bee2bee_engine_tokens_total",
+                  "bee2bee_engine_kv_free_blocks",
+                  "bee2bee_engine_engine_ms_per_step"):
+            assert g in text, g
+        # generated tokens actually counted
+        line = next(l for l in text.splitlines()
+                    if l.startswith("bee2bee_engine_tokens_total "))
+        assert float(line.split()[-1]) >= 4.0
+    finally:
+        svc.engine.shutdown()

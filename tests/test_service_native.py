@@ -76,3 +76,44 @@ def test_stop_strings(svc):
              "stop": [stopw]}
         )
         assert stopw not in res["text"]
+
+
+def test_generate_text_cancel_event():
+    """cancel stops generation at the next emitted token — a dead client
+    must not keep burning decode steps."""
+    import threading
+
+    from bee2bee_amd.engine.engine import InferenceEngine
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=480,
+                          seed=3)
+    try:
+        cancel = threading.Event()
+        res = eng.generate_text("a b c", max_new_tokens=400, temperature=0.0,
+                                on_text=lambda _t: cancel.set(),
+                                cancel=cancel)
+        assert 0 < res["tokens"] < 400, res["tokens"]
+    finally:
+        eng.shutdown()
+
+
+def test_execute_stream_close_cancels_engine(svc):
+    """Closing the JSON-lines generator mid-stream (consumer disconnected)
+    cancels the underlying engine request."""
+    import time
+
+    before = svc.engine.stats()["tokens_total"]
+    gen = svc.execute_stream({"prompt": "x y", "max_new_tokens": 100,
+                              "temperature": 0.0})
+    first = next(gen)
+    assert json.loads(first).get("text") is not None
+    gen.close()  # consumer gone
+
+    # the engine request winds down quickly instead of running to 100
+    for _ in range(100):
+        stats = svc.engine.stats()
+        if stats["active_requests"] == 0 and stats["queued_requests"] == 0:
+            break
+        time.sleep(0.05)
+    assert stats["active_requests"] == 0
+    assert stats["tokens_total"] - before < 100
