@@ -67,6 +67,21 @@ def main() -> None:
                 if line:
                     print("  chunk:", line.decode()[:80])
 
+        print("\n== GET /v1/models (OpenAI-compatible surface)")
+        print(requests.get(f"{BASE}/v1/models").json())
+
+        print("\n== POST /v1/chat/completions (SSE stream)")
+        with requests.post(f"{BASE}/v1/chat/completions", json={
+            "model": "tiny", "stream": True, "max_tokens": 12,
+            "messages": [{"role": "user", "content": "hello"}],
+        }, stream=True) as resp:
+            for line in resp.iter_lines():
+                if line:
+                    print("  sse:", line.decode()[:80])
+
+        print("\n== GET /metrics (Prometheus, first lines)")
+        print("\n".join(requests.get(f"{BASE}/metrics").text.splitlines()[:8]))
+
         print("\ndemo complete")
     finally:
         proc.terminate()
