@@ -19,9 +19,16 @@ from fastapi import APIRouter, Depends, HTTPException
 from fastapi.responses import StreamingResponse
 from pydantic import BaseModel
 
+from ..services.base import ServiceError
 from ..utils import new_id
 
 router = APIRouter()
+
+
+def _http_error(e: Exception) -> HTTPException:
+    msg = str(e)
+    code = 429 if "overloaded" in msg else 502
+    return HTTPException(status_code=code, detail=msg)
 
 
 class CompletionRequest(BaseModel):
@@ -224,7 +231,10 @@ async def completions(req: CompletionRequest):
         return StreamingResponse(_sse(events()),
                                  media_type="text/event-stream")
 
-    result = await _run_buffered(gw.node, model, req.prompt, req)
+    try:
+        result = await _run_buffered(gw.node, model, req.prompt, req)
+    except ServiceError as e:
+        raise _http_error(e) from e
     text = _truncate_at_stop(result.get("text", ""), _stop_list(req.stop))
     completion_tokens = result.get("tokens") or _est_tokens(text)
     return {
@@ -280,7 +290,10 @@ async def chat_completions(req: ChatCompletionRequest):
         return StreamingResponse(_sse(events()),
                                  media_type="text/event-stream")
 
-    result = await _run_buffered(gw.node, model, prompt, req)
+    try:
+        result = await _run_buffered(gw.node, model, prompt, req)
+    except ServiceError as e:
+        raise _http_error(e) from e
     text = _truncate_at_stop(result.get("text", ""), _stop_list(req.stop))
     completion_tokens = result.get("tokens") or _est_tokens(text)
     return {

@@ -13,6 +13,7 @@ from __future__ import annotations
 
 import asyncio
 import json
+import os
 import queue
 import threading
 import time
@@ -46,8 +47,6 @@ class NativeEngineService(BaseService):
 
     def load_sync(self) -> None:
         try:
-            import os
-
             from ..engine.engine import InferenceEngine
 
             self.engine = InferenceEngine(
@@ -83,6 +82,12 @@ class NativeEngineService(BaseService):
     def _check(self, params: Dict[str, Any]):
         if self.engine is None:
             raise ServiceError("Model not loaded")
+        # admission control: above this many queued requests, shed load with
+        # a typed error instead of letting the queue (and every client's
+        # latency) grow without bound
+        max_queue = int(os.environ.get("BEE2BEE_MAX_QUEUE", "512"))
+        if self.engine._pending.qsize() >= max_queue:
+            raise ServiceError("server_overloaded")
         prompt = params.get("prompt")
         if not prompt:
             raise ServiceError("Missing prompt")

@@ -253,3 +253,21 @@ def test_metrics_engine_gauges_with_native_service(client):
         assert float(line.split()[-1]) >= 4.0
     finally:
         svc.engine.shutdown()
+
+
+def test_openai_overloaded_returns_429(client, monkeypatch):
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.services.native import NativeEngineService
+
+    svc = NativeEngineService("tiny", device="cpu", max_batch=2,
+                              max_seq_len=128)
+    svc.load_sync()
+    try:
+        gateway_api.node.local_services["hf"] = svc
+        monkeypatch.setenv("BEE2BEE_MAX_QUEUE", "0")
+        r = client.post("/v1/completions", headers=KEY,
+                        json={"model": "tiny", "prompt": "x"})
+        assert r.status_code == 429
+        assert "overloaded" in r.json()["detail"]
+    finally:
+        svc.engine.shutdown()

@@ -117,3 +117,13 @@ def test_execute_stream_close_cancels_engine(svc):
         time.sleep(0.05)
     assert stats["active_requests"] == 0
     assert stats["tokens_total"] - before < 100
+
+
+def test_admission_control_sheds_load(svc, monkeypatch):
+    """BEE2BEE_MAX_QUEUE=0 makes every request shed with the typed
+    server_overloaded error (admission control for production serving)."""
+    monkeypatch.setenv("BEE2BEE_MAX_QUEUE", "0")
+    with pytest.raises(ServiceError, match="server_overloaded"):
+        svc.execute({"prompt": "x", "max_new_tokens": 2})
+    monkeypatch.delenv("BEE2BEE_MAX_QUEUE")
+    assert svc.execute({"prompt": "x", "max_new_tokens": 2})["tokens"] == 2
