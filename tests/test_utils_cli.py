@@ -119,3 +119,45 @@ def test_doctor_collect_checks_statuses():
     assert checks["ipc-env"][0] == "ok"  # exported in this image
     assert checks["mesh-port"][0] == "ok"
     assert "presets" in checks["model-catalog"][1]
+
+
+def test_debug_mesh_script_handshake_and_request():
+    """scripts/debug_mesh.py dials a live node, handshakes and runs a
+    streamed request end-to-end (reference scripts/test_connection +
+    debug_p2p_request shape)."""
+    import asyncio
+    import subprocess
+    import sys as _sys
+
+    async def start():
+        from bee2bee_amd.mesh.node import MeshNode
+        from tests.test_mesh import EchoService
+
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        await node.add_service(EchoService(model="dbg-model"))
+        return node
+
+    import threading
+
+    loop = asyncio.new_event_loop()
+    t = threading.Thread(target=loop.run_forever, daemon=True)
+    t.start()
+    node = asyncio.run_coroutine_threadsafe(start(), loop).result(15)
+    try:
+        out = subprocess.run(
+            [_sys.executable, "scripts/debug_mesh.py", node.addr,
+             "--model", "dbg-model", "--prompt", "ping pong",
+             "--max-new", "8", "--timeout", "20"],
+            capture_output=True, text=True, timeout=60,
+        )
+        assert out.returncode == 0, out.stdout + out.stderr
+        assert "HELLO from" in out.stdout
+        assert "dbg-model" in out.stdout
+        # streamed: chunks carry the payload, the terminal frame closes it
+        assert "chunk: 'echo:ping " in out.stdout
+        assert "RESULT:" in out.stdout
+    finally:
+        asyncio.run_coroutine_threadsafe(node.stop(), loop).result(15)
+        loop.call_soon_threadsafe(loop.stop)
+        t.join(timeout=5)
