@@ -1042,10 +1042,24 @@ async def run_mesh_node(
     if ready_event is not None:
         ready_event.set()
 
+    # graceful termination: SIGTERM (systemd/k8s stop) drains like Ctrl+C —
+    # peers see a clean close instead of a cut socket
+    stop_requested = asyncio.Event()
     try:
-        while True:
-            await asyncio.sleep(15)
+        import signal
+
+        asyncio.get_running_loop().add_signal_handler(
+            signal.SIGTERM, stop_requested.set)
+    except (NotImplementedError, RuntimeError):
+        pass  # non-unix loops / nested loops: Ctrl+C path still works
+
+    try:
+        await stop_requested.wait()
     except (asyncio.CancelledError, KeyboardInterrupt):
+        pass
+    finally:
+        if api_server is not None:
+            api_server.should_exit = True
         await node.stop()
 
 

@@ -161,3 +161,29 @@ def test_debug_mesh_script_handshake_and_request():
         asyncio.run_coroutine_threadsafe(node.stop(), loop).result(15)
         loop.call_soon_threadsafe(loop.stop)
         t.join(timeout=5)
+
+
+def test_run_mesh_node_sigterm_graceful():
+    """SIGTERM drains run_mesh_node cleanly (clean exit code, node stopped)
+    — the systemd/k8s stop path."""
+    import signal
+    import subprocess
+    import sys as _sys
+    import time
+
+    code = (
+        "import sys; sys.path.insert(0, '.')\n"
+        "import asyncio\n"
+        "from bee2bee_amd.mesh.node import run_mesh_node\n"
+        "asyncio.run(run_mesh_node(host='127.0.0.1', port=0,\n"
+        "                          enable_nat=False))\n"
+        "print('CLEAN_EXIT', flush=True)\n"
+    )
+    proc = subprocess.Popen([_sys.executable, "-c", code],
+                            stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                            text=True)
+    time.sleep(4)  # node up
+    proc.send_signal(signal.SIGTERM)
+    out, err = proc.communicate(timeout=30)
+    assert proc.returncode == 0, (proc.returncode, err[-800:])
+    assert "CLEAN_EXIT" in out
