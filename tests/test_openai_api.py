@@ -271,3 +271,56 @@ def test_openai_overloaded_returns_429(client, monkeypatch):
         assert "overloaded" in r.json()["detail"]
     finally:
         svc.engine.shutdown()
+
+
+def test_openai_concurrent_requests_soak(client):
+    """20 concurrent /v1 requests (mixed chat/completions, some streamed)
+    against a live tiny engine: all succeed, outputs non-empty, engine
+    drains back to idle."""
+    import concurrent.futures as cf
+    import time
+
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.services.native import NativeEngineService
+
+    svc = NativeEngineService("tiny", device="cpu", max_batch=8,
+                              max_seq_len=128)
+    svc.load_sync()
+    try:
+        gateway_api.node.local_services["hf"] = svc
+
+        def one(i):
+            if i % 3 == 0:
+                with client.stream(
+                    "POST", "/v1/completions", headers=KEY,
+                    json={"model": "tiny", "prompt": f"p{i}",
+                          "max_tokens": 6, "stream": True},
+                ) as r:
+                    assert r.status_code == 200
+                    lines = [l for l in r.iter_lines() if l]
+                return lines[-1] == "data: [DONE]" and len(lines) > 1
+            if i % 3 == 1:
+                r = client.post("/v1/chat/completions", headers=KEY,
+                                json={"model": "tiny", "max_tokens": 6,
+                                      "messages": [{"role": "user",
+                                                    "content": f"m{i}"}]})
+                return (r.status_code == 200 and
+                        r.json()["choices"][0]["message"]["content"] != "")
+            r = client.post("/v1/completions", headers=KEY,
+                            json={"model": "tiny", "prompt": f"q{i}",
+                                  "max_tokens": 6})
+            return (r.status_code == 200 and
+                    r.json()["usage"]["completion_tokens"] > 0)
+
+        with cf.ThreadPoolExecutor(max_workers=8) as pool:
+            results = list(pool.map(one, range(20)))
+        assert all(results), results
+
+        for _ in range(100):
+            st = svc.engine.stats()
+            if st["active_requests"] == 0 and st["queued_requests"] == 0:
+                break
+            time.sleep(0.05)
+        assert st["active_requests"] == 0
+    finally:
+        svc.engine.shutdown()
