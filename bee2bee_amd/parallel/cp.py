@@ -264,6 +264,13 @@ class CPEngine:
         return sample(self.runner.lm_head(hidden), sampling,
                       generator=self.sample_gen).cpu()
 
+    def reset(self) -> None:
+        """Free the batch's KV so repeated generate() calls (serving via
+        parallel/serve.py LockstepServer) never exhaust the local pool."""
+        for sid in self._seqs:
+            self.kv.free_seq(sid)
+        self._seqs, self._lens = [], []
+
     @torch.no_grad()
     def generate(self, prompts, max_new_tokens: int, sampling=None):
         ids = self.prefill(prompts, sampling)

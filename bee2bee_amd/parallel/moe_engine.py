@@ -143,6 +143,14 @@ class MoEEngine:
         return sample(self.runner.lm_head(hidden), sampling).cpu()
 
     @torch.no_grad()
+    def reset(self) -> None:
+        """Free the batch's KV so repeated generate() calls (serving via
+        parallel/serve.py LockstepServer) never exhaust the pool — prefill
+        allocates fresh sequence ids per call."""
+        for sid in self._seqs:
+            self.kv.free_seq(sid)
+        self._seqs, self._lens = [], []
+
     def generate(
         self,
         prompts: Sequence[Sequence[int]],

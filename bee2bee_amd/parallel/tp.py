@@ -219,6 +219,13 @@ class TPEngine:
                       generator=self.sample_gen).cpu()
 
     @torch.no_grad()
+    def reset(self) -> None:
+        """Free the batch's KV so repeated generate() calls (serving via
+        parallel/serve.py LockstepServer) never exhaust the pool."""
+        for sid in self._seqs:
+            self.kv.free_seq(sid)
+        self._seqs, self._lens = [], []
+
     def generate(
         self,
         prompts: Sequence[Sequence[int]],

@@ -114,3 +114,23 @@ def test_preflight_unit_checks():
     report = pf.run_preflight(2, skip=["gloo_wiring"])
     assert "gloo_wiring" not in report["checks"]
     assert report["preflight"] in ("ok", "failed")
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("mode", ["tp", "cp"])
+def test_serve_parallel_oneshot_contract(mode):
+    """Distributed-serving launch shape: torch.distributed.run world 2,
+    rank 0 drives one request through LockstepServer, followers serve it
+    and drain cleanly."""
+    port = 29881 if mode == "tp" else 29882
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "scripts/serve_parallel.py",
+         "--mode", mode, "--max-batch", "2", "--max-seq-len", "64",
+         "--oneshot-prompt", "hello mesh"],
+        cwd=REPO, capture_output=True, text=True, timeout=240,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "ONESHOT_RESULT tokens=8 world=2" in out.stdout
+    assert "rank 1: served 1 requests" in out.stdout
