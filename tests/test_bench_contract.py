@@ -117,12 +117,12 @@ def test_preflight_unit_checks():
 
 
 @pytest.mark.timeout(300)
-@pytest.mark.parametrize("mode", ["tp", "cp"])
+@pytest.mark.parametrize("mode", ["tp", "cp", "pp", "ep"])
 def test_serve_parallel_oneshot_contract(mode):
     """Distributed-serving launch shape: torch.distributed.run world 2,
     rank 0 drives one request through LockstepServer, followers serve it
     and drain cleanly."""
-    port = 29881 if mode == "tp" else 29882
+    port = {"tp": 29881, "cp": 29882, "pp": 29883, "ep": 29884}[mode]
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
