@@ -43,8 +43,13 @@ class TextStreamDecoder:
         self.tokenizer = tokenizer
         self.flush_every = flush_every
         self.decoded_upto = 0
-        self.held = 0
         self.n_seen = 0
+        # trailing-replacement hold-back state: a U+FFFD at the stream tail
+        # is usually a multibyte char split across the boundary — commit the
+        # valid prefix NOW, hold only the trailing replacements, and release
+        # them only if the same boundary persists (genuinely invalid bytes)
+        self._hold_pos = -1
+        self._held_flushes = 0
 
     def delta(self, output_ids: List[int], final: bool = False) -> str:
         if not final:
@@ -52,15 +57,26 @@ class TextStreamDecoder:
             if self.n_seen > 1 and self.n_seen % self.flush_every != 1:
                 return ""
         text = self.tokenizer.decode(output_ids)
-        d = text[self.decoded_upto :]
+        d = text[self.decoded_upto:]
+        if final:
+            self.decoded_upto = len(text)
+            return d
         if not d:
             return ""
-        if not final and d.endswith("\ufffd") and self.held < 3:
-            self.held += 1
-            return ""
-        self.held = 0
-        self.decoded_upto = len(text)
-        return d
+        stripped = d.rstrip("\ufffd")
+        if len(stripped) != len(d):
+            cut = self.decoded_upto + len(stripped)
+            if cut == self._hold_pos:
+                self._held_flushes += 1
+            else:
+                self._hold_pos = cut
+                self._held_flushes = 1
+            # a real utf-8 char completes within 3 more ids, so a boundary
+            # stuck longer than that is invalid data — emit it as-is
+            if self._held_flushes > 3:
+                stripped = d
+        self.decoded_upto += len(stripped)
+        return stripped
 
 
 @dataclass
@@ -305,35 +321,17 @@ class InferenceEngine:
         t0 = time.time()
         ids = self.tokenizer.encode(prompt)
         ids = ids[-(self.max_seq_len - max_new_tokens - 1) :]
-        emitted: List[str] = []
-        decoded_upto = 0
-        held = 0
-        n_seen = 0
         # emit the first token at once (TTFT), then flush every few tokens:
         # per-token queue/HTTP hops dominate the serving path at high
-        # concurrency (dozens of thread wakeups per decode step otherwise)
-        flush_every = 4
+        # concurrency; split multibyte chars are held back by the decoder
+        decoder = TextStreamDecoder(self.tokenizer, flush_every=4)
 
         def _on_token(tok: int) -> None:
-            nonlocal decoded_upto, held, n_seen
             if on_text is None:
                 return
-            n_seen += 1
-            if n_seen > 1 and n_seen % flush_every != 1:
-                return
-            text = self.tokenizer.decode(req.output_ids)
-            delta = text[decoded_upto:]
-            if not delta:
-                return
-            # hold back a possibly-incomplete multibyte char at the boundary,
-            # but never more than 3 flushes
-            if delta.endswith("�") and held < 3:
-                held += 1
-                return
-            held = 0
-            decoded_upto = len(text)
-            emitted.append(delta)
-            on_text(delta)
+            delta = decoder.delta(req.output_ids)
+            if delta:
+                on_text(delta)
 
         stop_ids = ()
         eos = getattr(self.tokenizer, "eos_token_id", None)
@@ -365,8 +363,8 @@ class InferenceEngine:
             )
             if cut >= 0:
                 text = text[:cut]
-        if on_text is not None and decoded_upto < len(text):
-            on_text(text[decoded_upto:])
+        if on_text is not None and decoder.decoded_upto < len(text):
+            on_text(text[decoder.decoded_upto:])
         now = time.time()
         ft = req.first_token_ts or now
         adm = req.admitted_ts or req.submit_ts
