@@ -162,9 +162,16 @@ async def _stream_deltas(node, model, prompt, req) -> AsyncIterator[str]:
                 loop.call_soon_threadsafe(queue.put_nowait, None)
 
         loop.run_in_executor(None, _pump)
+        # a stop sequence can straddle chunk boundaries: hold back the last
+        # max(len(stop))-1 chars until the stream proves they are not the
+        # start of a stop (streamed output must equal buffered truncation)
+        hold = max((len(x) for x in stops), default=1) - 1
+        pending = ""
         while True:
             line = await queue.get()
             if line is None:
+                if pending:
+                    yield pending
                 return
             try:
                 obj = json.loads(line)
@@ -173,14 +180,19 @@ async def _stream_deltas(node, model, prompt, req) -> AsyncIterator[str]:
             delta = obj.get("text") or ""
             if not delta:
                 continue
+            pending += delta
             if stops:
-                cut = _truncate_at_stop(emitted + delta, stops)
-                if len(cut) < len(emitted) + len(delta):
-                    if len(cut) > len(emitted):
-                        yield cut[len(emitted):]
+                cut = _truncate_at_stop(emitted + pending, stops)
+                if len(cut) < len(emitted) + len(pending):
+                    tail = cut[len(emitted):]
+                    if tail:
+                        yield tail
                     return
-            emitted += delta
-            yield delta
+            safe = len(pending) - hold
+            if safe > 0:
+                yield pending[:safe]
+                emitted += pending[:safe]
+                pending = pending[safe:]
     else:
         result = await _run_buffered(node, model, prompt, req)
         text = _truncate_at_stop(result.get("text", ""), stops)

@@ -229,3 +229,53 @@ class TestStreaming:
                 # flush_every=1 the hold budget covers every real utf-8 char,
                 # so a partial can only appear for genuinely invalid input
                 assert "�" in text or len(text.encode()) != len(ids)
+
+
+class TestOpenAIStreamStops:
+    @given(
+        words=st.lists(st.text(alphabet="abcXYZ ", min_size=1, max_size=6),
+                       min_size=1, max_size=12),
+        stop=st.text(alphabet="abcXYZ ", min_size=1, max_size=4),
+    )
+    @settings(max_examples=150, deadline=None)
+    def test_streamed_equals_buffered_truncation(self, words, stop):
+        """The SSE stream's incremental stop-sequence cut must emit exactly
+        the buffered truncation of the full text — a stop string split
+        across chunk boundaries included."""
+        import asyncio
+        import json
+
+        from bee2bee_amd.gateway.openai_compat import (_stream_deltas,
+                                                       _truncate_at_stop)
+
+        full = "".join(words)
+
+        class FakeSvc:
+            def get_metadata(self):
+                return {"models": ["fake"]}
+
+            def execute_stream(self, params):
+                for w in words:
+                    yield json.dumps({"text": w}) + "\n"
+                yield json.dumps({"done": True}) + "\n"
+
+        class FakeNode:
+            local_services = {"hf": FakeSvc()}
+
+        class Req:
+            model = "fake"
+            max_tokens = 64
+            temperature = 0.0
+            top_p = None
+            stream = True
+
+        Req.stop = [stop]
+
+        async def collect():
+            out = ""
+            async for d in _stream_deltas(FakeNode(), "fake", "p", Req()):
+                out += d
+            return out
+
+        got = asyncio.run(collect())
+        assert got == _truncate_at_stop(full, [stop])
