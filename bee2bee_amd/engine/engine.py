@@ -79,6 +79,57 @@ class TextStreamDecoder:
         return stripped
 
 
+class StopStringFilter:
+    """Incremental stop-string truncation for token streams.
+
+    Reference behavior (bee2bee/hf.py:111-136) is buffered: generate all,
+    then cut at the first stop string. Streaming must emit EXACTLY that
+    truncation without un-saying text, so the filter holds back the last
+    max(len(stop))-1 chars until the stream proves they do not start a
+    stop. `done` flips when a stop fired (callers cancel generation — an
+    improvement over the reference, which burns tokens past the stop)."""
+
+    __slots__ = ("stops", "hold", "emitted", "pending", "done")
+
+    def __init__(self, stops) -> None:
+        self.stops = [s for s in (stops or []) if s]
+        self.hold = max((len(s) for s in self.stops), default=1) - 1
+        self.emitted = ""
+        self.pending = ""
+        self.done = False
+
+    def feed(self, delta: str) -> str:
+        if self.done or not delta:
+            return ""
+        self.pending += delta
+        probe = self.emitted + self.pending
+        if self.stops:
+            cut = min((probe.find(s) for s in self.stops
+                       if probe.find(s) >= 0), default=-1)
+            if cut >= 0:
+                out = probe[:cut][len(self.emitted):]
+                self.emitted = probe[:cut]
+                self.pending = ""
+                self.done = True
+                return out
+        safe = len(self.pending) - self.hold
+        if safe <= 0:
+            return ""
+        out = self.pending[:safe]
+        self.emitted += out
+        self.pending = self.pending[safe:]
+        return out
+
+    def flush(self) -> str:
+        """End of stream: release the held tail (no stop can complete)."""
+        if self.done:
+            return ""
+        out = self.pending
+        self.emitted += out
+        self.pending = ""
+        return out
+
+
 @dataclass
 class GenerationRequest:
     prompt_ids: List[int]
@@ -323,15 +374,20 @@ class InferenceEngine:
         ids = ids[-(self.max_seq_len - max_new_tokens - 1) :]
         # emit the first token at once (TTFT), then flush every few tokens:
         # per-token queue/HTTP hops dominate the serving path at high
-        # concurrency; split multibyte chars are held back by the decoder
+        # concurrency; split multibyte chars are held back by the decoder,
+        # split stop strings by the stop filter (which also cancels
+        # generation the moment a stop completes)
         decoder = TextStreamDecoder(self.tokenizer, flush_every=4)
+        stop_filter = StopStringFilter(stop)
 
         def _on_token(tok: int) -> None:
             if on_text is None:
                 return
-            delta = decoder.delta(req.output_ids)
-            if delta:
-                on_text(delta)
+            out = stop_filter.feed(decoder.delta(req.output_ids))
+            if out:
+                on_text(out)
+            if stop_filter.done:
+                req.cancelled = True
 
         stop_ids = ()
         eos = getattr(self.tokenizer, "eos_token_id", None)
@@ -363,8 +419,11 @@ class InferenceEngine:
             )
             if cut >= 0:
                 text = text[:cut]
-        if on_text is not None and decoder.decoded_upto < len(text):
-            on_text(text[decoder.decoded_upto:])
+        if on_text is not None:
+            tail = stop_filter.feed(decoder.delta(req.output_ids, final=True))
+            tail += stop_filter.flush()
+            if tail:
+                on_text(tail)
         now = time.time()
         ft = req.first_token_ts or now
         adm = req.admitted_ts or req.submit_ts

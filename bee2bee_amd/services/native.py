@@ -171,7 +171,8 @@ class NativeEngineService(BaseService):
         straight into an asyncio queue via call_soon_threadsafe. The HTTP
         gateway prefers this over the sync generator (which costs two
         threads per request — measurable GIL churn at high concurrency)."""
-        from ..engine.engine import GenerationRequest, TextStreamDecoder
+        from ..engine.engine import (GenerationRequest, StopStringFilter,
+                                     TextStreamDecoder)
         from ..engine.sampler import SamplingParams
 
         prompt, max_new, temperature, extra = self._check(params)
@@ -200,27 +201,22 @@ class NativeEngineService(BaseService):
 
         req.on_emit = on_emit
         eng.submit(req)
-        stop_words = extra.get("stop") or []
-        emitted = ""
+        # split stop strings across deltas are handled by the filter (the
+        # streamed text equals the buffered truncation exactly)
+        stop_filter = StopStringFilter(extra.get("stop"))
         try:
             while True:
                 delta, done = await aq.get()
-                if delta:
-                    cut = -1
-                    probe = emitted + delta
-                    for w in stop_words:
-                        pos = probe.find(w)
-                        if pos >= 0 and (cut < 0 or pos < cut):
-                            cut = pos
-                    if cut >= 0:
-                        tail = probe[:cut][len(emitted):]
-                        if tail:
-                            yield json.dumps({"text": tail}) + "\n"
-                        req.cancelled = True
-                        break
-                    emitted = probe
-                    yield json.dumps({"text": delta}) + "\n"
+                out = stop_filter.feed(delta)
+                if out:
+                    yield json.dumps({"text": out}) + "\n"
+                if stop_filter.done:
+                    req.cancelled = True
+                    break
                 if done:
+                    tail = stop_filter.flush()
+                    if tail:
+                        yield json.dumps({"text": tail}) + "\n"
                     break
             if req.error:
                 yield json.dumps(

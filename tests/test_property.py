@@ -279,3 +279,31 @@ class TestOpenAIStreamStops:
 
         got = asyncio.run(collect())
         assert got == _truncate_at_stop(full, [stop])
+
+
+class TestStopStringFilter:
+    @given(
+        text=st.text(alphabet="abXY", max_size=40),
+        data=st.data(),
+        stops=st.lists(st.text(alphabet="abXY", min_size=1, max_size=4),
+                       max_size=2),
+    )
+    @settings(max_examples=200, deadline=None)
+    def test_any_chunking_equals_buffered(self, text, data, stops):
+        """feed() over ANY chunking of the text + flush() emits exactly the
+        buffered truncation; `done` iff a stop occurs."""
+        from bee2bee_amd.engine.engine import StopStringFilter
+        from bee2bee_amd.gateway.openai_compat import _truncate_at_stop
+
+        filt = StopStringFilter(stops)
+        out = ""
+        i = 0
+        while i < len(text):
+            n = data.draw(st.integers(min_value=1, max_value=6))
+            out += filt.feed(text[i:i + n])
+            i += n
+        out += filt.flush()
+        want = _truncate_at_stop(text, stops)
+        assert out == want
+        assert filt.emitted == want
+        assert filt.done == (any(s in text for s in stops))
