@@ -450,7 +450,14 @@ class MeshNode:
         rid = wire.request_id(data)
         svc_name = data.get("svc", "hf")
         model_name = data.get("model")
-        params = wire.request_params(data)
+        try:
+            params = wire.request_params(data)
+        except (TypeError, ValueError) as e:
+            # malformed knobs (non-numeric temperature, ...): answer a typed
+            # error NOW — silence would cost the requester the full timeout
+            await self._send(ws, wire.gen_error(rid, f"bad_request: {e}"))
+            await self._send(ws, wire.gen_result_error(rid, f"bad_request: {e}"))
+            return
         loop = asyncio.get_running_loop()
 
         # 1. local execution primary

@@ -782,3 +782,36 @@ def test_relay_hop_limit_prevents_loops():
             await b.stop()
 
     asyncio.run(run())
+
+
+def test_malformed_gen_request_gets_typed_error():
+    """Non-numeric sampling knobs answer bad_request immediately instead of
+    silently timing out the requester."""
+
+    async def run():
+        import aiohttp
+
+        node = await _start_node()
+        await node.add_service(EchoService())
+        session = aiohttp.ClientSession()
+        try:
+            ws = await session.ws_connect(node.addr)
+            await ws.send_str(json.dumps({
+                "type": "gen_request", "rid": "bad-1", "svc": "hf",
+                "prompt": "x", "temperature": "hot",
+            }))
+            data = None
+            for _ in range(20):
+                msg = await asyncio.wait_for(ws.receive(), timeout=5)
+                if msg.type != aiohttp.WSMsgType.TEXT:
+                    break
+                data = json.loads(msg.data)
+                if data.get("rid") == "bad-1":
+                    break
+            assert data and "bad_request" in (data.get("error") or "")
+            await ws.close()
+        finally:
+            await session.close()
+            await node.stop()
+
+    asyncio.run(run())
