@@ -384,6 +384,10 @@ class MeshNode:
             )
             if old_pid and old_pid != pid:
                 self.peers.pop(old_pid)
+                # the provider rows travel with the connection's identity:
+                # leaving them under the old pid would route requests at a
+                # peer id that no longer answers
+                self.providers.pop(old_pid, None)
             if pid not in self.peers:
                 first_contact = True
             existing = self.peers.get(pid)

@@ -815,3 +815,35 @@ def test_malformed_gen_request_gets_typed_error():
             await node.stop()
 
     asyncio.run(run())
+
+
+def test_hello_rename_reaps_old_provider_rows():
+    """A peer re-helloing with a NEW peer_id on the same connection must not
+    leave provider rows under the old id (they would route requests at a
+    dead identity)."""
+
+    async def run():
+        import aiohttp
+
+        node = await _start_node()
+        session = aiohttp.ClientSession()
+        try:
+            ws = await session.ws_connect(node.addr)
+            for pid in ("id-one", "id-two"):
+                await ws.send_str(json.dumps({
+                    "type": "hello", "peer_id": pid, "addr": "",
+                    "region": "T", "metrics": {},
+                    "services": {"hf": {"models": ["ren-model"],
+                                        "price_per_token": 0.001}},
+                }))
+                await asyncio.sleep(0.2)
+            assert "id-one" not in node.peers
+            assert "id-one" not in node.providers
+            assert "id-two" in node.peers
+            assert "ren-model" in node.providers["id-two"]["hf"]["models"]
+            await ws.close()
+        finally:
+            await session.close()
+            await node.stop()
+
+    asyncio.run(run())
