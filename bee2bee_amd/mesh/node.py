@@ -109,6 +109,9 @@ class MeshNode:
         self._pending: Dict[str, asyncio.Future] = {}
         self._pending_ws: Dict[str, Any] = {}  # rid -> ws it was sent over
         self._req_tasks: set = set()  # in-flight gen/piece request handlers
+        # short-lived tasks (per-connection readers, gossip dials) live in a
+        # self-pruning set; _tasks keeps only the persistent loops — a
+        # long-running node under churn must not accumulate task objects
         self._chunk_cbs: Dict[str, Callable[[str], None]] = {}
         self._running = False
         self._monitor_active = False
@@ -263,7 +266,7 @@ class MeshNode:
         # prune finished reader tasks so long-lived nodes with many
         # reconnects don't accumulate task objects forever
         self._tasks = [t for t in self._tasks if not t.done()]
-        self._tasks.append(asyncio.create_task(self._reader(ws)))
+        self._track(asyncio.create_task(self._reader(ws)))
 
     async def _reader(self, ws: Any) -> None:
         try:
@@ -358,12 +361,15 @@ class MeshNode:
             # long-running handlers must not block this connection's read
             # loop: a provider mid-generation still answers pings and serves
             # concurrent requests on the same link (chunks interleave by rid)
-            task = asyncio.create_task(self._run_request_handler(
-                handler, ws, data))
-            self._req_tasks.add(task)
-            task.add_done_callback(self._req_tasks.discard)
+            self._track(asyncio.create_task(self._run_request_handler(
+                handler, ws, data)))
             return
         await handler(ws, data)
+
+    def _track(self, task: "asyncio.Task") -> "asyncio.Task":
+        self._req_tasks.add(task)
+        task.add_done_callback(self._req_tasks.discard)
+        return task
 
     async def _run_request_handler(self, handler, ws: Any,
                                    data: Dict[str, Any]) -> None:
@@ -409,7 +415,7 @@ class MeshNode:
             if addr == self.addr:
                 continue
             if not any(p.addr == addr for p in self.peers.values()):
-                self._tasks.append(asyncio.create_task(self._safe_connect(addr)))
+                self._track(asyncio.create_task(self._safe_connect(addr)))
 
     async def _safe_connect(self, addr: str) -> None:
         try:
@@ -548,10 +554,8 @@ class MeshNode:
             want_stream = bool(data.get("stream"))
 
             def _forward_chunk(text: str) -> None:
-                task = asyncio.create_task(
-                    self._send(ws, wire.gen_chunk(rid, text)))
-                self._req_tasks.add(task)
-                task.add_done_callback(self._req_tasks.discard)
+                self._track(asyncio.create_task(
+                    self._send(ws, wire.gen_chunk(rid, text))))
 
             for pid, _meta in self.pick_providers(
                 model_name, limit=3, exclude=(requester_pid,)
