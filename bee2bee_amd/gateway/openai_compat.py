@@ -74,11 +74,12 @@ def _stop_list(stop: Any) -> List[str]:
 
 
 def _truncate_at_stop(text: str, stops: List[str]) -> str:
-    for s in stops:
-        i = text.find(s)
-        if i >= 0:
-            text = text[:i]
-    return text
+    """Cut at the EARLIEST occurrence of any stop (OpenAI semantics; also
+    what engine.generate_text and StopStringFilter compute — sequential
+    per-stop cuts would be order-dependent when stops overlap)."""
+    cut = min((text.find(s) for s in stops if s and text.find(s) >= 0),
+              default=-1)
+    return text[:cut] if cut >= 0 else text
 
 
 def _est_tokens(text: str) -> int:
