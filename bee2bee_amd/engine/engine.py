@@ -54,7 +54,8 @@ class TextStreamDecoder:
     def delta(self, output_ids: List[int], final: bool = False) -> str:
         if not final:
             self.n_seen += 1
-            if self.n_seen > 1 and self.n_seen % self.flush_every != 1:
+            if (self.flush_every > 1 and self.n_seen > 1
+                    and self.n_seen % self.flush_every != 1):
                 return ""
         text = self.tokenizer.decode(output_ids)
         d = text[self.decoded_upto:]

@@ -307,3 +307,17 @@ class TestStopStringFilter:
         assert out == want
         assert filt.emitted == want
         assert filt.done == (any(s in text for s in stops))
+
+
+def test_stream_decoder_flush_every_one_emits_per_token():
+    """flush_every=1 means a delta attempt on EVERY token (a modulo slip
+    made it emit only the first token, then nothing until final)."""
+    from bee2bee_amd.engine.engine import TextStreamDecoder
+    from bee2bee_amd.models.tokenizer import ByteTokenizer
+
+    tok = ByteTokenizer()
+    ids = tok.encode("abcdef", add_bos=False)
+    dec = TextStreamDecoder(tok, flush_every=1)
+    deltas = [dec.delta(ids[:n]) for n in range(1, len(ids) + 1)]
+    assert "".join(deltas) == "abcdef"
+    assert all(d for d in deltas)  # ascii: every token emits
