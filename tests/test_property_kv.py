@@ -155,3 +155,45 @@ class TestDHTAnnounceProperties:
         for r in set(ranks):
             last = max(i for i, x in enumerate(ranks) if x == r)
             assert table[f"peer{r}"]["rank"] == last
+
+
+class TestPlannerProperties:
+    @given(
+        n_layers=st.integers(min_value=1, max_value=96),
+        n_stages=st.integers(min_value=1, max_value=16),
+        budgets=st.one_of(
+            st.none(),
+            st.lists(st.integers(min_value=1, max_value=1000),
+                     min_size=1, max_size=16),
+        ),
+    )
+    @settings(max_examples=200, deadline=None)
+    def test_plan_partitions_layers_exactly(self, n_layers, n_stages,
+                                            budgets):
+        """Any (layers, stages, budgets) combo: stages are contiguous,
+        non-empty, cover [0, n_layers) exactly once; embed on the first,
+        head on the last."""
+        from dataclasses import replace
+
+        from bee2bee_amd.models.spec import PRESETS
+        from bee2bee_amd.parallel.planner import plan_stages
+
+        spec = replace(PRESETS["tiny"], n_layers=n_layers)
+        if budgets is not None and len(budgets) != n_stages:
+            budgets = (budgets * n_stages)[:n_stages]
+        if n_stages > n_layers:
+            with pytest.raises(ValueError):
+                plan_stages(spec, n_stages, budgets)
+            return
+        plans = plan_stages(spec, n_stages, budgets)
+        assert len(plans) == n_stages
+        lo = 0
+        for r, p in enumerate(plans):
+            assert p.rank == r
+            assert p.layer_range[0] == lo
+            assert p.layer_range[1] > p.layer_range[0], "empty stage"
+            lo = p.layer_range[1]
+            assert p.has_embed == (r == 0)
+            assert p.has_head == (r == n_stages - 1)
+            assert p.est_bytes > 0
+        assert lo == n_layers
