@@ -197,3 +197,26 @@ class TestPlannerProperties:
             assert p.has_head == (r == n_stages - 1)
             assert p.est_bytes > 0
         assert lo == n_layers
+
+
+class TestDeviceSlotMapping:
+    @given(lens=st.lists(st.integers(min_value=1, max_value=3 * BLOCK),
+                         min_size=1, max_size=5))
+    @settings(max_examples=100, deadline=None)
+    def test_device_mapping_equals_host_mapping(self, lens):
+        """graphs.decode_slot_mapping (the hipGraph-capturable tensor path)
+        must equal PagedKV.slot_mapping (the host path) for the LAST
+        position of every sequence — the decode-step contract."""
+        from bee2bee_amd.engine.graphs import decode_slot_mapping
+
+        kv = PagedKV(PRESETS["tiny"], torch.device("cpu"), torch.float32,
+                     n_blocks=32, block_size=BLOCK)
+        for sid, ln in enumerate(lens):
+            kv.new_seq(sid)
+            kv.extend_seq(sid, ln)
+        table = kv.block_table(range(len(lens)))
+        positions = torch.tensor([ln - 1 for ln in lens], dtype=torch.int32)
+        dev_slots = decode_slot_mapping(table, positions, BLOCK).tolist()
+        host_slots = [kv.slot_mapping(sid, [ln - 1])[0]
+                      for sid, ln in enumerate(lens)]
+        assert dev_slots == host_slots
