@@ -321,3 +321,26 @@ def test_stream_decoder_flush_every_one_emits_per_token():
     deltas = [dec.delta(ids[:n]) for n in range(1, len(ids) + 1)]
     assert "".join(deltas) == "abcdef"
     assert all(d for d in deltas)  # ascii: every token emits
+
+
+class TestStunParser:
+    @given(data=st.binary(max_size=64))
+    @settings(max_examples=200, deadline=None)
+    def test_binding_response_parser_total(self, data):
+        """The STUN response parser is total over arbitrary network bytes:
+        it returns None or an (ip, port) tuple, never raises."""
+        from bee2bee_amd.mesh.stun import (create_binding_request,
+                                           parse_binding_response)
+
+        _req, txn = create_binding_request()
+        out = parse_binding_response(data, txn)
+        assert out is None or (
+            isinstance(out, tuple) and len(out) == 2
+            and isinstance(out[0], str) and 0 <= out[1] <= 65535)
+
+    def test_own_request_roundtrip_shape(self):
+        from bee2bee_amd.mesh.stun import create_binding_request
+
+        req, txn = create_binding_request()
+        assert len(req) == 20 and len(txn) == 12
+        assert req[0:2] == b"\x00\x01"  # binding request type
