@@ -220,3 +220,34 @@ class TestDeviceSlotMapping:
         host_slots = [kv.slot_mapping(sid, [ln - 1])[0]
                       for sid, ln in enumerate(lens)]
         assert dev_slots == host_slots
+
+
+class TestLayerRangeLoads:
+    @given(data=st.data(),
+           arch=st.sampled_from(["tiny", "tiny-gpt2"]))
+    @settings(max_examples=20, deadline=None)
+    def test_any_layer_range_loads_exact_slice(self, data, arch, tmp_path_factory):
+        """Every [lo, hi) slice of a checkpoint loads exactly those layers
+        (PP stage loading), with embed/pos on the first stage and final
+        norm/head on the last — llama AND gpt2 formats."""
+        from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+        spec = PRESETS[arch]
+        n = spec.n_layers
+        lo = data.draw(st.integers(min_value=0, max_value=n - 1))
+        hi = data.draw(st.integers(min_value=lo + 1, max_value=n))
+        d = tmp_path_factory.mktemp("ckpt")
+        full = ModelWeights(spec, torch.device("cpu"),
+                            torch.float32).random_init(7)
+        save_hf(full, str(d))
+        part = ModelWeights(spec, torch.device("cpu"), torch.float32).load_hf(
+            str(d), layer_range=(lo, hi))
+        for i in range(n):
+            loaded = part.layers[i].wqkv is not None
+            assert loaded == (lo <= i < hi), (i, lo, hi)
+            if loaded:
+                assert torch.equal(part.layers[i].wqkv, full.layers[i].wqkv)
+        assert (part.embed is not None) == (lo == 0)
+        assert (part.final_norm is not None) == (hi == n)
+        if spec.pos_type == "learned":
+            assert (part.pos_embed is not None) == (lo == 0)
