@@ -251,3 +251,56 @@ class TestLayerRangeLoads:
         assert (part.final_norm is not None) == (hi == n)
         if spec.pos_type == "learned":
             assert (part.pos_embed is not None) == (lo == 0)
+
+
+class TestEngineSchedulingEquivalence:
+    @given(data=st.data())
+    @settings(max_examples=6, deadline=None)
+    def test_chunked_prefill_scheduling_is_output_invariant(self, data):
+        """Random request mixes: tiny chunked-prefill budget (forces chunk
+        scheduling + decode interleave) must emit token-identical outputs
+        to an unconstrained engine."""
+        from bee2bee_amd.engine.engine import InferenceEngine
+
+        n_req = data.draw(st.integers(min_value=1, max_value=4))
+        prompts = [
+            data.draw(st.lists(st.integers(min_value=5, max_value=400),
+                               min_size=1, max_size=40))
+            for _ in range(n_req)
+        ]
+        max_news = [data.draw(st.integers(min_value=1, max_value=8))
+                    for _ in range(n_req)]
+
+        def run(max_prefill_tokens):
+            eng = InferenceEngine("tiny", device="cpu", max_batch=4,
+                                  max_seq_len=128, seed=5,
+                                  max_prefill_tokens=max_prefill_tokens)
+            try:
+                reqs = [eng.generate_async(p, max_new_tokens=m,
+                                           temperature=0.0)
+                        if hasattr(eng, "generate_async") else None
+                        for p, m in zip(prompts, max_news)]
+                if reqs[0] is None:
+                    # no async helper: submit directly
+                    from bee2bee_amd.engine.engine import GenerationRequest
+                    from bee2bee_amd.engine.sampler import SamplingParams
+
+                    reqs = [GenerationRequest(
+                        prompt_ids=p, max_new_tokens=m,
+                        sampling=SamplingParams(greedy=True))
+                        for p, m in zip(prompts, max_news)]
+                    for r in reqs:
+                        eng.submit(r)
+                outs = []
+                for r in reqs:
+                    while True:
+                        item = r.out_queue.get(timeout=60)
+                        if not isinstance(item, int):
+                            break
+                    assert r.error is None, r.error
+                    outs.append(list(r.output_ids))
+                return outs
+            finally:
+                eng.shutdown()
+
+        assert run(max_prefill_tokens=16) == run(max_prefill_tokens=4096)
