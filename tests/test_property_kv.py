@@ -304,3 +304,47 @@ class TestEngineSchedulingEquivalence:
                 eng.shutdown()
 
         assert run(max_prefill_tokens=16) == run(max_prefill_tokens=4096)
+
+
+class TestSpecDecodeInvariance:
+    @given(data=st.data(), penalty=st.sampled_from([1.0, 1.15]))
+    @settings(max_examples=6, deadline=None)
+    def test_spec_decode_output_invariant_random_mixes(self, data, penalty):
+        """Speculative decoding must be token-identical to plain decode for
+        ANY prompt mix, greedy and penalized-greedy both (the round-2
+        penalized verification path)."""
+        from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+        from bee2bee_amd.engine.sampler import SamplingParams
+
+        n_req = data.draw(st.integers(min_value=1, max_value=3))
+        # small alphabet -> repetitive prompts -> the proposer actually fires
+        prompts = [
+            data.draw(st.lists(st.integers(min_value=5, max_value=12),
+                               min_size=2, max_size=24))
+            for _ in range(n_req)
+        ]
+
+        def run(spec):
+            eng = InferenceEngine("tiny", device="cpu", max_batch=4,
+                                  max_seq_len=128, seed=9, spec_decode=spec)
+            try:
+                reqs = [GenerationRequest(
+                    prompt_ids=p, max_new_tokens=10,
+                    sampling=SamplingParams(greedy=True,
+                                            repetition_penalty=penalty))
+                    for p in prompts]
+                for r in reqs:
+                    eng.submit(r)
+                outs = []
+                for r in reqs:
+                    while True:
+                        item = r.out_queue.get(timeout=60)
+                        if not isinstance(item, int):
+                            break
+                    assert r.error is None, r.error
+                    outs.append(list(r.output_ids))
+                return outs
+            finally:
+                eng.shutdown()
+
+        assert run(True) == run(False)
