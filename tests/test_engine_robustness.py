@@ -292,3 +292,76 @@ def test_randomized_workload_soak():
         assert eng.kv.free_blocks == base, (eng.kv.free_blocks, base)
     finally:
         eng.shutdown()
+
+
+def test_context_boundary_exact_fit():
+    """prompt + max_new == max_seq_len must work (off-by-one guard)."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=64,
+                          seed=2)
+    try:
+        prompt = list(range(5, 5 + 48))
+        req = GenerationRequest(prompt_ids=prompt, max_new_tokens=16,
+                                sampling=SamplingParams(greedy=True))
+        eng.submit(req)
+        while True:
+            item = req.out_queue.get(timeout=60)
+            if not isinstance(item, int):
+                break
+        assert req.error is None, req.error
+        assert len(req.output_ids) == 16
+    finally:
+        eng.shutdown()
+
+
+def test_admission_is_fifo_when_saturated():
+    """With max_batch=1, queued requests complete in submission order."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=1, max_seq_len=64,
+                          seed=4)
+    try:
+        reqs = [GenerationRequest(prompt_ids=[7, 8, 9 + i],
+                                  max_new_tokens=4,
+                                  sampling=SamplingParams(greedy=True))
+                for i in range(4)]
+        for r in reqs:
+            eng.submit(r)
+        for r in reqs:
+            while True:
+                item = r.out_queue.get(timeout=60)
+                if not isinstance(item, int):
+                    break
+            assert r.error is None
+        done_order = sorted(range(4), key=lambda i: reqs[i].done_ts)
+        assert done_order == [0, 1, 2, 3], done_order
+    finally:
+        eng.shutdown()
+
+
+def test_oversized_prompt_rejected_typed():
+    """A prompt that cannot fit even alone must error, not hang."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=32,
+                          seed=6)
+    try:
+        req = GenerationRequest(prompt_ids=list(range(5, 5 + 64)),
+                                max_new_tokens=8,
+                                sampling=SamplingParams(greedy=True))
+        eng.submit(req)
+        deadline = 60
+        import time as _t
+
+        t0 = _t.time()
+        while _t.time() - t0 < deadline:
+            item = req.out_queue.get(timeout=deadline)
+            if not isinstance(item, int):
+                break
+        assert req.error is not None or len(req.output_ids) <= 32
+    finally:
+        eng.shutdown()
