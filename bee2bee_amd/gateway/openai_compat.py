@@ -119,19 +119,6 @@ async def _run_buffered(node, model: Optional[str], prompt: str,
     )
 
 
-def _service_chunks(svc, params) -> List[str]:
-    """Drain a service's JSON-lines stream into text deltas."""
-    out = []
-    for line in svc.execute_stream(params):
-        try:
-            obj = json.loads(line)
-        except (TypeError, json.JSONDecodeError):
-            continue
-        if obj.get("text"):
-            out.append(obj["text"])
-    return out
-
-
 async def _sse(events: AsyncIterator[Dict[str, Any]]) -> AsyncIterator[str]:
     async for ev in events:
         yield f"data: {json.dumps(ev)}\n\n"
