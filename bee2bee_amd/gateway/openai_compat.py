@@ -33,7 +33,9 @@ def _http_error(e: Exception) -> HTTPException:
 
 class CompletionRequest(BaseModel):
     model: Optional[str] = None
-    prompt: str = ""
+    # str, or a LIST of prompts (OpenAI batch shape — one choice per prompt,
+    # run concurrently through the engine's continuous batching)
+    prompt: Any = ""
     max_tokens: Optional[int] = 256
     temperature: Optional[float] = None
     top_p: Optional[float] = None
@@ -217,10 +219,19 @@ async def completions(req: CompletionRequest):
     cid = new_id("cmpl")
     created = int(time.time())
     model = req.model
+    prompts = req.prompt if isinstance(req.prompt, list) else [req.prompt]
+    if not all(isinstance(p, str) for p in prompts):
+        raise HTTPException(status_code=400,
+                            detail="prompt must be a string or list of strings")
 
     if req.stream:
+        if len(prompts) != 1:
+            raise HTTPException(
+                status_code=400,
+                detail="stream=true supports a single prompt")
+
         async def events():
-            async for delta in _stream_deltas(gw.node, model, req.prompt, req):
+            async for delta in _stream_deltas(gw.node, model, prompts[0], req):
                 yield {
                     "id": cid, "object": "text_completion",
                     "created": created, "model": model or "auto",
@@ -232,22 +243,31 @@ async def completions(req: CompletionRequest):
                                  media_type="text/event-stream")
 
     try:
-        result = await _run_buffered(gw.node, model, req.prompt, req)
+        # one choice per prompt; concurrent calls ride the engine's
+        # continuous batching (they share decode steps)
+        results = await asyncio.gather(
+            *(_run_buffered(gw.node, model, p, req) for p in prompts))
     except ServiceError as e:
         raise _http_error(e) from e
-    text = _truncate_at_stop(result.get("text", ""), _stop_list(req.stop))
-    completion_tokens = result.get("tokens") or _est_tokens(text)
+    stops = _stop_list(req.stop)
+    choices = []
+    completion_total = 0
+    for i, (p, result) in enumerate(zip(prompts, results)):
+        text = _truncate_at_stop(result.get("text", ""), stops)
+        completion_total += result.get("tokens") or _est_tokens(text)
+        choices.append({"index": i, "text": text, "logprobs": None,
+                        "finish_reason": "stop"})
+    prompt_total = sum(_est_tokens(p) for p in prompts)
     return {
         "id": cid,
         "object": "text_completion",
         "created": created,
         "model": model or "auto",
-        "choices": [{"index": 0, "text": text, "logprobs": None,
-                     "finish_reason": "stop"}],
+        "choices": choices,
         "usage": {
-            "prompt_tokens": _est_tokens(req.prompt),
-            "completion_tokens": completion_tokens,
-            "total_tokens": _est_tokens(req.prompt) + completion_tokens,
+            "prompt_tokens": prompt_total,
+            "completion_tokens": completion_total,
+            "total_tokens": prompt_total + completion_total,
         },
     }
 

@@ -324,3 +324,25 @@ def test_openai_concurrent_requests_soak(client):
         assert st["active_requests"] == 0
     finally:
         svc.engine.shutdown()
+
+
+def test_completions_batched_prompts(client):
+    """OpenAI batch shape: prompt as a LIST yields one indexed choice per
+    prompt (concurrent through the engine's continuous batching)."""
+    _attach_echo()
+    r = client.post("/v1/completions", headers=KEY,
+                    json={"model": "oai-model",
+                          "prompt": ["alpha", "beta", "gamma"]})
+    assert r.status_code == 200
+    body = r.json()
+    assert [c["index"] for c in body["choices"]] == [0, 1, 2]
+    assert [c["text"] for c in body["choices"]] == [
+        "echo:alpha", "echo:beta", "echo:gamma"]
+    assert body["usage"]["completion_tokens"] > 0
+
+    # streaming rejects multi-prompt; non-string entries rejected
+    r = client.post("/v1/completions", headers=KEY,
+                    json={"prompt": ["a", "b"], "stream": True})
+    assert r.status_code == 400
+    r = client.post("/v1/completions", headers=KEY, json={"prompt": [1, 2]})
+    assert r.status_code == 400
