@@ -195,6 +195,12 @@ def _reject_n(n: Optional[int]) -> None:
         raise HTTPException(status_code=400, detail="only n=1 is supported")
 
 
+def _check_max_tokens(v: Optional[int]) -> None:
+    if v is not None and v < 1:
+        raise HTTPException(status_code=400,
+                            detail="max_tokens must be >= 1")
+
+
 @router.get("/v1/models")
 async def list_models():
     gw = _gateway()
@@ -216,6 +222,7 @@ async def list_models():
 async def completions(req: CompletionRequest):
     gw = _gateway()
     _reject_n(req.n)
+    _check_max_tokens(req.max_tokens)
     cid = new_id("cmpl")
     created = int(time.time())
     model = req.model
@@ -285,6 +292,7 @@ def _transcript(messages: List[ChatMessage]) -> str:
 async def chat_completions(req: ChatCompletionRequest):
     gw = _gateway()
     _reject_n(req.n)
+    _check_max_tokens(req.max_tokens)
     if not req.messages:
         raise HTTPException(status_code=400, detail="messages must not be empty")
     cid = new_id("chatcmpl")

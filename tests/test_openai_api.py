@@ -346,3 +346,6 @@ def test_completions_batched_prompts(client):
     assert r.status_code == 400
     r = client.post("/v1/completions", headers=KEY, json={"prompt": [1, 2]})
     assert r.status_code == 400
+    r = client.post("/v1/completions", headers=KEY,
+                    json={"prompt": "x", "max_tokens": 0})
+    assert r.status_code == 400
