@@ -410,11 +410,15 @@ def resolve_spec(name: str, model_path: Optional[str] = None) -> ModelSpec:
     # llama3-8b preset: a remainder that continues with digits is a
     # different model size, not a variant, so it must fall through to the demo
     # spec rather than silently serve the wrong architecture.
-    for alias in sorted(_ALIASES, key=len, reverse=True):
-        if key.startswith(alias):
-            rest = key[len(alias):]
+    # preset keys participate too: "llama3-8b-instruct" is a variant of
+    # the llama3-8b preset even though no alias spells it out
+    candidates = set(_ALIASES) | {p for p in PRESETS
+                                  if not p.startswith(("tiny", "demo"))}
+    for cand in sorted(candidates, key=len, reverse=True):
+        if key.startswith(cand):
+            rest = key[len(cand):]
             if rest and rest[0] in "-_/." and len(rest) > 1 and rest[1].isalpha():
-                return PRESETS[_ALIASES[alias]]
+                return PRESETS[_ALIASES.get(cand, cand)]
     spec = PRESETS["demo-125m"]
     return ModelSpec(**{**spec.__dict__, "name": name})
 

@@ -344,3 +344,30 @@ class TestStunParser:
         req, txn = create_binding_request()
         assert len(req) == 20 and len(txn) == 12
         assert req[0:2] == b"\x00\x01"  # binding request type
+
+
+class TestAliasResolution:
+    @given(
+        alias=st.sampled_from(["llama3-8b", "qwen/qwen2.5-14b",
+                               "meta-llama/llama-3.2-1b"]),
+        sep=st.sampled_from("-_/."),
+        suffix=st.text(alphabet="abcdefgh", min_size=1, max_size=8),
+        digits=st.text(alphabet="0123456789", min_size=1, max_size=3),
+    )
+    @settings(max_examples=100, deadline=None)
+    def test_variant_suffixes_resolve_size_suffixes_do_not(
+            self, alias, sep, suffix, digits):
+        """`<alias>-instructlike` resolves to the alias's preset; a
+        remainder starting with digits (a DIFFERENT model size) must fall
+        through to the demo spec (the ADVICE-r1 wrong-architecture bug)."""
+        from bee2bee_amd.models.spec import _ALIASES, PRESETS, resolve_spec
+
+        target = PRESETS[_ALIASES.get(alias, alias)].n_layers
+
+        variant = resolve_spec(alias + sep + suffix)
+        assert variant.n_layers == target, variant.name
+
+        sized = resolve_spec(alias + sep + digits + "b")
+        # digit remainder = different size: NEVER silently the alias target
+        assert sized.name == alias + sep + digits + "b"
+        assert sized.n_layers == PRESETS["demo-125m"].n_layers
