@@ -83,51 +83,76 @@ class TextStreamDecoder:
 class StopStringFilter:
     """Incremental stop-string truncation for token streams.
 
-    Reference behavior (bee2bee/hf.py:111-136) is buffered: generate all,
-    then cut at the first stop string. Streaming must emit EXACTLY that
-    truncation without un-saying text, so the filter holds back the last
-    max(len(stop))-1 chars until the stream proves they do not start a
-    stop. `done` flips when a stop fired (callers cancel generation — an
-    improvement over the reference, which burns tokens past the stop)."""
+    Emits EXACTLY the buffered truncation (cut at the EARLIEST occurrence
+    of any stop) without un-saying text. Two hazards make this more than a
+    tail hold-back: a stop can straddle chunk boundaries, and a LONGER
+    stop completing later can start at an EARLIER position than a short
+    stop already visible — so text is withheld from the earliest position
+    where any stop is complete OR could still complete, and the cut is
+    final only when no earlier candidate remains open. `done` flips when
+    the cut is final (callers cancel generation — an improvement over the
+    reference's generate-then-truncate, bee2bee/hf.py:111-136)."""
 
-    __slots__ = ("stops", "hold", "emitted", "pending", "done")
+    __slots__ = ("stops", "buf", "n_emitted", "emitted", "done")
 
     def __init__(self, stops) -> None:
         self.stops = [s for s in (stops or []) if s]
-        self.hold = max((len(s) for s in self.stops), default=1) - 1
+        self.buf = ""
+        self.n_emitted = 0
         self.emitted = ""
-        self.pending = ""
         self.done = False
+
+    def _earliest_completed(self) -> Optional[int]:
+        c = None
+        for s in self.stops:
+            i = self.buf.find(s, self.n_emitted)
+            if i >= 0 and (c is None or i < c):
+                c = i
+        return c
+
+    def _earliest_open(self) -> Optional[int]:
+        """Start of the earliest tail that is a proper prefix of a stop
+        (a stop that may still complete with future text)."""
+        L = len(self.buf)
+        p = None
+        for s in self.stops:
+            lo = max(self.n_emitted, L - len(s) + 1)
+            for k in range(lo, L):
+                if self.buf[k:] == s[: L - k]:
+                    if p is None or k < p:
+                        p = k
+                    break
+        return p
 
     def feed(self, delta: str) -> str:
         if self.done or not delta:
             return ""
-        self.pending += delta
-        probe = self.emitted + self.pending
-        if self.stops:
-            cut = min((probe.find(s) for s in self.stops
-                       if probe.find(s) >= 0), default=-1)
-            if cut >= 0:
-                out = probe[:cut][len(self.emitted):]
-                self.emitted = probe[:cut]
-                self.pending = ""
-                self.done = True
-                return out
-        safe = len(self.pending) - self.hold
-        if safe <= 0:
-            return ""
-        out = self.pending[:safe]
+        self.buf += delta
+        c = self._earliest_completed()
+        p = self._earliest_open()
+        safe = len(self.buf)
+        if c is not None:
+            safe = min(safe, c)
+        if p is not None:
+            safe = min(safe, p)
+        out = self.buf[self.n_emitted:safe]
+        self.n_emitted = max(self.n_emitted, safe)
         self.emitted += out
-        self.pending = self.pending[safe:]
+        if c is not None and (p is None or p >= c):
+            self.done = True
         return out
 
     def flush(self) -> str:
-        """End of stream: release the held tail (no stop can complete)."""
+        """End of stream: open candidates are dead; finalize the cut."""
         if self.done:
             return ""
-        out = self.pending
+        c = self._earliest_completed()
+        end = c if c is not None else len(self.buf)
+        out = self.buf[self.n_emitted:end]
+        self.n_emitted = end
         self.emitted += out
-        self.pending = ""
+        if c is not None:
+            self.done = True
         return out
 
 
