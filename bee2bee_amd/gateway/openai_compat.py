@@ -132,7 +132,6 @@ async def _stream_deltas(node, model, prompt, req) -> AsyncIterator[str]:
     when only the P2P path exists)."""
     svc = _pick_service(node, model)
     stops = _stop_list(req.stop)
-    emitted = ""
     if svc is not None:
         params = {
             "prompt": prompt,
@@ -152,37 +151,27 @@ async def _stream_deltas(node, model, prompt, req) -> AsyncIterator[str]:
                 loop.call_soon_threadsafe(queue.put_nowait, None)
 
         loop.run_in_executor(None, _pump)
-        # a stop sequence can straddle chunk boundaries: hold back the last
-        # max(len(stop))-1 chars until the stream proves they are not the
-        # start of a stop (streamed output must equal buffered truncation)
-        hold = max((len(x) for x in stops), default=1) - 1
-        pending = ""
+        # exact incremental truncation (straddling and overlapping stops
+        # included): engine.StopStringFilter is the single implementation
+        from ..engine.engine import StopStringFilter
+
+        filt = StopStringFilter(stops)
         while True:
             line = await queue.get()
             if line is None:
-                if pending:
-                    yield pending
+                tail = filt.flush()
+                if tail:
+                    yield tail
                 return
             try:
                 obj = json.loads(line)
             except (TypeError, json.JSONDecodeError):
                 continue
-            delta = obj.get("text") or ""
-            if not delta:
-                continue
-            pending += delta
-            if stops:
-                cut = _truncate_at_stop(emitted + pending, stops)
-                if len(cut) < len(emitted) + len(pending):
-                    tail = cut[len(emitted):]
-                    if tail:
-                        yield tail
-                    return
-            safe = len(pending) - hold
-            if safe > 0:
-                yield pending[:safe]
-                emitted += pending[:safe]
-                pending = pending[safe:]
+            out = filt.feed(obj.get("text") or "")
+            if out:
+                yield out
+            if filt.done:
+                return
     else:
         result = await _run_buffered(node, model, prompt, req)
         text = _truncate_at_stop(result.get("text", ""), stops)
