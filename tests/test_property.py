@@ -371,3 +371,32 @@ class TestAliasResolution:
         # digit remainder = different size: NEVER silently the alias target
         assert sized.name == alias + sep + digits + "b"
         assert sized.n_layers == PRESETS["demo-125m"].n_layers
+
+
+class TestRequestParams:
+    @given(
+        max_new=st.one_of(st.none(), st.integers(min_value=-10, max_value=10**9),
+                          st.text(max_size=6), st.floats(allow_nan=False)),
+        temp=st.one_of(st.none(), st.floats(allow_nan=False,
+                                            allow_infinity=False),
+                       st.text(max_size=6), st.booleans()),
+    )
+    @settings(max_examples=200, deadline=None)
+    def test_request_params_typed_or_typed_error(self, max_new, temp):
+        """request_params either returns properly-typed params or raises
+        TypeError/ValueError — exactly what _handle_gen_request converts
+        into the bad_request wire error. Nothing else may escape."""
+        from bee2bee_amd.mesh import wire
+
+        frame = {"type": "gen_request", "rid": "r", "prompt": "p"}
+        if max_new is not None:
+            frame["max_new_tokens"] = max_new
+        if temp is not None:
+            frame["temperature"] = temp
+        try:
+            params = wire.request_params(frame)
+        except (TypeError, ValueError):
+            return
+        assert isinstance(params["max_new_tokens"], int)
+        assert isinstance(params["temperature"], float)
+        assert isinstance(params["prompt"], str)
