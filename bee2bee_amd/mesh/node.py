@@ -462,7 +462,7 @@ class MeshNode:
         model_name = data.get("model")
         try:
             params = wire.request_params(data)
-        except (TypeError, ValueError) as e:
+        except (TypeError, ValueError, OverflowError) as e:
             # malformed knobs (non-numeric temperature, ...): answer a typed
             # error NOW — silence would cost the requester the full timeout
             await self._send(ws, wire.gen_error(rid, f"bad_request: {e}"))
