@@ -31,15 +31,19 @@ class SamplingParams:
         defaults (bee2bee/hf.py:94-103): temperature 0.7, top_p 0.95,
         repetition_penalty 1.15. temperature <= 1e-4 selects greedy (the
         penalty still applies — matching do_sample=False + penalty in HF)."""
+        import math
+
         t = 0.7 if temperature is None else float(temperature)
+        if not math.isfinite(t):
+            t = 0.7  # NaN/inf from hostile JSON: fall back to the default
         p = cls(temperature=max(t, 1e-4), greedy=t <= 1e-4)
-        if top_p is not None:
+        if top_p is not None and math.isfinite(float(top_p)):
             p.top_p = float(top_p)
         if top_k is not None:
-            p.top_k = int(top_k)
-        p.repetition_penalty = (
-            1.15 if repetition_penalty is None else float(repetition_penalty)
-        )
+            p.top_k = max(0, int(top_k))
+        rp = (1.15 if repetition_penalty is None
+              else float(repetition_penalty))
+        p.repetition_penalty = rp if math.isfinite(rp) and rp > 0 else 1.15
         return p
 
 

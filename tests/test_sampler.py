@@ -85,3 +85,20 @@ def test_sample_greedy_ignores_generator():
     logits = torch.randn(4, 32)
     sp = SamplingParams(greedy=True)
     assert torch.equal(sample(logits, sp), logits.argmax(-1))
+
+
+def test_from_request_hostile_values_sanitized():
+    """NaN/inf/negative knobs (json.loads accepts NaN/Infinity) fall back
+    to defaults instead of poisoning softmax/multinomial downstream."""
+    import math
+
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    p = SamplingParams.from_request(float("nan"), float("inf"),
+                                    -3, float("-inf"))
+    assert math.isfinite(p.temperature) and p.temperature > 0
+    assert 0.0 < p.top_p <= 1.0 or p.top_p == 0.95
+    assert p.top_k == 0
+    assert p.repetition_penalty == 1.15
+    p2 = SamplingParams.from_request(float("inf"))
+    assert math.isfinite(p2.temperature)
