@@ -646,7 +646,7 @@ class InferenceEngine:
         completed = [batch[i][0] for i in done_idx]
         last_rows = torch.tensor([cu[i + 1] - 1 for i in done_idx], dtype=torch.int64).to(dev, non_blocking=True)
         logits = self.runner.lm_head(hidden[last_rows])
-        self._sample_and_emit(completed, logits)
+        self._sample_and_emit(completed, logits, update_last=False)
         for a in completed:
             if a.req.done_ts is not None:
                 # finished AT prefill (max_new=1 / instant stop token):
@@ -946,7 +946,7 @@ class InferenceEngine:
             ).to(logits.device, non_blocking=True)
             for a in nong:
                 a.length += 1
-            self._sample_and_emit(nong, logits[rows])
+            self._sample_and_emit(nong, logits[rows], update_last=False)
         self.spec_stats["steps"] += 1
         self._dec_seqs = None  # device decode-state caches are stale
         done_acts = [a for a in acts if a.req.done_ts is not None]
@@ -990,7 +990,8 @@ class InferenceEngine:
             self._pen_free.append(a.pen_slot)
             a.pen_slot = None
 
-    def _sample_and_emit(self, acts: List[_Active], logits: torch.Tensor) -> None:
+    def _sample_and_emit(self, acts: List[_Active], logits: torch.Tensor,
+                         update_last: bool = True) -> None:
         # group rows by sampling params so each group is one sample() call
         groups: Dict[tuple, List[int]] = {}
         for i, a in enumerate(acts):
@@ -1064,7 +1065,15 @@ class InferenceEngine:
                 idx_t = torch.tensor(pen_idx, dtype=torch.int64,
                                      device=logits.device)
                 self._pen_pool[slot_t, next_dev[idx_t]] = True
-        self._last_sampled = next_dev  # feeds the next step without H2D
+        if update_last:
+            # feeds the next decode step without an H2D copy. Callers whose
+            # `acts` is NOT the current decode set (prefill completions,
+            # spec-decode subsets) must pass update_last=False: the decode
+            # stable-path copies _last_sampled[:B] as the next inputs, and a
+            # clobber from another batch feeds WRONG TOKENS (a request once
+            # decoded its neighbor's token — caught by the scheduling-
+            # invariance property test).
+            self._last_sampled = next_dev
         next_ids = next_dev.cpu()  # the one host sync per step (emission)
         now = time.time()
         n_emitted = 0
