@@ -365,3 +365,39 @@ def test_oversized_prompt_rejected_typed():
         assert req.error is not None or len(req.output_ids) <= 32
     finally:
         eng.shutdown()
+
+
+def test_prefill_completion_does_not_clobber_decode_inputs():
+    """Deterministic regression for the _last_sampled clobber: request B
+    finishes AT prefill completion (max_new=1) between two decode steps of
+    request A (membership unchanged) — A's next decode input must be A's
+    own last token, not B's. Found by the scheduling-invariance property."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    def run(mpt):
+        eng = InferenceEngine("tiny", device="cpu", max_batch=4,
+                              max_seq_len=128, seed=5,
+                              max_prefill_tokens=mpt)
+        try:
+            reqs = [GenerationRequest(prompt_ids=[6], max_new_tokens=3,
+                                      sampling=SamplingParams(greedy=True)),
+                    GenerationRequest(prompt_ids=[5] * 16, max_new_tokens=1,
+                                      sampling=SamplingParams(greedy=True))]
+            for r in reqs:
+                eng.submit(r)
+            outs = []
+            for r in reqs:
+                while True:
+                    item = r.out_queue.get(timeout=60)
+                    if not isinstance(item, int):
+                        break
+                assert r.error is None, r.error
+                outs.append(list(r.output_ids))
+            return outs
+        finally:
+            eng.shutdown()
+
+    # the tiny chunk budget forces B's final chunk + completion between A's
+    # decode steps; outputs must equal the unconstrained schedule
+    assert run(16) == run(4096)
