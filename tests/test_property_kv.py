@@ -273,10 +273,13 @@ class TestEngineSchedulingEquivalence:
         max_news = [data.draw(st.sampled_from([1, 1, 2, 4, 8]))
                     for _ in range(n_req)]
 
-        def run(max_prefill_tokens):
+        spec = data.draw(st.booleans())
+
+        def run(max_prefill_tokens, spec_decode=False):
             eng = InferenceEngine("tiny", device="cpu", max_batch=4,
                                   max_seq_len=128, seed=5,
-                                  max_prefill_tokens=max_prefill_tokens)
+                                  max_prefill_tokens=max_prefill_tokens,
+                                  spec_decode=spec_decode)
             try:
                 reqs = [eng.generate_async(p, max_new_tokens=m,
                                            temperature=0.0)
@@ -305,7 +308,10 @@ class TestEngineSchedulingEquivalence:
             finally:
                 eng.shutdown()
 
-        assert run(max_prefill_tokens=16) == run(max_prefill_tokens=4096)
+        base = run(max_prefill_tokens=4096)
+        assert run(max_prefill_tokens=16) == base
+        if spec:  # chunked prefill AND speculative decoding together
+            assert run(max_prefill_tokens=16, spec_decode=True) == base
 
 
 class TestSpecDecodeInvariance:
