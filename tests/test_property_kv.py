@@ -356,3 +356,68 @@ class TestSpecDecodeInvariance:
                 eng.shutdown()
 
         assert run(True) == run(False)
+
+
+class TestMixedSamplingBatch:
+    @given(data=st.data())
+    @settings(max_examples=8, deadline=None)
+    def test_greedy_rows_invariant_to_sampled_neighbors(self, data):
+        """Greedy (and penalized-greedy) requests co-batched with
+        temperature-sampled neighbors must emit exactly what they emit
+        alone — the sampling-group split and penalty-slot pool must never
+        leak across rows."""
+        from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+        from bee2bee_amd.engine.sampler import SamplingParams
+
+        n_greedy = data.draw(st.integers(min_value=1, max_value=2))
+        greedy_specs = []
+        for _ in range(n_greedy):
+            prompt = data.draw(st.lists(
+                st.integers(min_value=5, max_value=40),
+                min_size=1, max_size=12))
+            pen = data.draw(st.sampled_from([1.0, 1.15]))
+            greedy_specs.append((prompt, pen))
+        n_sampled = data.draw(st.integers(min_value=1, max_value=2))
+
+        def make_reqs():
+            reqs = [GenerationRequest(
+                prompt_ids=p, max_new_tokens=6,
+                sampling=SamplingParams(greedy=True, repetition_penalty=pen))
+                for p, pen in greedy_specs]
+            for i in range(n_sampled):
+                reqs.append(GenerationRequest(
+                    prompt_ids=[20 + i, 21], max_new_tokens=6,
+                    sampling=SamplingParams(temperature=0.9, top_p=0.9,
+                                            top_k=8)))
+            return reqs
+
+        def drain(eng, reqs):
+            for r in reqs:
+                eng.submit(r)
+            for r in reqs:
+                while True:
+                    item = r.out_queue.get(timeout=60)
+                    if not isinstance(item, int):
+                        break
+                assert r.error is None, r.error
+            return [list(r.output_ids) for r in reqs]
+
+        eng = InferenceEngine("tiny", device="cpu", max_batch=8,
+                              max_seq_len=128, seed=3)
+        try:
+            mixed = drain(eng, make_reqs())[:n_greedy]
+        finally:
+            eng.shutdown()
+
+        eng = InferenceEngine("tiny", device="cpu", max_batch=8,
+                              max_seq_len=128, seed=3)
+        try:
+            solo_reqs = [GenerationRequest(
+                prompt_ids=p, max_new_tokens=6,
+                sampling=SamplingParams(greedy=True, repetition_penalty=pen))
+                for p, pen in greedy_specs]
+            solo = drain(eng, solo_reqs)
+        finally:
+            eng.shutdown()
+
+        assert mixed == solo
