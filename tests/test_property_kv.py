@@ -421,3 +421,45 @@ class TestMixedSamplingBatch:
             eng.shutdown()
 
         assert mixed == solo
+
+
+class TestArrivalTimingInvariance:
+    @given(data=st.data())
+    @settings(max_examples=5, deadline=None)
+    def test_staggered_arrivals_match_burst(self, data):
+        """Per-request greedy outputs must not depend on WHEN other
+        requests arrive (mid-decode admissions vs one burst)."""
+        import time as _t
+
+        from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+        from bee2bee_amd.engine.sampler import SamplingParams
+
+        n = data.draw(st.integers(min_value=2, max_value=4))
+        prompts = [data.draw(st.lists(st.integers(min_value=5, max_value=60),
+                                      min_size=1, max_size=20))
+                   for _ in range(n)]
+
+        def run(stagger):
+            eng = InferenceEngine("tiny", device="cpu", max_batch=8,
+                                  max_seq_len=128, seed=11)
+            try:
+                reqs = [GenerationRequest(
+                    prompt_ids=p, max_new_tokens=5,
+                    sampling=SamplingParams(greedy=True)) for p in prompts]
+                for r in reqs:
+                    eng.submit(r)
+                    if stagger:
+                        _t.sleep(0.03)  # others decode while this queues
+                outs = []
+                for r in reqs:
+                    while True:
+                        item = r.out_queue.get(timeout=60)
+                        if not isinstance(item, int):
+                            break
+                    assert r.error is None, r.error
+                    outs.append(list(r.output_ids))
+                return outs
+            finally:
+                eng.shutdown()
+
+        assert run(True) == run(False)
