@@ -401,3 +401,40 @@ def test_prefill_completion_does_not_clobber_decode_inputs():
     # the tiny chunk budget forces B's final chunk + completion between A's
     # decode steps; outputs must equal the unconstrained schedule
     assert run(16) == run(4096)
+
+
+def test_kv_pressure_queues_and_completes_all():
+    """More concurrent requests than the KV pool can hold at once: the
+    engine must admit in waves (block reservation) and complete every
+    request correctly — never deadlock, never error."""
+    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
+    from bee2bee_amd.engine.sampler import SamplingParams
+
+    # max_batch 8 but a pool that only fits ~3 active sequences
+    eng = InferenceEngine("tiny", device="cpu", max_batch=8, max_seq_len=256,
+                          seed=8, kv_margin_blocks=0)
+    try:
+        total = eng.kv.n_blocks
+        reqs = [GenerationRequest(prompt_ids=[5 + i] * 100,
+                                  max_new_tokens=6,
+                                  sampling=SamplingParams(greedy=True))
+                for i in range(10)]
+        for r in reqs:
+            eng.submit(r)
+        for r in reqs:
+            while True:
+                item = r.out_queue.get(timeout=120)
+                if not isinstance(item, int):
+                    break
+            assert r.error is None, r.error
+            assert len(r.output_ids) == 6
+        # pool fully restored
+        import time as _t
+
+        for _ in range(100):
+            if eng.kv.free_blocks == total - 1:  # -1: scratch seq
+                break
+            _t.sleep(0.05)
+        assert eng.kv.free_blocks >= total - 1
+    finally:
+        eng.shutdown()
