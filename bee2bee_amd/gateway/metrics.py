@@ -79,6 +79,12 @@ def refresh(node) -> None:
             val = stats.get(key)
             if isinstance(val, (int, float)):
                 _engine_gauge(key, doc).set(val)
+        spec = stats.get("spec_decode")
+        if isinstance(spec, dict):
+            for k, v in spec.items():
+                if isinstance(v, (int, float)):
+                    _engine_gauge(f"spec_{k}",
+                                  f"speculative decoding: {k}").set(v)
 
 
 def render() -> bytes:

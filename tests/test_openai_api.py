@@ -349,3 +349,29 @@ def test_completions_batched_prompts(client):
     r = client.post("/v1/completions", headers=KEY,
                     json={"prompt": "x", "max_tokens": 0})
     assert r.status_code == 400
+
+
+def test_metrics_spec_decode_gauges(client):
+    from bee2bee_amd.gateway import api as gateway_api
+    from bee2bee_amd.services.native import NativeEngineService
+
+    import os
+
+    os.environ["BEE2BEE_SPEC_DECODE"] = "1"
+    try:
+        svc = NativeEngineService("tiny", device="cpu", max_batch=2,
+                                  max_seq_len=128)
+        svc.load_sync()
+    finally:
+        del os.environ["BEE2BEE_SPEC_DECODE"]
+    try:
+        gateway_api.node.local_services["hf"] = svc
+        client.post("/chat", headers=KEY,
+                    json={"prompt": "a b a b a b", "model": "tiny",
+                          "max_new_tokens": 6, "temperature": 0.0,
+                          "repetition_penalty": 1.0})
+        text = client.get("/metrics").text
+        assert "bee2bee_engine_spec_steps" in text
+        assert "bee2bee_engine_spec_accepted" in text
+    finally:
+        svc.engine.shutdown()
