@@ -343,6 +343,9 @@ class InferenceEngine:
     # -------------------------------------------------------------- serving
 
     def submit(self, req: GenerationRequest) -> GenerationRequest:
+        # prefill completion always emits one token, so <1 behaves as 1 —
+        # normalize for exact accounting
+        req.max_new_tokens = max(1, int(req.max_new_tokens))
         self.start()
         self._pending.put(req)
         self._wake.set()
