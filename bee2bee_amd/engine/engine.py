@@ -343,17 +343,8 @@ class InferenceEngine:
     # -------------------------------------------------------------- serving
 
     def submit(self, req: GenerationRequest) -> GenerationRequest:
-        if not req.prompt_ids:
-            # an empty prompt would index prefill row -1 (a NEIGHBOR'S
-            # logits in a batched prefill) — refuse with a typed error
-            req.error = "empty_prompt"
-            req.done_ts = time.time()
-            req.out_queue.put(_STREAM_END)
-            if req.on_emit is not None:
-                req.on_emit(-1, True)
-            return req
-        # prefill completion always emits one token, so <1 behaves as 1 —
-        # normalize for exact accounting
+        # empty prompts become [bos] at admission (engine loop); <1
+        # max_new behaves as 1 (prefill completion always emits one token)
         req.max_new_tokens = max(1, int(req.max_new_tokens))
         self.start()
         self._pending.put(req)

@@ -438,23 +438,3 @@ def test_kv_pressure_queues_and_completes_all():
         assert eng.kv.free_blocks >= total - 1
     finally:
         eng.shutdown()
-
-
-def test_empty_prompt_rejected_typed():
-    """Empty prompts would read prefill row -1 (a neighbor's logits in a
-    batched prefill) — submit() refuses with a typed error instead."""
-    from bee2bee_amd.engine.engine import GenerationRequest, InferenceEngine
-    from bee2bee_amd.engine.sampler import SamplingParams
-
-    eng = InferenceEngine("tiny", device="cpu", max_batch=2, max_seq_len=64,
-                          seed=1)
-    try:
-        req = GenerationRequest(prompt_ids=[], max_new_tokens=4,
-                                sampling=SamplingParams(greedy=True))
-        eng.submit(req)
-        item = req.out_queue.get(timeout=10)
-        assert not isinstance(item, int)
-        assert req.error == "empty_prompt"
-        assert req.output_ids == []
-    finally:
-        eng.shutdown()
