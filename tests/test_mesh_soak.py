@@ -83,7 +83,6 @@ def test_mesh_soak_mixed_operations():
         res = await relay.request_generation(
             provider.peer_id, "final", 8, "soak-model", timeout=20)
         assert res["text"] == "echo:final"
-        assert provider.kv_alive if hasattr(provider, "kv_alive") else True
 
         for node in churners:
             await node.stop()
