@@ -1,5 +1,7 @@
 import json
 
+import pytest
+
 from click.testing import CliRunner
 
 from bee2bee_amd.__main__ import cli
@@ -187,3 +189,34 @@ def test_run_mesh_node_sigterm_graceful():
     out, err = proc.communicate(timeout=30)
     assert proc.returncode == 0, (proc.returncode, err[-800:])
     assert "CLEAN_EXIT" in out
+
+
+def test_example_guide_runs_end_to_end():
+    """examples/guide.py (the live two-node walkthrough) must stay
+    runnable — examples rot silently otherwise."""
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run([_sys.executable, "examples/guide.py"],
+                         capture_output=True, text=True, timeout=180)
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    assert "gen_result" in out.stdout
+    assert "fetched + hash-verified" in out.stdout
+    assert "done" in out.stdout
+
+
+@pytest.mark.timeout(300)
+def test_example_parallel_demo_tp2():
+    """examples/parallel_demo.py runs under the documented torchrun shape
+    (gloo world 2, tiny model)."""
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run(
+        [_sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29897", "examples/parallel_demo.py",
+         "--mode", "tp"],
+        capture_output=True, text=True, timeout=280,
+    )
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
