@@ -92,6 +92,12 @@ async def seed_checkpoint(
 ) -> Dict:
     """Split every file of a checkpoint dir into pieces, register them with
     the node for serving, and publish a manifest + provider records."""
+    # piece_data frames carry base64 (4/3 overhead) inside the 32 MiB WS
+    # limit; leave headroom for the JSON envelope
+    if piece_size > 20 * 1024 * 1024:
+        raise ValueError(
+            f"piece_size {piece_size} exceeds the wire frame budget "
+            "(max 20 MiB: base64 must fit the 32 MiB WS limit)")
     import hashlib
 
     files = []

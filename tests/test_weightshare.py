@@ -354,3 +354,24 @@ def test_seeding_is_disk_backed(tmp_path):
         await node.stop()
 
     asyncio.run(run())
+
+
+def test_oversize_piece_size_rejected(tmp_path):
+    """piece_size beyond the wire-frame budget is refused up front (a
+    32 MiB+ base64 piece_data frame would be dropped by the WS layer and
+    look like a hung fetch)."""
+    import pytest
+
+    async def run():
+        dht = DHTNode()
+        await dht.start()
+        node = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await node.start()
+        try:
+            with pytest.raises(ValueError, match="frame budget"):
+                await seed_checkpoint(node, dht, "big", str(tmp_path),
+                                      piece_size=32 * 1024 * 1024)
+        finally:
+            await node.stop()
+
+    asyncio.run(run())
