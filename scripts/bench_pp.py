@@ -34,7 +34,7 @@ def main() -> None:
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--steps", type=int, default=16)
     ap.add_argument("--warmup", type=int, default=4)
-    ap.add_argument("--mode", default="pp", choices=["pp", "tp", "cp"])
+    ap.add_argument("--mode", default="pp", choices=["pp", "tp", "cp", "ep"])
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -52,6 +52,12 @@ def main() -> None:
         from bee2bee_amd.parallel.tp import TPEngine as Engine
     elif args.mode == "cp":
         from bee2bee_amd.parallel.cp import CPEngine as Engine
+    elif args.mode == "ep":
+        from bee2bee_amd.parallel.moe_engine import MoEEngine as Engine
+
+        if args.model == "llama3-70b":  # ep needs a MoE model
+            args.model = "mixtral-8x7b" if torch.cuda.is_available() \
+                else "tiny-moe"
     else:
         from bee2bee_amd.parallel.pp import PipelineEngine as Engine
 
@@ -89,7 +95,7 @@ def main() -> None:
     if rank == 0:
         print(json.dumps({
             "metric": f"output tokens/sec ({eng.spec.name} bf16 greedy decode, "
-                      f"{dict(tp='tensor', cp='context', pp='pipeline')[args.mode]}"
+                      f"{dict(tp='tensor', cp='context', pp='pipeline', ep='expert')[args.mode]}"
                       f"-parallel {args.mode}{world})",
             "value": round(args.batch * args.steps / elapsed, 1),
             "unit": "tokens/s",

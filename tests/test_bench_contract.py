@@ -60,11 +60,11 @@ def test_bench_multi_process_contract():
 
 
 @pytest.mark.timeout(300)
-@pytest.mark.parametrize("mode,model", [("pp", "tiny"), ("tp", "tiny"), ("cp", "tiny")])
+@pytest.mark.parametrize("mode,model", [("pp", "tiny"), ("tp", "tiny"), ("cp", "tiny"), ("ep", "tiny-moe")])
 def test_bench_pp_contract(mode, model):
     """scripts/bench_pp.py (the 8-GPU model-parallel bench) launches under
     torch.distributed.run and emits the JSON contract (gloo dry run)."""
-    port = {"pp": "29771", "tp": "29772", "cp": "29773"}[mode]
+    port = {"pp": "29771", "tp": "29772", "cp": "29773", "ep": "29774"}[mode]
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
