@@ -134,3 +134,18 @@ def test_serve_parallel_oneshot_contract(mode):
     assert out.returncode == 0, out.stderr[-2000:]
     assert "ONESHOT_RESULT tokens=8 world=2" in out.stdout
     assert "rank 1: served 1 requests" in out.stdout
+
+
+def test_gpu_test_files_are_properly_gated():
+    """Meta-test: every tests/*_gpu.py file must mark itself `gpu` AND
+    module-level-skip off-GPU — an unguarded file would break CPU CI, an
+    unmarked one would silently run (and fail) in the CPU lane."""
+    import glob
+
+    files = sorted(glob.glob(os.path.join(REPO, "tests", "test_*_gpu.py")))
+    assert files, "gpu test files disappeared?"
+    for f in files:
+        src = open(f).read()
+        assert "pytestmark = pytest.mark.gpu" in src, f
+        assert "allow_module_level=True" in src, f
+        assert "torch.cuda.is_available()" in src, f
