@@ -144,6 +144,40 @@ def serve_hf_remote(model, token, region, api_port):
     _serve("hf_remote", model_name=model, region=region, api_port=api_port)
 
 
+@cli.command("serve-parallel")
+@click.option("--mode", default="tp",
+              type=click.Choice(["pp", "tp", "ep", "cp"]))
+@click.option("--nproc", default=8, type=int, help="ranks (one per GPU)")
+@click.option("--model", default=None)
+@click.option("--model-path", default=None)
+@click.option("--api-port", default=8000, type=int)
+@click.option("--max-batch", default=8, type=int)
+@click.option("--max-seq-len", default=2048, type=int)
+@click.option("--master-port", default=29500, type=int)
+def serve_parallel(mode, nproc, model, model_path, api_port, max_batch,
+                   max_seq_len, master_port):
+    """Serve a SHARDED model (pipeline/tensor/expert/context parallel)
+    behind the standard node + gateway: wraps the torch.distributed.run
+    launch of bee2bee_amd.parallel.serve_main (one rank per GPU; rank 0
+    owns the mesh/API, ranks 1+ follow in lockstep)."""
+    import subprocess
+    import sys as _sys
+
+    cmd = [
+        _sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", str(nproc), "--master-addr", "127.0.0.1",
+        "--master-port", str(master_port),
+        "-m", "bee2bee_amd.parallel.serve_main",
+        "--mode", mode, "--api-port", str(api_port),
+        "--max-batch", str(max_batch), "--max-seq-len", str(max_seq_len),
+    ]
+    if model:
+        cmd += ["--model", model]
+    if model_path:
+        cmd += ["--model-path", model_path]
+    raise SystemExit(subprocess.run(cmd, check=False).returncode)
+
+
 @cli.command("doctor")
 @click.option("--port", default=None, type=int,
               help="Also probe a specific mesh port for bindability")

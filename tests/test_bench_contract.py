@@ -149,3 +149,19 @@ def test_gpu_test_files_are_properly_gated():
         assert "pytestmark = pytest.mark.gpu" in src, f
         assert "allow_module_level=True" in src, f
         assert "torch.cuda.is_available()" in src, f
+
+
+@pytest.mark.timeout(300)
+def test_serve_parallel_cli_launch():
+    """`python -m bee2bee_amd serve-parallel` wraps the torchrun launch of
+    the packaged entry (oneshot via module path, gloo world 2)."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29885", "-m", "bee2bee_amd.parallel.serve_main",
+         "--mode", "tp", "--max-batch", "2", "--max-seq-len", "64",
+         "--oneshot-prompt", "cli shim"],
+        cwd=REPO, capture_output=True, text=True, timeout=240,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "ONESHOT_RESULT tokens=8 world=2" in out.stdout
