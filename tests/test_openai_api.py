@@ -375,3 +375,27 @@ def test_metrics_spec_decode_gauges(client):
         assert "bee2bee_engine_spec_accepted" in text
     finally:
         svc.engine.shutdown()
+
+
+def test_open_mode_without_api_key(monkeypatch):
+    """With no BEE2BEE_API_KEY configured the /v1 surface is open (dev
+    mode), matching the reference's optional-auth behavior."""
+    monkeypatch.delenv("BEE2BEE_API_KEY", raising=False)
+    monkeypatch.setenv("BEE2BEE_DISABLE_NAT", "1")
+    monkeypatch.setenv("BEE2BEE_PORT", "0")
+    monkeypatch.setenv("BEE2BEE_HOST", "127.0.0.1")
+    monkeypatch.delenv("BEE2BEE_BOOTSTRAP", raising=False)
+    from bee2bee_amd.gateway import api as gateway_api
+
+    gateway_api.node = None
+    with TestClient(gateway_api.app) as c:
+        from tests.test_mesh import EchoService
+
+        svc = EchoService(model="open-model")
+        gateway_api.node.local_services[svc.name] = svc
+        assert c.get("/v1/models").status_code == 200  # no key needed
+        r = c.post("/v1/completions", json={"model": "open-model",
+                                            "prompt": "open"})
+        assert r.status_code == 200
+        assert r.json()["choices"][0]["text"] == "echo:open"
+    gateway_api.node = None
