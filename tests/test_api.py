@@ -254,3 +254,23 @@ def test_generate_forwards_sampling_knobs(client):
     assert seen["top_p"] == 0.5
     assert seen["top_k"] == 11
     assert seen["repetition_penalty"] == 1.3
+
+
+def test_connect_failure_is_typed(client):
+    r = client.get("/connect", headers={"X-API-KEY": "secret-key"},
+                   params={"addr": "ws://127.0.0.1:9"})  # nothing listens
+    body = r.json()
+    assert body["status"] == "error"
+    assert "message" in body
+
+
+def test_chat_unknown_model_p2p_error_shape(client):
+    """No local service and no providers: the P2P fallback returns the
+    typed error envelope, never a 500."""
+    r = client.post("/chat", headers={"X-API-KEY": "secret-key"},
+                    json={"prompt": "x", "model": "ghost-model"})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["status"] == "error"
+    assert "no_local_service" in body["message"] or \
+        "no_node_available" in body["message"]
