@@ -66,3 +66,43 @@ def test_bpe_vocab_merges_fallback(tmp_path):
     ids = tk.encode("hello")
     assert tk.decode(ids) == "hello"
     assert tk.eos_token_id == 0  # <|endoftext|>
+
+
+def test_stream_decoder_with_real_byte_level_bpe(tmp_path):
+    """TextStreamDecoder's prefix-slicing contract against a REAL
+    byte-level BPE (the tokenizer.json family llama3/gpt2 use): decode of a
+    growing id list must extend monotonically, and streamed deltas must
+    reassemble the text exactly."""
+    tokenizers = pytest.importorskip("tokenizers")
+    from tokenizers import Tokenizer, models, pre_tokenizers, decoders, trainers
+
+    tk = Tokenizer(models.BPE(unk_token=None))
+    tk.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tk.decoder = decoders.ByteLevel()
+    trainer = trainers.BpeTrainer(vocab_size=300, special_tokens=["<bos>"])
+    tk.train_from_iterator(
+        ["hello mesh world", "héllo wörld", "aaa bbb aaab", "😀 emoji test"],
+        trainer)
+    path = tmp_path / "tokenizer.json"
+    tk.save(str(path))
+
+    from bee2bee_amd.engine.engine import TextStreamDecoder
+    from bee2bee_amd.models.tokenizer import HFTokenizer
+
+    tok = HFTokenizer(str(path))
+    for text in ["hello mesh world", "héllo wörld 😀", "aaab aaa bbb"]:
+        ids = tok.encode(text, add_bos=False)
+        # prefix monotonicity of decode (byte-level property the streaming
+        # decoder relies on)
+        prev = ""
+        for n in range(1, len(ids) + 1):
+            cur = tok.decode(ids[:n])
+            assert cur.startswith(prev) or cur.rstrip("�").startswith(
+                prev.rstrip("�")), (prev, cur)
+            prev = cur
+        dec = TextStreamDecoder(tok, flush_every=1)
+        out = ""
+        for n in range(1, len(ids) + 1):
+            out += dec.delta(ids[:n])
+        out += dec.delta(ids, final=True)
+        assert out == tok.decode(ids)
