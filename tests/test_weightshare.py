@@ -375,3 +375,48 @@ def test_oversize_piece_size_rejected(tmp_path):
             await node.stop()
 
     asyncio.run(run())
+
+
+def test_fetched_checkpoint_serves_identically(tmp_path):
+    """The full torrent->serve loop: seed a checkpoint, fetch it over the
+    mesh on another peer, and the ENGINE serving the fetched copy emits
+    exactly the tokens the source copy emits."""
+    spec = PRESETS["tiny"]
+    src_dir = tmp_path / "src"
+    dst_dir = tmp_path / "dst"
+    w = ModelWeights(spec, torch.device("cpu"), torch.float32).random_init(21)
+    save_hf(w, str(src_dir))
+
+    async def run():
+        dht = DHTNode()
+        await dht.start()
+        seeder = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        leech = MeshNode(host="127.0.0.1", port=0, enable_nat=False)
+        await seeder.start()
+        await leech.start()
+        await seed_checkpoint(seeder, dht, "serve-ck", str(src_dir),
+                              piece_size=4096)
+        await leech.connect_bootstrap(seeder.addr)
+        for _ in range(200):
+            if seeder.peer_id in leech.peers:
+                break
+            await asyncio.sleep(0.02)
+        await fetch_checkpoint(leech, dht, "serve-ck", str(dst_dir))
+        await leech.stop()
+        await seeder.stop()
+
+    asyncio.run(run())
+
+    from bee2bee_amd.engine.engine import InferenceEngine
+
+    def toks(path):
+        eng = InferenceEngine("tiny", device="cpu", model_path=path,
+                              max_batch=2, max_seq_len=128, seed=4)
+        try:
+            req = eng.generate([7, 8, 9], max_new_tokens=8, temperature=0.0,
+                               repetition_penalty=1.0)
+            return list(req.output_ids)
+        finally:
+            eng.shutdown()
+
+    assert toks(str(src_dir)) == toks(str(dst_dir))
