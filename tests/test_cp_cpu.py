@@ -246,3 +246,31 @@ def test_cpengine_short_prompt_empty_shard(tmp_path):
     finally:
         eng.shutdown()
     assert cp_outs == [ref], (cp_outs, ref)
+
+
+def test_cp_page_ownership_partitions_exactly():
+    """Pure page-math invariants of CPEngine (no process group needed):
+    for any (context length, world size), the owned positions partition
+    [0, L) exactly across ranks, local positions are injective per rank,
+    and the per-rank local lengths sum to L."""
+    from bee2bee_amd.parallel.cp import CPEngine
+
+    page = 256
+    for world in (1, 2, 3, 5, 8):
+        for L in (1, page - 1, page, page + 1, 3 * page + 17, 8 * page):
+            ranks = []
+            for r in range(world):
+                eng = CPEngine.__new__(CPEngine)  # math only, no dist/init
+                eng.rank, eng.world, eng.page = r, world, page
+                ranks.append(eng)
+            owners = [sum(1 for e in ranks if e._owned(p))
+                      for p in range(L)]
+            assert owners == [1] * L, (world, L)
+            total = 0
+            for e in ranks:
+                mine = [p for p in range(L) if e._owned(p)]
+                locs = [e._local_pos(p) for p in mine]
+                assert len(set(locs)) == len(locs), "local slot collision"
+                assert e._local_len(L) == len(mine), (world, L, e.rank)
+                total += len(mine)
+            assert total == L
