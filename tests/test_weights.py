@@ -123,3 +123,21 @@ def test_tied_head_loads_on_pp_last_stage(tmp_path):
         assert last.embed is None, preset
         assert last.lm_head is not None, preset
         assert torch.equal(last.lm_head, w.embed), preset
+
+
+def test_save_load_bf16_exact(tmp_path):
+    """bf16 checkpoints round-trip bit-exactly (the serving dtype)."""
+    import torch
+
+    from bee2bee_amd.models.spec import PRESETS
+    from bee2bee_amd.models.weights import ModelWeights, save_hf
+
+    spec = PRESETS["tiny"]
+    w = ModelWeights(spec, torch.device("cpu"), torch.bfloat16).random_init(13)
+    save_hf(w, str(tmp_path))
+    w2 = ModelWeights(spec, torch.device("cpu"), torch.bfloat16).load_hf(
+        str(tmp_path))
+    assert w2.embed.dtype == torch.bfloat16
+    assert torch.equal(w.embed, w2.embed)
+    assert torch.equal(w.layers[1].w_down, w2.layers[1].w_down)
+    assert torch.equal(w.final_norm, w2.final_norm)
