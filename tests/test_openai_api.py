@@ -399,3 +399,21 @@ def test_open_mode_without_api_key(monkeypatch):
         assert r.status_code == 200
         assert r.json()["choices"][0]["text"] == "echo:open"
     gateway_api.node = None
+
+
+def test_metrics_keys_match_engine_stats_contract():
+    """Pin the coupling: every engine gauge key in gateway/metrics.py must
+    exist in InferenceEngine.stats() output (key drift silently zeroes the
+    dashboards — it happened once)."""
+    from bee2bee_amd.engine.engine import InferenceEngine
+    from bee2bee_amd.gateway.metrics import _ENGINE_STATS
+
+    eng = InferenceEngine("tiny", device="cpu", max_batch=1, max_seq_len=64,
+                          seed=1)
+    try:
+        stats = eng.stats()
+        for key, _doc in _ENGINE_STATS:
+            assert key in stats, key
+            assert isinstance(stats[key], (int, float)), key
+    finally:
+        eng.shutdown()
