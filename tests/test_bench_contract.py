@@ -165,3 +165,17 @@ def test_serve_parallel_cli_launch():
     )
     assert out.returncode == 0, out.stderr[-2000:]
     assert "ONESHOT_RESULT tokens=8 world=2" in out.stdout
+
+
+def test_bench_defaults_match_baseline_json():
+    """The driver benches the metric/config BASELINE.json names — pin the
+    defaults (llama3-8b, weak-scaling replicas, bf16) so they cannot drift
+    from the declared headline."""
+    import json
+
+    baseline = json.loads(open(os.path.join(REPO, "BASELINE.json")).read())
+    assert "Llama-3-8B" in baseline["metric"]
+    src = open(os.path.join(REPO, "bench.py")).read()
+    assert '"--model", default="llama3-8b"' in src.replace("'", '"')
+    assert '"scaling": "weak"' in src
+    assert "llama3-8b" in src
