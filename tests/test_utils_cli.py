@@ -220,3 +220,19 @@ def test_example_parallel_demo_tp2():
         capture_output=True, text=True, timeout=280,
     )
     assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+
+
+def test_version_consistent_everywhere():
+    """pyproject.toml, bee2bee_amd.__version__ and the CLI --version all
+    agree (the reference shipped 3.3.1 in code vs 3.7.1 in pyproject)."""
+    import re
+
+    import bee2bee_amd
+
+    py = open("pyproject.toml").read()
+    m = re.search(r'^version = "([^"]+)"', py, re.M)
+    assert m, "no version in pyproject"
+    assert m.group(1) == bee2bee_amd.__version__
+    res = CliRunner().invoke(cli, ["--version"])
+    assert res.exit_code == 0
+    assert bee2bee_amd.__version__ in res.output
