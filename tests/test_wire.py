@@ -73,3 +73,18 @@ def test_request_params_optional_sampling_knobs():
     assert p["top_p"] == 0.4 and p["repetition_penalty"] == 1.2
     p2 = wire.request_params({"prompt": "x"})
     assert "top_p" not in p2  # absent stays absent (engine defaults apply)
+
+
+def test_wire_constants_pinned_to_reference():
+    """Protocol constants are the compatibility surface (reference
+    p2p_runtime.py:176 frame size, :831 timeout, :658/:654/:837/:792 error
+    strings) — pin them so refactors cannot drift the wire."""
+    assert wire.MAX_FRAME == 32 * 1024 * 1024
+    assert wire.REQUEST_TIMEOUT == 300.0
+    assert wire.PING_INTERVAL == 15.0
+    assert wire.ERR_NO_NODE == "consensus_deadlock: no_node_available"
+    assert wire.ERR_RELAY == "relay_link_failure"
+    assert wire.ERR_TIMEOUT == "request_timed_out"
+    assert wire.ERR_NOT_CONNECTED == "provider_not_connected"
+    assert set(wire.TERMINAL_TYPES) == {"gen_result", "gen_success",
+                                        "gen_error"}
