@@ -55,3 +55,13 @@ def test_slot_mapping_spans_pages():
     assert slots[1] == int(blocks[0]) * BLOCK_SIZE + BLOCK_SIZE - 1
     assert slots[2] == int(blocks[1]) * BLOCK_SIZE
     assert slots[3] == int(blocks[1]) * BLOCK_SIZE + 4
+
+
+def test_block_size_pinned_to_measured_optimum():
+    """BLOCK_SIZE=256 is load-bearing: one decode chunk = one page =
+    64 KB-sequential KV streams (measured 5.19 -> 5.42-5.65 TB/s in round
+    2). A silent change here costs ~4% end to end — change it only with a
+    new measurement in BASELINE.md."""
+    from bee2bee_amd.engine import kv as kv_mod
+
+    assert kv_mod.BLOCK_SIZE == 256
