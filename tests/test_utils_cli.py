@@ -236,3 +236,18 @@ def test_version_consistent_everywhere():
     res = CliRunner().invoke(cli, ["--version"])
     assert res.exit_code == 0
     assert bee2bee_amd.__version__ in res.output
+
+
+@pytest.mark.timeout(300)
+def test_example_api_demo_runs_end_to_end():
+    """examples/api_demo.py: full node boot (mesh + engine + FastAPI) and
+    every endpoint walkthrough incl. the OpenAI SSE stream and /metrics."""
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run([_sys.executable, "examples/api_demo.py", "8219"],
+                         capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    assert "demo complete" in out.stdout
+    assert "data: [DONE]" in out.stdout  # OpenAI SSE leg ran
+    assert "bee2bee_http_requests_total" in out.stdout  # metrics leg
