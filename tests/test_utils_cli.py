@@ -251,3 +251,17 @@ def test_example_api_demo_runs_end_to_end():
     assert "demo complete" in out.stdout
     assert "data: [DONE]" in out.stdout  # OpenAI SSE leg ran
     assert "bee2bee_http_requests_total" in out.stdout  # metrics leg
+
+
+def test_cli_bench_wrapper_smoke():
+    """`bee2bee-amd bench` wraps bench.py and surfaces its JSON line."""
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run(
+        [_sys.executable, "-m", "bee2bee_amd", "bench", "--model", "tiny",
+         "--batch", "2", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=240,
+    )
+    assert out.returncode == 0, out.stderr[-800:]
+    assert '"metric"' in out.stdout and '"tokens/s"' in out.stdout
